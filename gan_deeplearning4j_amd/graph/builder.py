@@ -1,0 +1,342 @@
+"""GraphBuilder / ComputationGraph — the DL4J-style model API.
+
+Recreates the API surface the reference exercises:
+NeuralNetConfiguration.Builder().graphBuilder() with addInputs /
+setInputTypes / addLayer / inputPreProcessor / setOutputs / build, then
+init(), summary(), output(), fit(), getLayer(name).getParam/setParam
+(reference Java:118-165, 173-221, 228-310).
+
+Internally a ComputationGraph is an nn.Module executing its vertices in
+topological order on the MI355X op library.
+"""
+
+from __future__ import annotations
+
+from dataclasses import dataclass
+from typing import Iterable, Optional, Sequence, Union
+
+import torch
+import torch.nn as nn
+
+from ..ops import functional as OF
+from ..ops.optim import Updater
+from .layers import BaseLayer, OutputLayer
+
+
+@dataclass
+class InputType:
+    """Input type declaration (DL4J InputType analog)."""
+
+    kind: str               # "feedforward" | "convolutional_flat" | "convolutional"
+    height: int = 0
+    width: int = 0
+    channels: int = 0
+    size: int = 0
+
+    @classmethod
+    def feed_forward(cls, size: int) -> "InputType":
+        return cls("feedforward", size=size)
+
+    @classmethod
+    def convolutional_flat(cls, h: int, w: int, c: int) -> "InputType":
+        # reference Java:130-131: InputType.convolutionalFlat(28,28,1)
+        return cls("convolutional_flat", height=h, width=w, channels=c)
+
+    @classmethod
+    def convolutional(cls, h: int, w: int, c: int) -> "InputType":
+        return cls("convolutional", height=h, width=w, channels=c)
+
+    def shape(self, n: int = 1) -> tuple:
+        if self.kind == "feedforward":
+            return (n, self.size)
+        if self.kind == "convolutional_flat":
+            return (n, self.height * self.width * self.channels)
+        return (n, self.channels, self.height, self.width)
+
+
+class GraphBuilder:
+    """Builds a ComputationGraph.
+
+    Defaults (seed, optimizer kind, clip, l2, default activation) mirror the
+    reference's NeuralNetConfiguration.Builder chain (Java:118-128).
+    """
+
+    def __init__(
+        self,
+        seed: int = 666,
+        default_activation: str = "identity",
+        optim_cfg=None,
+    ):
+        from ..config import OptimConfig
+
+        self.seed = seed
+        self.default_activation = default_activation
+        self.optim_cfg = optim_cfg or OptimConfig()
+        self._inputs: list[str] = []
+        self._input_types: dict[str, InputType] = {}
+        self._outputs: list[str] = []
+        self._vertices: dict[str, tuple[BaseLayer, list[str]]] = {}
+        self._preprocessors: dict[str, BaseLayer] = {}
+
+    def add_inputs(self, *names: str) -> "GraphBuilder":
+        self._inputs.extend(names)
+        return self
+
+    def set_input_types(self, *types: InputType) -> "GraphBuilder":
+        for name, t in zip(self._inputs, types):
+            self._input_types[name] = t
+        return self
+
+    def add_layer(
+        self, name: str, layer: BaseLayer, *inputs: str,
+        preprocessor: Optional[BaseLayer] = None,
+    ) -> "GraphBuilder":
+        if name in self._vertices:
+            raise KeyError(f"duplicate vertex {name!r}")
+        layer.name = name
+        self._vertices[name] = (layer, list(inputs))
+        if preprocessor is not None:
+            self._preprocessors[name] = preprocessor
+        return self
+
+    # DL4J spelling (inputPreProcessor("layer", proc))
+    def input_preprocessor(self, layer_name: str, proc: BaseLayer) -> "GraphBuilder":
+        self._preprocessors[layer_name] = proc
+        return self
+
+    def set_outputs(self, *names: str) -> "GraphBuilder":
+        self._outputs = list(names)
+        return self
+
+    def build(self) -> "ComputationGraph":
+        return ComputationGraph(
+            inputs=self._inputs,
+            input_types=self._input_types,
+            outputs=self._outputs,
+            vertices=self._vertices,
+            preprocessors=self._preprocessors,
+            seed=self.seed,
+            optim_cfg=self.optim_cfg,
+        )
+
+
+class ComputationGraph(nn.Module):
+    def __init__(self, inputs, input_types, outputs, vertices, preprocessors,
+                 seed, optim_cfg):
+        super().__init__()
+        self.input_names = list(inputs)
+        self.input_types = dict(input_types)
+        self.output_names = list(outputs)
+        self.seed = seed
+        self.optim_cfg = optim_cfg
+        self._topo = self._toposort(vertices)
+        self._vertex_inputs = {k: v[1] for k, v in vertices.items()}
+        self.layers = nn.ModuleDict({k: vertices[k][0] for k in self._topo})
+        self.preprocessors = nn.ModuleDict(preprocessors)
+        self._updater: Optional[Updater] = None
+        self._initialized = False
+
+    # ------------------------------------------------------------ build
+    def _toposort(self, vertices) -> list[str]:
+        order, seen, temp = [], set(), set()
+        names = list(vertices)
+
+        def visit(n):
+            if n in seen or n in self.input_names:
+                return
+            if n in temp:
+                raise ValueError(f"cycle at {n}")
+            temp.add(n)
+            for dep in vertices[n][1]:
+                visit(dep)
+            temp.discard(n)
+            seen.add(n)
+            order.append(n)
+
+        for n in names:
+            visit(n)
+        return order
+
+    def init(self) -> "ComputationGraph":
+        """Allocate/initialize parameters (DL4J ComputationGraph.init())."""
+        gen = torch.Generator().manual_seed(self.seed)
+        for name in self._topo:
+            layer = self.layers[name]
+            if hasattr(layer, "reset_parameters"):
+                layer.reset_parameters(gen)
+        self._initialized = True
+        return self
+
+    # ---------------------------------------------------------- forward
+    def forward(self, *inputs: torch.Tensor) -> torch.Tensor:
+        acts: dict[str, torch.Tensor] = {}
+        for name, x, in zip(self.input_names, inputs):
+            t = self.input_types.get(name)
+            if t is not None and t.kind == "convolutional_flat" and x.dim() == 2:
+                x = x.reshape(x.shape[0], t.channels, t.height, t.width)
+            acts[name] = x
+        for name in self._topo:
+            layer = self.layers[name]
+            srcs = [acts[s] for s in self._vertex_inputs[name]]
+            if name in self.preprocessors:
+                srcs = [self.preprocessors[name](s) for s in srcs]
+            acts[name] = layer(*srcs)
+        outs = [acts[n] for n in self.output_names]
+        return outs[0] if len(outs) == 1 else tuple(outs)
+
+    @torch.no_grad()
+    def output(self, *inputs: torch.Tensor) -> torch.Tensor:
+        """Inference forward (reference gen.output(z), Java:420, 551):
+        applies output-layer activations."""
+        was_training = self.training
+        self.eval()
+        try:
+            y = self.forward(*inputs)
+        finally:
+            self.train(was_training)
+        outs = list(y) if isinstance(y, tuple) else [y]
+        for i, name in enumerate(self.output_names):
+            layer = self.layers[name]
+            if isinstance(layer, OutputLayer):
+                act = layer.inference_activation
+                if act == "softmax":
+                    outs[i] = torch.softmax(outs[i].float(), dim=1)
+                else:
+                    outs[i] = OF.activation(outs[i], act) if act != "identity" else outs[i]
+        return outs[0] if len(outs) == 1 else tuple(outs)
+
+    # ------------------------------------------------------------- loss
+    def loss(self, logits, labels) -> torch.Tensor:
+        """Loss of the (single) OutputLayer on logits."""
+        out_layer = self.layers[self.output_names[0]]
+        if not isinstance(out_layer, OutputLayer):
+            raise TypeError("graph output vertex is not an OutputLayer")
+        if out_layer.loss == "xent" and out_layer.inference_activation == "softmax":
+            return OF.softmax_cross_entropy(logits, labels)
+        return OF.LOSSES[out_layer.loss](logits, labels)
+
+    # -------------------------------------------------------------- fit
+    @property
+    def updater(self) -> Updater:
+        if self._updater is None:
+            self._updater = Updater.for_graph(self, self.optim_cfg)
+        return self._updater
+
+    def fit(self, data, epochs: int = 1) -> float:
+        """Train on DataSet(s)/iterator (SparkComputationGraph.fit analog).
+
+        Returns the last minibatch loss.
+        """
+        from ..data.csv_reader import DataSet
+
+        self.train()
+        last = 0.0
+        for _ in range(epochs):
+            batches: Iterable
+            if isinstance(data, DataSet):
+                batches = [data]
+            else:
+                batches = data
+            for ds in batches:
+                dev = next(self.parameters()).device
+                feats = ds.features.to(dev)
+                labels = ds.labels.to(dev)
+                self.updater.zero_grad()
+                logits = self.forward(feats)
+                loss = self.loss(logits, labels)
+                loss.backward()
+                self.updater.step()
+                last = float(loss.detach())
+        return last
+
+    # ----------------------------------------------------- param access
+    def get_layer(self, name: str) -> BaseLayer:
+        return self.layers[name]
+
+    def layer_names(self) -> list[str]:
+        return list(self._topo)
+
+    def n_params(self) -> int:
+        """Total param count incl. BN running stats (DL4J convention)."""
+        return sum(p.numel() for p in self.parameters()) + sum(
+            b.numel() for b in self.buffers()
+        )
+
+    def params_flat(self) -> torch.Tensor:
+        """Flattened fp32 param vector in layer order (DL4J coefficients.bin
+        ordering: per layer, W then b then gamma/beta/mean/var)."""
+        chunks = []
+        for name in self._topo:
+            layer = self.layers[name]
+            for key in ("W", "b", "gamma", "beta", "mean", "var"):
+                if key in layer.param_keys():
+                    chunks.append(layer.get_param(key).detach().float().reshape(-1).cpu())
+        if not chunks:
+            return torch.empty(0)
+        return torch.cat(chunks)
+
+    def load_params_flat(self, vec: torch.Tensor) -> None:
+        off = 0
+        for name in self._topo:
+            layer = self.layers[name]
+            for key in ("W", "b", "gamma", "beta", "mean", "var"):
+                if key in layer.param_keys():
+                    t = layer.get_param(key)
+                    n = t.numel()
+                    layer.set_param(key, vec[off : off + n].reshape(t.shape))
+                    off += n
+        if off != vec.numel():
+            raise ValueError(f"param vector size mismatch: used {off}, got {vec.numel()}")
+
+    # ---------------------------------------------------------- summary
+    def summary(self, batch: int = 1) -> str:
+        """Shape/param table (DL4J ComputationGraph.summary())."""
+        lines = [f"{'name':<28}{'type':<26}{'out shape':<22}{'params':>10}"]
+        lines.append("-" * 86)
+        shapes: dict[str, tuple] = {}
+        for name in self.input_names:
+            t = self.input_types.get(name)
+            if t is not None:
+                s = t.shape(batch)
+                if t.kind == "convolutional_flat":
+                    s = (batch, t.channels, t.height, t.width)
+                shapes[name] = s
+            else:
+                shapes[name] = (batch, -1)
+        total = 0
+        for name in self._topo:
+            layer = self.layers[name]
+            src = self._vertex_inputs[name][0]
+            in_shape = shapes.get(src, (batch, -1))
+            if name in self.preprocessors:
+                in_shape = self.preprocessors[name].out_shape(in_shape)
+            out_shape = layer.out_shape(in_shape)
+            shapes[name] = out_shape
+            n = layer.n_params()
+            total += n
+            lines.append(
+                f"{name:<28}{type(layer).__name__:<26}{str(out_shape):<22}{n:>10}"
+            )
+        lines.append("-" * 86)
+        lines.append(f"total params: {total}")
+        return "\n".join(lines)
+
+    # ------------------------------------------------------------ misc
+    def clone(self) -> "ComputationGraph":
+        import copy
+
+        g = copy.deepcopy(self)
+        g._updater = None
+        return g
+
+    def to_device(self, device, dtype=None) -> "ComputationGraph":
+        self.to(device)
+        if dtype is not None:
+            for layer in self.layers.values():
+                for p in layer.parameters(recurse=False):
+                    # BN affine/stats stay fp32 (SURVEY hard-part #3)
+                    from .layers import BatchNormLayer
+
+                    if not isinstance(layer, BatchNormLayer):
+                        p.data = p.data.to(dtype)
+        return self

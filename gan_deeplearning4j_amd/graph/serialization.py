@@ -1,0 +1,239 @@
+"""Checkpointing: DL4J ModelSerializer-compatible .zip + native format.
+
+The reference writes all four models each iteration via
+ModelSerializer.writeModel(graph, file, saveUpdater=true) (Java:605-618).
+The zip layout:
+    configuration.json  - graph topology + training configuration
+    coefficients.bin    - flattened fp32 parameter vector in layer order
+                          (per layer: W, b, gamma, beta, mean, var)
+    updaterState.bin    - flattened fp32 updater state (saveUpdater=true)
+
+The reference itself never calls restore (save-only); this framework
+implements BOTH save and resume (SURVEY.md §5 checkpoint/resume).
+"""
+
+from __future__ import annotations
+
+import io
+import json
+import zipfile
+from pathlib import Path
+from typing import Optional
+
+import numpy as np
+import torch
+
+from . import layers as L
+from .builder import ComputationGraph, GraphBuilder, InputType
+
+_LAYER_FIELDS = {
+    "DenseLayer": ("n_in", "n_out", "activation", "lr", "frozen", "slope"),
+    "OutputLayer": ("n_in", "n_out", "inference_activation", "loss", "lr", "frozen"),
+    "Conv2dLayer": ("c_in", "c_out", "kernel", "stride", "padding", "activation",
+                    "lr", "frozen", "slope"),
+    "ConvTranspose2dLayer": ("c_in", "c_out", "kernel", "stride", "padding",
+                             "activation", "lr", "frozen", "slope"),
+    "BatchNormLayer": ("num_features", "eps", "momentum", "lr", "frozen"),
+    "MaxPool2dLayer": ("kernel", "stride"),
+    "Upsampling2dLayer": ("scale",),
+    "ActivationLayer": ("activation", "slope"),
+    "FeedForwardToCnnPreProcessor": ("height", "width", "channels"),
+    "CnnToFeedForwardPreProcessor": (),
+    "ReshapeVertex": ("shape",),
+}
+
+_CTOR_ARGS = {
+    "DenseLayer": lambda d: L.DenseLayer(d["n_in"], d["n_out"], d["activation"],
+                                         d.get("lr"), d.get("frozen", False),
+                                         slope=d.get("slope", 0.2)),
+    "OutputLayer": lambda d: L.OutputLayer(d["n_in"], d["n_out"],
+                                           d["inference_activation"], d["loss"],
+                                           d.get("lr"), d.get("frozen", False)),
+    "Conv2dLayer": lambda d: L.Conv2dLayer(d["c_in"], d["c_out"], d["kernel"],
+                                           d["stride"], d["padding"],
+                                           d["activation"], d.get("lr"),
+                                           d.get("frozen", False),
+                                           slope=d.get("slope", 0.2)),
+    "ConvTranspose2dLayer": lambda d: L.ConvTranspose2dLayer(
+        d["c_in"], d["c_out"], d["kernel"], d["stride"], d["padding"],
+        d["activation"], d.get("lr"), d.get("frozen", False),
+        slope=d.get("slope", 0.2)),
+    "BatchNormLayer": lambda d: L.BatchNormLayer(d["num_features"], d["eps"],
+                                                 d["momentum"], d.get("lr"),
+                                                 d.get("frozen", False)),
+    "MaxPool2dLayer": lambda d: L.MaxPool2dLayer(d["kernel"], d["stride"]),
+    "Upsampling2dLayer": lambda d: L.Upsampling2dLayer(d["scale"]),
+    "ActivationLayer": lambda d: L.ActivationLayer(d["activation"],
+                                                   d.get("slope", 0.2)),
+    "FeedForwardToCnnPreProcessor": lambda d: L.FeedForwardToCnnPreProcessor(
+        d["height"], d["width"], d["channels"]),
+    "CnnToFeedForwardPreProcessor": lambda d: L.CnnToFeedForwardPreProcessor(),
+    "ReshapeVertex": lambda d: L.ReshapeVertex(*d["shape"]),
+}
+
+
+def _layer_to_dict(layer: L.BaseLayer) -> dict:
+    tname = type(layer).__name__
+    if tname not in _LAYER_FIELDS:
+        raise TypeError(f"cannot serialize layer type {tname}")
+    d = {"type": tname}
+    for f in _LAYER_FIELDS[tname]:
+        v = getattr(layer, f)
+        d[f] = list(v) if isinstance(v, tuple) else v
+    return d
+
+
+def _layer_from_dict(d: dict) -> L.BaseLayer:
+    return _CTOR_ARGS[d["type"]](d)
+
+
+def graph_config_dict(graph: ComputationGraph) -> dict:
+    import dataclasses
+
+    return {
+        "format": "gan_deeplearning4j_amd/ComputationGraph",
+        "version": 1,
+        "seed": graph.seed,
+        "optim": dataclasses.asdict(graph.optim_cfg),
+        "inputs": graph.input_names,
+        "input_types": {
+            k: dataclasses.asdict(v) for k, v in graph.input_types.items()
+        },
+        "outputs": graph.output_names,
+        "vertices": [
+            {
+                "name": name,
+                "inputs": graph._vertex_inputs[name],
+                "layer": _layer_to_dict(graph.layers[name]),
+                "preprocessor": (
+                    _layer_to_dict(graph.preprocessors[name])
+                    if name in graph.preprocessors
+                    else None
+                ),
+            }
+            for name in graph.layer_names()
+        ],
+    }
+
+
+def graph_from_config_dict(conf: dict) -> ComputationGraph:
+    from ..config import OptimConfig
+
+    optim = OptimConfig(**conf["optim"])
+    gb = GraphBuilder(seed=conf["seed"], optim_cfg=optim)
+    gb.add_inputs(*conf["inputs"])
+    types = [InputType(**conf["input_types"][n]) for n in conf["inputs"]
+             if n in conf["input_types"]]
+    gb.set_input_types(*types)
+    for v in conf["vertices"]:
+        proc = _layer_from_dict(v["preprocessor"]) if v["preprocessor"] else None
+        gb.add_layer(v["name"], _layer_from_dict(v["layer"]), *v["inputs"],
+                     preprocessor=proc)
+    gb.set_outputs(*conf["outputs"])
+    return gb.build()
+
+
+class ModelSerializer:
+    """DL4J ModelSerializer analog (writeModel / restoreComputationGraph)."""
+
+    @staticmethod
+    def write_model(graph: ComputationGraph, path: str | Path,
+                    save_updater: bool = True) -> Path:
+        path = Path(path)
+        path.parent.mkdir(parents=True, exist_ok=True)
+        params = graph.params_flat().numpy().astype("<f4")
+        with zipfile.ZipFile(path, "w", zipfile.ZIP_DEFLATED) as zf:
+            zf.writestr(
+                "configuration.json",
+                json.dumps(graph_config_dict(graph), indent=1),
+            )
+            zf.writestr("coefficients.bin", params.tobytes())
+            if save_updater and graph._updater is not None:
+                sd = graph.updater.state_dict()
+                chunks = []
+                for slot in sd["slots"]:
+                    for k in ("v", "m", "master"):
+                        if slot[k] is not None:
+                            chunks.append(slot[k].reshape(-1).numpy().astype("<f4"))
+                flat = (
+                    np.concatenate(chunks) if chunks else np.empty(0, dtype="<f4")
+                )
+                zf.writestr("updaterState.bin", flat.tobytes())
+                meta = {
+                    "kind": sd["kind"],
+                    "t": sd["t"],
+                    "slots": [
+                        {
+                            "lr": s["lr"],
+                            "has_v": s["v"] is not None,
+                            "has_m": s["m"] is not None,
+                            "has_master": s["master"] is not None,
+                            "numel": int(s["v"].numel()) if s["v"] is not None else 0,
+                        }
+                        for s in sd["slots"]
+                    ],
+                }
+                zf.writestr("updater.json", json.dumps(meta))
+        return path
+
+    @staticmethod
+    def restore_computation_graph(path: str | Path,
+                                  load_updater: bool = True) -> ComputationGraph:
+        path = Path(path)
+        with zipfile.ZipFile(path) as zf:
+            conf = json.loads(zf.read("configuration.json"))
+            graph = graph_from_config_dict(conf)
+            vec = torch.from_numpy(
+                np.frombuffer(zf.read("coefficients.bin"), dtype="<f4").copy()
+            )
+            graph.load_params_flat(vec)
+            graph._initialized = True
+            if load_updater and "updater.json" in zf.namelist():
+                meta = json.loads(zf.read("updater.json"))
+                flat = np.frombuffer(
+                    zf.read("updaterState.bin"), dtype="<f4"
+                ).copy()
+                upd = graph.updater
+                upd.t = meta["t"]
+                off = 0
+                for slot, sm in zip(upd.slots, meta["slots"]):
+                    slot.lr = sm["lr"]
+                    n = slot.param.numel()
+                    shape = slot.param.shape
+                    if sm["has_v"]:
+                        slot.v = torch.from_numpy(flat[off:off + n]).reshape(shape).clone()
+                        off += n
+                    if sm["has_m"]:
+                        slot.m = torch.from_numpy(flat[off:off + n]).reshape(shape).clone()
+                        off += n
+                    if sm["has_master"]:
+                        slot.master = torch.from_numpy(flat[off:off + n]).reshape(shape).clone()
+                        off += n
+        return graph
+
+    # ------------------------------------------------ native fast format
+    @staticmethod
+    def save_native(graph: ComputationGraph, path: str | Path) -> Path:
+        path = Path(path)
+        path.parent.mkdir(parents=True, exist_ok=True)
+        torch.save(
+            {
+                "config": graph_config_dict(graph),
+                "state_dict": graph.state_dict(),
+                "updater": (
+                    graph.updater.state_dict() if graph._updater else None
+                ),
+            },
+            path,
+        )
+        return path
+
+    @staticmethod
+    def load_native(path: str | Path) -> ComputationGraph:
+        blob = torch.load(path, map_location="cpu", weights_only=False)
+        graph = graph_from_config_dict(blob["config"])
+        graph.load_state_dict(blob["state_dict"])
+        graph._initialized = True
+        if blob.get("updater") is not None:
+            graph.updater.load_state_dict(blob["updater"])
+        return graph

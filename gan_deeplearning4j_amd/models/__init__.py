@@ -1,0 +1,25 @@
+from .reference_protocol import (
+    build_discriminator,
+    build_frozen_generator,
+    build_stacked_gan,
+    build_transfer_classifier,
+    DIS_SYNC_KEYS,
+    GAN_TO_GEN_SYNC,
+    DIS_TO_GAN_SYNC,
+    DIS_TO_CV_SYNC,
+)
+from .dcgan import build_dcgan
+from .mlp_gan import build_mlp_gan
+
+__all__ = [
+    "build_discriminator",
+    "build_frozen_generator",
+    "build_stacked_gan",
+    "build_transfer_classifier",
+    "build_dcgan",
+    "build_mlp_gan",
+    "DIS_SYNC_KEYS",
+    "GAN_TO_GEN_SYNC",
+    "DIS_TO_GAN_SYNC",
+    "DIS_TO_CV_SYNC",
+]

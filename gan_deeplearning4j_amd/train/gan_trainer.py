@@ -1,0 +1,305 @@
+"""GAN trainers.
+
+Two protocols over the same op library:
+
+- `GanTrainer` — the idiomatic MI355X fast path used by the flagship
+  bench: shared G/D graphs, proper stop-gradient freezing during the
+  G-step (no frozen twin graphs, no weight-copy passes), fused updaters,
+  bucketed RCCL gradient all-reduce overlapped with backward.
+
+- `ReferenceProtocolTrainer` — the reference's exact alternating protocol
+  (Java:408-621): three graphs (D, frozen gen, stacked gan), one-shot
+  label softening, manual per-tensor weight sync between twins,
+  transfer-learned classifier trained alongside, CSV artifact dumps and
+  4x DL4J-zip checkpoints per iteration, optional parameter-averaging
+  distributed semantics (averaging_frequency).
+"""
+
+from __future__ import annotations
+
+import logging
+from pathlib import Path
+from typing import Optional
+
+import numpy as np
+import torch
+
+from ..config import GanConfig
+from ..data.csv_reader import DataSet
+from ..graph.serialization import ModelSerializer
+from ..models import (
+    DIS_TO_CV_SYNC,
+    DIS_TO_GAN_SYNC,
+    GAN_TO_GEN_SYNC,
+    build_discriminator,
+    build_frozen_generator,
+    build_stacked_gan,
+    build_transfer_classifier,
+)
+from ..models.reference_protocol import sync_params
+from ..ops import functional as OF
+from ..parallel.ddp import GradReducer, average_parameters, broadcast_parameters
+from .metrics import MetricsLogger
+
+log = logging.getLogger("gan_deeplearning4j_amd")
+
+
+def latent_grid(n: int, z_size: int, device="cpu", dtype=torch.float32):
+    """n x n grid over linspace(-1,1)^2 (reference Java:382-389).
+    For z_size > 2 the remaining dims are zero."""
+    lin = torch.linspace(-1, 1, n)
+    zs = torch.zeros(n * n, z_size)
+    for i in range(n):
+        for j in range(n):
+            zs[i * n + j, 0] = lin[i]
+            zs[i * n + j, min(1, z_size - 1)] = lin[j]
+    return zs.to(device=device, dtype=dtype)
+
+
+class GanTrainer:
+    """Idiomatic alternating GAN trainer (flagship/bench path)."""
+
+    def __init__(
+        self,
+        gen,
+        dis,
+        cfg: GanConfig,
+        device: Optional[torch.device] = None,
+        dtype: Optional[torch.dtype] = None,
+        bucket_cap_mb: Optional[int] = None,
+    ):
+        self.cfg = cfg
+        self.device = device or torch.device(
+            "cuda" if torch.cuda.is_available() and cfg.train.use_gpu else "cpu"
+        )
+        if dtype is None:
+            dtype = (
+                torch.bfloat16
+                if self.device.type == "cuda" and cfg.model.dtype in ("bf16", "fp8")
+                else torch.float32
+            )
+        self.dtype = dtype
+        self.gen = gen.to_device(self.device, dtype)
+        self.dis = dis.to_device(self.device, dtype)
+        broadcast_parameters(self.gen)
+        broadcast_parameters(self.dis)
+        cap = bucket_cap_mb or cfg.parallel.bucket_cap_mb
+        self.g_reducer = GradReducer([self.gen], cap, cfg.parallel.local_steps)
+        self.d_reducer = GradReducer([self.dis], cap, cfg.parallel.local_steps)
+        self.z_size = cfg.model.z_size
+        from ..parallel.launch import get_rank
+
+        self._g = torch.Generator(device="cpu").manual_seed(
+            cfg.train.seed + 1000 * get_rank()
+        )
+        # one-sided label softening, drawn once (reference Java:405-406)
+        std = cfg.train.label_noise_std
+        self._soft_real = None
+        self._soft_fake = None
+        self._noise_std = std
+        self.it = 0
+
+    def _labels(self, n: int):
+        if self._soft_real is None or self._soft_real.shape[0] != n:
+            std = self._noise_std
+            self._soft_real = (
+                1.0 + std * torch.randn(n, 1, generator=self._g)
+            ).to(self.device)
+            self._soft_fake = (std * torch.randn(n, 1, generator=self._g)).to(
+                self.device
+            )
+        return self._soft_real, self._soft_fake
+
+    def sample_z(self, n: int) -> torch.Tensor:
+        z = torch.randn(n, self.z_size, generator=self._g)
+        return z.to(self.device, self.dtype)
+
+    def step(self, real: torch.Tensor) -> dict:
+        """One alternating D+G update on a batch of real images."""
+        self.it += 1
+        n = real.shape[0]
+        real = real.to(self.device, self.dtype)
+        soft_real, soft_fake = self._labels(n)
+
+        z = self.sample_z(n)
+        self.gen.train()
+        self.dis.train()
+        fake = self.gen(z)
+
+        # ---- D step -------------------------------------------------
+        self.dis.updater.zero_grad()
+        self.d_reducer.prepare()
+        d_real = self.dis(real)
+        d_fake = self.dis(fake.detach())
+        loss_d = OF.bce_with_logits_loss(d_real, soft_real) + \
+            OF.bce_with_logits_loss(d_fake, soft_fake)
+        loss_d.backward()
+        self.d_reducer.finish()
+        self.dis.updater.step()
+
+        # ---- G step (stop-gradient freeze of D: no D wgrad compute) --
+        for p in self.dis.parameters():
+            p.requires_grad_(False)
+        self.gen.updater.zero_grad()
+        self.g_reducer.prepare()
+        g_logits = self.dis(fake)
+        loss_g = OF.bce_with_logits_loss(
+            g_logits, torch.ones(n, 1, device=self.device)
+        )
+        loss_g.backward()
+        self.g_reducer.finish()
+        self.gen.updater.step()
+        for p in self.dis.parameters():
+            p.requires_grad_(True)
+
+        return {
+            "loss_d": float(loss_d.detach()),
+            "loss_g": float(loss_g.detach()),
+            "images": n,
+        }
+
+    @torch.no_grad()
+    def sample_grid(self, n: int = 10) -> torch.Tensor:
+        z = latent_grid(n, self.z_size, self.device, self.dtype)
+        return self.gen.output(z).float().cpu()
+
+
+class ReferenceProtocolTrainer:
+    """The reference's exact 6-phase alternating loop (Java:408-621)."""
+
+    def __init__(self, cfg: GanConfig, device: Optional[torch.device] = None,
+                 out_dir: Optional[str] = None):
+        self.cfg = cfg
+        self.device = device or torch.device(
+            "cuda" if torch.cuda.is_available() and cfg.train.use_gpu else "cpu"
+        )
+        self.out_dir = Path(out_dir or cfg.train.out_dir)
+        self.out_dir.mkdir(parents=True, exist_ok=True)
+
+        # graphs (Java:118-314, 337-368)
+        self.dis = build_discriminator(cfg).to_device(self.device)
+        self.gen = build_frozen_generator(cfg).to_device(self.device)
+        self.gan = build_stacked_gan(cfg).to_device(self.device)
+        self.cv = build_transfer_classifier(self.dis, cfg).to_device(self.device)
+        for g in (self.dis, self.gen, self.gan, self.cv):
+            broadcast_parameters(g)
+
+        self.metrics = MetricsLogger(str(self.out_dir),
+                                     cfg.train.print_every)
+        torch.manual_seed(cfg.train.seed)
+        self._cpu_gen = torch.Generator().manual_seed(cfg.train.seed)
+        # label softening drawn ONCE before the loop (Java:405-406)
+        b = cfg.data.batch_size_per_worker
+        std = cfg.train.label_noise_std
+        self.soft_fake = (std * torch.randn(b, 1, generator=self._cpu_gen))
+        self.soft_real = (1 + std * torch.randn(b, 1, generator=self._cpu_gen))
+        self.batch_counter = 0
+
+    # ------------------------------------------------------------------
+    def _uniform_z(self, n: int) -> torch.Tensor:
+        # Nd4j.rand.muli(2).subi(1): U(-1,1) (Java:420, 465)
+        return (torch.rand(n, self.cfg.model.z_size,
+                           generator=self._cpu_gen) * 2 - 1)
+
+    def train_iteration(self, real: DataSet) -> dict:
+        """One full reference iteration on a real labeled batch."""
+        cfg = self.cfg
+        dev = self.device
+        n = real.num_examples()
+        feats = real.features.to(dev)
+
+        # (a) D-step on {real -> soft 1, G(z) -> soft 0} (Java:408-426)
+        z = self._uniform_z(n).to(dev)
+        fake_imgs = self.gen.output(z)
+        fake_flat = fake_imgs.reshape(n, -1)
+        d_data = [
+            DataSet(feats, self.soft_real[:n].to(dev)),
+            DataSet(fake_flat, self.soft_fake[:n].to(dev)),
+        ]
+        loss_d = self.dis.fit(d_data)
+
+        # (b) copy fresh D into gan's frozen D (Java:429-460)
+        sync_params(self.dis, self.gan, DIS_TO_GAN_SYNC)
+
+        # (c) G-step through the stacked gan: {z~U(-1,1) -> 1} (Java:462-471)
+        z2 = self._uniform_z(n).to(dev)
+        loss_g = self.gan.fit(DataSet(z2, torch.ones(n, 1, device=dev)))
+
+        # (d) copy trained G from gan into frozen gen (Java:474-510)
+        sync_params(self.gan, self.gen, GAN_TO_GEN_SYNC)
+
+        # (e) copy D backbone into classifier (Java:516-542)
+        sync_params(self.dis, self.cv, DIS_TO_CV_SYNC)
+
+        # (f) classifier step on the real labeled batch (Java:544-545)
+        loss_cv = (
+            self.cv.fit(DataSet(feats, real.labels.to(dev)))
+            if cfg.train.train_classifier
+            else 0.0
+        )
+
+        self.batch_counter += 1
+        # parameter averaging every averaging_frequency batches (Java:326)
+        if cfg.train.averaging_frequency > 0 and \
+                self.batch_counter % cfg.train.averaging_frequency == 0:
+            for g in (self.dis, self.gan, self.cv):
+                average_parameters(g)
+            sync_params(self.gan, self.gen, GAN_TO_GEN_SYNC)
+
+        out = {"loss_d": loss_d, "loss_g": loss_g, "loss_cv": loss_cv,
+               "images": n}
+        self.metrics.step(self.batch_counter, **out)
+        return out
+
+    # ------------------------------------------------- artifacts (L5)
+    @torch.no_grad()
+    def dump_sample_grid(self, idx: int) -> Path:
+        """10x10 latent-manifold grid -> mnist_out_<i>.csv (Java:550-570).
+
+        The reference had a writer bug (fileWriter.close() inside the row
+        loop, Java:568-569) — intent replicated, bug not."""
+        n = self.cfg.train.num_gen_samples
+        z = latent_grid(n, self.cfg.model.z_size, self.device)
+        imgs = self.gen.output(z).float().cpu().reshape(n * n, -1)
+        path = self.out_dir / f"mnist_out_{idx}.csv"
+        np.savetxt(path, imgs.numpy(), delimiter=",", fmt="%.6f")
+        return path
+
+    @torch.no_grad()
+    def dump_test_predictions(self, idx: int, test_iter) -> Path:
+        """Classifier softmax on the test fold -> CSV (Java:572-598)."""
+        rows = []
+        for ds in test_iter:
+            probs = self.cv.output(ds.features.to(self.device))
+            rows.append(probs.float().cpu().numpy())
+        path = self.out_dir / f"mnist_test_predictions_{idx}.csv"
+        np.savetxt(path, np.concatenate(rows), delimiter=",", fmt="%.6f")
+        return path
+
+    def save_checkpoints(self, save_updater: bool = True) -> list[Path]:
+        """4x DL4J-zip checkpoints (Java:605-618)."""
+        ckpts = []
+        for name, g in (("dis", self.dis), ("gan", self.gan),
+                        ("gen", self.gen), ("CV", self.cv)):
+            p = self.out_dir / f"mnist_{name}_model.zip"
+            ModelSerializer.write_model(g, p, save_updater=save_updater)
+            ckpts.append(p)
+        return ckpts
+
+    # ---------------------------------------------------------- driver
+    def run(self, train_iter, test_iter=None) -> dict:
+        """The while-loop (Java:408): numIterations batches."""
+        cfg = self.cfg
+        last = {}
+        for ds in train_iter:
+            if self.batch_counter >= cfg.train.num_iterations:
+                break
+            last = self.train_iteration(ds)
+            i = self.batch_counter
+            if i % cfg.train.print_every == 0:
+                self.dump_sample_grid(i)
+                if test_iter is not None and cfg.train.train_classifier:
+                    self.dump_test_predictions(i, test_iter)
+            if i % cfg.train.save_every == 0:
+                self.save_checkpoints()
+        return last

@@ -1,0 +1,50 @@
+import numpy as np
+import torch
+
+from gan_deeplearning4j_amd.data import (
+    CSVRecordReader,
+    RecordReaderDataSetIterator,
+    pixel_lattice_images,
+    transactions_tabular,
+    write_synthetic_csv,
+)
+
+
+def test_pixel_lattice_shapes():
+    x, y = pixel_lattice_images(32, 28, 28, 1, 10, seed=1)
+    assert x.shape == (32, 1, 28, 28)
+    assert x.min() >= 0 and x.max() <= 1
+    assert y.shape == (32,) and y.max() < 10
+
+
+def test_transactions_shapes():
+    x, y = transactions_tabular(64, 32, 2, seed=1)
+    assert x.shape == (64, 32)
+    assert x.min() >= 0 and x.max() <= 1.0001
+
+
+def test_csv_roundtrip(tmp_path):
+    # CSV format matches the reference notebook writer (cell 2):
+    # 784 floats + integer label per row
+    p = write_synthetic_csv(tmp_path / "train.csv", "pixel_lattice", n=50,
+                            height=28, width=28, channels=1)
+    reader = CSVRecordReader().initialize(p)
+    assert reader.records.shape == (50, 785)
+    it = RecordReaderDataSetIterator(reader, batch_size=20, label_index=784,
+                                     num_classes=10)
+    batches = list(it)
+    assert len(batches) == 3
+    assert batches[0].features.shape == (20, 784)
+    assert batches[0].labels.shape == (20, 10)
+    # one-hot labels
+    assert torch.all(batches[0].labels.sum(dim=1) == 1)
+    assert batches[2].features.shape == (10, 784)
+
+
+def test_iterator_shuffle_determinism(tmp_path):
+    p = write_synthetic_csv(tmp_path / "t.csv", "transactions", n=40,
+                            num_features=8)
+    r = CSVRecordReader().initialize(p)
+    a = list(RecordReaderDataSetIterator(r, 16, 8, 2, shuffle=True, seed=3))
+    b = list(RecordReaderDataSetIterator(r, 16, 8, 2, shuffle=True, seed=3))
+    assert torch.equal(a[0].features, b[0].features)

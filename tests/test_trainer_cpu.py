@@ -1,0 +1,110 @@
+import numpy as np
+import torch
+
+from gan_deeplearning4j_amd.config import GanConfig, preset
+from gan_deeplearning4j_amd.data import (
+    CSVRecordReader,
+    RecordReaderDataSetIterator,
+    write_synthetic_csv,
+)
+from gan_deeplearning4j_amd.models import build_dcgan, build_mlp_gan
+from gan_deeplearning4j_amd.train import GanTrainer, ReferenceProtocolTrainer
+from gan_deeplearning4j_amd.train.gan_trainer import latent_grid
+
+
+def small_cfg():
+    cfg = GanConfig()
+    cfg.train.use_gpu = False
+    cfg.data.batch_size_per_worker = 16
+    cfg.data.batch_size_pred = 32
+    cfg.train.num_iterations = 2
+    return cfg
+
+
+def test_latent_grid():
+    z = latent_grid(10, 2)
+    assert z.shape == (100, 2)
+    assert z.min() == -1 and z.max() == 1
+
+
+def test_fast_trainer_mlp():
+    cfg = preset("mlp_tabular_cpu")
+    cfg.train.use_gpu = False
+    gen, dis = build_mlp_gan(cfg, hidden=32)
+    tr = GanTrainer(gen, dis, cfg, device=torch.device("cpu"))
+    x = torch.rand(8, cfg.data.num_features)
+    out1 = tr.step(x)
+    out2 = tr.step(x)
+    assert np.isfinite(out1["loss_d"]) and np.isfinite(out2["loss_g"])
+    grid = tr.sample_grid(4)
+    assert grid.shape[0] == 16
+
+
+def test_fast_trainer_dcgan28_step():
+    cfg = preset("dcgan28")
+    cfg.train.use_gpu = False
+    cfg.model.base_width = 8  # small for CPU test speed
+    gen, dis = build_dcgan(cfg)
+    tr = GanTrainer(gen, dis, cfg, device=torch.device("cpu"))
+    real = torch.rand(4, 1, 28, 28) * 2 - 1
+    out = tr.step(real)
+    assert np.isfinite(out["loss_d"])
+    # G step must not have updated D
+    # (D was stop-grad frozen during G step; its updater ran only once)
+    assert tr.dis.updater.t == 1
+    assert tr.gen.updater.t == 1
+
+
+def test_reference_protocol_two_iterations(tmp_path):
+    """Fixed-seed 2-iteration end-to-end replica of the reference run
+    (numIterations=2, Java:72), asserting artifact shapes."""
+    cfg = small_cfg()
+    train_csv = write_synthetic_csv(tmp_path / "train.csv", "pixel_lattice",
+                                    n=64, height=28, width=28, channels=1)
+    test_csv = write_synthetic_csv(tmp_path / "test.csv", "pixel_lattice",
+                                   n=32, height=28, width=28, channels=1)
+    train_it = RecordReaderDataSetIterator(
+        CSVRecordReader().initialize(train_csv),
+        cfg.data.batch_size_per_worker, cfg.data.label_index,
+        cfg.data.num_classes)
+    test_it = RecordReaderDataSetIterator(
+        CSVRecordReader().initialize(test_csv),
+        cfg.data.batch_size_pred, cfg.data.label_index, cfg.data.num_classes)
+
+    tr = ReferenceProtocolTrainer(cfg, device=torch.device("cpu"),
+                                  out_dir=str(tmp_path / "out"))
+    last = tr.run(train_it, test_it)
+    assert tr.batch_counter == 2
+    assert np.isfinite(last["loss_d"]) and np.isfinite(last["loss_g"])
+
+    # artifacts: 10x10 grid of 784-float rows (mnist_out_<i>.csv)
+    grid = np.loadtxt(tmp_path / "out" / "mnist_out_2.csv", delimiter=",")
+    assert grid.shape == (100, 784)
+    preds = np.loadtxt(tmp_path / "out" / "mnist_test_predictions_2.csv",
+                       delimiter=",")
+    assert preds.shape == (32, 10)
+    assert np.allclose(preds.sum(axis=1), 1.0, atol=1e-4)
+    # 4 model zips
+    for name in ("dis", "gan", "gen", "CV"):
+        assert (tmp_path / "out" / f"mnist_{name}_model.zip").exists()
+
+
+def test_reference_weight_sync_semantics(tmp_path):
+    """After an iteration, gan's frozen D == dis, gen == gan's G."""
+    cfg = small_cfg()
+    cfg.train.train_classifier = False
+    p = write_synthetic_csv(tmp_path / "t.csv", "pixel_lattice", n=32,
+                            height=28, width=28, channels=1)
+    it = RecordReaderDataSetIterator(CSVRecordReader().initialize(p),
+                                     16, 784, 10)
+    tr = ReferenceProtocolTrainer(cfg, device=torch.device("cpu"),
+                                  out_dir=str(tmp_path / "o"))
+    tr.train_iteration(next(iter(it)))
+    assert torch.equal(
+        tr.dis.get_layer("dis_conv2d_layer_2").get_param("W"),
+        tr.gan.get_layer("gan_dis_conv2d_layer_10").get_param("W"),
+    )
+    assert torch.equal(
+        tr.gen.get_layer("gen_dense_layer_2").get_param("W"),
+        tr.gan.get_layer("gan_dense_layer_2").get_param("W"),
+    )
