@@ -1,0 +1,122 @@
+"""Flagship benchmark: DCGAN 64x64 bf16 images/sec (whole node).
+
+Driver contract:
+    python bench.py --gpus N --steps K --warmup W
+launched for N>1 as one rank per GPU via torch.distributed.run; reads
+RANK/LOCAL_RANK/WORLD_SIZE/MASTER_* from env. One "step" = one full
+alternating GAN iteration (D update on real+fake and G update) on the
+per-rank batch; value = whole-job real images consumed per second
+(global_batch * steps / elapsed), weak scaling.
+
+Prints exactly one JSON line from rank 0.
+"""
+
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import sys
+import time
+
+import torch
+
+sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+
+from gan_deeplearning4j_amd.config import preset  # noqa: E402
+from gan_deeplearning4j_amd.models import build_dcgan, build_mlp_gan  # noqa: E402
+from gan_deeplearning4j_amd.parallel.launch import init_distributed  # noqa: E402
+from gan_deeplearning4j_amd.train import GanTrainer  # noqa: E402
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=20)
+    ap.add_argument("--warmup", type=int, default=5)
+    ap.add_argument("--batch", type=int, default=0,
+                    help="per-GPU batch (0 = arch default)")
+    ap.add_argument("--arch", type=str, default="dcgan64")
+    args = ap.parse_args()
+
+    rank, world, local_rank, device = init_distributed()
+    use_gpu = device.type == "cuda"
+
+    cfg = preset(args.arch) if args.arch != "mlp" else preset("mlp_tabular_cpu")
+    per_gpu_batch = args.batch
+    if per_gpu_batch == 0:
+        per_gpu_batch = {"dcgan28": 2048, "dcgan64": 1024,
+                         "dcgan128": 256}.get(args.arch, 512)
+        if not use_gpu:
+            per_gpu_batch = 16
+    cfg.train.use_gpu = use_gpu
+
+    if args.arch == "mlp":
+        gen, dis = build_mlp_gan(cfg)
+    else:
+        gen, dis = build_dcgan(cfg)
+    dtype = torch.bfloat16 if use_gpu else torch.float32
+    tr = GanTrainer(gen, dis, cfg, device=device, dtype=dtype)
+
+    m = cfg.model
+    # synthetic pixel-lattice-shaped data, random-init weights (no-network
+    # environment; BASELINE.json data contract)
+    g = torch.Generator().manual_seed(1234 + rank)
+    real = (torch.rand(per_gpu_batch, m.image_channels, m.image_height,
+                       m.image_width, generator=g) * 2 - 1).to(device, dtype)
+    if args.arch == "mlp":
+        real = torch.rand(per_gpu_batch, cfg.data.num_features,
+                          generator=g).to(device, dtype)
+
+    def barrier_sync():
+        if torch.distributed.is_initialized():
+            torch.distributed.barrier()
+        if use_gpu:
+            torch.cuda.synchronize(device)
+
+    for _ in range(args.warmup):
+        tr.step(real)
+
+    barrier_sync()
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        tr.step(real)
+    barrier_sync()
+    elapsed = time.perf_counter() - t0
+
+    # MAX over ranks
+    if torch.distributed.is_initialized():
+        t = torch.tensor([elapsed], dtype=torch.float64,
+                         device=device if use_gpu else "cpu")
+        torch.distributed.all_reduce(t, op=torch.distributed.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    n_gpus = world if use_gpu else 1
+    global_batch = per_gpu_batch * world
+    images_per_sec = global_batch * args.steps / elapsed
+    if rank == 0:
+        print(json.dumps({
+            "metric": "images/sec (whole node) DCGAN 64x64 bf16",
+            "value": round(images_per_sec, 2),
+            "unit": "images/s",
+            "n_gpus": n_gpus,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(elapsed / args.steps * 1e3, 3),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "bf16" if use_gpu else "fp32",
+            "data": "synthetic",
+            "config": {
+                "model": args.arch,
+                "global_batch": global_batch,
+                "image": [m.image_channels, m.image_height, m.image_width],
+                "z_size": m.z_size,
+                "parallelism": f"dp{world}",
+            },
+        }))
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,430 @@
+"""GPU op layer: custom autograd Functions over the gfx950 HIP kernels.
+
+Layout strategy (MI355X-first): convolution activations live in NHWC
+physical layout (channels innermost) — the natural layout for MFMA
+implicit-GEMM convs. Tensors stay logically NCHW; outputs are returned as
+channels-last VIEWS of our NHWC buffers, so chains of conv/BN/pool ops
+pass layout through with zero copies (only graph edges that leave the conv
+world pay one permute-copy).
+
+Contraction (K) dims are zero-padded to multiples of 64 to match the MFMA
+GEMM tile (TN_BK); im2col emits padded K directly, dense inputs/weights are
+padded with one small copy.
+
+Every Function here raises if the _C extension is missing — on a GPU box
+the HIP path is the only path (no silent eager fallback).
+"""
+
+from __future__ import annotations
+
+from typing import Optional
+
+import torch
+
+from .backend import hip_ext
+
+ACT_CODES = {"identity": 0, None: 0, "tanh": 1, "sigmoid": 2, "lrelu": 3,
+             "relu": 4}
+
+
+def _rup64(k: int) -> int:
+    return (k + 63) // 64 * 64
+
+
+def _bf(t: torch.Tensor) -> torch.Tensor:
+    return t.to(torch.bfloat16).contiguous()
+
+
+def _pad_k(t: torch.Tensor) -> torch.Tensor:
+    """Zero-pad the last dim of a 2D tensor to a multiple of 64."""
+    k = t.shape[-1]
+    kp = _rup64(k)
+    if kp == k:
+        return t.contiguous()
+    out = t.new_zeros(*t.shape[:-1], kp)
+    out[..., :k] = t
+    return out
+
+
+def _nhwc(x: torch.Tensor) -> torch.Tensor:
+    """[N,C,H,W] logical -> [N,H,W,C] contiguous (free if x is already a
+    channels-last view from an upstream op)."""
+    return x.permute(0, 2, 3, 1).contiguous()
+
+
+def _as_nchw_view(y_nhwc: torch.Tensor) -> torch.Tensor:
+    """[N,H,W,C] buffer -> logical [N,C,H,W] channels-last view (no copy)."""
+    return y_nhwc.permute(0, 3, 1, 2)
+
+
+def _splitk_for(m_tiles: int, n_tiles: int, kchunks: int) -> int:
+    tiles = max(1, m_tiles * n_tiles)
+    want = max(1, 512 // tiles)  # ~2 blocks per CU target
+    return int(min(want, kchunks, 64))
+
+
+# ===================================================================== linear
+class _Linear(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, w, b, act: int, slope: float):
+        ext = hip_ext()
+        xp = _pad_k(_bf(x))
+        wp = _pad_k(_bf(w))
+        bias = b.detach().float().contiguous() if b is not None else None
+        y = ext.gemm_tn(xp, wp, bias, act, slope, False)
+        ctx.save_for_backward(xp, wp, y)
+        ctx.act, ctx.slope = act, slope
+        ctx.nin = x.shape[-1]
+        ctx.has_bias = b is not None
+        ctx.dtypes = (x.dtype, w.dtype, b.dtype if b is not None else None)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        ext = hip_ext()
+        xp, wp, y = ctx.saved_tensors
+        dy = _bf(dy)
+        dpre = ext.act_bwd(dy, y, ctx.act, ctx.slope) if ctx.act else dy
+        dx = dw = db = None
+        if ctx.needs_input_grad[0]:
+            # dgrad: dx = dpre @ w ;  B = w^T padded over nout
+            wt = _pad_k(wp[:, : ctx.nin].t().contiguous())
+            dprep = _pad_k(dpre)
+            dx = ext.gemm_tn(dprep, wt, None, 0, 0.0, False)
+            if dx.shape[1] != ctx.nin:
+                dx = dx[:, : ctx.nin].contiguous()
+        if ctx.needs_input_grad[1]:
+            # wgrad: dw = dpre^T-contracted-with-x (rows = batch); padded
+            # x cols are zero so the sliced grad region is exact
+            nout = wp.shape[0]
+            kchunks = (xp.shape[0] + 63) // 64
+            sk = _splitk_for((nout + 63) // 64, (ctx.nin + 63) // 64, kchunks)
+            dw = ext.gemm_nt(dpre, xp, sk)[:, : ctx.nin]
+            dw = dw.contiguous().to(ctx.dtypes[1])
+        if ctx.has_bias and ctx.needs_input_grad[2]:
+            db = ext.col_sum(dpre).to(ctx.dtypes[2])
+        if dx is not None:
+            dx = dx.to(ctx.dtypes[0])
+        return dx, dw, db, None, None
+
+
+def linear(x, w, b=None, act="identity", slope=0.2):
+    x2 = x.reshape(-1, x.shape[-1]) if x.dim() > 2 else x
+    y = _Linear.apply(_bf(x2), w, b, ACT_CODES[act], slope)
+    if x.dim() > 2:
+        y = y.reshape(*x.shape[:-1], y.shape[-1])
+    return y
+
+
+# ====================================================================== conv
+def _conv_out(h, k, stride, pad):
+    return (h + 2 * pad - k) // stride + 1
+
+
+class _Conv2d(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, w, b, stride: int, pad: int, act: int, slope: float):
+        ext = hip_ext()
+        N, C, H, W = x.shape
+        Kout, _, R, S = w.shape
+        Ho, Wo = _conv_out(H, R, stride, pad), _conv_out(W, S, stride, pad)
+        kpad = _rup64(R * S * C)
+        xh = _nhwc(_bf(x))
+        col = ext.im2col(xh, N, H, W, C, Ho, Wo, R, S, stride, pad, kpad)
+        w_cl = _bf(w.permute(0, 2, 3, 1)).reshape(Kout, R * S * C)
+        wp = _pad_k(w_cl)
+        bias = b.detach().float().contiguous() if b is not None else None
+        y2d = ext.gemm_tn(col, wp, bias, act, slope, False)
+        ctx.save_for_backward(xh, wp, y2d)
+        ctx.geom = (N, C, H, W, Kout, R, S, Ho, Wo, stride, pad, kpad)
+        ctx.act, ctx.slope = act, slope
+        ctx.has_bias = b is not None
+        ctx.dtypes = (x.dtype, w.dtype, b.dtype if b is not None else None)
+        return _as_nchw_view(y2d.view(N, Ho, Wo, Kout))
+
+    @staticmethod
+    def backward(ctx, dy):
+        ext = hip_ext()
+        xh, wp, y2d = ctx.saved_tensors
+        N, C, H, W, Kout, R, S, Ho, Wo, stride, pad, kpad = ctx.geom
+        rsc = R * S * C
+        dy2d = _bf(dy.permute(0, 2, 3, 1)).reshape(-1, Kout)
+        dpre = ext.act_bwd(dy2d, y2d, ctx.act, ctx.slope) if ctx.act else dy2d
+
+        dx = dw = db = None
+        if ctx.needs_input_grad[1]:
+            # wgrad (rows = N*Ho*Wo)
+            col = ext.im2col(xh, N, H, W, C, Ho, Wo, R, S, stride, pad, kpad)
+            npq = dpre.shape[0]
+            sk = _splitk_for((Kout + 63) // 64, (kpad + 63) // 64,
+                             (npq + 63) // 64)
+            dw = ext.gemm_nt(dpre, col, sk)[:, :rsc]
+            dw = (dw.reshape(Kout, R, S, C).permute(0, 3, 1, 2)
+                  .contiguous().to(ctx.dtypes[1]))
+        if ctx.needs_input_grad[0]:
+            # dgrad: dcol = dpre @ w_cl ; then col2im
+            wt = _pad_k(wp[:, :rsc].t().contiguous())  # [rsc, kout_pad]
+            dprep = _pad_k(dpre)
+            dcol = ext.gemm_tn(dprep, wt, None, 0, 0.0, False)  # [NP, rsc]
+            dxh = ext.col2im(dcol, N, H, W, C, Ho, Wo, R, S, stride, pad,
+                             rsc, None, 0, 0.0)
+            dx = _as_nchw_view(dxh).to(ctx.dtypes[0])
+        if ctx.has_bias and ctx.needs_input_grad[2]:
+            db = ext.col_sum(dpre).to(ctx.dtypes[2])
+        return dx, dw, db, None, None, None, None
+
+
+def conv2d(x, w, b=None, stride=1, padding=0, act="identity", slope=0.2):
+    return _Conv2d.apply(x, w, b, stride, padding, ACT_CODES[act], slope)
+
+
+# ============================================================ conv transpose
+class _ConvTranspose2d(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, w, b, stride: int, pad: int, act: int, slope: float):
+        ext = hip_ext()
+        N, Cin, Hi, Wi = x.shape
+        _, Cout, R, S = w.shape
+        Ho = (Hi - 1) * stride - 2 * pad + R
+        Wo = (Wi - 1) * stride - 2 * pad + S
+        xh = _nhwc(_bf(x))                       # [N,Hi,Wi,Cin]
+        x2d = _pad_k(xh.reshape(-1, Cin))
+        # B operand: [R*S*Cout, Cin] (k-contiguous over Cin)
+        w2a = _pad_k(_bf(w.permute(2, 3, 1, 0)).reshape(R * S * Cout, Cin))
+        col = ext.gemm_tn(x2d, w2a, None, 0, 0.0, False)  # [NPin, RS*Cout]
+        bias = b.detach().float().contiguous() if b is not None else None
+        # col2im over the OUTPUT image grid; patches at the input grid
+        yh = ext.col2im(col, N, Ho, Wo, Cout, Hi, Wi, R, S, stride, pad,
+                        R * S * Cout, bias, act, slope)
+        ctx.save_for_backward(x2d, w2a, yh)
+        ctx.geom = (N, Cin, Hi, Wi, Cout, R, S, Ho, Wo, stride, pad)
+        ctx.act, ctx.slope = act, slope
+        ctx.has_bias = b is not None
+        ctx.dtypes = (x.dtype, w.dtype, b.dtype if b is not None else None)
+        return _as_nchw_view(yh)
+
+    @staticmethod
+    def backward(ctx, dy):
+        ext = hip_ext()
+        x2d, w2a, yh = ctx.saved_tensors
+        N, Cin, Hi, Wi, Cout, R, S, Ho, Wo, stride, pad = ctx.geom
+        rsco = R * S * Cout
+        rscop = _rup64(rsco)
+        dyh = _nhwc(_bf(dy))                       # [N,Ho,Wo,Cout]
+        if ctx.act:
+            dpre = ext.act_bwd(dyh.reshape(-1, Cout),
+                               yh.reshape(-1, Cout), ctx.act, ctx.slope)
+            dpre_img = dpre.view(N, Ho, Wo, Cout)
+        else:
+            dpre_img = dyh
+        # dcol = im2col(dpre_img) over the input grid
+        dcol = ext.im2col(dpre_img.contiguous(), N, Ho, Wo, Cout, Hi, Wi, R,
+                          S, stride, pad, rscop)   # [NPin, rscop]
+        dx = dw = db = None
+        if ctx.needs_input_grad[0]:
+            # dx = dcol @ w2b ; w2b = [Cin, R*S*Cout]
+            w2b = _pad_k(
+                w2a[:, :Cin].reshape(R, S, Cout, Cin)
+                .permute(3, 0, 1, 2).reshape(Cin, rsco)
+            )
+            dx2d = ext.gemm_tn(dcol, w2b, None, 0, 0.0, False)  # [NPin, Cin]
+            dx = _as_nchw_view(dx2d.view(N, Hi, Wi, Cin)).to(ctx.dtypes[0])
+        if ctx.needs_input_grad[1]:
+            # wgrad: dW2a[rsco][cin] = sum_np dcol[np][rsco]*x2d[np][cin];
+            # padded x2d cols are zero -> sliced grad exact
+            npq = dcol.shape[0]
+            sk = _splitk_for((rscop + 63) // 64, (Cin + 63) // 64,
+                             (npq + 63) // 64)
+            dw2a = ext.gemm_nt(dcol, x2d, sk)[:rsco, :Cin]
+            dw = (dw2a.reshape(R, S, Cout, Cin).permute(3, 2, 0, 1)
+                  .contiguous().to(ctx.dtypes[1]))
+        if ctx.has_bias and ctx.needs_input_grad[2]:
+            db = ext.col_sum(dpre_img.reshape(-1, Cout)).to(ctx.dtypes[2])
+        return dx, dw, db, None, None, None, None
+
+
+def conv_transpose2d(x, w, b=None, stride=1, padding=0, act="identity",
+                     slope=0.2):
+    return _ConvTranspose2d.apply(x, w, b, stride, padding, ACT_CODES[act],
+                                  slope)
+
+
+# ================================================================ batch norm
+class _BatchNorm(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x2, gamma, beta, rm, rv, momentum, eps):
+        ext = hip_ext()
+        y, mean, istd = ext.bn_fwd_train(x2, gamma.detach().float(),
+                                         beta.detach().float(), rm, rv,
+                                         momentum, eps)
+        ctx.save_for_backward(x2, mean, istd, gamma)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        ext = hip_ext()
+        x2, mean, istd, gamma = ctx.saved_tensors
+        dx, dgamma, dbeta = ext.bn_bwd(x2, _bf(dy), mean, istd,
+                                       gamma.detach().float())
+        return dx, dgamma.to(gamma.dtype), dbeta.to(gamma.dtype), None, \
+            None, None, None
+
+
+def batch_norm(x, weight, bias, running_mean, running_var, training,
+               momentum=0.1, eps=1e-5):
+    is4d = x.dim() == 4
+    if is4d:
+        xh = _nhwc(_bf(x))
+        x2 = xh.reshape(-1, xh.shape[-1])
+    else:
+        x2 = _bf(x)
+    if training:
+        y2 = _BatchNorm.apply(x2, weight, bias, running_mean, running_var,
+                              momentum, eps)
+    else:
+        y2 = hip_ext().bn_fwd_eval(x2, weight.detach().float(),
+                                   bias.detach().float(), running_mean,
+                                   running_var, eps)
+    if is4d:
+        return _as_nchw_view(y2.view(*xh.shape))
+    return y2
+
+
+# ================================================================== pooling
+class _MaxPool2d(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, kernel: int, stride: int):
+        ext = hip_ext()
+        N, C, H, W = x.shape
+        xh = _nhwc(_bf(x))
+        out, argmax = ext.maxpool_fwd(xh, N, H, W, C, kernel, stride)
+        ctx.save_for_backward(argmax)
+        ctx.geom = (N, C, H, W, kernel, stride)
+        ctx.x_dtype = x.dtype
+        return _as_nchw_view(out)
+
+    @staticmethod
+    def backward(ctx, dy):
+        ext = hip_ext()
+        (argmax,) = ctx.saved_tensors
+        N, C, H, W, kernel, stride = ctx.geom
+        dyh = _nhwc(_bf(dy))
+        din = ext.maxpool_bwd(dyh, argmax, N, H, W, C, kernel, stride)
+        return _as_nchw_view(din).to(ctx.x_dtype), None, None
+
+
+def max_pool2d(x, kernel, stride):
+    return _MaxPool2d.apply(x, kernel, stride)
+
+
+class _Upsample2d(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, scale: int):
+        ext = hip_ext()
+        N, C, H, W = x.shape
+        xh = _nhwc(_bf(x))
+        out = ext.upsample_fwd(xh, N, H, W, C, scale)
+        ctx.geom = (N, C, H, W, scale)
+        ctx.x_dtype = x.dtype
+        return _as_nchw_view(out)
+
+    @staticmethod
+    def backward(ctx, dy):
+        ext = hip_ext()
+        N, C, H, W, scale = ctx.geom
+        dyh = _nhwc(_bf(dy))
+        din = ext.upsample_bwd(dyh, N, H, W, C, scale)
+        return _as_nchw_view(din).to(ctx.x_dtype), None
+
+
+def upsample_nearest2d(x, scale):
+    return _Upsample2d.apply(x, scale)
+
+
+# =============================================================== activations
+class _Act(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, act: int, slope: float):
+        y = hip_ext().act_fwd(_bf(x), act, slope)
+        ctx.save_for_backward(y)
+        ctx.act, ctx.slope = act, slope
+        ctx.x_dtype = x.dtype
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        (y,) = ctx.saved_tensors
+        dx = hip_ext().act_bwd(_bf(dy), y, ctx.act, ctx.slope)
+        return dx.to(ctx.x_dtype), None, None
+
+
+def activation(x, act, slope=0.2):
+    code = ACT_CODES[act]
+    if code == 0:
+        return x
+    return _Act.apply(x, code, slope)
+
+
+# ==================================================================== losses
+class _BceWithLogits(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, logits, labels):
+        ext = hip_ext()
+        lb = _bf(logits)
+        yb = _bf(labels)
+        s = ext.bce_fwd(lb, yb)
+        ctx.save_for_backward(lb, yb)
+        ctx.x_dtype = logits.dtype
+        return (s / lb.numel()).squeeze(0)
+
+    @staticmethod
+    def backward(ctx, grad):
+        ext = hip_ext()
+        lb, yb = ctx.saved_tensors
+        d = ext.bce_bwd(lb, yb, 1.0 / lb.numel())
+        return (d * grad).to(ctx.x_dtype), None
+
+
+def bce_with_logits(logits, labels):
+    return _BceWithLogits.apply(logits, labels)
+
+
+class _SoftmaxXent(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, logits, onehot):
+        ext = hip_ext()
+        lb = _bf(logits)
+        ob = _bf(onehot)
+        s, probs = ext.softmax_xent_fwd(lb, ob)
+        ctx.save_for_backward(probs, ob)
+        ctx.x_dtype = logits.dtype
+        return (s / lb.shape[0]).squeeze(0)
+
+    @staticmethod
+    def backward(ctx, grad):
+        ext = hip_ext()
+        probs, ob = ctx.saved_tensors
+        d = ext.softmax_xent_bwd(probs, ob, 1.0 / probs.shape[0])
+        return (d * grad).to(ctx.x_dtype), None
+
+
+def softmax_cross_entropy(logits, onehot):
+    return _SoftmaxXent.apply(logits, onehot)
+
+
+# =================================================================== updater
+def fused_update(kind, param, grad, master, m, v, lr, beta1, beta2, rms_decay,
+                 eps, clip, l2, t):
+    ext = hip_ext()
+    g = grad.contiguous()
+    if kind == "adam":
+        ext.fused_adam(param.view(-1), g.view(-1),
+                       None if master is None else master.view(-1),
+                       m.view(-1), v.view(-1), lr, beta1, beta2, eps, clip,
+                       l2, t)
+    elif kind == "rmsprop":
+        ext.fused_rmsprop(param.view(-1), g.view(-1),
+                          None if master is None else master.view(-1),
+                          v.view(-1), lr, rms_decay, eps, clip, l2)
+    else:
+        raise KeyError(kind)

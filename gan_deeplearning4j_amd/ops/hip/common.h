@@ -1,0 +1,82 @@
+// Common helpers for the gfx950 (CDNA4) kernel library.
+// Wavefront = 64; block sizes are multiples of 64 throughout.
+#pragma once
+
+#include <hip/hip_runtime.h>
+#include <hip/hip_bf16.h>
+
+#define DEV_INLINE __device__ __forceinline__
+
+// ---------------------------------------------------------------- vectors
+typedef __attribute__((ext_vector_type(2))) float    f32x2;
+typedef __attribute__((ext_vector_type(4))) float    f32x4;
+typedef __attribute__((ext_vector_type(16))) float   f32x16;
+typedef __attribute__((ext_vector_type(4))) short    s16x4;
+typedef __attribute__((ext_vector_type(8))) short    s16x8;
+typedef __attribute__((ext_vector_type(4))) int      i32x4;
+
+using bf16 = __hip_bfloat16;
+
+DEV_INLINE float bf2f(unsigned short u) {
+  union { unsigned int i; float f; } v;
+  v.i = (unsigned int)u << 16;
+  return v.f;
+}
+
+DEV_INLINE unsigned short f2bf(float f) {
+  union { float f; unsigned int i; } v;
+  v.f = f;
+  // round-to-nearest-even
+  unsigned int lsb = (v.i >> 16) & 1;
+  v.i += 0x7fffu + lsb;
+  return (unsigned short)(v.i >> 16);
+}
+
+// -------------------------------------------------------------- activations
+// act codes: 0 identity, 1 tanh, 2 sigmoid, 3 leaky-relu, 4 relu
+DEV_INLINE float act_fwd(float x, int act, float slope) {
+  switch (act) {
+    case 1: return tanhf(x);
+    case 2: return 1.0f / (1.0f + __expf(-x));
+    case 3: return x > 0.0f ? x : slope * x;
+    case 4: return x > 0.0f ? x : 0.0f;
+    default: return x;
+  }
+}
+
+// derivative expressed in terms of the OUTPUT y (valid for all five)
+DEV_INLINE float act_bwd_from_y(float y, int act, float slope) {
+  switch (act) {
+    case 1: return 1.0f - y * y;
+    case 2: return y * (1.0f - y);
+    case 3: return y > 0.0f ? 1.0f : slope;
+    case 4: return y > 0.0f ? 1.0f : 0.0f;
+    default: return 1.0f;
+  }
+}
+
+// ------------------------------------------------------------- reductions
+DEV_INLINE float wave_reduce_sum(float v) {
+  #pragma unroll
+  for (int off = 32; off > 0; off >>= 1)
+    v += __shfl_down(v, off, 64);
+  return v;  // valid in lane 0
+}
+
+DEV_INLINE float wave_reduce_max(float v) {
+  #pragma unroll
+  for (int off = 32; off > 0; off >>= 1)
+    v = fmaxf(v, __shfl_down(v, off, 64));
+  return v;
+}
+
+constexpr int ceil_div(int a, int b) { return (a + b - 1) / b; }
+
+#define HIP_CHECK_LAST()                                                     \
+  do {                                                                       \
+    hipError_t e = hipGetLastError();                                        \
+    if (e != hipSuccess) {                                                   \
+      printf("HIP error %s at %s:%d\n", hipGetErrorString(e), __FILE__,      \
+             __LINE__);                                                      \
+    }                                                                        \
+  } while (0)

@@ -1,0 +1,391 @@
+// Torch bindings for the gfx950 kernel library (_C extension).
+// Thin layer: checks, allocation, stream plumbing; all compute is in the
+// .hip kernels. Built in-tree by setup.py (hipcc, --offload-arch=gfx950).
+
+#include <hip/hip_runtime.h>
+#include <torch/extension.h>
+#include <ATen/cuda/CUDAContext.h>
+
+namespace {
+
+struct ConvGeom {
+  int N, H, W, C, Ho, Wo, R, S, stride, pad, kpad;
+};
+struct PoolGeom {
+  int N, H, W, C, Ho, Wo, k, stride;
+};
+
+extern "C" {
+void launch_gemm_tn(const void*, const void*, void*, float*, const float*,
+                    int, int, int, long, long, int, float, hipStream_t);
+void launch_gemm_nt(const void*, const void*, float*, int, int, int, long,
+                    long, int, hipStream_t);
+void launch_im2col(const void*, void*, ConvGeom, hipStream_t);
+void launch_col2im(const void*, void*, ConvGeom, const float*, int, float,
+                   hipStream_t);
+void launch_maxpool_fwd(const void*, void*, void*, PoolGeom, hipStream_t);
+void launch_maxpool_bwd(const void*, const void*, void*, PoolGeom,
+                        hipStream_t);
+void launch_upsample_fwd(const void*, void*, int, int, int, int, int,
+                         hipStream_t);
+void launch_upsample_bwd(const void*, void*, int, int, int, int, int,
+                         hipStream_t);
+void launch_act_fwd(const void*, void*, long, int, float, hipStream_t);
+void launch_act_bwd(const void*, const void*, void*, long, int, float,
+                    hipStream_t);
+void launch_col_sum(const void*, float*, int, int, hipStream_t);
+void launch_bce_fwd(const void*, const void*, float*, long, hipStream_t);
+void launch_bce_bwd(const void*, const void*, void*, float, long, hipStream_t);
+void launch_softmax_xent_fwd(const void*, const void*, float*, void*, int,
+                             int, hipStream_t);
+void launch_softmax_xent_bwd(const void*, const void*, void*, float, long,
+                             hipStream_t);
+void launch_bn_stats(const void*, long, int, float*, float*, hipStream_t);
+void launch_bn_finalize(const float*, const float*, long, int, float, float,
+                        float*, float*, float*, float*, hipStream_t);
+void launch_bn_apply(const void*, void*, long, int, const float*,
+                     const float*, const float*, const float*, hipStream_t);
+void launch_bn_apply_eval(const void*, void*, long, int, const float*,
+                          const float*, const float*, const float*, float,
+                          hipStream_t);
+void launch_bn_bwd_reduce(const void*, const void*, long, int, const float*,
+                          const float*, float*, float*, hipStream_t);
+void launch_bn_bwd_apply(const void*, const void*, void*, long, int,
+                         const float*, const float*, const float*,
+                         const float*, const float*, hipStream_t);
+void launch_fused_adam(void*, const void*, float*, float*, float*, long, int,
+                       int, float, float, float, float, float, float, int,
+                       hipStream_t);
+void launch_fused_rmsprop(void*, const void*, float*, float*, long, int, int,
+                          float, float, float, float, float, hipStream_t);
+}
+
+hipStream_t cur_stream() {
+  return (hipStream_t)at::cuda::getCurrentCUDAStream().stream();
+}
+
+void check_bf16(const torch::Tensor& t, const char* name) {
+  TORCH_CHECK(t.is_cuda(), name, " must be on GPU");
+  TORCH_CHECK(t.scalar_type() == torch::kBFloat16, name, " must be bf16");
+  TORCH_CHECK(t.is_contiguous(), name, " must be contiguous");
+}
+
+void check_f32(const torch::Tensor& t, const char* name) {
+  TORCH_CHECK(t.is_cuda(), name, " must be on GPU");
+  TORCH_CHECK(t.scalar_type() == torch::kFloat32, name, " must be fp32");
+  TORCH_CHECK(t.is_contiguous(), name, " must be contiguous");
+}
+
+// ------------------------------------------------------------------ GEMM
+torch::Tensor gemm_tn(torch::Tensor A, torch::Tensor B,
+                      c10::optional<torch::Tensor> bias, int64_t act,
+                      double slope, bool out_f32) {
+  check_bf16(A, "A");
+  check_bf16(B, "B");
+  int64_t M = A.size(0), K = A.size(1), N = B.size(0);
+  TORCH_CHECK(B.size(1) == K, "K mismatch");
+  TORCH_CHECK(K % 64 == 0, "K must be padded to a multiple of 64, got ", K);
+  const float* bias_p = nullptr;
+  if (bias.has_value() && bias->defined() && bias->numel() > 0) {
+    check_f32(*bias, "bias");
+    TORCH_CHECK(bias->numel() == N, "bias size");
+    bias_p = bias->data_ptr<float>();
+  }
+  auto opts = A.options();
+  torch::Tensor C = torch::empty(
+      {M, N}, out_f32 ? opts.dtype(torch::kFloat32) : opts);
+  launch_gemm_tn(A.data_ptr(), B.data_ptr(),
+                 out_f32 ? nullptr : C.data_ptr(),
+                 out_f32 ? C.data_ptr<float>() : nullptr, bias_p, (int)M,
+                 (int)N, (int)K, K, K, (int)act, (float)slope, cur_stream());
+  return C;
+}
+
+torch::Tensor gemm_nt(torch::Tensor A, torch::Tensor B, int64_t splitk) {
+  // C[M][N] = sum_k A[k][M]*B[k][N], fp32 out
+  check_bf16(A, "A");
+  check_bf16(B, "B");
+  int64_t K = A.size(0), M = A.size(1), N = B.size(1);
+  TORCH_CHECK(B.size(0) == K, "K mismatch");
+  torch::Tensor C = torch::zeros({M, N}, A.options().dtype(torch::kFloat32));
+  launch_gemm_nt(A.data_ptr(), B.data_ptr(), C.data_ptr<float>(), (int)M,
+                 (int)N, (int)K, M, N, (int)splitk, cur_stream());
+  return C;
+}
+
+// ------------------------------------------------------------------ conv
+torch::Tensor im2col(torch::Tensor x, int64_t N, int64_t H, int64_t W,
+                     int64_t C, int64_t Ho, int64_t Wo, int64_t R, int64_t S,
+                     int64_t stride, int64_t pad, int64_t kpad) {
+  check_bf16(x, "x");
+  ConvGeom g{(int)N, (int)H, (int)W, (int)C, (int)Ho, (int)Wo,
+             (int)R, (int)S, (int)stride, (int)pad, (int)kpad};
+  torch::Tensor col = torch::empty({N * Ho * Wo, kpad}, x.options());
+  launch_im2col(x.data_ptr(), col.data_ptr(), g, cur_stream());
+  return col;
+}
+
+torch::Tensor col2im(torch::Tensor dcol, int64_t N, int64_t H, int64_t W,
+                     int64_t C, int64_t Ho, int64_t Wo, int64_t R, int64_t S,
+                     int64_t stride, int64_t pad, int64_t kpad,
+                     c10::optional<torch::Tensor> bias, int64_t act,
+                     double slope) {
+  check_bf16(dcol, "dcol");
+  ConvGeom g{(int)N, (int)H, (int)W, (int)C, (int)Ho, (int)Wo,
+             (int)R, (int)S, (int)stride, (int)pad, (int)kpad};
+  const float* bias_p = nullptr;
+  if (bias.has_value() && bias->defined() && bias->numel() > 0) {
+    check_f32(*bias, "bias");
+    bias_p = bias->data_ptr<float>();
+  }
+  torch::Tensor out = torch::empty({N, H, W, C}, dcol.options());
+  launch_col2im(dcol.data_ptr(), out.data_ptr(), g, bias_p, (int)act,
+                (float)slope, cur_stream());
+  return out;
+}
+
+// ------------------------------------------------------------------ pool
+std::vector<torch::Tensor> maxpool_fwd(torch::Tensor x, int64_t N, int64_t H,
+                                       int64_t W, int64_t C, int64_t k,
+                                       int64_t stride) {
+  check_bf16(x, "x");
+  int Ho = (int)((H - k) / stride + 1), Wo = (int)((W - k) / stride + 1);
+  PoolGeom g{(int)N, (int)H, (int)W, (int)C, Ho, Wo, (int)k, (int)stride};
+  torch::Tensor out = torch::empty({N, Ho, Wo, C}, x.options());
+  torch::Tensor am = torch::empty({N, Ho, Wo, C},
+                                  x.options().dtype(torch::kUInt8));
+  launch_maxpool_fwd(x.data_ptr(), out.data_ptr(), am.data_ptr(), g,
+                     cur_stream());
+  return {out, am};
+}
+
+torch::Tensor maxpool_bwd(torch::Tensor dout, torch::Tensor argmax, int64_t N,
+                          int64_t H, int64_t W, int64_t C, int64_t k,
+                          int64_t stride) {
+  check_bf16(dout, "dout");
+  int Ho = (int)((H - k) / stride + 1), Wo = (int)((W - k) / stride + 1);
+  PoolGeom g{(int)N, (int)H, (int)W, (int)C, Ho, Wo, (int)k, (int)stride};
+  torch::Tensor din = torch::empty({N, H, W, C}, dout.options());
+  launch_maxpool_bwd(dout.data_ptr(), argmax.data_ptr(), din.data_ptr(), g,
+                     cur_stream());
+  return din;
+}
+
+torch::Tensor upsample_fwd(torch::Tensor x, int64_t N, int64_t H, int64_t W,
+                           int64_t C, int64_t scale) {
+  check_bf16(x, "x");
+  torch::Tensor out = torch::empty({N, H * scale, W * scale, C}, x.options());
+  launch_upsample_fwd(x.data_ptr(), out.data_ptr(), (int)N, (int)H, (int)W,
+                      (int)C, (int)scale, cur_stream());
+  return out;
+}
+
+torch::Tensor upsample_bwd(torch::Tensor dout, int64_t N, int64_t H,
+                           int64_t W, int64_t C, int64_t scale) {
+  check_bf16(dout, "dout");
+  torch::Tensor din = torch::empty({N, H, W, C}, dout.options());
+  launch_upsample_bwd(dout.data_ptr(), din.data_ptr(), (int)N, (int)H, (int)W,
+                      (int)C, (int)scale, cur_stream());
+  return din;
+}
+
+// ----------------------------------------------------------- elementwise
+torch::Tensor act_fwd(torch::Tensor x, int64_t act, double slope) {
+  check_bf16(x, "x");
+  torch::Tensor y = torch::empty_like(x);
+  launch_act_fwd(x.data_ptr(), y.data_ptr(), x.numel(), (int)act,
+                 (float)slope, cur_stream());
+  return y;
+}
+
+torch::Tensor act_bwd(torch::Tensor dy, torch::Tensor y, int64_t act,
+                      double slope) {
+  check_bf16(dy, "dy");
+  check_bf16(y, "y");
+  torch::Tensor dx = torch::empty_like(dy);
+  launch_act_bwd(dy.data_ptr(), y.data_ptr(), dx.data_ptr(), dy.numel(),
+                 (int)act, (float)slope, cur_stream());
+  return dx;
+}
+
+torch::Tensor col_sum(torch::Tensor a) {
+  check_bf16(a, "a");
+  int64_t m = a.size(0), n = a.size(1);
+  torch::Tensor out = torch::zeros({n}, a.options().dtype(torch::kFloat32));
+  launch_col_sum(a.data_ptr(), out.data_ptr<float>(), (int)m, (int)n,
+                 cur_stream());
+  return out;
+}
+
+// ----------------------------------------------------------------- losses
+torch::Tensor bce_fwd(torch::Tensor logits, torch::Tensor labels) {
+  check_bf16(logits, "logits");
+  check_bf16(labels, "labels");
+  torch::Tensor s = torch::zeros({1}, logits.options().dtype(torch::kFloat32));
+  launch_bce_fwd(logits.data_ptr(), labels.data_ptr(), s.data_ptr<float>(),
+                 logits.numel(), cur_stream());
+  return s;
+}
+
+torch::Tensor bce_bwd(torch::Tensor logits, torch::Tensor labels,
+                      double gscale) {
+  check_bf16(logits, "logits");
+  torch::Tensor d = torch::empty_like(logits);
+  launch_bce_bwd(logits.data_ptr(), labels.data_ptr(), d.data_ptr(),
+                 (float)gscale, logits.numel(), cur_stream());
+  return d;
+}
+
+std::vector<torch::Tensor> softmax_xent_fwd(torch::Tensor logits,
+                                            torch::Tensor onehot) {
+  check_bf16(logits, "logits");
+  check_bf16(onehot, "onehot");
+  int64_t rows = logits.size(0), cols = logits.size(1);
+  torch::Tensor s = torch::zeros({1}, logits.options().dtype(torch::kFloat32));
+  torch::Tensor probs = torch::empty_like(logits);
+  launch_softmax_xent_fwd(logits.data_ptr(), onehot.data_ptr(),
+                          s.data_ptr<float>(), probs.data_ptr(), (int)rows,
+                          (int)cols, cur_stream());
+  return {s, probs};
+}
+
+torch::Tensor softmax_xent_bwd(torch::Tensor probs, torch::Tensor onehot,
+                               double gscale) {
+  check_bf16(probs, "probs");
+  torch::Tensor d = torch::empty_like(probs);
+  launch_softmax_xent_bwd(probs.data_ptr(), onehot.data_ptr(), d.data_ptr(),
+                          (float)gscale, probs.numel(), cur_stream());
+  return d;
+}
+
+// -------------------------------------------------------------------- BN
+std::vector<torch::Tensor> bn_fwd_train(torch::Tensor x, torch::Tensor gamma,
+                                        torch::Tensor beta,
+                                        torch::Tensor running_mean,
+                                        torch::Tensor running_var,
+                                        double momentum, double eps) {
+  check_bf16(x, "x");
+  check_f32(gamma, "gamma");
+  int64_t c = x.size(-1);
+  long m = x.numel() / c;
+  auto f32 = x.options().dtype(torch::kFloat32);
+  torch::Tensor sum = torch::zeros({c}, f32), sumsq = torch::zeros({c}, f32);
+  torch::Tensor mean = torch::empty({c}, f32), istd = torch::empty({c}, f32);
+  torch::Tensor y = torch::empty_like(x);
+  auto s = cur_stream();
+  launch_bn_stats(x.data_ptr(), m, (int)c, sum.data_ptr<float>(),
+                  sumsq.data_ptr<float>(), s);
+  launch_bn_finalize(sum.data_ptr<float>(), sumsq.data_ptr<float>(), m,
+                     (int)c, (float)eps, (float)momentum,
+                     mean.data_ptr<float>(), istd.data_ptr<float>(),
+                     running_mean.defined() && running_mean.numel() > 0
+                         ? running_mean.data_ptr<float>()
+                         : nullptr,
+                     running_var.defined() && running_var.numel() > 0
+                         ? running_var.data_ptr<float>()
+                         : nullptr,
+                     s);
+  launch_bn_apply(x.data_ptr(), y.data_ptr(), m, (int)c,
+                  mean.data_ptr<float>(), istd.data_ptr<float>(),
+                  gamma.data_ptr<float>(), beta.data_ptr<float>(), s);
+  return {y, mean, istd};
+}
+
+torch::Tensor bn_fwd_eval(torch::Tensor x, torch::Tensor gamma,
+                          torch::Tensor beta, torch::Tensor rm,
+                          torch::Tensor rv, double eps) {
+  check_bf16(x, "x");
+  int64_t c = x.size(-1);
+  long m = x.numel() / c;
+  torch::Tensor y = torch::empty_like(x);
+  launch_bn_apply_eval(x.data_ptr(), y.data_ptr(), m, (int)c,
+                       rm.data_ptr<float>(), rv.data_ptr<float>(),
+                       gamma.data_ptr<float>(), beta.data_ptr<float>(),
+                       (float)eps, cur_stream());
+  return y;
+}
+
+std::vector<torch::Tensor> bn_bwd(torch::Tensor x, torch::Tensor dy,
+                                  torch::Tensor mean, torch::Tensor istd,
+                                  torch::Tensor gamma) {
+  check_bf16(x, "x");
+  check_bf16(dy, "dy");
+  int64_t c = x.size(-1);
+  long m = x.numel() / c;
+  auto f32 = x.options().dtype(torch::kFloat32);
+  torch::Tensor dgamma = torch::zeros({c}, f32);
+  torch::Tensor dbeta = torch::zeros({c}, f32);
+  torch::Tensor dx = torch::empty_like(x);
+  auto s = cur_stream();
+  launch_bn_bwd_reduce(x.data_ptr(), dy.data_ptr(), m, (int)c,
+                       mean.data_ptr<float>(), istd.data_ptr<float>(),
+                       dgamma.data_ptr<float>(), dbeta.data_ptr<float>(), s);
+  launch_bn_bwd_apply(x.data_ptr(), dy.data_ptr(), dx.data_ptr(), m, (int)c,
+                      mean.data_ptr<float>(), istd.data_ptr<float>(),
+                      gamma.data_ptr<float>(), dgamma.data_ptr<float>(),
+                      dbeta.data_ptr<float>(), s);
+  return {dx, dgamma, dbeta};
+}
+
+// ------------------------------------------------------------------ optim
+void fused_adam(torch::Tensor param, torch::Tensor grad,
+                c10::optional<torch::Tensor> master, torch::Tensor m,
+                torch::Tensor v, double lr, double b1, double b2, double eps,
+                double clip, double l2, int64_t t) {
+  TORCH_CHECK(param.is_cuda() && param.is_contiguous());
+  int param_bf16 = param.scalar_type() == torch::kBFloat16;
+  int grad_bf16 = grad.scalar_type() == torch::kBFloat16;
+  float* master_p = nullptr;
+  if (param_bf16) {
+    TORCH_CHECK(master.has_value(), "bf16 param needs fp32 master");
+    master_p = master->data_ptr<float>();
+  }
+  launch_fused_adam(param.data_ptr(), grad.data_ptr(), master_p,
+                    m.data_ptr<float>(), v.data_ptr<float>(), param.numel(),
+                    grad_bf16, param_bf16, (float)lr, (float)b1, (float)b2,
+                    (float)eps, (float)clip, (float)l2, (int)t, cur_stream());
+}
+
+void fused_rmsprop(torch::Tensor param, torch::Tensor grad,
+                   c10::optional<torch::Tensor> master, torch::Tensor v,
+                   double lr, double decay, double eps, double clip,
+                   double l2) {
+  TORCH_CHECK(param.is_cuda() && param.is_contiguous());
+  int param_bf16 = param.scalar_type() == torch::kBFloat16;
+  int grad_bf16 = grad.scalar_type() == torch::kBFloat16;
+  float* master_p = nullptr;
+  if (param_bf16) {
+    TORCH_CHECK(master.has_value(), "bf16 param needs fp32 master");
+    master_p = master->data_ptr<float>();
+  }
+  launch_fused_rmsprop(param.data_ptr(), grad.data_ptr(), master_p,
+                       v.data_ptr<float>(), param.numel(), grad_bf16,
+                       param_bf16, (float)lr, (float)decay, (float)eps,
+                       (float)clip, (float)l2, cur_stream());
+}
+
+}  // namespace
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
+  mod.def("gemm_tn", &gemm_tn, "C = act(A.B^T + bias) bf16 MFMA");
+  mod.def("gemm_nt", &gemm_nt, "C = A^T.B (contraction over rows) fp32 out");
+  mod.def("im2col", &im2col);
+  mod.def("col2im", &col2im);
+  mod.def("maxpool_fwd", &maxpool_fwd);
+  mod.def("maxpool_bwd", &maxpool_bwd);
+  mod.def("upsample_fwd", &upsample_fwd);
+  mod.def("upsample_bwd", &upsample_bwd);
+  mod.def("act_fwd", &act_fwd);
+  mod.def("act_bwd", &act_bwd);
+  mod.def("col_sum", &col_sum);
+  mod.def("bce_fwd", &bce_fwd);
+  mod.def("bce_bwd", &bce_bwd);
+  mod.def("softmax_xent_fwd", &softmax_xent_fwd);
+  mod.def("softmax_xent_bwd", &softmax_xent_bwd);
+  mod.def("bn_fwd_train", &bn_fwd_train);
+  mod.def("bn_fwd_eval", &bn_fwd_eval);
+  mod.def("bn_bwd", &bn_bwd);
+  mod.def("fused_adam", &fused_adam);
+  mod.def("fused_rmsprop", &fused_rmsprop);
+  mod.attr("arch") = "gfx950";
+}

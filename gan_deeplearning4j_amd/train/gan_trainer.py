@@ -176,11 +176,13 @@ class ReferenceProtocolTrainer:
         self.out_dir = Path(out_dir or cfg.train.out_dir)
         self.out_dir.mkdir(parents=True, exist_ok=True)
 
-        # graphs (Java:118-314, 337-368)
-        self.dis = build_discriminator(cfg).to_device(self.device)
-        self.gen = build_frozen_generator(cfg).to_device(self.device)
-        self.gan = build_stacked_gan(cfg).to_device(self.device)
-        self.cv = build_transfer_classifier(self.dis, cfg).to_device(self.device)
+        # graphs (Java:118-314, 337-368); bf16 compute on GPU, fp32 on CPU
+        dt = torch.bfloat16 if self.device.type == "cuda" else None
+        self.dis = build_discriminator(cfg).to_device(self.device, dt)
+        self.gen = build_frozen_generator(cfg).to_device(self.device, dt)
+        self.gan = build_stacked_gan(cfg).to_device(self.device, dt)
+        self.cv = build_transfer_classifier(self.dis, cfg).to_device(
+            self.device, dt)
         for g in (self.dis, self.gen, self.gan, self.cv):
             broadcast_parameters(g)
 

@@ -1,0 +1,385 @@
+"""GPU numerics: every HIP kernel vs a plain PyTorch fp32 reference.
+
+bf16 inputs -> fp32 torch reference on the SAME bf16-rounded values;
+tolerances sized for bf16 I/O with fp32 accumulation.
+"""
+
+import pytest
+import torch
+import torch.nn.functional as F
+
+pytestmark = pytest.mark.gpu
+
+DEV = "cuda:0"
+
+
+def ext():
+    from gan_deeplearning4j_amd.ops.backend import hip_ext
+
+    return hip_ext()
+
+
+def mk(shape, seed=0, scale=1.0):
+    g = torch.Generator().manual_seed(seed)
+    t = torch.randn(*shape, generator=g) * scale
+    return t.to(DEV, torch.bfloat16)
+
+
+def relerr(a, b):
+    a = a.float().cpu()
+    b = b.float().cpu()
+    denom = b.abs().max().clamp_min(1e-6)
+    return ((a - b).abs().max() / denom).item()
+
+
+# ------------------------------------------------------------------- GEMM
+def test_gemm_tn_identity_asymmetric():
+    # A = I -> C = B^T ; asymmetric B catches operand/output transposes
+    e = ext()
+    A = torch.eye(128, device=DEV, dtype=torch.bfloat16)
+    B = torch.arange(128 * 128, device=DEV, dtype=torch.float32)
+    B = ((B % 251) / 251.0 - 0.5).reshape(128, 128).to(torch.bfloat16)
+    C = e.gemm_tn(A, B, None, 0, 0.0, False)
+    assert torch.allclose(C.float(), B.t().float(), atol=1e-2)
+
+
+@pytest.mark.parametrize("m,n,k", [(128, 128, 64), (200, 64, 128),
+                                   (1000, 1, 1024), (513, 130, 192),
+                                   (100, 1024, 6272)])
+def test_gemm_tn_shapes(m, n, k):
+    e = ext()
+    A, B = mk((m, k), 1), mk((n, k), 2)
+    C = e.gemm_tn(A, B, None, 0, 0.0, False)
+    ref = A.float().cpu() @ B.float().cpu().t()
+    assert relerr(C, ref) < 0.02
+
+
+def test_gemm_tn_bias_act():
+    e = ext()
+    A, B = mk((256, 128), 3), mk((64, 128), 4)
+    bias = torch.randn(64, device=DEV)
+    C = e.gemm_tn(A, B, bias, 1, 0.0, False)  # tanh
+    ref = torch.tanh(A.float().cpu() @ B.float().cpu().t() +
+                     bias.float().cpu())
+    assert relerr(C, ref) < 0.03
+
+
+@pytest.mark.parametrize("splitk", [1, 4])
+@pytest.mark.parametrize("m,n,k", [(64, 64, 64), (130, 70, 1000),
+                                   (512, 1152, 4096)])
+def test_gemm_nt(m, n, k, splitk):
+    e = ext()
+    A, B = mk((k, m), 5, 0.5), mk((k, n), 6, 0.5)
+    C = e.gemm_nt(A, B, splitk)
+    ref = A.float().cpu().t() @ B.float().cpu()
+    assert relerr(C, ref) < 0.02
+
+
+# ----------------------------------------------------------- im2col/col2im
+def test_im2col_matches_unfold():
+    e = ext()
+    N, C, H, W, R, stride, pad = 4, 3, 9, 9, 4, 2, 1
+    Ho = (H + 2 * pad - R) // stride + 1
+    x = mk((N, C, H, W), 7)
+    xh = x.permute(0, 2, 3, 1).contiguous()
+    kpad = (R * R * C + 63) // 64 * 64
+    col = e.im2col(xh, N, H, W, C, Ho, Ho, R, R, stride, pad, kpad)
+    # torch unfold gives [N, C*R*S, L] with C-major k: ours is (r,s,c)
+    ref = F.unfold(x.float().cpu(), R, padding=pad, stride=stride)
+    ref = ref.view(N, C, R, R, -1).permute(0, 4, 2, 3, 1)  # n,L,r,s,c
+    ref = ref.reshape(N * Ho * Ho, R * R * C)
+    assert relerr(col[:, : R * R * C], ref) < 1e-2
+    assert col[:, R * R * C:].abs().max().item() == 0  # zero K-pad
+
+
+def test_col2im_adjoint_of_im2col():
+    # <im2col(x), y> == <x, col2im(y)> (adjointness on random fixtures)
+    e = ext()
+    N, C, H, W, R, stride, pad = 2, 8, 8, 8, 3, 2, 1
+    Ho = (H + 2 * pad - R) // stride + 1
+    kpad = (R * R * C + 63) // 64 * 64
+    x = mk((N, H, W, C), 8)
+    y = mk((N * Ho * Ho, kpad), 9)
+    y[:, R * R * C:] = 0
+    col = e.im2col(x, N, H, W, C, Ho, Ho, R, R, stride, pad, kpad)
+    back = e.col2im(y, N, H, W, C, Ho, Ho, R, R, stride, pad, kpad,
+                    None, 0, 0.0)
+    lhs = (col.float() * y.float()).sum().item()
+    rhs = (x.float() * back.float()).sum().item()
+    assert abs(lhs - rhs) / (abs(rhs) + 1e-6) < 2e-2
+
+
+# ------------------------------------------------------------- conv layers
+def test_conv2d_fwd_bwd_vs_torch():
+    from gan_deeplearning4j_amd.ops import gpu_ops
+
+    N, Cin, H, Cout, R, stride, pad = 8, 16, 16, 32, 4, 2, 1
+    x = mk((N, Cin, H, H), 10, 0.5).requires_grad_(True)
+    w = mk((Cout, Cin, R, R), 11, 0.2).requires_grad_(True)
+    b = torch.randn(Cout, device=DEV, dtype=torch.bfloat16,
+                    requires_grad=True)
+    y = gpu_ops.conv2d(x, w, b, stride, pad, "lrelu", 0.2)
+    gout = mk(y.shape, 12)
+    y.backward(gout)
+
+    xr = x.detach().float().cpu().requires_grad_(True)
+    wr = w.detach().float().cpu().requires_grad_(True)
+    br = b.detach().float().cpu().requires_grad_(True)
+    yr = F.leaky_relu(F.conv2d(xr, wr, br, stride=stride, padding=pad), 0.2)
+    yr.backward(gout.float().cpu())
+
+    assert relerr(y, yr) < 0.03
+    assert relerr(x.grad, xr.grad) < 0.04
+    assert relerr(w.grad, wr.grad) < 0.04
+    assert relerr(b.grad, br.grad) < 0.04
+
+
+def test_conv_transpose2d_fwd_bwd_vs_torch():
+    from gan_deeplearning4j_amd.ops import gpu_ops
+
+    N, Cin, H, Cout, R, stride, pad = 4, 32, 8, 16, 4, 2, 1
+    x = mk((N, Cin, H, H), 13, 0.5).requires_grad_(True)
+    w = mk((Cin, Cout, R, R), 14, 0.2).requires_grad_(True)
+    b = torch.randn(Cout, device=DEV, dtype=torch.bfloat16,
+                    requires_grad=True)
+    y = gpu_ops.conv_transpose2d(x, w, b, stride, pad, "tanh")
+    gout = mk(y.shape, 15)
+    y.backward(gout)
+
+    xr = x.detach().float().cpu().requires_grad_(True)
+    wr = w.detach().float().cpu().requires_grad_(True)
+    br = b.detach().float().cpu().requires_grad_(True)
+    yr = torch.tanh(F.conv_transpose2d(xr, wr, br, stride=stride,
+                                       padding=pad))
+    yr.backward(gout.float().cpu())
+
+    assert relerr(y, yr) < 0.03
+    assert relerr(x.grad, xr.grad) < 0.04
+    assert relerr(w.grad, wr.grad) < 0.04
+    assert relerr(b.grad, br.grad) < 0.04
+
+
+def test_linear_fwd_bwd_vs_torch():
+    from gan_deeplearning4j_amd.ops import gpu_ops
+
+    x = mk((64, 784), 16).requires_grad_(True)
+    w = mk((256, 784), 17, 0.1).requires_grad_(True)
+    b = torch.randn(256, device=DEV, dtype=torch.bfloat16,
+                    requires_grad=True)
+    y = gpu_ops.linear(x, w, b, "tanh")
+    gout = mk(y.shape, 18)
+    y.backward(gout)
+
+    xr = x.detach().float().cpu().requires_grad_(True)
+    wr = w.detach().float().cpu().requires_grad_(True)
+    br = b.detach().float().cpu().requires_grad_(True)
+    yr = torch.tanh(F.linear(xr, wr, br))
+    yr.backward(gout.float().cpu())
+    assert relerr(y, yr) < 0.03
+    assert relerr(x.grad, xr.grad) < 0.04
+    assert relerr(w.grad, wr.grad) < 0.04
+    assert relerr(b.grad, br.grad) < 0.04
+
+
+# --------------------------------------------------------------------- BN
+def test_batchnorm_train_vs_torch():
+    from gan_deeplearning4j_amd.ops import gpu_ops
+
+    N, C, H = 16, 32, 8
+    x = mk((N, C, H, H), 19).requires_grad_(True)
+    gamma = torch.rand(C, device=DEV) + 0.5
+    beta = torch.randn(C, device=DEV)
+    gamma.requires_grad_(True)
+    beta.requires_grad_(True)
+    rm = torch.zeros(C, device=DEV)
+    rv = torch.ones(C, device=DEV)
+    y = gpu_ops.batch_norm(x, gamma, beta, rm, rv, True, 0.1, 1e-5)
+    gout = mk(y.shape, 20)
+    y.backward(gout)
+
+    xr = x.detach().float().cpu().requires_grad_(True)
+    gr = gamma.detach().float().cpu().requires_grad_(True)
+    br = beta.detach().float().cpu().requires_grad_(True)
+    rmr = torch.zeros(C)
+    rvr = torch.ones(C)
+    yr = F.batch_norm(xr, rmr, rvr, gr, br, True, 0.1, 1e-5)
+    yr.backward(gout.float().cpu())
+
+    assert relerr(y, yr) < 0.05
+    assert relerr(rm, rmr) < 0.02
+    assert relerr(rv, rvr) < 0.02
+    assert relerr(gamma.grad, gr.grad) < 0.05
+    assert relerr(beta.grad, br.grad) < 0.05
+    assert relerr(x.grad, xr.grad) < 0.08
+
+
+def test_batchnorm_eval_matches_running_stats():
+    from gan_deeplearning4j_amd.ops import gpu_ops
+
+    x = mk((32, 16), 21)
+    gamma = torch.rand(16, device=DEV) + 0.5
+    beta = torch.randn(16, device=DEV)
+    rm = torch.randn(16, device=DEV) * 0.1
+    rv = torch.rand(16, device=DEV) + 0.5
+    with torch.no_grad():
+        y = gpu_ops.batch_norm(x, gamma, beta, rm, rv, False)
+    ref = F.batch_norm(x.float().cpu(), rm.cpu(), rv.cpu(),
+                       gamma.cpu(), beta.cpu(), False)
+    assert relerr(y, ref) < 0.02
+
+
+# ------------------------------------------------------------ pool/upsample
+def test_maxpool_fwd_bwd():
+    from gan_deeplearning4j_amd.ops import gpu_ops
+
+    x = mk((4, 8, 11, 11), 22).requires_grad_(True)
+    y = gpu_ops.max_pool2d(x, 2, 1)
+    gout = mk(y.shape, 23)
+    y.backward(gout)
+    xr = x.detach().float().cpu().requires_grad_(True)
+    yr = F.max_pool2d(xr, 2, 1)
+    yr.backward(gout.float().cpu())
+    assert relerr(y, yr) < 1e-2
+    assert relerr(x.grad, xr.grad) < 2e-2
+
+
+def test_upsample_fwd_bwd():
+    from gan_deeplearning4j_amd.ops import gpu_ops
+
+    x = mk((4, 8, 7, 7), 24).requires_grad_(True)
+    y = gpu_ops.upsample_nearest2d(x, 2)
+    gout = mk(y.shape, 25)
+    y.backward(gout)
+    xr = x.detach().float().cpu().requires_grad_(True)
+    yr = F.interpolate(xr, scale_factor=2, mode="nearest")
+    yr.backward(gout.float().cpu())
+    assert relerr(y, yr) < 1e-2
+    assert relerr(x.grad, xr.grad) < 2e-2
+
+
+# ---------------------------------------------------------------- act/loss
+@pytest.mark.parametrize("act,fn", [
+    ("tanh", torch.tanh),
+    ("sigmoid", torch.sigmoid),
+    ("lrelu", lambda t: F.leaky_relu(t, 0.2)),
+    ("relu", F.relu),
+])
+def test_activations(act, fn):
+    from gan_deeplearning4j_amd.ops import gpu_ops
+
+    x = mk((333,), 26).requires_grad_(True)  # odd size exercises the tail
+    y = gpu_ops.activation(x, act, 0.2)
+    gout = mk(y.shape, 27)
+    y.backward(gout)
+    xr = x.detach().float().cpu().requires_grad_(True)
+    yr = fn(xr)
+    yr.backward(gout.float().cpu())
+    assert relerr(y, yr) < 2e-2
+    assert relerr(x.grad, xr.grad) < 3e-2
+
+
+def test_bce_with_logits():
+    from gan_deeplearning4j_amd.ops import gpu_ops
+
+    logits = mk((200, 1), 28).requires_grad_(True)
+    labels = (torch.rand(200, 1) > 0.5).float().to(DEV, torch.bfloat16)
+    loss = gpu_ops.bce_with_logits(logits, labels)
+    loss.backward()
+    lr = logits.detach().float().cpu().requires_grad_(True)
+    ref = F.binary_cross_entropy_with_logits(lr, labels.float().cpu())
+    ref.backward()
+    assert abs(loss.item() - ref.item()) / ref.item() < 2e-2
+    assert relerr(logits.grad, lr.grad) < 3e-2
+
+
+def test_softmax_xent():
+    from gan_deeplearning4j_amd.ops import gpu_ops
+
+    logits = mk((128, 10), 29).requires_grad_(True)
+    onehot = torch.eye(10)[torch.randint(0, 10, (128,))].to(DEV,
+                                                            torch.bfloat16)
+    loss = gpu_ops.softmax_cross_entropy(logits, onehot)
+    loss.backward()
+    lr = logits.detach().float().cpu().requires_grad_(True)
+    ref = -(onehot.float().cpu() * F.log_softmax(lr, dim=1)).sum(1).mean()
+    ref.backward()
+    assert abs(loss.item() - ref.item()) / ref.item() < 2e-2
+    assert relerr(logits.grad, lr.grad) < 3e-2
+
+
+# ----------------------------------------------------------------- updater
+def test_fused_adam_matches_cpu_updater():
+    from gan_deeplearning4j_amd.ops.optim import ParamSlot, Updater
+
+    torch.manual_seed(0)
+    w0 = torch.randn(1000)
+    grads = [torch.randn(1000) * 0.1 for _ in range(5)]
+
+    # CPU fp32 reference
+    p_cpu = torch.nn.Parameter(w0.clone())
+    s_cpu = ParamSlot(p_cpu, 0.01)
+    u_cpu = Updater([s_cpu], kind="adam", grad_clip=1.0, l2=1e-4)
+    # GPU bf16 param + fp32 master through the fused kernel
+    p_gpu = torch.nn.Parameter(w0.clone().to(DEV, torch.bfloat16))
+    s_gpu = ParamSlot(p_gpu, 0.01)
+    u_gpu = Updater([s_gpu], kind="adam", grad_clip=1.0, l2=1e-4)
+
+    for g in grads:
+        p_cpu.grad = g.clone()
+        p_gpu.grad = g.clone().to(DEV, torch.bfloat16)
+        u_cpu.step()
+        u_gpu.step()
+    torch.cuda.synchronize()
+    assert relerr(s_gpu.master, p_cpu.detach()) < 2e-2
+
+
+def test_fused_rmsprop_runs():
+    from gan_deeplearning4j_amd.ops.optim import ParamSlot, Updater
+
+    p = torch.nn.Parameter(torch.randn(512, device=DEV,
+                                       dtype=torch.bfloat16))
+    s = ParamSlot(p, 0.01)
+    u = Updater([s], kind="rmsprop")
+    p.grad = torch.randn(512, device=DEV, dtype=torch.bfloat16)
+    before = p.detach().float().clone()
+    u.step()
+    torch.cuda.synchronize()
+    assert not torch.equal(before, p.detach().float())
+
+
+# ----------------------------------------------------------- end-to-end GPU
+def test_dcgan28_trainer_step_gpu():
+    from gan_deeplearning4j_amd.config import preset
+    from gan_deeplearning4j_amd.models import build_dcgan
+    from gan_deeplearning4j_amd.train import GanTrainer
+
+    cfg = preset("dcgan28")
+    gen, dis = build_dcgan(cfg)
+    tr = GanTrainer(gen, dis, cfg, device=torch.device(DEV),
+                    dtype=torch.bfloat16)
+    real = (torch.rand(64, 1, 28, 28, device=DEV, dtype=torch.bfloat16)
+            * 2 - 1)
+    out1 = tr.step(real)
+    out2 = tr.step(real)
+    torch.cuda.synchronize()
+    import math
+
+    for v in (out1["loss_d"], out1["loss_g"], out2["loss_d"], out2["loss_g"]):
+        assert math.isfinite(v)
+
+
+def test_reference_protocol_iteration_gpu(tmp_path):
+    from gan_deeplearning4j_amd.config import GanConfig
+    from gan_deeplearning4j_amd.data.csv_reader import DataSet
+    from gan_deeplearning4j_amd.train import ReferenceProtocolTrainer
+
+    cfg = GanConfig()
+    cfg.data.batch_size_per_worker = 32
+    tr = ReferenceProtocolTrainer(cfg, device=torch.device(DEV),
+                                  out_dir=str(tmp_path))
+    feats = torch.rand(32, 784)
+    labels = torch.eye(10)[torch.randint(0, 10, (32,))]
+    out = tr.train_iteration(DataSet(feats, labels))
+    import math
+
+    assert math.isfinite(out["loss_d"]) and math.isfinite(out["loss_g"])
