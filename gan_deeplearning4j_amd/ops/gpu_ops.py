@@ -63,14 +63,33 @@ def _splitk_for(m_tiles: int, n_tiles: int, kchunks: int) -> int:
     return int(min(want, kchunks, 64))
 
 
+def _packed(w: torch.Tensor, key: str, builder):
+    """Per-parameter cache of derived weight layouts (padded/transposed/
+    permuted copies), invalidated by the tensor's in-place version counter.
+    Saves a handful of aten copies per layer per forward/backward."""
+    cache = getattr(w, "_gdlj_cache", None)
+    ver = w._version
+    if cache is None or cache[0] != ver:
+        cache = (ver, {})
+        try:
+            w._gdlj_cache = cache
+        except AttributeError:  # non-leaf / view tensors: skip caching
+            return builder()
+    d = cache[1]
+    if key not in d:
+        d[key] = builder()
+    return d[key]
+
+
 # ===================================================================== linear
 class _Linear(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, w, b, act: int, slope: float):
         ext = hip_ext()
         xp = _pad_k(_bf(x))
-        wp = _pad_k(_bf(w))
-        bias = b.detach().float().contiguous() if b is not None else None
+        wp = _packed(w, "wp", lambda: _pad_k(_bf(w.detach())))
+        bias = (_packed(b, "f32", lambda: b.detach().float().contiguous())
+                if b is not None else None)
         y = ext.gemm_tn(xp, wp, bias, act, slope, False)
         ctx.save_for_backward(xp, wp, y)
         ctx.act, ctx.slope = act, slope
@@ -88,7 +107,9 @@ class _Linear(torch.autograd.Function):
         dx = dw = db = None
         if ctx.needs_input_grad[0]:
             # dgrad: dx = dpre @ w ;  B = w^T padded over nout
-            wt = _pad_k(wp[:, : ctx.nin].t().contiguous())
+            nin = ctx.nin
+            wt = _packed(wp, "wt",
+                         lambda: _pad_k(wp[:, :nin].t().contiguous()))
             dprep = _pad_k(dpre)
             dx = ext.gemm_tn(dprep, wt, None, 0, 0.0, False)
             if dx.shape[1] != ctx.nin:
@@ -131,9 +152,10 @@ class _Conv2d(torch.autograd.Function):
         kpad = _rup64(R * S * C)
         xh = _nhwc(_bf(x))
         col = ext.im2col(xh, N, H, W, C, Ho, Wo, R, S, stride, pad, kpad)
-        w_cl = _bf(w.permute(0, 2, 3, 1)).reshape(Kout, R * S * C)
-        wp = _pad_k(w_cl)
-        bias = b.detach().float().contiguous() if b is not None else None
+        wp = _packed(w, "conv_wp", lambda: _pad_k(
+            _bf(w.detach().permute(0, 2, 3, 1)).reshape(Kout, R * S * C)))
+        bias = (_packed(b, "f32", lambda: b.detach().float().contiguous())
+                if b is not None else None)
         y2d = ext.gemm_tn(col, wp, bias, act, slope, False)
         ctx.save_for_backward(xh, wp, y2d)
         ctx.geom = (N, C, H, W, Kout, R, S, Ho, Wo, stride, pad, kpad)
@@ -163,7 +185,8 @@ class _Conv2d(torch.autograd.Function):
                   .contiguous().to(ctx.dtypes[1]))
         if ctx.needs_input_grad[0]:
             # dgrad: dcol = dpre @ w_cl ; then col2im
-            wt = _pad_k(wp[:, :rsc].t().contiguous())  # [rsc, kout_pad]
+            wt = _packed(wp, "wt", lambda: _pad_k(
+                wp[:, :rsc].t().contiguous()))       # [rsc, kout_pad]
             dprep = _pad_k(dpre)
             dcol = ext.gemm_tn(dprep, wt, None, 0, 0.0, False)  # [NP, rsc]
             dxh = ext.col2im(dcol, N, H, W, C, Ho, Wo, R, S, stride, pad,
@@ -190,9 +213,11 @@ class _ConvTranspose2d(torch.autograd.Function):
         xh = _nhwc(_bf(x))                       # [N,Hi,Wi,Cin]
         x2d = _pad_k(xh.reshape(-1, Cin))
         # B operand: [R*S*Cout, Cin] (k-contiguous over Cin)
-        w2a = _pad_k(_bf(w.permute(2, 3, 1, 0)).reshape(R * S * Cout, Cin))
+        w2a = _packed(w, "w2a", lambda: _pad_k(
+            _bf(w.detach().permute(2, 3, 1, 0)).reshape(R * S * Cout, Cin)))
         col = ext.gemm_tn(x2d, w2a, None, 0, 0.0, False)  # [NPin, RS*Cout]
-        bias = b.detach().float().contiguous() if b is not None else None
+        bias = (_packed(b, "f32", lambda: b.detach().float().contiguous())
+                if b is not None else None)
         # col2im over the OUTPUT image grid; patches at the input grid
         yh = ext.col2im(col, N, Ho, Wo, Cout, Hi, Wi, R, S, stride, pad,
                         R * S * Cout, bias, act, slope)
@@ -223,10 +248,10 @@ class _ConvTranspose2d(torch.autograd.Function):
         dx = dw = db = None
         if ctx.needs_input_grad[0]:
             # dx = dcol @ w2b ; w2b = [Cin, R*S*Cout]
-            w2b = _pad_k(
+            w2b = _packed(w2a, "w2b", lambda: _pad_k(
                 w2a[:, :Cin].reshape(R, S, Cout, Cin)
                 .permute(3, 0, 1, 2).reshape(Cin, rsco)
-            )
+            ))
             dx2d = ext.gemm_tn(dcol, w2b, None, 0, 0.0, False)  # [NPin, Cin]
             dx = _as_nchw_view(dx2d.view(N, Hi, Wi, Cin)).to(ctx.dtypes[0])
         if ctx.needs_input_grad[1]:

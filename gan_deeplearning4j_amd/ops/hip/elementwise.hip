@@ -57,16 +57,31 @@ __global__ void act_bwd_tail(const unsigned short* __restrict__ dy,
 
 // --------------------------------------------------------------- bias ops
 // column sum over C[M][N] -> out[N] fp32 (bias gradient).
-// One workgroup per column tile; each wave owns 64 columns, loops rows.
+// Row-major coalesced: a block covers a 256-wide column window and walks
+// rows with blockDim.x/colsW row-lanes in parallel, then LDS-reduces the
+// row-lanes and atomically adds one partial per column. N is typically
+// small (64..1024) and M huge (batch x spatial), so the row axis carries
+// the parallelism (grid.x row-chunks).
 __global__ void col_sum_bf16(const unsigned short* __restrict__ a,
-                             float* __restrict__ out, int m, int n) {
-  int col = blockIdx.x * blockDim.x + threadIdx.x;
-  if (col >= n) return;
+                             float* __restrict__ out, long m, int n) {
+  __shared__ float ls[256];
+  int c0 = blockIdx.y * 256;
+  int colsW = min(256, n - c0);
+  int lanes = (int)blockDim.x / colsW;          // row-lanes per block
+  int sub = (int)threadIdx.x / colsW;
+  int c = c0 + (int)threadIdx.x % colsW;
   float acc = 0.f;
-  // rows split over gridDim.y, partial atomics
-  for (int r = blockIdx.y; r < m; r += gridDim.y)
-    acc += bf2f(a[(long)r * n + col]);
-  atomicAdd(&out[col], acc);
+  if (sub < lanes) {
+    for (long r = (long)blockIdx.x * lanes + sub; r < m;
+         r += (long)gridDim.x * lanes)
+      acc += bf2f(a[r * n + c]);
+  }
+  ls[threadIdx.x] = acc;
+  __syncthreads();
+  if (sub == 0) {
+    for (int s = 1; s < lanes; ++s) acc += ls[s * colsW + c - c0];
+    atomicAdd(&out[c], acc);
+  }
 }
 
 // ------------------------------------------------------------------ losses
@@ -178,8 +193,12 @@ void launch_act_bwd(const void* dy, const void* y, void* dx, long n, int act,
                        (unsigned short*)dx, n8 * 8, n, act, slope);
 }
 
-void launch_col_sum(const void* a, float* out, int m, int n, hipStream_t s) {
-  dim3 grid(ceil_div(n, 256), min(64, max(1, m / 16)));
+void launch_col_sum(const void* a, float* out, long m, int n, hipStream_t s) {
+  int colsW = min(256, n);
+  int lanes = 256 / colsW;
+  long chunks = (m + lanes - 1) / lanes;
+  dim3 grid((unsigned)min((long)1024, max((long)1, chunks)),
+            (unsigned)ceil_div(n, 256));
   hipLaunchKernelGGL(col_sum_bf16, grid, dim3(256), 0, s,
                      (const unsigned short*)a, out, m, n);
 }

@@ -33,7 +33,7 @@ void launch_upsample_bwd(const void*, void*, int, int, int, int, int,
 void launch_act_fwd(const void*, void*, long, int, float, hipStream_t);
 void launch_act_bwd(const void*, const void*, void*, long, int, float,
                     hipStream_t);
-void launch_col_sum(const void*, float*, int, int, hipStream_t);
+void launch_col_sum(const void*, float*, long, int, hipStream_t);
 void launch_bce_fwd(const void*, const void*, float*, long, hipStream_t);
 void launch_bce_bwd(const void*, const void*, void*, float, long, hipStream_t);
 void launch_softmax_xent_fwd(const void*, const void*, float*, void*, int,
@@ -212,7 +212,7 @@ torch::Tensor col_sum(torch::Tensor a) {
   check_bf16(a, "a");
   int64_t m = a.size(0), n = a.size(1);
   torch::Tensor out = torch::zeros({n}, a.options().dtype(torch::kFloat32));
-  launch_col_sum(a.data_ptr(), out.data_ptr<float>(), (int)m, (int)n,
+  launch_col_sum(a.data_ptr(), out.data_ptr<float>(), (long)m, (int)n,
                  cur_stream());
   return out;
 }

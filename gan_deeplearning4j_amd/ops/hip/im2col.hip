@@ -53,6 +53,37 @@ __global__ void im2col_nhwc(const unsigned short* __restrict__ in,
   }
 }
 
+// fast path: C % 8 == 0 -> move 8 bf16 (16B) per thread-iteration
+__global__ void im2col_nhwc_v8(const s16x8* __restrict__ in,
+                               s16x8* __restrict__ col, ConvGeom g) {
+  int kp8 = g.kpad / 8;
+  long total8 = (long)g.N * g.Ho * g.Wo * kp8;
+  long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  long stride = (long)gridDim.x * blockDim.x;
+  int rsc = g.R * g.S * g.C;
+  int c8 = g.C / 8;
+  for (; i < total8; i += stride) {
+    int k8 = (int)(i % kp8);
+    long np = i / kp8;
+    int k = k8 * 8;
+    s16x8 v = {0, 0, 0, 0, 0, 0, 0, 0};
+    if (k < rsc) {
+      int c = k % g.C;
+      int rs = k / g.C;
+      int s_ = rs % g.S, r = rs / g.S;
+      int wo = (int)(np % g.Wo);
+      long t = np / g.Wo;
+      int ho = (int)(t % g.Ho);
+      int n = (int)(t / g.Ho);
+      int hi = ho * g.stride - g.pad + r;
+      int wi = wo * g.stride - g.pad + s_;
+      if (hi >= 0 && hi < g.H && wi >= 0 && wi < g.W)
+        v = in[(((long)n * g.H + hi) * g.W + wi) * c8 + c / 8];
+    }
+    col[i] = v;
+  }
+}
+
 // dIn[n][h][w][c] = sum over (r,s) with ho = (h + pad - r)/stride integral
 // and in range of dcol[np(ho,wo)][(r*S+s)*C + c]
 // With fuse_bias_act != 0 this is the convT forward epilogue:
@@ -91,9 +122,60 @@ __global__ void col2im_nhwc(const unsigned short* __restrict__ dcol,
   }
 }
 
+// fast col2im: C % 8 == 0
+__global__ void col2im_nhwc_v8(const unsigned short* __restrict__ dcol,
+                               s16x8* __restrict__ din, ConvGeom g,
+                               const float* __restrict__ bias, int act,
+                               float slope) {
+  long total8 = (long)g.N * g.H * g.W * (g.C / 8);
+  int c8 = g.C / 8;
+  long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  long stride = (long)gridDim.x * blockDim.x;
+  for (; i < total8; i += stride) {
+    int cg = (int)(i % c8);
+    long t = i / c8;
+    int w = (int)(t % g.W);
+    t /= g.W;
+    int h = (int)(t % g.H);
+    int n = (int)(t / g.H);
+    float acc[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+    for (int r = 0; r < g.R; ++r) {
+      int hop = h + g.pad - r;
+      if (hop < 0 || hop % g.stride) continue;
+      int ho = hop / g.stride;
+      if (ho >= g.Ho) continue;
+      for (int s_ = 0; s_ < g.S; ++s_) {
+        int wop = w + g.pad - s_;
+        if (wop < 0 || wop % g.stride) continue;
+        int wo = wop / g.stride;
+        if (wo >= g.Wo) continue;
+        long np = ((long)n * g.Ho + ho) * g.Wo + wo;
+        s16x8 v = *(const s16x8*)(&dcol[np * g.kpad +
+                                        (r * g.S + s_) * g.C + cg * 8]);
+        #pragma unroll
+        for (int j = 0; j < 8; ++j) acc[j] += bf2f((unsigned short)v[j]);
+      }
+    }
+    s16x8 o;
+    #pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float b = bias != nullptr ? bias[cg * 8 + j] : 0.f;
+      o[j] = (short)f2bf(act_fwd(acc[j] + b, act, slope));
+    }
+    din[i] = o;
+  }
+}
+
 extern "C" {
 
 void launch_im2col(const void* in, void* col, ConvGeom g, hipStream_t s) {
+  if (g.C % 8 == 0) {
+    long total8 = (long)g.N * g.Ho * g.Wo * (g.kpad / 8);
+    int grid = (int)min((long)2048, (total8 + 255) / 256 + 1);
+    hipLaunchKernelGGL(im2col_nhwc_v8, dim3(grid), dim3(256), 0, s,
+                       (const s16x8*)in, (s16x8*)col, g);
+    return;
+  }
   long total = (long)g.N * g.Ho * g.Wo * g.kpad;
   int grid = (int)min((long)2048, (total / 4 + 255) / 256 + 1);
   hipLaunchKernelGGL(im2col_nhwc, dim3(grid), dim3(256), 0, s,
@@ -102,6 +184,16 @@ void launch_im2col(const void* in, void* col, ConvGeom g, hipStream_t s) {
 
 void launch_col2im(const void* dcol, void* din, ConvGeom g, const float* bias,
                    int act, float slope, hipStream_t s) {
+  // v8 path needs 16B-aligned channel groups in dcol rows: kpad and the
+  // (r*S+s)*C offsets must be multiples of 8 -> C % 8 == 0 && kpad % 8 == 0
+  if (g.C % 8 == 0 && g.kpad % 8 == 0) {
+    long total8 = (long)g.N * g.H * g.W * (g.C / 8);
+    int grid = (int)min((long)2048, (total8 + 255) / 256 + 1);
+    hipLaunchKernelGGL(col2im_nhwc_v8, dim3(grid), dim3(256), 0, s,
+                       (const unsigned short*)dcol, (s16x8*)din, g, bias, act,
+                       slope);
+    return;
+  }
   long total = (long)g.N * g.H * g.W * g.C;
   int grid = (int)min((long)2048, (total + 255) / 256 + 1);
   hipLaunchKernelGGL(col2im_nhwc, dim3(grid), dim3(256), 0, s,
