@@ -119,7 +119,7 @@ class _Linear(torch.autograd.Function):
             # x cols are zero so the sliced grad region is exact
             nout = wp.shape[0]
             kchunks = (xp.shape[0] + 63) // 64
-            sk = _splitk_for((nout + 63) // 64, (ctx.nin + 63) // 64, kchunks)
+            sk = _splitk_for((nout + 127) // 128, (ctx.nin + 127) // 128, kchunks)
             dw = ext.gemm_nt(dpre, xp, sk)[:, : ctx.nin]
             dw = dw.contiguous().to(ctx.dtypes[1])
         if ctx.has_bias and ctx.needs_input_grad[2]:
@@ -178,7 +178,7 @@ class _Conv2d(torch.autograd.Function):
             # wgrad (rows = N*Ho*Wo)
             col = ext.im2col(xh, N, H, W, C, Ho, Wo, R, S, stride, pad, kpad)
             npq = dpre.shape[0]
-            sk = _splitk_for((Kout + 63) // 64, (kpad + 63) // 64,
+            sk = _splitk_for((Kout + 127) // 128, (kpad + 127) // 128,
                              (npq + 63) // 64)
             dw = ext.gemm_nt(dpre, col, sk)[:, :rsc]
             dw = (dw.reshape(Kout, R, S, C).permute(0, 3, 1, 2)
@@ -258,7 +258,7 @@ class _ConvTranspose2d(torch.autograd.Function):
             # wgrad: dW2a[rsco][cin] = sum_np dcol[np][rsco]*x2d[np][cin];
             # padded x2d cols are zero -> sliced grad exact
             npq = dcol.shape[0]
-            sk = _splitk_for((rscop + 63) // 64, (Cin + 63) // 64,
+            sk = _splitk_for((rscop + 127) // 128, (Cin + 127) // 128,
                              (npq + 63) // 64)
             dw2a = ext.gemm_nt(dcol, x2d, sk)[:rsco, :Cin]
             dw = (dw2a.reshape(R, S, Cout, Cin).permute(3, 2, 0, 1)

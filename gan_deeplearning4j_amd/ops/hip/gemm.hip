@@ -157,86 +157,125 @@ __global__ __launch_bounds__(256, 2) void gemm_tn_bf16(
 // LDS: transposed tiles [64 rows][72 k-pitch] bf16 (pitch 144 B: 16B-aligned
 // b128 fragment reads, conflict-free across the 16-lane read groups).
 
-constexpr int NT_BM = 64, NT_BN = 64, NT_BK = 64, NT_PITCH = 72;
+constexpr int NT_BM = 128, NT_BN = 128, NT_BK = 64, NT_PITCH = 72;
 constexpr int NT_TILE_E = NT_BM * NT_PITCH;  // elements per LDS tile
+constexpr int NT_THREADS = 512;              // 8 waves: 2(m) x 4(n)
 
-// stage [k0..k0+63] x [c0..c0+63] of src[K][L] into ldsT[c][k] (zero-filled
-// outside K). Each thread: 8 passes of (2 scalar loads -> one b32 write).
-DEV_INLINE void nt_stage(const unsigned short* __restrict__ g, int k0, int K,
-                         int c0, int L, long ldl,
-                         unsigned short* __restrict__ lds) {
+// Register-staged transpose load of [k0..k0+63] x [c0..c0+127] of src[K][L]
+// into regs (zero-filled outside K), written later as ldsT[c][k] b32 pairs.
+// Thread t owns column c = t&127 and k-pairs {kq + 8p}, p=0..7.
+struct NtStageRegs {
+  ushort2 v[8];
+};
+
+DEV_INLINE void nt_load(const unsigned short* __restrict__ g, int k0, int K,
+                        int c0, int L, long ldl, NtStageRegs& r) {
   const int t = threadIdx.x;
-  const int c = t & 63;             // column within tile (-> LDS row)
-  const int k2 = (t >> 6) * 2;      // k pair base
+  const int c = t & 127;
+  const int kq = (t >> 7) * 2;      // 4 k-pair groups across 512 threads
   int gc = min(c0 + c, L - 1);
   #pragma unroll
   for (int p = 0; p < 8; ++p) {
-    int k = k2 + p * 8;
+    int k = kq + p * 8;
     unsigned short v0 = 0, v1 = 0;
     if (k0 + k < K) v0 = g[(long)(k0 + k) * ldl + gc];
     if (k0 + k + 1 < K) v1 = g[(long)(k0 + k + 1) * ldl + gc];
-    *(ushort2*)(&lds[c * NT_PITCH + k]) = make_ushort2(v0, v1);
+    r.v[p] = make_ushort2(v0, v1);
   }
 }
 
-__global__ __launch_bounds__(256, 2) void gemm_nt_bf16(
+DEV_INLINE void nt_write(const NtStageRegs& r,
+                         unsigned short* __restrict__ lds) {
+  const int t = threadIdx.x;
+  const int c = t & 127;
+  const int kq = (t >> 7) * 2;
+  #pragma unroll
+  for (int p = 0; p < 8; ++p)
+    *(ushort2*)(&lds[c * NT_PITCH + kq + p * 8]) = r.v[p];
+}
+
+__global__ __launch_bounds__(NT_THREADS, 1) void gemm_nt_bf16(
     const unsigned short* __restrict__ A, const unsigned short* __restrict__ B,
     float* __restrict__ C, int M, int N, int K, long lda, long ldb,
     int kchunks_per_block, int use_atomic) {
-  __shared__ unsigned short lds[2 * NT_TILE_E];
-  unsigned short* Al = lds;
-  unsigned short* Bl = lds + NT_TILE_E;
+  __shared__ unsigned short lds[4 * NT_TILE_E];  // [buf][A|B]
+  auto albuf = [&](int i) -> unsigned short* {
+    return lds + (i ? 2 * NT_TILE_E : 0);
+  };
+  auto blbuf = [&](int i) -> unsigned short* {
+    return lds + NT_TILE_E + (i ? 2 * NT_TILE_E : 0);
+  };
 
   const int m0 = blockIdx.x * NT_BM;
   const int n0 = blockIdx.y * NT_BN;
   const int wid = threadIdx.x >> 6;
   const int lane = threadIdx.x & 63;
-  const int wr = wid >> 1, wc = wid & 1;    // wave -> 32x32 sub-tile
+  const int wr = wid >> 2, wc = wid & 3;    // wave -> 64(m) x 32(n) sub-tile
   const int fr = lane & 15, fq = lane >> 4;
 
-  f32x4 acc[2][2];
+  f32x4 acc[4][2];
   #pragma unroll
-  for (int i = 0; i < 2; ++i)
+  for (int i = 0; i < 4; ++i)
     #pragma unroll
     for (int j = 0; j < 2; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
 
+  const int kchunks = (K + NT_BK - 1) / NT_BK;
   int kt0 = blockIdx.z * kchunks_per_block;
-  int kt1 = min(kt0 + kchunks_per_block, (K + NT_BK - 1) / NT_BK);
+  int kt1 = min(kt0 + kchunks_per_block, kchunks);
+  if (kt0 >= kt1) return;
 
+  NtStageRegs ra, rb;
+  nt_load(A, kt0 * NT_BK, K, m0, M, lda, ra);
+  nt_load(B, kt0 * NT_BK, K, n0, N, ldb, rb);
+  nt_write(ra, albuf(0));
+  nt_write(rb, blbuf(0));
+  __syncthreads();
+
+  int cur = 0;
   for (int kt = kt0; kt < kt1; ++kt) {
-    nt_stage(A, kt * NT_BK, K, m0, M, lda, Al);
-    nt_stage(B, kt * NT_BK, K, n0, N, ldb, Bl);
-    __syncthreads();
+    // issue next tile's loads early: HBM latency hides under the MFMAs
+    if (kt + 1 < kt1) {
+      nt_load(A, (kt + 1) * NT_BK, K, m0, M, lda, ra);
+      nt_load(B, (kt + 1) * NT_BK, K, n0, N, ldb, rb);
+    }
+    const unsigned short* Al = albuf(cur);
+    const unsigned short* Bl = blbuf(cur);
     #pragma unroll
     for (int kc = 0; kc < 2; ++kc) {
-      bf16x8 a[2], b[2];
+      bf16x8 a[4], b[2];
       #pragma unroll
-      for (int mi = 0; mi < 2; ++mi)
-        a[mi] = *(const bf16x8*)(&Al[(wr * 32 + mi * 16 + fr) * NT_PITCH +
+      for (int mi = 0; mi < 4; ++mi)
+        a[mi] = *(const bf16x8*)(&Al[(wr * 64 + mi * 16 + fr) * NT_PITCH +
                                      kc * 32 + fq * 8]);
       #pragma unroll
       for (int ni = 0; ni < 2; ++ni)
         b[ni] = *(const bf16x8*)(&Bl[(wc * 32 + ni * 16 + fr) * NT_PITCH +
                                      kc * 32 + fq * 8]);
       #pragma unroll
-      for (int mi = 0; mi < 2; ++mi)
+      for (int mi = 0; mi < 4; ++mi)
         #pragma unroll
         for (int ni = 0; ni < 2; ++ni)
           acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               a[mi], b[ni], acc[mi][ni], 0, 0, 0);
     }
+    if (kt + 1 < kt1) {
+      // disjoint buffer: no barrier needed before the write pass
+      nt_write(ra, albuf(cur ^ 1));
+      nt_write(rb, blbuf(cur ^ 1));
+    }
     __syncthreads();
+    cur ^= 1;
   }
 
   #pragma unroll
-  for (int mi = 0; mi < 2; ++mi) {
+  for (int mi = 0; mi < 4; ++mi) {
     #pragma unroll
     for (int ni = 0; ni < 2; ++ni) {
       int col = n0 + wc * 32 + ni * 16 + fr;
       if (col >= N) continue;
       #pragma unroll
       for (int r = 0; r < 4; ++r) {
-        int row = m0 + wr * 32 + mi * 16 + fq * 4 + r;
+        int row = m0 + wr * 64 + mi * 16 + fq * 4 + r;
         if (row >= M) continue;
         if (use_atomic)
           atomicAdd(&C[(long)row * N + col], acc[mi][ni][r]);
@@ -263,9 +302,10 @@ void launch_gemm_tn(const void* A, const void* B, void* C_bf16, float* C_f32,
 void launch_gemm_nt(const void* A, const void* B, float* C, int M, int N,
                     int K, long lda, long ldb, int splitk, hipStream_t s) {
   int kchunks = (K + NT_BK - 1) / NT_BK;
+  if (splitk > kchunks) splitk = kchunks;
   int per_block = (kchunks + splitk - 1) / splitk;
   dim3 grid(ceil_div(M, NT_BM), ceil_div(N, NT_BN), splitk);
-  hipLaunchKernelGGL(gemm_nt_bf16, grid, dim3(256), 0, s,
+  hipLaunchKernelGGL(gemm_nt_bf16, grid, dim3(NT_THREADS), 0, s,
                      (const unsigned short*)A, (const unsigned short*)B, C, M,
                      N, K, lda, ldb, per_block, splitk > 1 ? 1 : 0);
 }

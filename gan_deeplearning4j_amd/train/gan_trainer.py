@@ -152,9 +152,11 @@ class GanTrainer:
         for p in self.dis.parameters():
             p.requires_grad_(True)
 
+        # losses stay on-device (no .item() sync in the hot loop); callers
+        # float() them when they actually need host values
         return {
-            "loss_d": float(loss_d.detach()),
-            "loss_g": float(loss_g.detach()),
+            "loss_d": loss_d.detach(),
+            "loss_g": loss_g.detach(),
             "images": n,
         }
 

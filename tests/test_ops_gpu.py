@@ -365,7 +365,7 @@ def test_dcgan28_trainer_step_gpu():
     import math
 
     for v in (out1["loss_d"], out1["loss_g"], out2["loss_d"], out2["loss_g"]):
-        assert math.isfinite(v)
+        assert math.isfinite(float(v))
 
 
 def test_reference_protocol_iteration_gpu(tmp_path):

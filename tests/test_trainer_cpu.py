@@ -35,7 +35,7 @@ def test_fast_trainer_mlp():
     x = torch.rand(8, cfg.data.num_features)
     out1 = tr.step(x)
     out2 = tr.step(x)
-    assert np.isfinite(out1["loss_d"]) and np.isfinite(out2["loss_g"])
+    assert np.isfinite(float(out1["loss_d"])) and np.isfinite(float(out2["loss_g"]))
     grid = tr.sample_grid(4)
     assert grid.shape[0] == 16
 
@@ -48,7 +48,7 @@ def test_fast_trainer_dcgan28_step():
     tr = GanTrainer(gen, dis, cfg, device=torch.device("cpu"))
     real = torch.rand(4, 1, 28, 28) * 2 - 1
     out = tr.step(real)
-    assert np.isfinite(out["loss_d"])
+    assert np.isfinite(float(out["loss_d"]))
     # G step must not have updated D
     # (D was stop-grad frozen during G step; its updater ran only once)
     assert tr.dis.updater.t == 1
