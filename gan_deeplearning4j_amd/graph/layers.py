@@ -280,9 +280,21 @@ class FeedForwardToCnnPreProcessor(BaseLayer):
 
 
 class CnnToFeedForwardPreProcessor(BaseLayer):
-    """[N, C, H, W] -> [N, C*H*W] (DL4J inserts this before dense layers)."""
+    """[N, C, H, W] -> [N, C*H*W] (DL4J inserts this before dense layers).
+
+    channels_last=True flattens in NHWC order instead (a free view when the
+    activations live in channels-last memory, as on the MFMA conv path) —
+    same semantics up to a fixed permutation of the dense layer's input
+    features. The reference-protocol graphs keep DL4J's C-major order.
+    """
+
+    def __init__(self, channels_last: bool = False):
+        super().__init__()
+        self.channels_last = channels_last
 
     def forward(self, x):
+        if self.channels_last and x.dim() == 4:
+            return x.permute(0, 2, 3, 1).reshape(x.shape[0], -1)
         return x.reshape(x.shape[0], -1)
 
     def out_shape(self, in_shape):

@@ -38,7 +38,7 @@ _LAYER_FIELDS = {
     "Upsampling2dLayer": ("scale",),
     "ActivationLayer": ("activation", "slope"),
     "FeedForwardToCnnPreProcessor": ("height", "width", "channels"),
-    "CnnToFeedForwardPreProcessor": (),
+    "CnnToFeedForwardPreProcessor": ("channels_last",),
     "ReshapeVertex": ("shape",),
 }
 
@@ -67,7 +67,8 @@ _CTOR_ARGS = {
                                                    d.get("slope", 0.2)),
     "FeedForwardToCnnPreProcessor": lambda d: L.FeedForwardToCnnPreProcessor(
         d["height"], d["width"], d["channels"]),
-    "CnnToFeedForwardPreProcessor": lambda d: L.CnnToFeedForwardPreProcessor(),
+    "CnnToFeedForwardPreProcessor": lambda d: L.CnnToFeedForwardPreProcessor(
+        d.get("channels_last", False)),
     "ReshapeVertex": lambda d: L.ReshapeVertex(*d["shape"]),
 }
 

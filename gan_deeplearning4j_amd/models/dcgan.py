@@ -78,7 +78,7 @@ def _dis_graph(cfg: GanConfig, stages: list[int], s_last: int) -> ComputationGra
         c_prev = c
     gb.add_layer("d_dense_feat", DenseLayer(c_prev * s_last * s_last, 1024,
                                             activation="lrelu", lr=lr), prev,
-                 preprocessor=CnnToFeedForwardPreProcessor())
+                 preprocessor=CnnToFeedForwardPreProcessor(channels_last=True))
     gb.add_layer("d_out", OutputLayer(1024, 1, activation="sigmoid",
                                       loss="xent", lr=lr), "d_dense_feat")
     gb.set_outputs("d_out")

@@ -47,9 +47,11 @@ def _pad_k(t: torch.Tensor) -> torch.Tensor:
 
 
 def _nhwc(x: torch.Tensor) -> torch.Tensor:
-    """[N,C,H,W] logical -> [N,H,W,C] contiguous (free if x is already a
-    channels-last view from an upstream op)."""
-    return x.permute(0, 2, 3, 1).contiguous()
+    """[N,C,H,W] logical -> [N,H,W,C] bf16 contiguous (free if x is already
+    a channels-last view from an upstream op). The cast happens BEFORE the
+    permute-contiguous so a channels-last input never takes an NCHW
+    round-trip copy."""
+    return x.to(torch.bfloat16).permute(0, 2, 3, 1).contiguous()
 
 
 def _as_nchw_view(y_nhwc: torch.Tensor) -> torch.Tensor:
@@ -150,7 +152,7 @@ class _Conv2d(torch.autograd.Function):
         Kout, _, R, S = w.shape
         Ho, Wo = _conv_out(H, R, stride, pad), _conv_out(W, S, stride, pad)
         kpad = _rup64(R * S * C)
-        xh = _nhwc(_bf(x))
+        xh = _nhwc(x)
         col = ext.im2col(xh, N, H, W, C, Ho, Wo, R, S, stride, pad, kpad)
         wp = _packed(w, "conv_wp", lambda: _pad_k(
             _bf(w.detach().permute(0, 2, 3, 1)).reshape(Kout, R * S * C)))
@@ -210,7 +212,7 @@ class _ConvTranspose2d(torch.autograd.Function):
         _, Cout, R, S = w.shape
         Ho = (Hi - 1) * stride - 2 * pad + R
         Wo = (Wi - 1) * stride - 2 * pad + S
-        xh = _nhwc(_bf(x))                       # [N,Hi,Wi,Cin]
+        xh = _nhwc(x)                       # [N,Hi,Wi,Cin]
         x2d = _pad_k(xh.reshape(-1, Cin))
         # B operand: [R*S*Cout, Cin] (k-contiguous over Cin)
         w2a = _packed(w, "w2a", lambda: _pad_k(
@@ -235,7 +237,7 @@ class _ConvTranspose2d(torch.autograd.Function):
         N, Cin, Hi, Wi, Cout, R, S, Ho, Wo, stride, pad = ctx.geom
         rsco = R * S * Cout
         rscop = _rup64(rsco)
-        dyh = _nhwc(_bf(dy))                       # [N,Ho,Wo,Cout]
+        dyh = _nhwc(dy)                       # [N,Ho,Wo,Cout]
         if ctx.act:
             dpre = ext.act_bwd(dyh.reshape(-1, Cout),
                                yh.reshape(-1, Cout), ctx.act, ctx.slope)
@@ -299,7 +301,7 @@ def batch_norm(x, weight, bias, running_mean, running_var, training,
                momentum=0.1, eps=1e-5):
     is4d = x.dim() == 4
     if is4d:
-        xh = _nhwc(_bf(x))
+        xh = _nhwc(x)
         x2 = xh.reshape(-1, xh.shape[-1])
     else:
         x2 = _bf(x)
@@ -321,7 +323,7 @@ class _MaxPool2d(torch.autograd.Function):
     def forward(ctx, x, kernel: int, stride: int):
         ext = hip_ext()
         N, C, H, W = x.shape
-        xh = _nhwc(_bf(x))
+        xh = _nhwc(x)
         out, argmax = ext.maxpool_fwd(xh, N, H, W, C, kernel, stride)
         ctx.save_for_backward(argmax)
         ctx.geom = (N, C, H, W, kernel, stride)
@@ -333,7 +335,7 @@ class _MaxPool2d(torch.autograd.Function):
         ext = hip_ext()
         (argmax,) = ctx.saved_tensors
         N, C, H, W, kernel, stride = ctx.geom
-        dyh = _nhwc(_bf(dy))
+        dyh = _nhwc(dy)
         din = ext.maxpool_bwd(dyh, argmax, N, H, W, C, kernel, stride)
         return _as_nchw_view(din).to(ctx.x_dtype), None, None
 
@@ -347,7 +349,7 @@ class _Upsample2d(torch.autograd.Function):
     def forward(ctx, x, scale: int):
         ext = hip_ext()
         N, C, H, W = x.shape
-        xh = _nhwc(_bf(x))
+        xh = _nhwc(x)
         out = ext.upsample_fwd(xh, N, H, W, C, scale)
         ctx.geom = (N, C, H, W, scale)
         ctx.x_dtype = x.dtype
@@ -357,7 +359,7 @@ class _Upsample2d(torch.autograd.Function):
     def backward(ctx, dy):
         ext = hip_ext()
         N, C, H, W, scale = ctx.geom
-        dyh = _nhwc(_bf(dy))
+        dyh = _nhwc(dy)
         din = ext.upsample_bwd(dyh, N, H, W, C, scale)
         return _as_nchw_view(din).to(ctx.x_dtype), None
 

@@ -8,19 +8,38 @@
 
 #include "common.h"
 
-// pass 1: per-channel sum and sum-of-squares
+// pass 1: per-channel sum and sum-of-squares.
+// Row-major coalesced with row-lanes (same scheme as col_sum_bf16):
+// a block covers a <=256-wide channel window; blockDim.x/colsW row-lanes
+// walk rows in parallel, LDS-reduce, one atomic pair per channel.
 __global__ void bn_stats(const unsigned short* __restrict__ x, long m, int c,
                          float* __restrict__ sum, float* __restrict__ sumsq) {
-  int col = blockIdx.y * blockDim.x + threadIdx.x;
-  if (col >= c) return;
+  __shared__ float ls[512];
+  int c0 = blockIdx.y * 256;
+  int colsW = min(256, c - c0);
+  int lanes = (int)blockDim.x / colsW;
+  int sub = (int)threadIdx.x / colsW;
+  int col = c0 + (int)threadIdx.x % colsW;
   float s = 0.f, ss = 0.f;
-  for (long r = blockIdx.x; r < m; r += gridDim.x) {
-    float v = bf2f(x[r * c + col]);
-    s += v;
-    ss += v * v;
+  if (sub < lanes) {
+    for (long r = (long)blockIdx.x * lanes + sub; r < m;
+         r += (long)gridDim.x * lanes) {
+      float v = bf2f(x[r * c + col]);
+      s += v;
+      ss += v * v;
+    }
   }
-  atomicAdd(&sum[col], s);
-  atomicAdd(&sumsq[col], ss);
+  ls[threadIdx.x] = s;
+  ls[256 + threadIdx.x] = ss;
+  __syncthreads();
+  if (sub == 0) {
+    for (int q = 1; q < lanes; ++q) {
+      s += ls[q * colsW + col - c0];
+      ss += ls[256 + q * colsW + col - c0];
+    }
+    atomicAdd(&sum[col], s);
+    atomicAdd(&sumsq[col], ss);
+  }
 }
 
 // finalize: mean/istd from sums; update running stats in-place (fp32)
@@ -77,24 +96,41 @@ __global__ void bn_apply_eval(const unsigned short* __restrict__ x,
   }
 }
 
-// backward pass 1: dgamma = sum dy*xhat, dbeta = sum dy
+// backward pass 1: dgamma = sum dy*xhat, dbeta = sum dy (coalesced scheme)
 __global__ void bn_bwd_reduce(const unsigned short* __restrict__ x,
                               const unsigned short* __restrict__ dy, long m,
                               int c, const float* __restrict__ mean,
                               const float* __restrict__ istd,
                               float* __restrict__ dgamma,
                               float* __restrict__ dbeta) {
-  int col = blockIdx.y * blockDim.x + threadIdx.x;
-  if (col >= c) return;
+  __shared__ float ls[512];
+  int c0 = blockIdx.y * 256;
+  int colsW = min(256, c - c0);
+  int lanes = (int)blockDim.x / colsW;
+  int sub = (int)threadIdx.x / colsW;
+  int col = c0 + (int)threadIdx.x % colsW;
   float dg = 0.f, db = 0.f;
-  for (long r = blockIdx.x; r < m; r += gridDim.x) {
-    float g = bf2f(dy[r * c + col]);
-    float xh = (bf2f(x[r * c + col]) - mean[col]) * istd[col];
-    dg += g * xh;
-    db += g;
+  if (sub < lanes) {
+    float mu = mean[col], is = istd[col];
+    for (long r = (long)blockIdx.x * lanes + sub; r < m;
+         r += (long)gridDim.x * lanes) {
+      float g = bf2f(dy[r * c + col]);
+      float xh = (bf2f(x[r * c + col]) - mu) * is;
+      dg += g * xh;
+      db += g;
+    }
   }
-  atomicAdd(&dgamma[col], dg);
-  atomicAdd(&dbeta[col], db);
+  ls[threadIdx.x] = dg;
+  ls[256 + threadIdx.x] = db;
+  __syncthreads();
+  if (sub == 0) {
+    for (int q = 1; q < lanes; ++q) {
+      dg += ls[q * colsW + col - c0];
+      db += ls[256 + q * colsW + col - c0];
+    }
+    atomicAdd(&dgamma[col], dg);
+    atomicAdd(&dbeta[col], db);
+  }
 }
 
 // backward pass 2 (training):
@@ -122,7 +158,10 @@ __global__ void bn_bwd_apply(const unsigned short* __restrict__ x,
 extern "C" {
 
 static dim3 _colgrid(long m, int c) {
-  return dim3((unsigned)min((long)256, max((long)1, m / 8)),
+  int colsW = c < 256 ? c : 256;
+  int lanes = 256 / colsW;
+  long chunks = (m + lanes - 1) / lanes;
+  return dim3((unsigned)min((long)1024, max((long)1, chunks)),
               (unsigned)((c + 255) / 256));
 }
 
