@@ -47,9 +47,12 @@ class BaseLayer(nn.Module):
     def set_param(self, key: str, value: torch.Tensor) -> None:
         t = self.get_param(key)
         with torch.no_grad():
-            getattr(self, self.PARAM_KEYS[key]).data.copy_(
-                value.to(t.device, t.dtype)
-            )
+            target = getattr(self, self.PARAM_KEYS[key])
+            target.data.copy_(value.to(t.device, t.dtype))
+            # .data.copy_ does not bump the version counter; drop any
+            # packed-weight cache so GPU GEMMs see the new values
+            if hasattr(target, "_gdlj_cache"):
+                del target._gdlj_cache
 
     def param_keys(self) -> list[str]:
         out = []

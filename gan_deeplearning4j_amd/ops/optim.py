@@ -100,6 +100,11 @@ class Updater:
             s._lazy_state(self.kind)
             if s.param.is_cuda:
                 self._step_gpu(s, g)
+                # the fused kernel writes params behind torch's back: the
+                # version counter does not move, so drop any packed-weight
+                # cache (gpu_ops._packed) explicitly
+                if hasattr(s.param, "_gdlj_cache"):
+                    del s.param._gdlj_cache
             else:
                 self._step_cpu(s, g)
 

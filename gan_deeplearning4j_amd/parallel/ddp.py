@@ -147,6 +147,8 @@ def average_parameters(module: torch.nn.Module):
         t = p.data.float()
         dist.all_reduce(t, op=dist.ReduceOp.SUM)
         p.data.copy_((t / world).to(p.dtype))
+        if hasattr(p, "_gdlj_cache"):
+            del p._gdlj_cache
 
 
 @torch.no_grad()
