@@ -56,7 +56,8 @@ def main():
     else:
         gen, dis = build_dcgan(cfg)
     dtype = torch.bfloat16 if use_gpu else torch.float32
-    tr = GanTrainer(gen, dis, cfg, device=device, dtype=dtype)
+    tr = GanTrainer(gen, dis, cfg, device=device, dtype=dtype,
+                    capture=use_gpu and os.environ.get("GDLJ_NO_CAPTURE") != "1")
 
     m = cfg.model
     # synthetic pixel-lattice-shaped data, random-init weights (no-network

@@ -61,8 +61,8 @@ def _as_nchw_view(y_nhwc: torch.Tensor) -> torch.Tensor:
 
 def _splitk_for(m_tiles: int, n_tiles: int, kchunks: int) -> int:
     tiles = max(1, m_tiles * n_tiles)
-    want = max(1, 512 // tiles)  # ~2 blocks per CU target
-    return int(min(want, kchunks, 64))
+    want = max(1, 512 // tiles)  # ~2 blocks per CU target (512-thread blocks)
+    return int(min(want, kchunks, 512))
 
 
 def _packed(w: torch.Tensor, key: str, builder):
@@ -441,14 +441,14 @@ def softmax_cross_entropy(logits, onehot):
 
 # =================================================================== updater
 def fused_update(kind, param, grad, master, m, v, lr, beta1, beta2, rms_decay,
-                 eps, clip, l2, t):
+                 eps, clip, l2, t, t_dev=None):
     ext = hip_ext()
     g = grad.contiguous()
     if kind == "adam":
         ext.fused_adam(param.view(-1), g.view(-1),
                        None if master is None else master.view(-1),
                        m.view(-1), v.view(-1), lr, beta1, beta2, eps, clip,
-                       l2, t)
+                       l2, t, t_dev)
     elif kind == "rmsprop":
         ext.fused_rmsprop(param.view(-1), g.view(-1),
                           None if master is None else master.view(-1),

@@ -55,7 +55,7 @@ void launch_bn_bwd_apply(const void*, const void*, void*, long, int,
                          const float*, const float*, hipStream_t);
 void launch_fused_adam(void*, const void*, float*, float*, float*, long, int,
                        int, float, float, float, float, float, float, int,
-                       hipStream_t);
+                       const int*, hipStream_t);
 void launch_fused_rmsprop(void*, const void*, float*, float*, long, int, int,
                           float, float, float, float, float, hipStream_t);
 }
@@ -107,7 +107,10 @@ torch::Tensor gemm_nt(torch::Tensor A, torch::Tensor B, int64_t splitk) {
   check_bf16(B, "B");
   int64_t K = A.size(0), M = A.size(1), N = B.size(1);
   TORCH_CHECK(B.size(0) == K, "K mismatch");
-  torch::Tensor C = torch::zeros({M, N}, A.options().dtype(torch::kFloat32));
+  // splitk==1 writes every element exactly once: skip the zero-fill pass
+  torch::Tensor C =
+      splitk > 1 ? torch::zeros({M, N}, A.options().dtype(torch::kFloat32))
+                 : torch::empty({M, N}, A.options().dtype(torch::kFloat32));
   launch_gemm_nt(A.data_ptr(), B.data_ptr(), C.data_ptr<float>(), (int)M,
                  (int)N, (int)K, M, N, (int)splitk, cur_stream());
   return C;
@@ -331,7 +334,8 @@ std::vector<torch::Tensor> bn_bwd(torch::Tensor x, torch::Tensor dy,
 void fused_adam(torch::Tensor param, torch::Tensor grad,
                 c10::optional<torch::Tensor> master, torch::Tensor m,
                 torch::Tensor v, double lr, double b1, double b2, double eps,
-                double clip, double l2, int64_t t) {
+                double clip, double l2, int64_t t,
+                c10::optional<torch::Tensor> t_dev) {
   TORCH_CHECK(param.is_cuda() && param.is_contiguous());
   int param_bf16 = param.scalar_type() == torch::kBFloat16;
   int grad_bf16 = grad.scalar_type() == torch::kBFloat16;
@@ -340,10 +344,16 @@ void fused_adam(torch::Tensor param, torch::Tensor grad,
     TORCH_CHECK(master.has_value(), "bf16 param needs fp32 master");
     master_p = master->data_ptr<float>();
   }
+  const int* t_p = nullptr;
+  if (t_dev.has_value() && t_dev->defined()) {
+    TORCH_CHECK(t_dev->scalar_type() == torch::kInt32, "t_dev must be int32");
+    t_p = t_dev->data_ptr<int>();
+  }
   launch_fused_adam(param.data_ptr(), grad.data_ptr(), master_p,
                     m.data_ptr<float>(), v.data_ptr<float>(), param.numel(),
                     grad_bf16, param_bf16, (float)lr, (float)b1, (float)b2,
-                    (float)eps, (float)clip, (float)l2, (int)t, cur_stream());
+                    (float)eps, (float)clip, (float)l2, (int)t, t_p,
+                    cur_stream());
 }
 
 void fused_rmsprop(torch::Tensor param, torch::Tensor grad,

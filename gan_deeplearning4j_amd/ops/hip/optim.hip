@@ -25,7 +25,15 @@ __global__ void fused_adam_k(void* __restrict__ param,
                              float* __restrict__ master,
                              float* __restrict__ m, float* __restrict__ v,
                              long n, float lr, float b1, float b2, float eps,
-                             float clip, float l2, float bc1, float bc2) {
+                             float clip, float l2, float bc1, float bc2,
+                             const int* __restrict__ t_dev) {
+  if (t_dev != nullptr) {
+    // hipGraph-replayable: step count lives on-device (host scalars are
+    // frozen at capture), so bias correction is computed here
+    float t = (float)*t_dev;
+    bc1 = 1.f - __powf(b1, t);
+    bc2 = 1.f - __powf(b2, t);
+  }
   long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
   for (; i < n; i += (long)gridDim.x * blockDim.x) {
     float w = BF16_PARAM ? master[i] : ((float*)param)[i];
@@ -75,26 +83,26 @@ extern "C" {
 void launch_fused_adam(void* param, const void* grad, float* master, float* m,
                        float* v, long n, int grad_is_bf16, int param_is_bf16,
                        float lr, float b1, float b2, float eps, float clip,
-                       float l2, int t, hipStream_t s) {
+                       float l2, int t, const int* t_dev, hipStream_t s) {
   float bc1 = 1.f - powf(b1, (float)t);
   float bc2 = 1.f - powf(b2, (float)t);
   int grid = (int)min((long)2048, (n + 255) / 256 + 1);
   if (grad_is_bf16 && param_is_bf16)
     hipLaunchKernelGGL((fused_adam_k<unsigned short, true>), dim3(grid),
                        dim3(256), 0, s, param, (const unsigned short*)grad,
-                       master, m, v, n, lr, b1, b2, eps, clip, l2, bc1, bc2);
+                       master, m, v, n, lr, b1, b2, eps, clip, l2, bc1, bc2, t_dev);
   else if (!grad_is_bf16 && param_is_bf16)
     hipLaunchKernelGGL((fused_adam_k<float, true>), dim3(grid), dim3(256), 0,
                        s, param, (const float*)grad, master, m, v, n, lr, b1,
-                       b2, eps, clip, l2, bc1, bc2);
+                       b2, eps, clip, l2, bc1, bc2, t_dev);
   else if (grad_is_bf16)
     hipLaunchKernelGGL((fused_adam_k<unsigned short, false>), dim3(grid),
                        dim3(256), 0, s, param, (const unsigned short*)grad,
-                       master, m, v, n, lr, b1, b2, eps, clip, l2, bc1, bc2);
+                       master, m, v, n, lr, b1, b2, eps, clip, l2, bc1, bc2, t_dev);
   else
     hipLaunchKernelGGL((fused_adam_k<float, false>), dim3(grid), dim3(256), 0,
                        s, param, (const float*)grad, master, m, v, n, lr, b1,
-                       b2, eps, clip, l2, bc1, bc2);
+                       b2, eps, clip, l2, bc1, bc2, t_dev);
 }
 
 void launch_fused_rmsprop(void* param, const void* grad, float* master,

@@ -60,6 +60,8 @@ class Updater:
         self.grad_clip = grad_clip
         self.l2 = l2
         self.t = 0
+        # device-side step counter (hipGraph-replayable bias correction)
+        self._t_dev: Optional[torch.Tensor] = None
 
     # ------------------------------------------------------------------
     @classmethod
@@ -93,6 +95,12 @@ class Updater:
     @torch.no_grad()
     def step(self):
         self.t += 1
+        if self.kind == "adam" and any(s.param.is_cuda for s in self.slots):
+            if self._t_dev is None:
+                dev = next(s.param.device for s in self.slots
+                           if s.param.is_cuda)
+                self._t_dev = torch.zeros(1, dtype=torch.int32, device=dev)
+            self._t_dev += 1
         for s in self.slots:
             g = s.param.grad
             if g is None or s.lr == 0.0:
@@ -148,6 +156,7 @@ class Updater:
             clip=self.grad_clip,
             l2=self.l2,
             t=self.t,
+            t_dev=self._t_dev,
         )
 
     # -------------------------------------------------------- state io

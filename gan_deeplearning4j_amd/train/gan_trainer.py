@@ -67,6 +67,7 @@ class GanTrainer:
         device: Optional[torch.device] = None,
         dtype: Optional[torch.dtype] = None,
         bucket_cap_mb: Optional[int] = None,
+        capture: bool = False,
     ):
         self.cfg = cfg
         self.device = device or torch.device(
@@ -98,6 +99,16 @@ class GanTrainer:
         self._soft_fake = None
         self._noise_std = std
         self.it = 0
+        # hipGraph capture of the whole training step (one replay per step;
+        # removes launch/dispatch host overhead). GPU-only; falls back to
+        # eager if capture fails.
+        self.capture = capture and self.device.type == "cuda"
+        self._graph = None
+        self._graph_failed = False
+        self._static_real = None
+        self._graph_out = None
+        if self.device.type == "cuda":
+            torch.cuda.manual_seed(cfg.train.seed + 1000 * get_rank())
 
     def _labels(self, n: int):
         if self._soft_real is None or self._soft_real.shape[0] != n:
@@ -111,12 +122,21 @@ class GanTrainer:
         return self._soft_real, self._soft_fake
 
     def sample_z(self, n: int) -> torch.Tensor:
+        if self.device.type == "cuda":
+            # device RNG: capture-aware (philox offset advances per replay)
+            return torch.randn(n, self.z_size, device=self.device,
+                               dtype=self.dtype)
         z = torch.randn(n, self.z_size, generator=self._g)
         return z.to(self.device, self.dtype)
 
     def step(self, real: torch.Tensor) -> dict:
         """One alternating D+G update on a batch of real images."""
         self.it += 1
+        if self.capture and not self._graph_failed:
+            return self._step_graphed(real)
+        return self._step_eager(real)
+
+    def _step_eager(self, real: torch.Tensor) -> dict:
         n = real.shape[0]
         real = real.to(self.device, self.dtype)
         soft_real, soft_fake = self._labels(n)
@@ -159,6 +179,47 @@ class GanTrainer:
             "loss_g": loss_g.detach(),
             "images": n,
         }
+
+    # ----------------------------------------------------- hipGraph path
+    def _invalidate_packed(self):
+        for g in (self.gen, self.dis):
+            for p in g.parameters():
+                if hasattr(p, "_gdlj_cache"):
+                    del p._gdlj_cache
+
+    def _step_graphed(self, real: torch.Tensor) -> dict:
+        if self._graph is None:
+            try:
+                self._capture(real)
+            except Exception as e:  # fall back to eager permanently
+                log.warning("hipGraph capture failed (%s); eager fallback", e)
+                self._graph_failed = True
+                self._graph = None
+                return self._step_eager(real)
+        self._static_real.copy_(real.to(self.device, self.dtype),
+                                non_blocking=True)
+        self._graph.replay()
+        out = dict(self._graph_out)
+        out["images"] = real.shape[0]
+        return out
+
+    def _capture(self, real: torch.Tensor):
+        # warm up state (updater slots, labels, allocator) on a side stream
+        self._static_real = real.to(self.device, self.dtype).clone()
+        s = torch.cuda.Stream()
+        s.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(s):
+            for _ in range(2):
+                self._step_eager(self._static_real)
+        torch.cuda.current_stream().wait_stream(s)
+        torch.cuda.synchronize()
+        # caches must be rebuilt INSIDE the capture so the re-pack kernels
+        # (reading freshly-updated params) are part of the recorded stream
+        self._invalidate_packed()
+        self._graph = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(self._graph):
+            self._graph_out = self._step_eager(self._static_real)
+        log.info("training step captured as hipGraph")
 
     @torch.no_grad()
     def sample_grid(self, n: int = 10) -> torch.Tensor:
