@@ -66,7 +66,7 @@ __global__ __launch_bounds__(256, 2) void gemm_tn_bf16(
     unsigned short* __restrict__ C, float* __restrict__ Cf,
     const float* __restrict__ bias, int M, int N, int K, long lda, long ldb,
     int act, float slope) {
-  __shared__ char lds[4 * TN_TILE_B];  // [buf][A|B] double-buffered
+  __shared__ __attribute__((aligned(128))) char lds[4 * TN_TILE_B];
   // buffer offsets (avoid LDS pointer arrays: static-initializer limitation)
   auto abuf = [&](int i) -> char* { return lds + (i ? 2 * TN_TILE_B : 0); };
   auto bbuf = [&](int i) -> char* {
@@ -129,7 +129,50 @@ __global__ __launch_bounds__(256, 2) void gemm_tn_bf16(
     __syncthreads();
   }
 
-  // epilogue: bias + activation, predicated stores
+  // ---- epilogue: bias + activation --------------------------------
+  // MFMA C-fragments are (16-col x 4-row) slivers per store: direct global
+  // stores touch 32 B per 128 B line (25% write efficiency). Stage the
+  // whole 128x128 tile through LDS (reusing the operand buffers) and store
+  // full 256 B rows, 16 B per lane, fully coalesced.
+  if (C != nullptr && (N & 7) == 0) {
+    unsigned short* ctile = (unsigned short*)lds;  // [128][128] bf16, 32 KiB
+    __syncthreads();  // K-loop LDS reads are done (post-barrier), reuse lds
+    #pragma unroll
+    for (int mi = 0; mi < 4; ++mi) {
+      #pragma unroll
+      for (int ni = 0; ni < 4; ++ni) {
+        int lc = wc * 64 + ni * 16 + fr;
+        float bv = bias != nullptr ? bias[min(n0 + lc, N - 1)] : 0.f;
+        #pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          int lr = wr * 64 + mi * 16 + fq * 4 + r;
+          ctile[lr * 128 + lc] = f2bf(act_fwd(acc[mi][ni][r] + bv, act,
+                                              slope));
+        }
+      }
+    }
+    __syncthreads();
+    const int t = threadIdx.x;
+    #pragma unroll
+    for (int i = 0; i < 8; ++i) {
+      int piece = i * 256 + t;       // 2048 16B pieces = 128 rows x 16
+      int row = piece >> 4;
+      int seg = piece & 15;
+      int grow = m0 + row;
+      int gcol = n0 + seg * 8;
+      if (grow < M && gcol < N) {
+        s16x8 v = *(const s16x8*)(ctile + row * 128 + seg * 8);
+        if (gcol + 8 <= N) {
+          *(s16x8*)(&C[(long)grow * N + gcol]) = v;
+        } else {
+          for (int j = 0; j < 8 && gcol + j < N; ++j)
+            C[(long)grow * N + gcol + j] = (unsigned short)v[j];
+        }
+      }
+    }
+    return;
+  }
+  // fallback (fp32 out or N not a multiple of 8): direct predicated stores
   #pragma unroll
   for (int mi = 0; mi < 4; ++mi) {
     #pragma unroll
@@ -198,7 +241,7 @@ __global__ __launch_bounds__(NT_THREADS, 1) void gemm_nt_bf16(
     const unsigned short* __restrict__ A, const unsigned short* __restrict__ B,
     float* __restrict__ C, int M, int N, int K, long lda, long ldb,
     int kchunks_per_block, int use_atomic) {
-  __shared__ unsigned short lds[4 * NT_TILE_E];  // [buf][A|B]
+  __shared__ __attribute__((aligned(128))) unsigned short lds[4 * NT_TILE_E];
   auto albuf = [&](int i) -> unsigned short* {
     return lds + (i ? 2 * NT_TILE_E : 0);
   };
