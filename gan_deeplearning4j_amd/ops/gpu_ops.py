@@ -65,6 +65,26 @@ def _splitk_for(m_tiles: int, n_tiles: int, kchunks: int) -> int:
     return int(min(want, kchunks, 512))
 
 
+_zero_pages: dict = {}
+_dims_cache: dict = {}
+
+
+def _zp(device) -> torch.Tensor:
+    """16B zero page for gathered global_load_lds (OOB/pad redirect)."""
+    key = str(device)
+    if key not in _zero_pages:
+        _zero_pages[key] = torch.zeros(16, dtype=torch.bfloat16,
+                                       device=device)
+    return _zero_pages[key]
+
+
+def _dims(*vals) -> torch.Tensor:
+    """Cached CPU int64 tensor of conv gather dims."""
+    if vals not in _dims_cache:
+        _dims_cache[vals] = torch.tensor(vals, dtype=torch.int64)
+    return _dims_cache[vals]
+
+
 def _packed(w: torch.Tensor, key: str, builder):
     """Per-parameter cache of derived weight layouts (padded/transposed/
     permuted copies), invalidated by the tensor's in-place version counter.
@@ -153,12 +173,18 @@ class _Conv2d(torch.autograd.Function):
         Ho, Wo = _conv_out(H, R, stride, pad), _conv_out(W, S, stride, pad)
         kpad = _rup64(R * S * C)
         xh = _nhwc(x)
-        col = ext.im2col(xh, N, H, W, C, Ho, Wo, R, S, stride, pad, kpad)
         wp = _packed(w, "conv_wp", lambda: _pad_k(
             _bf(w.detach().permute(0, 2, 3, 1)).reshape(Kout, R * S * C)))
         bias = (_packed(b, "f32", lambda: b.detach().float().contiguous())
                 if b is not None else None)
-        y2d = ext.gemm_tn(col, wp, bias, act, slope, False)
+        if C % 8 == 0:
+            # implicit GEMM: im2col gather fused into the MFMA staging
+            y2d = ext.conv_fwd_implicit(xh, wp, bias, _zp(x.device), N, H, W,
+                                        C, Ho, Wo, R, S, stride, pad, act,
+                                        slope)
+        else:
+            col = ext.im2col(xh, N, H, W, C, Ho, Wo, R, S, stride, pad, kpad)
+            y2d = ext.gemm_tn(col, wp, bias, act, slope, False)
         ctx.save_for_backward(xh, wp, y2d)
         ctx.geom = (N, C, H, W, Kout, R, S, Ho, Wo, stride, pad, kpad)
         ctx.act, ctx.slope = act, slope
@@ -178,11 +204,18 @@ class _Conv2d(torch.autograd.Function):
         dx = dw = db = None
         if ctx.needs_input_grad[1]:
             # wgrad (rows = N*Ho*Wo)
-            col = ext.im2col(xh, N, H, W, C, Ho, Wo, R, S, stride, pad, kpad)
             npq = dpre.shape[0]
             sk = _splitk_for((Kout + 127) // 128, (kpad + 127) // 128,
                              (npq + 63) // 64)
-            dw = ext.gemm_nt(dpre, col, sk)[:, :rsc]
+            if C % 8 == 0:
+                dw = ext.gemm_nt_implicit(
+                    dpre, xh, 2, Kout, kpad, npq,
+                    _dims(N, H, W, C, Ho, Wo, R, S, stride, pad), sk)
+                dw = dw[:, :rsc]
+            else:
+                col = ext.im2col(xh, N, H, W, C, Ho, Wo, R, S, stride, pad,
+                                 kpad)
+                dw = ext.gemm_nt(dpre, col, sk)[:, :rsc]
             dw = (dw.reshape(Kout, R, S, C).permute(0, 3, 1, 2)
                   .contiguous().to(ctx.dtypes[1]))
         if ctx.needs_input_grad[0]:
@@ -244,9 +277,15 @@ class _ConvTranspose2d(torch.autograd.Function):
             dpre_img = dpre.view(N, Ho, Wo, Cout)
         else:
             dpre_img = dyh
-        # dcol = im2col(dpre_img) over the input grid
-        dcol = ext.im2col(dpre_img.contiguous(), N, Ho, Wo, Cout, Hi, Wi, R,
-                          S, stride, pad, rscop)   # [NPin, rscop]
+        # dcol = im2col(dpre_img) over the input grid (fused into the GEMM
+        # staging when Cout is vectorizable; explicit buffer otherwise)
+        dpre_img = dpre_img.contiguous()
+        use_impl = Cout % 8 == 0
+        dcol = None
+        if not use_impl:
+            dcol = ext.im2col(dpre_img, N, Ho, Wo, Cout, Hi, Wi, R, S,
+                              stride, pad, rscop)   # [NPin, rscop]
+        npq = N * Hi * Wi
         dx = dw = db = None
         if ctx.needs_input_grad[0]:
             # dx = dcol @ w2b ; w2b = [Cin, R*S*Cout]
@@ -254,15 +293,26 @@ class _ConvTranspose2d(torch.autograd.Function):
                 w2a[:, :Cin].reshape(R, S, Cout, Cin)
                 .permute(3, 0, 1, 2).reshape(Cin, rsco)
             ))
-            dx2d = ext.gemm_tn(dcol, w2b, None, 0, 0.0, False)  # [NPin, Cin]
+            if use_impl:
+                dx2d = ext.conv_fwd_implicit(dpre_img, w2b, None,
+                                             _zp(dpre_img.device), N, Ho, Wo,
+                                             Cout, Hi, Wi, R, S, stride, pad,
+                                             0, 0.0)
+            else:
+                dx2d = ext.gemm_tn(dcol, w2b, None, 0, 0.0, False)
             dx = _as_nchw_view(dx2d.view(N, Hi, Wi, Cin)).to(ctx.dtypes[0])
         if ctx.needs_input_grad[1]:
             # wgrad: dW2a[rsco][cin] = sum_np dcol[np][rsco]*x2d[np][cin];
             # padded x2d cols are zero -> sliced grad exact
-            npq = dcol.shape[0]
             sk = _splitk_for((rscop + 127) // 128, (Cin + 127) // 128,
                              (npq + 63) // 64)
-            dw2a = ext.gemm_nt(dcol, x2d, sk)[:rsco, :Cin]
+            if use_impl:
+                dw2a = ext.gemm_nt_implicit(
+                    dpre_img, x2d, 1, rscop, x2d.shape[1], npq,
+                    _dims(N, Ho, Wo, Cout, Hi, Wi, R, S, stride, pad),
+                    sk)[:rsco, :Cin]
+            else:
+                dw2a = ext.gemm_nt(dcol, x2d, sk)[:rsco, :Cin]
             dw = (dw2a.reshape(R, S, Cout, Cin).permute(3, 2, 0, 1)
                   .contiguous().to(ctx.dtypes[1]))
         if ctx.has_bias and ctx.needs_input_grad[2]:

@@ -72,6 +72,42 @@ DEV_INLINE float wave_reduce_max(float v) {
 
 constexpr int ceil_div(int a, int b) { return (a + b - 1) / b; }
 
+// ------------------------------------------------- fast integer division
+// Granlund-Montgomery round-up magic; exact for 0 <= n < 2^31, 1 <= d < 2^16.
+struct FastDiv {
+  unsigned mul;
+  int shift;  // = 32 + ceil(log2 d)
+  int d;
+};
+
+inline FastDiv make_fastdiv(int d) {
+  int l = 0;
+  while ((1 << l) < d) ++l;
+  FastDiv f;
+  f.shift = 32 + l;
+  f.mul = (unsigned)(((1ULL << (32 + l)) + d - 1) / (unsigned long long)d);
+  f.d = d;
+  return f;
+}
+
+DEV_INLINE unsigned fdiv(unsigned n, FastDiv f) {
+  return (unsigned)(((unsigned long long)n * f.mul) >> f.shift);
+}
+
+DEV_INLINE unsigned fmod_(unsigned n, FastDiv f, unsigned q) {
+  return n - q * (unsigned)f.d;
+}
+
+// Conv gather geometry for implicit-GEMM staging: maps an im2col
+// coordinate (np, k) to an input-image address without materializing col.
+struct ConvGather {
+  int N, H, W, C;        // source image dims (NHWC)
+  int Ho, Wo;            // patch grid
+  int R, S, stride, pad;
+  int rsc;               // R*S*C (valid k range; >= rsc is zero padding)
+  FastDiv fC, fS, fWo, fHo;
+};
+
 #define HIP_CHECK_LAST()                                                     \
   do {                                                                       \
     hipError_t e = hipGetLastError();                                        \

@@ -14,12 +14,47 @@ struct ConvGeom {
 struct PoolGeom {
   int N, H, W, C, Ho, Wo, k, stride;
 };
+struct FastDiv {
+  unsigned mul;
+  int shift;
+  int d;
+};
+struct ConvGather {
+  int N, H, W, C, Ho, Wo, R, S, stride, pad, rsc;
+  FastDiv fC, fS, fWo, fHo;
+};
+
+static FastDiv make_fastdiv_h(int d) {
+  int l = 0;
+  while ((1 << l) < d) ++l;
+  FastDiv f;
+  f.shift = 32 + l;
+  f.mul = (unsigned)(((1ULL << (32 + l)) + d - 1) / (unsigned long long)d);
+  f.d = d;
+  return f;
+}
+
+static ConvGather make_gather(int N, int H, int W, int C, int Ho, int Wo,
+                              int R, int S, int stride, int pad) {
+  ConvGather g;
+  g.N = N; g.H = H; g.W = W; g.C = C; g.Ho = Ho; g.Wo = Wo;
+  g.R = R; g.S = S; g.stride = stride; g.pad = pad;
+  g.rsc = R * S * C;
+  g.fC = make_fastdiv_h(C);
+  g.fS = make_fastdiv_h(S);
+  g.fWo = make_fastdiv_h(Wo);
+  g.fHo = make_fastdiv_h(Ho);
+  return g;
+}
 
 extern "C" {
 void launch_gemm_tn(const void*, const void*, void*, float*, const float*,
                     int, int, int, long, long, int, float, hipStream_t);
+void launch_gemm_tn_gather(const void*, const void*, void*, const float*,
+                           int, int, int, long, int, float, ConvGather,
+                           const void*, hipStream_t);
 void launch_gemm_nt(const void*, const void*, float*, int, int, int, long,
-                    long, int, hipStream_t);
+                    long, int, int, ConvGather, hipStream_t);
 void launch_im2col(const void*, void*, ConvGeom, hipStream_t);
 void launch_col2im(const void*, void*, ConvGeom, const float*, int, float,
                    hipStream_t);
@@ -111,8 +146,63 @@ torch::Tensor gemm_nt(torch::Tensor A, torch::Tensor B, int64_t splitk) {
   torch::Tensor C =
       splitk > 1 ? torch::zeros({M, N}, A.options().dtype(torch::kFloat32))
                  : torch::empty({M, N}, A.options().dtype(torch::kFloat32));
+  ConvGather dummy{};
   launch_gemm_nt(A.data_ptr(), B.data_ptr(), C.data_ptr<float>(), (int)M,
-                 (int)N, (int)K, M, N, (int)splitk, cur_stream());
+                 (int)N, (int)K, M, N, (int)splitk, 0, dummy, cur_stream());
+  return C;
+}
+
+// Implicit-GEMM conv forward: y2d[NP][Kout] = act(im2col(x).Wp^T + bias)
+// without materializing col. x NHWC [Nb,H,W,C] (C % 8 == 0), Wp [Kout][kpad].
+torch::Tensor conv_fwd_implicit(torch::Tensor x, torch::Tensor Wp,
+                                c10::optional<torch::Tensor> bias,
+                                torch::Tensor zero_page, int64_t Nb,
+                                int64_t H, int64_t W, int64_t C, int64_t Ho,
+                                int64_t Wo, int64_t R, int64_t S,
+                                int64_t stride, int64_t pad, int64_t act,
+                                double slope) {
+  check_bf16(x, "x");
+  check_bf16(Wp, "Wp");
+  TORCH_CHECK(C % 8 == 0, "implicit conv needs C % 8 == 0");
+  int64_t Kout = Wp.size(0), kpad = Wp.size(1);
+  int64_t M = Nb * Ho * Wo;
+  const float* bias_p = nullptr;
+  if (bias.has_value() && bias->defined() && bias->numel() > 0) {
+    check_f32(*bias, "bias");
+    bias_p = bias->data_ptr<float>();
+  }
+  torch::Tensor y = torch::empty({M, Kout}, x.options());
+  ConvGather g = make_gather((int)Nb, (int)H, (int)W, (int)C, (int)Ho,
+                             (int)Wo, (int)R, (int)S, (int)stride, (int)pad);
+  launch_gemm_tn_gather(x.data_ptr(), Wp.data_ptr(), y.data_ptr(), bias_p,
+                        (int)M, (int)Kout, (int)kpad, kpad, (int)act,
+                        (float)slope, g, zero_page.data_ptr(), cur_stream());
+  return y;
+}
+
+// Implicit weight-grad: C[M][N] += sum_np A'[np][M] * B'[np][N] where the
+// operand selected by gmode (1=A, 2=B) is the im2col of an NHWC image.
+torch::Tensor gemm_nt_implicit(torch::Tensor A, torch::Tensor B,
+                               int64_t gmode, int64_t Mdim, int64_t Ndim,
+                               int64_t Kdim, torch::Tensor img_dims,
+                               int64_t splitk) {
+  check_bf16(A, "A");
+  check_bf16(B, "B");
+  auto d = img_dims.cpu().contiguous();
+  TORCH_CHECK(d.numel() == 10, "img_dims = [N,H,W,C,Ho,Wo,R,S,stride,pad]");
+  const int64_t* p = d.data_ptr<int64_t>();
+  ConvGather g = make_gather((int)p[0], (int)p[1], (int)p[2], (int)p[3],
+                             (int)p[4], (int)p[5], (int)p[6], (int)p[7],
+                             (int)p[8], (int)p[9]);
+  torch::Tensor C =
+      splitk > 1
+          ? torch::zeros({Mdim, Ndim}, A.options().dtype(torch::kFloat32))
+          : torch::empty({Mdim, Ndim}, A.options().dtype(torch::kFloat32));
+  long lda = gmode == 1 ? 0 : Mdim;
+  long ldb = gmode == 2 ? 0 : Ndim;
+  launch_gemm_nt(A.data_ptr(), B.data_ptr(), C.data_ptr<float>(), (int)Mdim,
+                 (int)Ndim, (int)Kdim, lda, ldb, (int)splitk, (int)gmode, g,
+                 cur_stream());
   return C;
 }
 
@@ -379,6 +469,10 @@ void fused_rmsprop(torch::Tensor param, torch::Tensor grad,
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("gemm_tn", &gemm_tn, "C = act(A.B^T + bias) bf16 MFMA");
   mod.def("gemm_nt", &gemm_nt, "C = A^T.B (contraction over rows) fp32 out");
+  mod.def("conv_fwd_implicit", &conv_fwd_implicit,
+          "implicit-GEMM conv forward (gathered im2col A)");
+  mod.def("gemm_nt_implicit", &gemm_nt_implicit,
+          "weight-grad GEMM with one operand gathered as im2col");
   mod.def("im2col", &im2col);
   mod.def("col2im", &col2im);
   mod.def("maxpool_fwd", &maxpool_fwd);

@@ -7,15 +7,23 @@
 //            (A=im2col, B=weights) and data-grad (B=W^T).
 //            128x128x64 tile, 4 waves, double-buffered LDS staged with
 //            global_load_lds (16B), XOR-swizzled via the SOURCE address
-//            (guide rule 21), mfma_f32_16x16x32_bf16 inner loop.
+//            (guide rule 21), mfma_f32_16x16x32_bf16 inner loop, LDS-staged
+//            coalesced epilogue with fused bias+activation.
+//            GATHER_A variant = implicit-GEMM convolution: the A operand is
+//            gathered straight from the NHWC image (im2col fused into the
+//            staging address computation; out-of-bounds/padded chunks are
+//            redirected to a 16B zero page so global_load_lds stays
+//            unconditional).
 //
 //  gemm_nt:  C[M][N] (fp32) += sum_k A[k][M] * B[k][N]
 //            contraction over ROWS of both operands: weight-grad
-//            (A=dOut, B=im2col). 64x64x64 tile, register-staged LDS
-//            transpose, split-K over blockIdx.z with fp32 atomics.
+//            (A=dOut, B=im2col). 128x128x64 tile, 8 waves, double-buffered
+//            register staging (loads for tile t+1 issued before tile t's
+//            MFMAs), split-K over blockIdx.z with fp32 atomics. Either
+//            operand may be gathered from an NHWC image (implicit wgrad).
 //
-// K (contraction) must be a multiple of 64 for gemm_tn (callers pad via
-// im2col kpad / weight padding); gemm_nt handles arbitrary K by zero-fill.
+// K must be a multiple of 64 for gemm_tn (callers pad; gathered A pads by
+// zero-page); gemm_nt handles arbitrary K by zero-fill.
 
 #include "common.h"
 
@@ -26,22 +34,37 @@ typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
       (const __attribute__((address_space(1))) unsigned int*)(gsrc),          \
       (__attribute__((address_space(3))) unsigned int*)(ldst), 16, 0, 0)
 
+// decode np -> (n, ho, wo)
+DEV_INLINE void np_decode(const ConvGather& g, unsigned np, int& n, int& ho,
+                          int& wo) {
+  unsigned q1 = fdiv(np, g.fWo);
+  wo = (int)(np - q1 * g.Wo);
+  unsigned q2 = fdiv(q1, g.fHo);
+  ho = (int)(q1 - q2 * g.Ho);
+  n = (int)q2;
+}
+
+// decode k -> (r, s, c); caller checks k < rsc
+DEV_INLINE void k_decode(const ConvGather& g, unsigned k, int& r, int& s,
+                         int& c) {
+  unsigned rs = fdiv(k, g.fC);
+  c = (int)(k - rs * g.C);
+  unsigned rr = fdiv(rs, g.fS);
+  s = (int)(rs - rr * g.S);
+  r = (int)rr;
+}
+
 // ---------------------------------------------------------------------------
 // TN kernel
 // ---------------------------------------------------------------------------
-// LDS image per operand tile: [128 rows][64 k] bf16, rows 128 B, stored
-// lane-linear by glds; the XOR swizzle slot' = slot ^ (row&7) lives on the
-// global SOURCE address and on the ds_read byte offset (both-sides rule).
-
 constexpr int TN_BM = 128, TN_BN = 128, TN_BK = 64;
 constexpr int TN_TILE_B = TN_BM * TN_BK * 2;  // 16 KiB per operand tile
 
-// stage one operand tile (A or B): 8 chunks of 16B per thread? ->
-// tile = 128*64*2B = 16KB = 1024 chunks; 256 threads -> 4 glds each.
+// stage one plain operand tile: 1024 16B chunks, 4 glds per thread
 DEV_INLINE void tn_stage(const unsigned short* __restrict__ g, int row0,
                          int nrows, long ldk, int k0, char* lds) {
   const int t = threadIdx.x;
-  const int wid = t >> 6;   // wave id (LDS dst base must be wave-uniform)
+  const int wid = t >> 6;
   #pragma unroll
   for (int i = 0; i < 4; ++i) {
     int chunk = i * 256 + t;         // = row*8 + slot
@@ -55,19 +78,49 @@ DEV_INLINE void tn_stage(const unsigned short* __restrict__ g, int row0,
   }
 }
 
-// fragment read: lane holds rows (l&15), k-octet (l>>4) of a 16x32 subtile
+// implicit-GEMM stage: A rows are im2col rows gathered from the NHWC image
+DEV_INLINE void tn_stage_gather(const unsigned short* __restrict__ img,
+                                const ConvGather& g,
+                                const unsigned short* __restrict__ zp,
+                                int row0, int nrows, int k0, char* lds) {
+  const int t = threadIdx.x;
+  const int wid = t >> 6;
+  #pragma unroll
+  for (int i = 0; i < 4; ++i) {
+    int chunk = i * 256 + t;
+    int row = chunk >> 3;
+    int slot = chunk & 7;
+    int gslot = slot ^ (row & 7);
+    int np = min(row0 + row, nrows - 1);
+    int k = k0 + gslot * 8;
+    const unsigned short* src = zp;
+    if (k < g.rsc) {
+      int r, s, c, n, ho, wo;
+      k_decode(g, (unsigned)k, r, s, c);
+      np_decode(g, (unsigned)np, n, ho, wo);
+      int hi = ho * g.stride - g.pad + r;
+      int wi = wo * g.stride - g.pad + s;
+      if (hi >= 0 && hi < g.H && wi >= 0 && wi < g.W)
+        src = img + (((long)n * g.H + hi) * g.W + wi) * g.C + c;
+    }
+    char* dst = lds + (i * 256 + wid * 64) * 16;
+    GLDS16(src, dst);
+  }
+}
+
 DEV_INLINE bf16x8 tn_frag(const char* lds, int row, int kslot) {
   int byte = row * 128 + ((kslot ^ (row & 7)) * 16);
   return *(const bf16x8*)(lds + byte);
 }
 
-__global__ __launch_bounds__(256, 2) void gemm_tn_bf16(
+template <bool GATHER_A>
+__global__ __launch_bounds__(256, 2) void gemm_tn_core(
     const unsigned short* __restrict__ A, const unsigned short* __restrict__ B,
     unsigned short* __restrict__ C, float* __restrict__ Cf,
     const float* __restrict__ bias, int M, int N, int K, long lda, long ldb,
-    int act, float slope) {
+    int act, float slope, ConvGather ga,
+    const unsigned short* __restrict__ zp) {
   __shared__ __attribute__((aligned(128))) char lds[4 * TN_TILE_B];
-  // buffer offsets (avoid LDS pointer arrays: static-initializer limitation)
   auto abuf = [&](int i) -> char* { return lds + (i ? 2 * TN_TILE_B : 0); };
   auto bbuf = [&](int i) -> char* {
     return lds + TN_TILE_B + (i ? 2 * TN_TILE_B : 0);
@@ -86,8 +139,8 @@ __global__ __launch_bounds__(256, 2) void gemm_tn_bf16(
 
   const int wid = threadIdx.x >> 6;
   const int lane = threadIdx.x & 63;
-  const int wr = wid >> 1, wc = wid & 1;      // wave -> 64x64 sub-tile
-  const int fr = lane & 15, fq = lane >> 4;   // fragment row / k-octet
+  const int wr = wid >> 1, wc = wid & 1;
+  const int fr = lane & 15, fq = lane >> 4;
 
   f32x4 acc[4][4];
   #pragma unroll
@@ -96,7 +149,10 @@ __global__ __launch_bounds__(256, 2) void gemm_tn_bf16(
     for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
 
   const int ntiles = K / TN_BK;
-  tn_stage(A, m0, M, lda, 0, abuf(0));
+  if (GATHER_A)
+    tn_stage_gather(A, ga, zp, m0, M, 0, abuf(0));
+  else
+    tn_stage(A, m0, M, lda, 0, abuf(0));
   tn_stage(B, n0, N, ldb, 0, bbuf(0));
   asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
   __syncthreads();
@@ -104,13 +160,16 @@ __global__ __launch_bounds__(256, 2) void gemm_tn_bf16(
   for (int t = 0; t < ntiles; ++t) {
     int cur = t & 1;
     if (t + 1 < ntiles) {
-      tn_stage(A, m0, M, lda, (t + 1) * TN_BK, abuf(cur ^ 1));
+      if (GATHER_A)
+        tn_stage_gather(A, ga, zp, m0, M, (t + 1) * TN_BK, abuf(cur ^ 1));
+      else
+        tn_stage(A, m0, M, lda, (t + 1) * TN_BK, abuf(cur ^ 1));
       tn_stage(B, n0, N, ldb, (t + 1) * TN_BK, bbuf(cur ^ 1));
     }
     const char* Al = abuf(cur);
     const char* Bl = bbuf(cur);
     #pragma unroll
-    for (int kc = 0; kc < 2; ++kc) {      // two K=32 chunks per BK=64
+    for (int kc = 0; kc < 2; ++kc) {
       bf16x8 a[4], b[4];
       #pragma unroll
       for (int mi = 0; mi < 4; ++mi)
@@ -129,14 +188,10 @@ __global__ __launch_bounds__(256, 2) void gemm_tn_bf16(
     __syncthreads();
   }
 
-  // ---- epilogue: bias + activation --------------------------------
-  // MFMA C-fragments are (16-col x 4-row) slivers per store: direct global
-  // stores touch 32 B per 128 B line (25% write efficiency). Stage the
-  // whole 128x128 tile through LDS (reusing the operand buffers) and store
-  // full 256 B rows, 16 B per lane, fully coalesced.
+  // ---- epilogue: bias + activation, LDS-staged coalesced stores ------
   if (C != nullptr && (N & 7) == 0) {
-    unsigned short* ctile = (unsigned short*)lds;  // [128][128] bf16, 32 KiB
-    __syncthreads();  // K-loop LDS reads are done (post-barrier), reuse lds
+    unsigned short* ctile = (unsigned short*)lds;  // [128][128] bf16
+    __syncthreads();
     #pragma unroll
     for (int mi = 0; mi < 4; ++mi) {
       #pragma unroll
@@ -197,25 +252,20 @@ __global__ __launch_bounds__(256, 2) void gemm_tn_bf16(
 // ---------------------------------------------------------------------------
 // NT kernel (weight grad): C[M][N] += sum_k A[k][M]*B[k][N], fp32 out.
 // ---------------------------------------------------------------------------
-// LDS: transposed tiles [64 rows][72 k-pitch] bf16 (pitch 144 B: 16B-aligned
-// b128 fragment reads, conflict-free across the 16-lane read groups).
-
 constexpr int NT_BM = 128, NT_BN = 128, NT_BK = 64, NT_PITCH = 72;
-constexpr int NT_TILE_E = NT_BM * NT_PITCH;  // elements per LDS tile
+constexpr int NT_TILE_E = NT_BM * NT_PITCH;
 constexpr int NT_THREADS = 512;              // 8 waves: 2(m) x 4(n)
 
-// Register-staged transpose load of [k0..k0+63] x [c0..c0+127] of src[K][L]
-// into regs (zero-filled outside K), written later as ldsT[c][k] b32 pairs.
-// Thread t owns column c = t&127 and k-pairs {kq + 8p}, p=0..7.
 struct NtStageRegs {
   ushort2 v[8];
 };
 
+// plain: thread owns column c = t&127 (L dim), k-pairs {kq + 8p}
 DEV_INLINE void nt_load(const unsigned short* __restrict__ g, int k0, int K,
                         int c0, int L, long ldl, NtStageRegs& r) {
   const int t = threadIdx.x;
   const int c = t & 127;
-  const int kq = (t >> 7) * 2;      // 4 k-pair groups across 512 threads
+  const int kq = (t >> 7) * 2;
   int gc = min(c0 + c, L - 1);
   #pragma unroll
   for (int p = 0; p < 8; ++p) {
@@ -224,6 +274,40 @@ DEV_INLINE void nt_load(const unsigned short* __restrict__ g, int k0, int K,
     if (k0 + k < K) v0 = g[(long)(k0 + k) * ldl + gc];
     if (k0 + k + 1 < K) v1 = g[(long)(k0 + k + 1) * ldl + gc];
     r.v[p] = make_ushort2(v0, v1);
+  }
+}
+
+// gathered: operand rows are im2col rows of an NHWC image.
+// k (contraction) = np; c (tile column) = rsc coordinate, decoded once.
+DEV_INLINE void nt_load_gather(const unsigned short* __restrict__ img,
+                               const ConvGather& g, int k0, int K, int c0,
+                               NtStageRegs& out) {
+  const int t = threadIdx.x;
+  const int c = t & 127;
+  const int kq = (t >> 7) * 2;
+  int kk = c0 + c;                       // rsc coordinate (column)
+  int r = 0, s = 0, cc = 0;
+  bool valid_col = kk < g.rsc;
+  if (valid_col) k_decode(g, (unsigned)kk, r, s, cc);
+  #pragma unroll
+  for (int p = 0; p < 8; ++p) {
+    int k = kq + p * 8;
+    unsigned short v0 = 0, v1 = 0;
+    #pragma unroll
+    for (int u = 0; u < 2; ++u) {
+      int np = k0 + k + u;
+      unsigned short v = 0;
+      if (np < K && valid_col) {
+        int n, ho, wo;
+        np_decode(g, (unsigned)np, n, ho, wo);
+        int hi = ho * g.stride - g.pad + r;
+        int wi = wo * g.stride - g.pad + s;
+        if (hi >= 0 && hi < g.H && wi >= 0 && wi < g.W)
+          v = img[(((long)n * g.H + hi) * g.W + wi) * g.C + cc];
+      }
+      if (u == 0) v0 = v; else v1 = v;
+    }
+    out.v[p] = make_ushort2(v0, v1);
   }
 }
 
@@ -237,10 +321,12 @@ DEV_INLINE void nt_write(const NtStageRegs& r,
     *(ushort2*)(&lds[c * NT_PITCH + kq + p * 8]) = r.v[p];
 }
 
-__global__ __launch_bounds__(NT_THREADS, 1) void gemm_nt_bf16(
+// GMODE: 0 = plain/plain, 1 = A gathered, 2 = B gathered
+template <int GMODE>
+__global__ __launch_bounds__(NT_THREADS, 1) void gemm_nt_core(
     const unsigned short* __restrict__ A, const unsigned short* __restrict__ B,
     float* __restrict__ C, int M, int N, int K, long lda, long ldb,
-    int kchunks_per_block, int use_atomic) {
+    int kchunks_per_block, int use_atomic, ConvGather gg) {
   __shared__ __attribute__((aligned(128))) unsigned short lds[4 * NT_TILE_E];
   auto albuf = [&](int i) -> unsigned short* {
     return lds + (i ? 2 * NT_TILE_E : 0);
@@ -253,7 +339,7 @@ __global__ __launch_bounds__(NT_THREADS, 1) void gemm_nt_bf16(
   const int n0 = blockIdx.y * NT_BN;
   const int wid = threadIdx.x >> 6;
   const int lane = threadIdx.x & 63;
-  const int wr = wid >> 2, wc = wid & 3;    // wave -> 64(m) x 32(n) sub-tile
+  const int wr = wid >> 2, wc = wid & 3;
   const int fr = lane & 15, fq = lane >> 4;
 
   f32x4 acc[4][2];
@@ -268,18 +354,30 @@ __global__ __launch_bounds__(NT_THREADS, 1) void gemm_nt_bf16(
   if (kt0 >= kt1) return;
 
   NtStageRegs ra, rb;
-  nt_load(A, kt0 * NT_BK, K, m0, M, lda, ra);
-  nt_load(B, kt0 * NT_BK, K, n0, N, ldb, rb);
+  auto load_a = [&](int kt, NtStageRegs& r) {
+    if (GMODE == 1)
+      nt_load_gather(A, gg, kt * NT_BK, K, m0, r);
+    else
+      nt_load(A, kt * NT_BK, K, m0, M, lda, r);
+  };
+  auto load_b = [&](int kt, NtStageRegs& r) {
+    if (GMODE == 2)
+      nt_load_gather(B, gg, kt * NT_BK, K, n0, r);
+    else
+      nt_load(B, kt * NT_BK, K, n0, N, ldb, r);
+  };
+
+  load_a(kt0, ra);
+  load_b(kt0, rb);
   nt_write(ra, albuf(0));
   nt_write(rb, blbuf(0));
   __syncthreads();
 
   int cur = 0;
   for (int kt = kt0; kt < kt1; ++kt) {
-    // issue next tile's loads early: HBM latency hides under the MFMAs
     if (kt + 1 < kt1) {
-      nt_load(A, (kt + 1) * NT_BK, K, m0, M, lda, ra);
-      nt_load(B, (kt + 1) * NT_BK, K, n0, N, ldb, rb);
+      load_a(kt + 1, ra);
+      load_b(kt + 1, rb);
     }
     const unsigned short* Al = albuf(cur);
     const unsigned short* Bl = blbuf(cur);
@@ -302,7 +400,6 @@ __global__ __launch_bounds__(NT_THREADS, 1) void gemm_nt_bf16(
               a[mi], b[ni], acc[mi][ni], 0, 0, 0);
     }
     if (kt + 1 < kt1) {
-      // disjoint buffer: no barrier needed before the write pass
       nt_write(ra, albuf(cur ^ 1));
       nt_write(rb, blbuf(cur ^ 1));
     }
@@ -336,21 +433,46 @@ void launch_gemm_tn(const void* A, const void* B, void* C_bf16, float* C_f32,
                     const float* bias, int M, int N, int K, long lda, long ldb,
                     int act, float slope, hipStream_t s) {
   dim3 grid(ceil_div(M, TN_BM), ceil_div(N, TN_BN));
-  hipLaunchKernelGGL(gemm_tn_bf16, grid, dim3(256), 0, s,
+  ConvGather dummy{};
+  hipLaunchKernelGGL((gemm_tn_core<false>), grid, dim3(256), 0, s,
                      (const unsigned short*)A, (const unsigned short*)B,
                      (unsigned short*)C_bf16, C_f32, bias, M, N, K, lda, ldb,
-                     act, slope);
+                     act, slope, dummy, nullptr);
+}
+
+// implicit-GEMM conv forward / gathered-A TN: A is an NHWC image; M = the
+// number of patch positions; K = kpad (zero page covers k >= rsc).
+void launch_gemm_tn_gather(const void* img, const void* B, void* C_bf16,
+                           const float* bias, int M, int N, int K, long ldb,
+                           int act, float slope, ConvGather ga,
+                           const void* zero_page, hipStream_t s) {
+  dim3 grid(ceil_div(M, TN_BM), ceil_div(N, TN_BN));
+  hipLaunchKernelGGL((gemm_tn_core<true>), grid, dim3(256), 0, s,
+                     (const unsigned short*)img, (const unsigned short*)B,
+                     (unsigned short*)C_bf16, nullptr, bias, M, N, K, 0, ldb,
+                     act, slope, ga, (const unsigned short*)zero_page);
 }
 
 void launch_gemm_nt(const void* A, const void* B, float* C, int M, int N,
-                    int K, long lda, long ldb, int splitk, hipStream_t s) {
+                    int K, long lda, long ldb, int splitk, int gmode,
+                    ConvGather gg, hipStream_t s) {
   int kchunks = (K + NT_BK - 1) / NT_BK;
   if (splitk > kchunks) splitk = kchunks;
   int per_block = (kchunks + splitk - 1) / splitk;
   dim3 grid(ceil_div(M, NT_BM), ceil_div(N, NT_BN), splitk);
-  hipLaunchKernelGGL(gemm_nt_bf16, grid, dim3(NT_THREADS), 0, s,
-                     (const unsigned short*)A, (const unsigned short*)B, C, M,
-                     N, K, lda, ldb, per_block, splitk > 1 ? 1 : 0);
+  int ua = splitk > 1 ? 1 : 0;
+  if (gmode == 1)
+    hipLaunchKernelGGL((gemm_nt_core<1>), grid, dim3(NT_THREADS), 0, s,
+                       (const unsigned short*)A, (const unsigned short*)B, C,
+                       M, N, K, lda, ldb, per_block, ua, gg);
+  else if (gmode == 2)
+    hipLaunchKernelGGL((gemm_nt_core<2>), grid, dim3(NT_THREADS), 0, s,
+                       (const unsigned short*)A, (const unsigned short*)B, C,
+                       M, N, K, lda, ldb, per_block, ua, gg);
+  else
+    hipLaunchKernelGGL((gemm_nt_core<0>), grid, dim3(NT_THREADS), 0, s,
+                       (const unsigned short*)A, (const unsigned short*)B, C,
+                       M, N, K, lda, ldb, per_block, ua, gg);
 }
 
 }  // extern "C"
