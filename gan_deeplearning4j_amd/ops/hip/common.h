@@ -73,9 +73,11 @@ DEV_INLINE float wave_reduce_max(float v) {
 constexpr int ceil_div(int a, int b) { return (a + b - 1) / b; }
 
 // ------------------------------------------------- fast integer division
-// Granlund-Montgomery round-up magic; exact for 0 <= n < 2^31, 1 <= d < 2^16.
+// Granlund-Montgomery round-up magic; exact for 0 <= n < 2^30, 1 <= d < 2^16.
+// mul is 64-bit: for power-of-two (and small) divisors ceil(2^(32+l)/d)
+// exceeds 32 bits.
 struct FastDiv {
-  unsigned mul;
+  unsigned long long mul;
   int shift;  // = 32 + ceil(log2 d)
   int d;
 };
@@ -85,7 +87,8 @@ inline FastDiv make_fastdiv(int d) {
   while ((1 << l) < d) ++l;
   FastDiv f;
   f.shift = 32 + l;
-  f.mul = (unsigned)(((1ULL << (32 + l)) + d - 1) / (unsigned long long)d);
+  f.mul = ((1ULL << (32 + l)) + (unsigned long long)d - 1) /
+          (unsigned long long)d;
   f.d = d;
   return f;
 }

@@ -15,7 +15,7 @@ struct PoolGeom {
   int N, H, W, C, Ho, Wo, k, stride;
 };
 struct FastDiv {
-  unsigned mul;
+  unsigned long long mul;  // 64-bit: pow2 divisors overflow 32-bit magic
   int shift;
   int d;
 };
@@ -29,7 +29,8 @@ static FastDiv make_fastdiv_h(int d) {
   while ((1 << l) < d) ++l;
   FastDiv f;
   f.shift = 32 + l;
-  f.mul = (unsigned)(((1ULL << (32 + l)) + d - 1) / (unsigned long long)d);
+  f.mul = ((1ULL << (32 + l)) + (unsigned long long)d - 1) /
+          (unsigned long long)d;
   f.d = d;
   return f;
 }
