@@ -185,6 +185,34 @@ class ComputationGraph(nn.Module):
         return outs[0] if len(outs) == 1 else tuple(outs)
 
     @torch.no_grad()
+    def feed_forward(self, *inputs: torch.Tensor, upto: Optional[str] = None):
+        """Run the graph and return the activation of vertex `upto`
+        (feature extraction; DL4J ComputationGraph.feedForward analog).
+        With upto=None returns the dict of ALL vertex activations."""
+        was_training = self.training
+        self.eval()
+        try:
+            acts: dict[str, torch.Tensor] = {}
+            for name, x in zip(self.input_names, inputs):
+                t = self.input_types.get(name)
+                if t is not None and t.kind == "convolutional_flat" and x.dim() == 2:
+                    x = x.reshape(x.shape[0], t.channels, t.height, t.width)
+                acts[name] = x
+            for name in self._topo:
+                layer = self.layers[name]
+                srcs = [acts[s] for s in self._vertex_inputs[name]]
+                if name in self.preprocessors:
+                    srcs = [self.preprocessors[name](s) for s in srcs]
+                acts[name] = layer(*srcs)
+                if name == upto:
+                    return acts[name]
+        finally:
+            self.train(was_training)
+        if upto is not None:
+            raise KeyError(f"vertex {upto!r} not found")
+        return acts
+
+    @torch.no_grad()
     def output(self, *inputs: torch.Tensor) -> torch.Tensor:
         """Inference forward (reference gen.output(z), Java:420, 551):
         applies output-layer activations."""
