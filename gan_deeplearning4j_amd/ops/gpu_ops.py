@@ -68,6 +68,17 @@ def _splitk_for(m_tiles: int, n_tiles: int, kchunks: int) -> int:
 _zero_pages: dict = {}
 _dims_cache: dict = {}
 
+# FP8 conv-forward mode (BASELINE config 4: DCGAN-128 fp8 MFMA path).
+# When enabled, conv forwards with C % 16 == 0 quantize activations and
+# weights to e4m3 (per-tensor dynamic scale) and run the fp8 MFMA implicit
+# GEMM; backward stays bf16.
+FP8_CONV = False
+
+
+def set_fp8_conv(enabled: bool) -> None:
+    global FP8_CONV
+    FP8_CONV = bool(enabled)
+
 
 def _zp(device) -> torch.Tensor:
     """16B zero page for gathered global_load_lds (OOB/pad redirect)."""
@@ -177,7 +188,16 @@ class _Conv2d(torch.autograd.Function):
             _bf(w.detach().permute(0, 2, 3, 1)).reshape(Kout, R * S * C)))
         bias = (_packed(b, "f32", lambda: b.detach().float().contiguous())
                 if b is not None else None)
-        if C % 8 == 0:
+        if FP8_CONV and C % 16 == 0:
+            xq, _, ix = ext.fp8_quantize(xh)
+            wq, _, iw = _packed(w, "fp8", lambda: tuple(
+                ext.fp8_quantize(wp)))
+            zp8 = _packed(wp, "zp8", lambda: torch.zeros(
+                32, dtype=torch.uint8, device=x.device))
+            y2d = ext.conv_fwd_implicit_fp8(xq, wq, bias, ix, iw, zp8, N, H,
+                                            W, C, Ho, Wo, R, S, stride, pad,
+                                            act, slope)
+        elif C % 8 == 0:
             # implicit GEMM: im2col gather fused into the MFMA staging
             y2d = ext.conv_fwd_implicit(xh, wp, bias, _zp(x.device), N, H, W,
                                         C, Ho, Wo, R, S, stride, pad, act,

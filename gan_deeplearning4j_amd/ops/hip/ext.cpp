@@ -92,6 +92,13 @@ void launch_bn_bwd_apply(const void*, const void*, void*, long, int,
 void launch_fused_adam(void*, const void*, float*, float*, float*, long, int,
                        int, float, float, float, float, float, float, int,
                        const int*, hipStream_t);
+void launch_amax(const void*, long, unsigned*, hipStream_t);
+void launch_fp8_make_scale(const unsigned*, float*, float*, hipStream_t);
+void launch_quant_fp8(const void*, void*, long, const float*, hipStream_t);
+void launch_gemm_tn_fp8(const void*, const void*, void*, const float*,
+                        const float*, const float*, int, int, int, long,
+                        long, int, float, int, ConvGather, const void*,
+                        hipStream_t);
 void launch_fused_rmsprop(void*, const void*, float*, float*, long, int, int,
                           float, float, float, float, float, hipStream_t);
 }
@@ -421,6 +428,52 @@ std::vector<torch::Tensor> bn_bwd(torch::Tensor x, torch::Tensor dy,
   return {dx, dgamma, dbeta};
 }
 
+// -------------------------------------------------------------------- fp8
+std::vector<torch::Tensor> fp8_quantize(torch::Tensor x) {
+  // bf16 -> e4m3 with per-tensor dynamic amax scaling.
+  // Returns {u8 data, scale, inv_scale} (scale/inv stay on device).
+  check_bf16(x, "x");
+  TORCH_CHECK(x.numel() % 8 == 0, "fp8 quantize needs numel % 8 == 0");
+  auto f32 = x.options().dtype(torch::kFloat32);
+  torch::Tensor amax = torch::zeros({1}, x.options().dtype(torch::kInt32));
+  torch::Tensor scale = torch::empty({1}, f32), inv = torch::empty({1}, f32);
+  torch::Tensor y = torch::empty(x.sizes(), x.options().dtype(torch::kUInt8));
+  auto s = cur_stream();
+  launch_amax(x.data_ptr(), x.numel(), (unsigned*)amax.data_ptr(), s);
+  launch_fp8_make_scale((const unsigned*)amax.data_ptr(),
+                        scale.data_ptr<float>(), inv.data_ptr<float>(), s);
+  launch_quant_fp8(x.data_ptr(), y.data_ptr(), x.numel(),
+                   scale.data_ptr<float>(), s);
+  return {y, scale, inv};
+}
+
+torch::Tensor conv_fwd_implicit_fp8(
+    torch::Tensor xq, torch::Tensor wq, c10::optional<torch::Tensor> bias,
+    torch::Tensor inv_qx, torch::Tensor inv_qw, torch::Tensor zero_page,
+    int64_t Nb, int64_t H, int64_t W, int64_t C, int64_t Ho, int64_t Wo,
+    int64_t R, int64_t S, int64_t stride, int64_t pad, int64_t act,
+    double slope) {
+  TORCH_CHECK(xq.scalar_type() == torch::kUInt8 && xq.is_contiguous());
+  TORCH_CHECK(wq.scalar_type() == torch::kUInt8 && wq.is_contiguous());
+  TORCH_CHECK(C % 16 == 0, "fp8 implicit conv needs C % 16 == 0");
+  int64_t Kout = wq.size(0), kpad = wq.size(1);
+  int64_t M = Nb * Ho * Wo;
+  const float* bias_p = nullptr;
+  if (bias.has_value() && bias->defined() && bias->numel() > 0) {
+    check_f32(*bias, "bias");
+    bias_p = bias->data_ptr<float>();
+  }
+  torch::Tensor y = torch::empty({M, Kout},
+                                 xq.options().dtype(torch::kBFloat16));
+  ConvGather g = make_gather((int)Nb, (int)H, (int)W, (int)C, (int)Ho,
+                             (int)Wo, (int)R, (int)S, (int)stride, (int)pad);
+  launch_gemm_tn_fp8(xq.data_ptr(), wq.data_ptr(), y.data_ptr(), bias_p,
+                     inv_qx.data_ptr<float>(), inv_qw.data_ptr<float>(),
+                     (int)M, (int)Kout, (int)kpad, 0, kpad, (int)act,
+                     (float)slope, 1, g, zero_page.data_ptr(), cur_stream());
+  return y;
+}
+
 // ------------------------------------------------------------------ optim
 void fused_adam(torch::Tensor param, torch::Tensor grad,
                 c10::optional<torch::Tensor> master, torch::Tensor m,
@@ -474,6 +527,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
           "implicit-GEMM conv forward (gathered im2col A)");
   mod.def("gemm_nt_implicit", &gemm_nt_implicit,
           "weight-grad GEMM with one operand gathered as im2col");
+  mod.def("fp8_quantize", &fp8_quantize,
+          "bf16 -> e4m3 (OCP) with dynamic per-tensor scale");
+  mod.def("conv_fwd_implicit_fp8", &conv_fwd_implicit_fp8,
+          "implicit-GEMM conv forward on fp8 MFMA");
   mod.def("im2col", &im2col);
   mod.def("col2im", &col2im);
   mod.def("maxpool_fwd", &maxpool_fwd);

@@ -1,0 +1,318 @@
+// FP8 (OCP e4m3fn — gfx950 native, NOT the MI300X fnuz variant) conv path.
+//
+// Mixed-precision scheme: bf16 master activations/weights are quantized
+// per-tensor (dynamic amax scaling) to e4m3, the conv forward runs on
+// mfma_f32_16x16x32_fp8_fp8 with fp32 accumulation, and the epilogue
+// de-scales + bias + activation back to bf16. Backward stays bf16.
+//
+// Same implicit-GEMM structure as the bf16 TN kernel: 128x128x64 tiles,
+// double-buffered LDS staged by 16B global_load_lds with the im2col gather
+// fused into the source address (zero-page redirect for OOB/pad).
+
+#include "common.h"
+
+typedef __attribute__((ext_vector_type(8))) float f32x8;
+typedef long fp8x8;  // 8 e4m3 bytes = 2 VGPRs (MFMA operand)
+
+#define GLDS16(gsrc, ldst)                                                    \
+  __builtin_amdgcn_global_load_lds(                                          \
+      (const __attribute__((address_space(1))) unsigned int*)(gsrc),          \
+      (__attribute__((address_space(3))) unsigned int*)(ldst), 16, 0, 0)
+
+// ----------------------------------------------------------- quantization
+// amax via atomicMax on the positive-float bit pattern
+__global__ void amax_bf16(const s16x8* __restrict__ x, long n8,
+                          unsigned* __restrict__ amax_bits) {
+  long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  float m = 0.f;
+  for (; i < n8; i += (long)gridDim.x * blockDim.x) {
+    s16x8 v = x[i];
+    #pragma unroll
+    for (int j = 0; j < 8; ++j) m = fmaxf(m, fabsf(bf2f((unsigned short)v[j])));
+  }
+  m = wave_reduce_max(m);
+  if ((threadIdx.x & 63) == 0) {
+    union { float f; unsigned u; } c;
+    c.f = m;
+    atomicMax(amax_bits, c.u);  // positive floats order like their bits
+  }
+}
+
+// scale = 448/amax (e4m3 max normal 448); inv = 1/scale
+__global__ void fp8_make_scale(const unsigned* __restrict__ amax_bits,
+                               float* __restrict__ scale,
+                               float* __restrict__ inv) {
+  union { float f; unsigned u; } c;
+  c.u = *amax_bits;
+  float a = c.f;
+  float s = (a > 1e-20f) ? 448.f / a : 1.f;
+  *scale = s;
+  *inv = 1.f / s;
+}
+
+// bf16 -> e4m3 with scale (device scalar), vectorized by 8
+__global__ void quant_fp8(const s16x8* __restrict__ x,
+                          unsigned long long* __restrict__ y, long n8,
+                          const float* __restrict__ scale) {
+  float s = *scale;
+  long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  for (; i < n8; i += (long)gridDim.x * blockDim.x) {
+    s16x8 v = x[i];
+    unsigned lo = 0, hi = 0;
+    lo = __builtin_amdgcn_cvt_pk_fp8_f32(bf2f((unsigned short)v[0]) * s,
+                                         bf2f((unsigned short)v[1]) * s, lo,
+                                         false);
+    lo = __builtin_amdgcn_cvt_pk_fp8_f32(bf2f((unsigned short)v[2]) * s,
+                                         bf2f((unsigned short)v[3]) * s, lo,
+                                         true);
+    hi = __builtin_amdgcn_cvt_pk_fp8_f32(bf2f((unsigned short)v[4]) * s,
+                                         bf2f((unsigned short)v[5]) * s, hi,
+                                         false);
+    hi = __builtin_amdgcn_cvt_pk_fp8_f32(bf2f((unsigned short)v[6]) * s,
+                                         bf2f((unsigned short)v[7]) * s, hi,
+                                         true);
+    y[i] = ((unsigned long long)hi << 32) | lo;
+  }
+}
+
+// ------------------------------------------------------------ fp8 TN GEMM
+// A gathered from an fp8 NHWC image (implicit conv) or plain [M][K];
+// B plain fp8 [N][K]. LDS tiles [128 rows][64 k] fp8 = 8 KiB each.
+constexpr int F8_BM = 128, F8_BN = 128, F8_BK = 64;
+constexpr int F8_TILE_B = F8_BM * F8_BK;  // bytes (1B/elem)
+
+DEV_INLINE void f8_stage(const unsigned char* __restrict__ g, int row0,
+                         int nrows, long ldk, int k0, char* lds) {
+  const int t = threadIdx.x;
+  const int wid = t >> 6;
+  #pragma unroll
+  for (int i = 0; i < 2; ++i) {
+    int chunk = i * 256 + t;       // 512 chunks = row*4 + slot16
+    int row = chunk >> 2;
+    int slot = chunk & 3;
+    int gslot = slot ^ (row & 3);
+    int grow = min(row0 + row, nrows - 1);
+    const unsigned char* src = g + (long)grow * ldk + k0 + gslot * 16;
+    char* dst = lds + (i * 256 + wid * 64) * 16;
+    GLDS16(src, dst);
+  }
+}
+
+DEV_INLINE void k_decode_f8(const ConvGather& g, unsigned k, int& r, int& s,
+                            int& c) {
+  unsigned rs = fdiv(k, g.fC);
+  c = (int)(k - rs * g.C);
+  unsigned rr = fdiv(rs, g.fS);
+  s = (int)(rs - rr * g.S);
+  r = (int)rr;
+}
+
+// NOTE: 16-fp8 chunks must not straddle (r,s) boundaries -> requires
+// C % 16 == 0 (enforced by the binding).
+DEV_INLINE void f8_stage_gather(const unsigned char* __restrict__ img,
+                                const ConvGather& g,
+                                const unsigned char* __restrict__ zp,
+                                int row0, int nrows, int k0, char* lds) {
+  const int t = threadIdx.x;
+  const int wid = t >> 6;
+  #pragma unroll
+  for (int i = 0; i < 2; ++i) {
+    int chunk = i * 256 + t;
+    int row = chunk >> 2;
+    int slot = chunk & 3;
+    int gslot = slot ^ (row & 3);
+    int np = min(row0 + row, nrows - 1);
+    int k = k0 + gslot * 16;       // 16 fp8 channels per chunk
+    const unsigned char* src = zp;
+    if (k < g.rsc) {
+      int r, s, c, n, ho, wo;
+      k_decode_f8(g, (unsigned)k, r, s, c);
+      unsigned q1 = fdiv((unsigned)np, g.fWo);
+      wo = (int)((unsigned)np - q1 * g.Wo);
+      unsigned q2 = fdiv(q1, g.fHo);
+      ho = (int)(q1 - q2 * g.Ho);
+      n = (int)q2;
+      int hi = ho * g.stride - g.pad + r;
+      int wi = wo * g.stride - g.pad + s;
+      if (hi >= 0 && hi < g.H && wi >= 0 && wi < g.W)
+        src = img + (((long)n * g.H + hi) * g.W + wi) * g.C + c;
+    }
+    char* dst = lds + (i * 256 + wid * 64) * 16;
+    GLDS16(src, dst);
+  }
+}
+
+DEV_INLINE fp8x8 f8_frag(const char* lds, int row, int kslot8) {
+  // fragment = 8 fp8 at k-octet kslot8; 16B swizzle on slot16 = kslot8>>1
+  int s16 = (kslot8 >> 1) ^ (row & 3);
+  int byte = row * 64 + s16 * 16 + (kslot8 & 1) * 8;
+  return *(const fp8x8*)(lds + byte);
+}
+
+template <bool GATHER_A>
+__global__ __launch_bounds__(256, 2) void gemm_tn_fp8_core(
+    const unsigned char* __restrict__ A, const unsigned char* __restrict__ B,
+    unsigned short* __restrict__ C, const float* __restrict__ bias,
+    const float* __restrict__ inv_qa, const float* __restrict__ inv_qb,
+    int M, int N, int K, long lda, long ldb, int act, float slope,
+    ConvGather ga, const unsigned char* __restrict__ zp) {
+  __shared__ __attribute__((aligned(128))) char lds[4 * F8_TILE_B];
+  auto abuf = [&](int i) -> char* { return lds + (i ? 2 * F8_TILE_B : 0); };
+  auto bbuf = [&](int i) -> char* {
+    return lds + F8_TILE_B + (i ? 2 * F8_TILE_B : 0);
+  };
+
+  int nwgx = gridDim.x;
+  int bidx = blockIdx.x;
+  if (nwgx >= 8) {
+    int q = nwgx / 8, r = nwgx % 8;
+    int xcd = bidx % 8, idx = bidx / 8;
+    bidx = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + idx;
+  }
+  const int m0 = bidx * F8_BM;
+  const int n0 = blockIdx.y * F8_BN;
+  const int wid = threadIdx.x >> 6;
+  const int lane = threadIdx.x & 63;
+  const int wr = wid >> 1, wc = wid & 1;
+  const int fr = lane & 15, fq = lane >> 4;
+
+  f32x4 acc[4][4];
+  #pragma unroll
+  for (int i = 0; i < 4; ++i)
+    #pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  const int ntiles = K / F8_BK;
+  if (GATHER_A)
+    f8_stage_gather(A, ga, zp, m0, M, 0, abuf(0));
+  else
+    f8_stage(A, m0, M, lda, 0, abuf(0));
+  f8_stage(B, n0, N, ldb, 0, bbuf(0));
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __syncthreads();
+
+  for (int t = 0; t < ntiles; ++t) {
+    int cur = t & 1;
+    if (t + 1 < ntiles) {
+      if (GATHER_A)
+        f8_stage_gather(A, ga, zp, m0, M, (t + 1) * F8_BK, abuf(cur ^ 1));
+      else
+        f8_stage(A, m0, M, lda, (t + 1) * F8_BK, abuf(cur ^ 1));
+      f8_stage(B, n0, N, ldb, (t + 1) * F8_BK, bbuf(cur ^ 1));
+    }
+    const char* Al = abuf(cur);
+    const char* Bl = bbuf(cur);
+    #pragma unroll
+    for (int kc = 0; kc < 2; ++kc) {
+      fp8x8 a[4], b[4];
+      #pragma unroll
+      for (int mi = 0; mi < 4; ++mi)
+        a[mi] = f8_frag(Al, wr * 64 + mi * 16 + fr, kc * 4 + fq);
+      #pragma unroll
+      for (int ni = 0; ni < 4; ++ni)
+        b[ni] = f8_frag(Bl, wc * 64 + ni * 16 + fr, kc * 4 + fq);
+      #pragma unroll
+      for (int mi = 0; mi < 4; ++mi)
+        #pragma unroll
+        for (int ni = 0; ni < 4; ++ni)
+          acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8(
+              a[mi], b[ni], acc[mi][ni], 0, 0, 0);
+    }
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __syncthreads();
+  }
+
+  // de-scale + bias + act, LDS-staged coalesced stores (N % 8 == 0 assumed
+  // for the fast path; scalar fallback otherwise)
+  float descale = (*inv_qa) * (*inv_qb);
+  if ((N & 7) == 0) {
+    unsigned short* ctile = (unsigned short*)lds;
+    __syncthreads();
+    #pragma unroll
+    for (int mi = 0; mi < 4; ++mi) {
+      #pragma unroll
+      for (int ni = 0; ni < 4; ++ni) {
+        int lc = wc * 64 + ni * 16 + fr;
+        float bv = bias != nullptr ? bias[min(n0 + lc, N - 1)] : 0.f;
+        #pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          int lr = wr * 64 + mi * 16 + fq * 4 + r;
+          ctile[lr * 128 + lc] =
+              f2bf(act_fwd(acc[mi][ni][r] * descale + bv, act, slope));
+        }
+      }
+    }
+    __syncthreads();
+    const int t = threadIdx.x;
+    #pragma unroll
+    for (int i = 0; i < 8; ++i) {
+      int piece = i * 256 + t;
+      int row = piece >> 4;
+      int seg = piece & 15;
+      int grow = m0 + row;
+      int gcol = n0 + seg * 8;
+      if (grow < M && gcol + 8 <= N)
+        *(s16x8*)(&C[(long)grow * N + gcol]) =
+            *(const s16x8*)(ctile + row * 128 + seg * 8);
+    }
+    return;
+  }
+  #pragma unroll
+  for (int mi = 0; mi < 4; ++mi)
+    #pragma unroll
+    for (int ni = 0; ni < 4; ++ni) {
+      int col = n0 + wc * 64 + ni * 16 + fr;
+      if (col >= N) continue;
+      float bv = bias != nullptr ? bias[col] : 0.f;
+      #pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int row = m0 + wr * 64 + mi * 16 + fq * 4 + r;
+        if (row >= M) continue;
+        C[(long)row * N + col] =
+            f2bf(act_fwd(acc[mi][ni][r] * descale + bv, act, slope));
+      }
+    }
+}
+
+extern "C" {
+
+void launch_amax(const void* x, long n, unsigned* amax_bits, hipStream_t s) {
+  long n8 = n / 8;
+  int grid = (int)min((long)1024, n8 / 256 + 1);
+  hipLaunchKernelGGL(amax_bf16, dim3(grid), dim3(256), 0, s, (const s16x8*)x,
+                     n8, amax_bits);
+}
+
+void launch_fp8_make_scale(const unsigned* amax_bits, float* scale,
+                           float* inv, hipStream_t s) {
+  hipLaunchKernelGGL(fp8_make_scale, dim3(1), dim3(1), 0, s, amax_bits,
+                     scale, inv);
+}
+
+void launch_quant_fp8(const void* x, void* y, long n, const float* scale,
+                      hipStream_t s) {
+  long n8 = n / 8;
+  int grid = (int)min((long)2048, n8 / 256 + 1);
+  hipLaunchKernelGGL(quant_fp8, dim3(grid), dim3(256), 0, s, (const s16x8*)x,
+                     (unsigned long long*)y, n8, scale);
+}
+
+void launch_gemm_tn_fp8(const void* A, const void* B, void* C,
+                        const float* bias, const float* inv_qa,
+                        const float* inv_qb, int M, int N, int K, long lda,
+                        long ldb, int act, float slope, int gather,
+                        ConvGather ga, const void* zp, hipStream_t s) {
+  dim3 grid(ceil_div(M, F8_BM), ceil_div(N, F8_BN));
+  if (gather)
+    hipLaunchKernelGGL((gemm_tn_fp8_core<true>), grid, dim3(256), 0, s,
+                       (const unsigned char*)A, (const unsigned char*)B,
+                       (unsigned short*)C, bias, inv_qa, inv_qb, M, N, K, lda,
+                       ldb, act, slope, ga, (const unsigned char*)zp);
+  else
+    hipLaunchKernelGGL((gemm_tn_fp8_core<false>), grid, dim3(256), 0, s,
+                       (const unsigned char*)A, (const unsigned char*)B,
+                       (unsigned short*)C, bias, inv_qa, inv_qb, M, N, K, lda,
+                       ldb, act, slope, ga, (const unsigned char*)zp);
+}
+
+}  // extern "C"

@@ -109,6 +109,10 @@ class GanTrainer:
         self._graph_out = None
         if self.device.type == "cuda":
             torch.cuda.manual_seed(cfg.train.seed + 1000 * get_rank())
+            if cfg.model.dtype == "fp8":
+                from ..ops.gpu_ops import set_fp8_conv
+
+                set_fp8_conv(True)
 
     def _labels(self, n: int):
         if self._soft_real is None or self._soft_real.shape[0] != n:
