@@ -95,9 +95,15 @@ def main():
     n_gpus = world if use_gpu else 1
     global_batch = per_gpu_batch * world
     images_per_sec = global_batch * args.steps / elapsed
+    size = m.image_height
+    dt = "fp8-fwd/bf16-bwd" if (use_gpu and cfg.model.dtype == "fp8") else (
+        "bf16" if use_gpu else "fp32")
+    metric = (f"images/sec (whole node) DCGAN {size}x{size} "
+              f"{'bf16' if cfg.model.dtype != 'fp8' else 'fp8'}"
+              if args.arch != "mlp" else "samples/sec MLP GAN")
     if rank == 0:
         print(json.dumps({
-            "metric": "images/sec (whole node) DCGAN 64x64 bf16",
+            "metric": metric,
             "value": round(images_per_sec, 2),
             "unit": "images/s",
             "n_gpus": n_gpus,
@@ -107,7 +113,7 @@ def main():
             "higher_is_better": True,
             "scaling": "weak",
             "vs_baseline": None,
-            "dtype": "bf16" if use_gpu else "fp32",
+            "dtype": dt,
             "data": "synthetic",
             "config": {
                 "model": args.arch,

@@ -329,9 +329,18 @@ class ReferenceProtocolTrainer:
         loop, Java:568-569) — intent replicated, bug not."""
         n = self.cfg.train.num_gen_samples
         z = latent_grid(n, self.cfg.model.z_size, self.device)
-        imgs = self.gen.output(z).float().cpu().reshape(n * n, -1)
+        img4d = self.gen.output(z).float().cpu()
+        imgs = img4d.reshape(n * n, -1)
         path = self.out_dir / f"mnist_out_{idx}.csv"
         np.savetxt(path, imgs.numpy(), delimiter=",", fmt="%.6f")
+        # the notebook's tiled PNG (DCGAN_Generated_Images.png analog)
+        try:
+            from ..utils.imaging import save_image_grid
+
+            save_image_grid(img4d, self.out_dir / f"generated_grid_{idx}.png",
+                            nrow=n)
+        except Exception:  # pragma: no cover - matplotlib optional
+            pass
         return path
 
     @torch.no_grad()
