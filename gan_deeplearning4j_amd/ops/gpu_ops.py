@@ -46,6 +46,17 @@ def _pad_k(t: torch.Tensor) -> torch.Tensor:
     return out
 
 
+def _pad8(t: torch.Tensor) -> torch.Tensor:
+    """Zero-pad the last dim to a multiple of 8 (gemm_nt row alignment)."""
+    k = t.shape[-1]
+    kp = (k + 7) // 8 * 8
+    if kp == k:
+        return t.contiguous()
+    out = t.new_zeros(*t.shape[:-1], kp)
+    out[..., :k] = t
+    return out
+
+
 def _nhwc(x: torch.Tensor) -> torch.Tensor:
     """[N,C,H,W] logical -> [N,H,W,C] bf16 contiguous (free if x is already
     a channels-last view from an upstream op). The cast happens BEFORE the
@@ -153,8 +164,9 @@ class _Linear(torch.autograd.Function):
             nout = wp.shape[0]
             kchunks = (xp.shape[0] + 63) // 64
             sk = _splitk_for((nout + 127) // 128, (ctx.nin + 127) // 128, kchunks)
-            dw = ext.gemm_nt(dpre, xp, sk)[:, : ctx.nin]
-            dw = dw.contiguous().to(ctx.dtypes[1])
+            dpre8 = _pad8(dpre)
+            dw = ext.gemm_nt(dpre8, xp, sk, _zp(xp.device))
+            dw = dw[: nout, : ctx.nin].contiguous().to(ctx.dtypes[1])
         if ctx.has_bias and ctx.needs_input_grad[2]:
             db = ext.col_sum(dpre).to(ctx.dtypes[2])
         if dx is not None:
@@ -230,12 +242,13 @@ class _Conv2d(torch.autograd.Function):
             if C % 8 == 0:
                 dw = ext.gemm_nt_implicit(
                     dpre, xh, 2, Kout, kpad, npq,
-                    _dims(N, H, W, C, Ho, Wo, R, S, stride, pad), sk)
+                    _dims(N, H, W, C, Ho, Wo, R, S, stride, pad), sk,
+                    _zp(xh.device))
                 dw = dw[:, :rsc]
             else:
                 col = ext.im2col(xh, N, H, W, C, Ho, Wo, R, S, stride, pad,
                                  kpad)
-                dw = ext.gemm_nt(dpre, col, sk)[:, :rsc]
+                dw = ext.gemm_nt(dpre, col, sk, _zp(col.device))[:, :rsc]
             dw = (dw.reshape(Kout, R, S, C).permute(0, 3, 1, 2)
                   .contiguous().to(ctx.dtypes[1]))
         if ctx.needs_input_grad[0]:
@@ -330,9 +343,10 @@ class _ConvTranspose2d(torch.autograd.Function):
                 dw2a = ext.gemm_nt_implicit(
                     dpre_img, x2d, 1, rscop, x2d.shape[1], npq,
                     _dims(N, Ho, Wo, Cout, Hi, Wi, R, S, stride, pad),
-                    sk)[:rsco, :Cin]
+                    sk, _zp(x2d.device))[:rsco, :Cin]
             else:
-                dw2a = ext.gemm_nt(dcol, x2d, sk)[:rsco, :Cin]
+                dw2a = ext.gemm_nt(dcol, x2d, sk,
+                                   _zp(x2d.device))[:rsco, :Cin]
             dw = (dw2a.reshape(R, S, Cout, Cin).permute(3, 2, 0, 1)
                   .contiguous().to(ctx.dtypes[1]))
         if ctx.has_bias and ctx.needs_input_grad[2]:

@@ -55,7 +55,7 @@ void launch_gemm_tn_gather(const void*, const void*, void*, const float*,
                            int, int, int, long, int, float, ConvGather,
                            const void*, hipStream_t);
 void launch_gemm_nt(const void*, const void*, float*, int, int, int, long,
-                    long, int, int, ConvGather, hipStream_t);
+                    long, int, int, ConvGather, const void*, hipStream_t);
 void launch_im2col(const void*, void*, ConvGeom, hipStream_t);
 void launch_col2im(const void*, void*, ConvGeom, const float*, int, float,
                    hipStream_t);
@@ -144,19 +144,28 @@ torch::Tensor gemm_tn(torch::Tensor A, torch::Tensor B,
   return C;
 }
 
-torch::Tensor gemm_nt(torch::Tensor A, torch::Tensor B, int64_t splitk) {
-  // C[M][N] = sum_k A[k][M]*B[k][N], fp32 out
+torch::Tensor gemm_nt(torch::Tensor A, torch::Tensor B, int64_t splitk,
+                      c10::optional<torch::Tensor> zero_page) {
+  // C[M][N] = sum_k A[k][M]*B[k][N], fp32 out.
+  // Operand row pitches must be multiples of 8 elements (16B-aligned rows
+  // for global_load_lds staging).
   check_bf16(A, "A");
   check_bf16(B, "B");
   int64_t K = A.size(0), M = A.size(1), N = B.size(1);
   TORCH_CHECK(B.size(0) == K, "K mismatch");
+  TORCH_CHECK(M % 8 == 0 && N % 8 == 0,
+              "gemm_nt operand widths must be multiples of 8 (pad)");
+  torch::Tensor zp = zero_page.has_value() && zero_page->defined()
+                         ? *zero_page
+                         : torch::zeros({16}, A.options());
   // splitk==1 writes every element exactly once: skip the zero-fill pass
   torch::Tensor C =
       splitk > 1 ? torch::zeros({M, N}, A.options().dtype(torch::kFloat32))
                  : torch::empty({M, N}, A.options().dtype(torch::kFloat32));
   ConvGather dummy{};
   launch_gemm_nt(A.data_ptr(), B.data_ptr(), C.data_ptr<float>(), (int)M,
-                 (int)N, (int)K, M, N, (int)splitk, 0, dummy, cur_stream());
+                 (int)N, (int)K, M, N, (int)splitk, 0, dummy, zp.data_ptr(),
+                 cur_stream());
   return C;
 }
 
@@ -193,7 +202,7 @@ torch::Tensor conv_fwd_implicit(torch::Tensor x, torch::Tensor Wp,
 torch::Tensor gemm_nt_implicit(torch::Tensor A, torch::Tensor B,
                                int64_t gmode, int64_t Mdim, int64_t Ndim,
                                int64_t Kdim, torch::Tensor img_dims,
-                               int64_t splitk) {
+                               int64_t splitk, torch::Tensor zero_page) {
   check_bf16(A, "A");
   check_bf16(B, "B");
   auto d = img_dims.cpu().contiguous();
@@ -210,7 +219,7 @@ torch::Tensor gemm_nt_implicit(torch::Tensor A, torch::Tensor B,
   long ldb = gmode == 2 ? 0 : Ndim;
   launch_gemm_nt(A.data_ptr(), B.data_ptr(), C.data_ptr<float>(), (int)Mdim,
                  (int)Ndim, (int)Kdim, lda, ldb, (int)splitk, (int)gmode, g,
-                 cur_stream());
+                 zero_page.data_ptr(), cur_stream());
   return C;
 }
 

@@ -251,82 +251,107 @@ __global__ __launch_bounds__(256, 2) void gemm_tn_core(
 
 // ---------------------------------------------------------------------------
 // NT kernel (weight grad): C[M][N] += sum_k A[k][M]*B[k][N], fp32 out.
+//
+// Contraction runs over operand ROWS, so rows stage straight into LDS with
+// global_load_lds (fully coalesced) in a [kb][mb][4k][16m] window layout,
+// and MFMA fragments come out via ds_read_b64_tr_b16 (gfx950 hardware
+// transpose read: per 16-lane group, lane l receives column l of a [4][16]
+// row-major window — verified by tools/probe_tr16.hip).
 // ---------------------------------------------------------------------------
-constexpr int NT_BM = 128, NT_BN = 128, NT_BK = 64, NT_PITCH = 72;
-constexpr int NT_TILE_E = NT_BM * NT_PITCH;
+constexpr int NT_BM = 128, NT_BN = 128, NT_BK = 64;
+constexpr int NT_TILE_E = NT_BM * NT_BK;     // elements per operand tile
 constexpr int NT_THREADS = 512;              // 8 waves: 2(m) x 4(n)
 
-struct NtStageRegs {
-  ushort2 v[8];
-};
+typedef __attribute__((ext_vector_type(4))) __bf16 bf16x4;
 
-// plain: thread owns column c = t&127 (L dim), k-pairs {kq + 8p}
-DEV_INLINE void nt_load(const unsigned short* __restrict__ g, int k0, int K,
-                        int c0, int L, long ldl, NtStageRegs& r) {
+// chunk c (16B = 8 elems) of the window-linear LDS image:
+//   window w = c>>3 (64 elems: [4 rows k][16 cols m]), j = (c>>1)&3, h = c&1
+//   w = kb*(BM/16) + mb
+// source = operand row (k0 + kb*4 + j), cols (c0 + mb*16 + h*8 .. +8)
+DEV_INLINE void nt_stage(const unsigned short* __restrict__ g, int k0, int K,
+                         int c0, int L, long ldl,
+                         const unsigned short* __restrict__ zp, char* lds) {
   const int t = threadIdx.x;
-  const int c = t & 127;
-  const int kq = (t >> 7) * 2;
-  int gc = min(c0 + c, L - 1);
+  const int wid = t >> 6;
   #pragma unroll
-  for (int p = 0; p < 8; ++p) {
-    int k = kq + p * 8;
-    unsigned short v0 = 0, v1 = 0;
-    if (k0 + k < K) v0 = g[(long)(k0 + k) * ldl + gc];
-    if (k0 + k + 1 < K) v1 = g[(long)(k0 + k + 1) * ldl + gc];
-    r.v[p] = make_ushort2(v0, v1);
-  }
-}
-
-// gathered: operand rows are im2col rows of an NHWC image.
-// k (contraction) = np; c (tile column) = rsc coordinate, decoded once.
-DEV_INLINE void nt_load_gather(const unsigned short* __restrict__ img,
-                               const ConvGather& g, int k0, int K, int c0,
-                               NtStageRegs& out) {
-  const int t = threadIdx.x;
-  const int c = t & 127;
-  const int kq = (t >> 7) * 2;
-  int kk = c0 + c;                       // rsc coordinate (column)
-  int r = 0, s = 0, cc = 0;
-  bool valid_col = kk < g.rsc;
-  if (valid_col) k_decode(g, (unsigned)kk, r, s, cc);
-  #pragma unroll
-  for (int p = 0; p < 8; ++p) {
-    int k = kq + p * 8;
-    unsigned short v0 = 0, v1 = 0;
-    #pragma unroll
-    for (int u = 0; u < 2; ++u) {
-      int np = k0 + k + u;
-      unsigned short v = 0;
-      if (np < K && valid_col) {
-        int n, ho, wo;
-        np_decode(g, (unsigned)np, n, ho, wo);
-        int hi = ho * g.stride - g.pad + r;
-        int wi = wo * g.stride - g.pad + s;
-        if (hi >= 0 && hi < g.H && wi >= 0 && wi < g.W)
-          v = img[(((long)n * g.H + hi) * g.W + wi) * g.C + cc];
-      }
-      if (u == 0) v0 = v; else v1 = v;
+  for (int i = 0; i < 2; ++i) {
+    int c = i * 512 + t;
+    int w = c >> 3;
+    int j = (c >> 1) & 3;
+    int h = c & 1;
+    int kb = w >> 3;
+    int mb = w & 7;
+    int row = k0 + kb * 4 + j;
+    const unsigned short* src = zp;
+    if (row < K) {
+      int col = c0 + mb * 16 + h * 8;
+      col = min(col, max(0, L - 8));  // M-edge clamp (rows >= M not stored)
+      src = g + (long)row * ldl + col;
     }
-    out.v[p] = make_ushort2(v0, v1);
+    char* dst = lds + (i * 512 + wid * 64) * 16;
+    GLDS16(src, dst);
   }
 }
 
-DEV_INLINE void nt_write(const NtStageRegs& r,
-                         unsigned short* __restrict__ lds) {
+// gathered flavor: operand is the im2col of an NHWC image
+// (contraction row = np; tile column = rsc coordinate)
+DEV_INLINE void nt_stage_gather(const unsigned short* __restrict__ img,
+                                const ConvGather& g, int k0, int K, int c0,
+                                const unsigned short* __restrict__ zp,
+                                char* lds) {
   const int t = threadIdx.x;
-  const int c = t & 127;
-  const int kq = (t >> 7) * 2;
+  const int wid = t >> 6;
   #pragma unroll
-  for (int p = 0; p < 8; ++p)
-    *(ushort2*)(&lds[c * NT_PITCH + kq + p * 8]) = r.v[p];
+  for (int i = 0; i < 2; ++i) {
+    int c = i * 512 + t;
+    int w = c >> 3;
+    int j = (c >> 1) & 3;
+    int h = c & 1;
+    int kb = w >> 3;
+    int mb = w & 7;
+    int np = k0 + kb * 4 + j;
+    int kk = c0 + mb * 16 + h * 8;
+    const unsigned short* src = zp;
+    if (np < K && kk < g.rsc) {
+      int r, s, cc, n, ho, wo;
+      k_decode(g, (unsigned)kk, r, s, cc);
+      np_decode(g, (unsigned)np, n, ho, wo);
+      int hi = ho * g.stride - g.pad + r;
+      int wi = wo * g.stride - g.pad + s;
+      if (hi >= 0 && hi < g.H && wi >= 0 && wi < g.W)
+        src = img + (((long)n * g.H + hi) * g.W + wi) * g.C + cc;
+    }
+    char* dst = lds + (i * 512 + wid * 64) * 16;
+    GLDS16(src, dst);
+  }
+}
+
+// fragment: lane holds col m = (mb*16 + fr), k = kc*32 + fq*8 + (0..7)
+DEV_INLINE bf16x8 nt_tr_frag(const unsigned short* lds, int kc, int mb) {
+  const int l = threadIdx.x & 63;
+  const int g = l >> 4;
+  int kb = kc * 8 + 2 * g;
+  const unsigned short* a0 = lds + ((kb * 8 + mb) * 64 + (l & 15) * 4);
+  bf16x4 lo = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
+      (__attribute__((address_space(3))) bf16x4*)a0);
+  bf16x4 hi = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
+      (__attribute__((address_space(3))) bf16x4*)(a0 + 512));
+  bf16x8 r;
+  #pragma unroll
+  for (int j = 0; j < 4; ++j) {
+    r[j] = lo[j];
+    r[4 + j] = hi[j];
+  }
+  return r;
 }
 
 // GMODE: 0 = plain/plain, 1 = A gathered, 2 = B gathered
 template <int GMODE>
-__global__ __launch_bounds__(NT_THREADS, 1) void gemm_nt_core(
+__global__ __launch_bounds__(NT_THREADS, 2) void gemm_nt_core(
     const unsigned short* __restrict__ A, const unsigned short* __restrict__ B,
     float* __restrict__ C, int M, int N, int K, long lda, long ldb,
-    int kchunks_per_block, int use_atomic, ConvGather gg) {
+    int kchunks_per_block, int use_atomic, ConvGather gg,
+    const unsigned short* __restrict__ zp) {
   __shared__ __attribute__((aligned(128))) unsigned short lds[4 * NT_TILE_E];
   auto albuf = [&](int i) -> unsigned short* {
     return lds + (i ? 2 * NT_TILE_E : 0);
@@ -353,31 +378,29 @@ __global__ __launch_bounds__(NT_THREADS, 1) void gemm_nt_core(
   int kt1 = min(kt0 + kchunks_per_block, kchunks);
   if (kt0 >= kt1) return;
 
-  NtStageRegs ra, rb;
-  auto load_a = [&](int kt, NtStageRegs& r) {
+  auto stage_a = [&](int kt, char* buf) {
     if (GMODE == 1)
-      nt_load_gather(A, gg, kt * NT_BK, K, m0, r);
+      nt_stage_gather(A, gg, kt * NT_BK, K, m0, zp, buf);
     else
-      nt_load(A, kt * NT_BK, K, m0, M, lda, r);
+      nt_stage(A, kt * NT_BK, K, m0, M, lda, zp, buf);
   };
-  auto load_b = [&](int kt, NtStageRegs& r) {
+  auto stage_b = [&](int kt, char* buf) {
     if (GMODE == 2)
-      nt_load_gather(B, gg, kt * NT_BK, K, n0, r);
+      nt_stage_gather(B, gg, kt * NT_BK, K, n0, zp, buf);
     else
-      nt_load(B, kt * NT_BK, K, n0, N, ldb, r);
+      nt_stage(B, kt * NT_BK, K, n0, N, ldb, zp, buf);
   };
 
-  load_a(kt0, ra);
-  load_b(kt0, rb);
-  nt_write(ra, albuf(0));
-  nt_write(rb, blbuf(0));
+  stage_a(kt0, (char*)albuf(0));
+  stage_b(kt0, (char*)blbuf(0));
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
   __syncthreads();
 
-  int cur = 0;
   for (int kt = kt0; kt < kt1; ++kt) {
+    int cur = (kt - kt0) & 1;
     if (kt + 1 < kt1) {
-      load_a(kt + 1, ra);
-      load_b(kt + 1, rb);
+      stage_a(kt + 1, (char*)albuf(cur ^ 1));
+      stage_b(kt + 1, (char*)blbuf(cur ^ 1));
     }
     const unsigned short* Al = albuf(cur);
     const unsigned short* Bl = blbuf(cur);
@@ -386,12 +409,10 @@ __global__ __launch_bounds__(NT_THREADS, 1) void gemm_nt_core(
       bf16x8 a[4], b[2];
       #pragma unroll
       for (int mi = 0; mi < 4; ++mi)
-        a[mi] = *(const bf16x8*)(&Al[(wr * 64 + mi * 16 + fr) * NT_PITCH +
-                                     kc * 32 + fq * 8]);
+        a[mi] = nt_tr_frag(Al, kc, wr * 4 + mi);
       #pragma unroll
       for (int ni = 0; ni < 2; ++ni)
-        b[ni] = *(const bf16x8*)(&Bl[(wc * 32 + ni * 16 + fr) * NT_PITCH +
-                                     kc * 32 + fq * 8]);
+        b[ni] = nt_tr_frag(Bl, kc, wc * 2 + ni);
       #pragma unroll
       for (int mi = 0; mi < 4; ++mi)
         #pragma unroll
@@ -399,12 +420,8 @@ __global__ __launch_bounds__(NT_THREADS, 1) void gemm_nt_core(
           acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               a[mi], b[ni], acc[mi][ni], 0, 0, 0);
     }
-    if (kt + 1 < kt1) {
-      nt_write(ra, albuf(cur ^ 1));
-      nt_write(rb, blbuf(cur ^ 1));
-    }
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
     __syncthreads();
-    cur ^= 1;
   }
 
   #pragma unroll
@@ -455,24 +472,25 @@ void launch_gemm_tn_gather(const void* img, const void* B, void* C_bf16,
 
 void launch_gemm_nt(const void* A, const void* B, float* C, int M, int N,
                     int K, long lda, long ldb, int splitk, int gmode,
-                    ConvGather gg, hipStream_t s) {
+                    ConvGather gg, const void* zp, hipStream_t s) {
   int kchunks = (K + NT_BK - 1) / NT_BK;
   if (splitk > kchunks) splitk = kchunks;
   int per_block = (kchunks + splitk - 1) / splitk;
   dim3 grid(ceil_div(M, NT_BM), ceil_div(N, NT_BN), splitk);
   int ua = splitk > 1 ? 1 : 0;
+  const unsigned short* z = (const unsigned short*)zp;
   if (gmode == 1)
     hipLaunchKernelGGL((gemm_nt_core<1>), grid, dim3(NT_THREADS), 0, s,
                        (const unsigned short*)A, (const unsigned short*)B, C,
-                       M, N, K, lda, ldb, per_block, ua, gg);
+                       M, N, K, lda, ldb, per_block, ua, gg, z);
   else if (gmode == 2)
     hipLaunchKernelGGL((gemm_nt_core<2>), grid, dim3(NT_THREADS), 0, s,
                        (const unsigned short*)A, (const unsigned short*)B, C,
-                       M, N, K, lda, ldb, per_block, ua, gg);
+                       M, N, K, lda, ldb, per_block, ua, gg, z);
   else
     hipLaunchKernelGGL((gemm_nt_core<0>), grid, dim3(NT_THREADS), 0, s,
                        (const unsigned short*)A, (const unsigned short*)B, C,
-                       M, N, K, lda, ldb, per_block, ua, gg);
+                       M, N, K, lda, ldb, per_block, ua, gg, z);
 }
 
 }  // extern "C"

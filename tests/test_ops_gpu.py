@@ -65,12 +65,13 @@ def test_gemm_tn_bias_act():
 
 
 @pytest.mark.parametrize("splitk", [1, 4])
-@pytest.mark.parametrize("m,n,k", [(64, 64, 64), (130, 70, 1000),
-                                   (512, 1152, 4096)])
+@pytest.mark.parametrize("m,n,k", [(64, 64, 64), (128, 72, 1000),
+                                   (512, 1152, 4096), (8, 64, 2048)])
 def test_gemm_nt(m, n, k, splitk):
+    # operand widths must be multiples of 8 (glds row staging)
     e = ext()
     A, B = mk((k, m), 5, 0.5), mk((k, n), 6, 0.5)
-    C = e.gemm_nt(A, B, splitk)
+    C = e.gemm_nt(A, B, splitk, None)
     ref = A.float().cpu().t() @ B.float().cpu()
     assert relerr(C, ref) < 0.02
 
