@@ -187,6 +187,16 @@ def _conv_out(h, k, stride, pad):
     return (h + 2 * pad - k) // stride + 1
 
 
+def _pad_channels(xh: torch.Tensor, c8: int) -> torch.Tensor:
+    """[N,H,W,C] -> [N,H,W,c8] zero-padded channels (gather needs C%8==0)."""
+    n, h, w, c = xh.shape
+    if c == c8:
+        return xh
+    out = xh.new_zeros(n, h, w, c8)
+    out[..., :c] = xh
+    return out
+
+
 class _Conv2d(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, w, b, stride: int, pad: int, act: int, slope: float):
@@ -194,72 +204,75 @@ class _Conv2d(torch.autograd.Function):
         N, C, H, W = x.shape
         Kout, _, R, S = w.shape
         Ho, Wo = _conv_out(H, R, stride, pad), _conv_out(W, S, stride, pad)
-        kpad = _rup64(R * S * C)
-        xh = _nhwc(x)
-        wp = _packed(w, "conv_wp", lambda: _pad_k(
-            _bf(w.detach().permute(0, 2, 3, 1)).reshape(Kout, R * S * C)))
+        # channel dim padded to 8 so every conv runs the implicit path
+        C8 = (C + 7) // 8 * 8
+        xh = _pad_channels(_nhwc(x), C8)
+        kpad = _rup64(R * S * C8)
+
+        def build_wp():
+            wc = _bf(w.detach().permute(0, 2, 3, 1))     # [Kout,R,S,C]
+            wc = _pad_channels(wc.reshape(Kout * R * S, 1, 1, C),
+                               C8).reshape(Kout, R * S * C8)
+            return _pad_k(wc)
+
+        wp = _packed(w, "conv_wp", build_wp)
         bias = (_packed(b, "f32", lambda: b.detach().float().contiguous())
                 if b is not None else None)
-        if FP8_CONV and C % 16 == 0:
+        if FP8_CONV and C8 % 16 == 0:
             xq, _, ix = ext.fp8_quantize(xh)
             wq, _, iw = _packed(w, "fp8", lambda: tuple(
                 ext.fp8_quantize(wp)))
             zp8 = _packed(wp, "zp8", lambda: torch.zeros(
                 32, dtype=torch.uint8, device=x.device))
             y2d = ext.conv_fwd_implicit_fp8(xq, wq, bias, ix, iw, zp8, N, H,
-                                            W, C, Ho, Wo, R, S, stride, pad,
+                                            W, C8, Ho, Wo, R, S, stride, pad,
                                             act, slope)
-        elif C % 8 == 0:
+        else:
             # implicit GEMM: im2col gather fused into the MFMA staging
             y2d = ext.conv_fwd_implicit(xh, wp, bias, _zp(x.device), N, H, W,
-                                        C, Ho, Wo, R, S, stride, pad, act,
-                                        slope)
-        else:
-            col = ext.im2col(xh, N, H, W, C, Ho, Wo, R, S, stride, pad, kpad)
-            y2d = ext.gemm_tn(col, wp, bias, act, slope, False)
+                                        C8, Ho, Wo, R, S, stride, pad, act,
+                                        slope, 0)
         ctx.save_for_backward(xh, wp, y2d)
-        ctx.geom = (N, C, H, W, Kout, R, S, Ho, Wo, stride, pad, kpad)
+        ctx.geom = (N, C, H, W, Kout, R, S, Ho, Wo, stride, pad, kpad, C8)
         ctx.act, ctx.slope = act, slope
         ctx.has_bias = b is not None
         ctx.dtypes = (x.dtype, w.dtype, b.dtype if b is not None else None)
+        ctx.wref = w
         return _as_nchw_view(y2d.view(N, Ho, Wo, Kout))
 
     @staticmethod
     def backward(ctx, dy):
         ext = hip_ext()
         xh, wp, y2d = ctx.saved_tensors
-        N, C, H, W, Kout, R, S, Ho, Wo, stride, pad, kpad = ctx.geom
-        rsc = R * S * C
+        N, C, H, W, Kout, R, S, Ho, Wo, stride, pad, kpad, C8 = ctx.geom
         dy2d = _bf(dy.permute(0, 2, 3, 1)).reshape(-1, Kout)
         dpre = ext.act_bwd(dy2d, y2d, ctx.act, ctx.slope) if ctx.act else dy2d
 
         dx = dw = db = None
         if ctx.needs_input_grad[1]:
-            # wgrad (rows = N*Ho*Wo)
+            # wgrad: implicit gathered-B NT (rows = N*Ho*Wo)
             npq = dpre.shape[0]
             sk = _splitk_for((Kout + 127) // 128, (kpad + 127) // 128,
                              (npq + 63) // 64)
-            if C % 8 == 0:
-                dw = ext.gemm_nt_implicit(
-                    dpre, xh, 2, Kout, kpad, npq,
-                    _dims(N, H, W, C, Ho, Wo, R, S, stride, pad), sk,
-                    _zp(xh.device))
-                dw = dw[:, :rsc]
-            else:
-                col = ext.im2col(xh, N, H, W, C, Ho, Wo, R, S, stride, pad,
-                                 kpad)
-                dw = ext.gemm_nt(dpre, col, sk, _zp(col.device))[:, :rsc]
-            dw = (dw.reshape(Kout, R, S, C).permute(0, 3, 1, 2)
-                  .contiguous().to(ctx.dtypes[1]))
+            dw = ext.gemm_nt_implicit(
+                dpre, xh, 2, Kout, kpad, npq,
+                _dims(N, H, W, C8, Ho, Wo, R, S, stride, pad), sk,
+                _zp(xh.device))
+            dw = (dw[:, :R * S * C8].reshape(Kout, R, S, C8)[..., :C]
+                  .permute(0, 3, 1, 2).contiguous().to(ctx.dtypes[1]))
         if ctx.needs_input_grad[0]:
-            # dgrad: dcol = dpre @ w_cl ; then col2im
-            wt = _packed(wp, "wt", lambda: _pad_k(
-                wp[:, :rsc].t().contiguous()))       # [rsc, kout_pad]
-            dprep = _pad_k(dpre)
-            dcol = ext.gemm_tn(dprep, wt, None, 0, 0.0, False)  # [NP, rsc]
-            dxh = ext.col2im(dcol, N, H, W, C, Ho, Wo, R, S, stride, pad,
-                             rsc, None, 0, 0.0)
-            dx = _as_nchw_view(dxh).to(ctx.dtypes[0])
+            # dgrad as ONE transposed-gather GEMM (no dcol, no col2im):
+            # dx[npix][cin] = sum_{r,s,kout} dpre[(hi+pad-r)/stride..][kout]
+            #                 * W[kout][r][s][cin]
+            w = ctx.wref
+            wd = _packed(w, "dgrad_w", lambda: _pad_k(
+                _bf(w.detach().permute(1, 2, 3, 0))
+                .reshape(C, R * S * Kout)))          # [C][rs*Kout] pad64
+            dpre_img = dpre.view(N, Ho, Wo, Kout)
+            dx2d = ext.conv_fwd_implicit(
+                dpre_img, wd, None, _zp(dpre.device), N, Ho, Wo, Kout,
+                H, W, R, S, stride, pad, 0, 0.0, 1)   # mode 1
+            dx = _as_nchw_view(dx2d.view(N, H, W, C)).to(ctx.dtypes[0])
         if ctx.has_bias and ctx.needs_input_grad[2]:
             db = ext.col_sum(dpre).to(ctx.dtypes[2])
         return dx, dw, db, None, None, None, None
@@ -278,28 +291,32 @@ class _ConvTranspose2d(torch.autograd.Function):
         _, Cout, R, S = w.shape
         Ho = (Hi - 1) * stride - 2 * pad + R
         Wo = (Wi - 1) * stride - 2 * pad + S
-        xh = _nhwc(x)                       # [N,Hi,Wi,Cin]
-        x2d = _pad_k(xh.reshape(-1, Cin))
-        # B operand: [R*S*Cout, Cin] (k-contiguous over Cin)
-        w2a = _packed(w, "w2a", lambda: _pad_k(
-            _bf(w.detach().permute(2, 3, 1, 0)).reshape(R * S * Cout, Cin)))
-        col = ext.gemm_tn(x2d, w2a, None, 0, 0.0, False)  # [NPin, RS*Cout]
+        xh = _nhwc(x)                       # [N,Hi,Wi,Cin]; Cin % 8 == 0
         bias = (_packed(b, "f32", lambda: b.detach().float().contiguous())
                 if b is not None else None)
-        # col2im over the OUTPUT image grid; patches at the input grid
-        yh = ext.col2im(col, N, Ho, Wo, Cout, Hi, Wi, R, S, stride, pad,
-                        R * S * Cout, bias, act, slope)
-        ctx.save_for_backward(x2d, w2a, yh)
+        # ONE transposed-gather GEMM with fused bias+activation:
+        # y[ho,wo,cout] = act( sum_{r,s,cin}[valid hi=(ho+pad-r)/stride]
+        #                      x[hi,wi,cin] * W[cin][cout][r][s] + b )
+        wt = _packed(w, "convt_fwd_w", lambda: _pad_k(
+            _bf(w.detach().permute(1, 2, 3, 0))      # [Cout,R,S,Cin]
+            .reshape(Cout, R * S * Cin)))
+        y2d = ext.conv_fwd_implicit(xh, wt, bias, _zp(x.device), N, Hi, Wi,
+                                    Cin, Ho, Wo, R, S, stride, pad, act,
+                                    slope, 1)         # mode 1
+        yh = y2d.view(N, Ho, Wo, Cout)
+        ctx.save_for_backward(xh, yh)
         ctx.geom = (N, Cin, Hi, Wi, Cout, R, S, Ho, Wo, stride, pad)
         ctx.act, ctx.slope = act, slope
         ctx.has_bias = b is not None
         ctx.dtypes = (x.dtype, w.dtype, b.dtype if b is not None else None)
+        ctx.wref = w
         return _as_nchw_view(yh)
 
     @staticmethod
     def backward(ctx, dy):
         ext = hip_ext()
-        x2d, w2a, yh = ctx.saved_tensors
+        xh, yh = ctx.saved_tensors
+        w = ctx.wref
         N, Cin, Hi, Wi, Cout, R, S, Ho, Wo, stride, pad = ctx.geom
         rsco = R * S * Cout
         rscop = _rup64(rsco)
@@ -310,45 +327,42 @@ class _ConvTranspose2d(torch.autograd.Function):
             dpre_img = dpre.view(N, Ho, Wo, Cout)
         else:
             dpre_img = dyh
-        # dcol = im2col(dpre_img) over the input grid (fused into the GEMM
-        # staging when Cout is vectorizable; explicit buffer otherwise)
         dpre_img = dpre_img.contiguous()
-        use_impl = Cout % 8 == 0
-        dcol = None
-        if not use_impl:
-            dcol = ext.im2col(dpre_img, N, Ho, Wo, Cout, Hi, Wi, R, S,
-                              stride, pad, rscop)   # [NPin, rscop]
+        # forward-gather geometry over dOut (chunks of 8 channels):
+        # pad Cout to 8 if needed (g_out has Cout = image channels = 3)
+        Co8 = (Cout + 7) // 8 * 8
+        dpre8 = _pad_channels(dpre_img, Co8)
         npq = N * Hi * Wi
         dx = dw = db = None
         if ctx.needs_input_grad[0]:
-            # dx = dcol @ w2b ; w2b = [Cin, R*S*Cout]
-            w2b = _packed(w2a, "w2b", lambda: _pad_k(
-                w2a[:, :Cin].reshape(R, S, Cout, Cin)
-                .permute(3, 0, 1, 2).reshape(Cin, rsco)
-            ))
-            if use_impl:
-                dx2d = ext.conv_fwd_implicit(dpre_img, w2b, None,
-                                             _zp(dpre_img.device), N, Ho, Wo,
-                                             Cout, Hi, Wi, R, S, stride, pad,
-                                             0, 0.0)
-            else:
-                dx2d = ext.gemm_tn(dcol, w2b, None, 0, 0.0, False)
+            # dx[np_in][cin] = sum_{r,s,cout} dpre[ho=hi*s-p+r..][cout]
+            #                  * W[cin][cout][r][s]  (forward-gather, mode 0)
+            def build_w2b():
+                wc = _bf(w.detach().permute(0, 2, 3, 1))   # [Cin,R,S,Cout]
+                wc = _pad_channels(wc.reshape(Cin * R * S, 1, 1, Cout),
+                                   Co8).reshape(Cin, R * S * Co8)
+                return _pad_k(wc)
+
+            w2b = _packed(w, "convt_dgrad_w", build_w2b)
+            dx2d = ext.conv_fwd_implicit(dpre8, w2b, None,
+                                         _zp(dpre8.device), N, Ho, Wo,
+                                         Co8, Hi, Wi, R, S, stride, pad,
+                                         0, 0.0, 0)
             dx = _as_nchw_view(dx2d.view(N, Hi, Wi, Cin)).to(ctx.dtypes[0])
         if ctx.needs_input_grad[1]:
-            # wgrad: dW2a[rsco][cin] = sum_np dcol[np][rsco]*x2d[np][cin];
-            # padded x2d cols are zero -> sliced grad exact
-            sk = _splitk_for((rscop + 127) // 128, (Cin + 127) // 128,
-                             (npq + 63) // 64)
-            if use_impl:
-                dw2a = ext.gemm_nt_implicit(
-                    dpre_img, x2d, 1, rscop, x2d.shape[1], npq,
-                    _dims(N, Ho, Wo, Cout, Hi, Wi, R, S, stride, pad),
-                    sk, _zp(x2d.device))[:rsco, :Cin]
-            else:
-                dw2a = ext.gemm_nt(dcol, x2d, sk,
-                                   _zp(x2d.device))[:rsco, :Cin]
-            dw = (dw2a.reshape(R, S, Cout, Cin).permute(3, 2, 0, 1)
-                  .contiguous().to(ctx.dtypes[1]))
+            # wgrad: dW[(r,s,cout)][cin] = sum_np im2col(dpre)[np][rs*cout]
+            #        * x[np][cin]  (gathered-A NT over dOut)
+            rsco8 = R * S * Co8
+            rscop8 = _rup64(rsco8)
+            x2d = _pad_k(xh.reshape(-1, Cin))
+            sk = _splitk_for((rscop8 + 127) // 128,
+                             (x2d.shape[1] + 127) // 128, (npq + 63) // 64)
+            dw2a = ext.gemm_nt_implicit(
+                dpre8, x2d, 1, rscop8, x2d.shape[1], npq,
+                _dims(N, Ho, Wo, Co8, Hi, Wi, R, S, stride, pad),
+                sk, _zp(x2d.device))
+            dw = (dw2a[:rsco8, :Cin].reshape(R, S, Co8, Cin)[:, :, :Cout]
+                  .permute(3, 2, 0, 1).contiguous().to(ctx.dtypes[1]))
         if ctx.has_bias and ctx.needs_input_grad[2]:
             db = ext.col_sum(dpre_img.reshape(-1, Cout)).to(ctx.dtypes[2])
         return dx, dw, db, None, None, None, None

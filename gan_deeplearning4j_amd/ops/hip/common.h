@@ -103,12 +103,16 @@ DEV_INLINE unsigned fmod_(unsigned n, FastDiv f, unsigned q) {
 
 // Conv gather geometry for implicit-GEMM staging: maps an im2col
 // coordinate (np, k) to an input-image address without materializing col.
+// mode 0 (forward):    src pixel = grid_pos * stride - pad + (r,s)
+// mode 1 (transposed): src pixel = (grid_pos + pad - (r,s)) / stride,
+//                      valid only when divisible (conv dgrad / convT fwd)
 struct ConvGather {
   int N, H, W, C;        // source image dims (NHWC)
   int Ho, Wo;            // patch grid
   int R, S, stride, pad;
   int rsc;               // R*S*C (valid k range; >= rsc is zero padding)
-  FastDiv fC, fS, fWo, fHo;
+  int mode;
+  FastDiv fC, fS, fWo, fHo, fStride;
 };
 
 #define HIP_CHECK_LAST()                                                     \

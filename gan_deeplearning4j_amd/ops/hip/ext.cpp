@@ -20,8 +20,8 @@ struct FastDiv {
   int d;
 };
 struct ConvGather {
-  int N, H, W, C, Ho, Wo, R, S, stride, pad, rsc;
-  FastDiv fC, fS, fWo, fHo;
+  int N, H, W, C, Ho, Wo, R, S, stride, pad, rsc, mode;
+  FastDiv fC, fS, fWo, fHo, fStride;
 };
 
 static FastDiv make_fastdiv_h(int d) {
@@ -36,15 +36,18 @@ static FastDiv make_fastdiv_h(int d) {
 }
 
 static ConvGather make_gather(int N, int H, int W, int C, int Ho, int Wo,
-                              int R, int S, int stride, int pad) {
+                              int R, int S, int stride, int pad,
+                              int mode = 0) {
   ConvGather g;
   g.N = N; g.H = H; g.W = W; g.C = C; g.Ho = Ho; g.Wo = Wo;
   g.R = R; g.S = S; g.stride = stride; g.pad = pad;
   g.rsc = R * S * C;
+  g.mode = mode;
   g.fC = make_fastdiv_h(C);
   g.fS = make_fastdiv_h(S);
   g.fWo = make_fastdiv_h(Wo);
   g.fHo = make_fastdiv_h(Ho);
+  g.fStride = make_fastdiv_h(stride);
   return g;
 }
 
@@ -190,7 +193,8 @@ torch::Tensor conv_fwd_implicit(torch::Tensor x, torch::Tensor Wp,
   }
   torch::Tensor y = torch::empty({M, Kout}, x.options());
   ConvGather g = make_gather((int)Nb, (int)H, (int)W, (int)C, (int)Ho,
-                             (int)Wo, (int)R, (int)S, (int)stride, (int)pad);
+                             (int)Wo, (int)R, (int)S, (int)stride, (int)pad,
+                             (int)mode);
   launch_gemm_tn_gather(x.data_ptr(), Wp.data_ptr(), y.data_ptr(), bias_p,
                         (int)M, (int)Kout, (int)kpad, kpad, (int)act,
                         (float)slope, g, zero_page.data_ptr(), cur_stream());
@@ -206,11 +210,13 @@ torch::Tensor gemm_nt_implicit(torch::Tensor A, torch::Tensor B,
   check_bf16(A, "A");
   check_bf16(B, "B");
   auto d = img_dims.cpu().contiguous();
-  TORCH_CHECK(d.numel() == 10, "img_dims = [N,H,W,C,Ho,Wo,R,S,stride,pad]");
+  TORCH_CHECK(d.numel() == 10 || d.numel() == 11,
+              "img_dims = [N,H,W,C,Ho,Wo,R,S,stride,pad(,mode)]");
   const int64_t* p = d.data_ptr<int64_t>();
   ConvGather g = make_gather((int)p[0], (int)p[1], (int)p[2], (int)p[3],
                              (int)p[4], (int)p[5], (int)p[6], (int)p[7],
-                             (int)p[8], (int)p[9]);
+                             (int)p[8], (int)p[9],
+                             d.numel() == 11 ? (int)p[10] : 0);
   torch::Tensor C =
       splitk > 1
           ? torch::zeros({Mdim, Ndim}, A.options().dtype(torch::kFloat32))

@@ -54,6 +54,34 @@ DEV_INLINE void k_decode(const ConvGather& g, unsigned k, int& r, int& s,
   r = (int)rr;
 }
 
+// source address for grid position (ho,wo) + kernel offset (r,s), channel c;
+// returns nullptr-sentinel via 'valid'
+DEV_INLINE const unsigned short* gather_addr(
+    const ConvGather& g, const unsigned short* img, int n, int ho, int wo,
+    int r, int s, int c, bool& valid) {
+  int hi, wi;
+  if (g.mode == 0) {
+    hi = ho * g.stride - g.pad + r;
+    wi = wo * g.stride - g.pad + s;
+    valid = hi >= 0 && hi < g.H && wi >= 0 && wi < g.W;
+  } else {
+    int hop = ho + g.pad - r;
+    int wop = wo + g.pad - s;
+    if (hop < 0 || wop < 0) {
+      valid = false;
+      hi = wi = 0;
+    } else {
+      unsigned qh = fdiv((unsigned)hop, g.fStride);
+      unsigned qw = fdiv((unsigned)wop, g.fStride);
+      valid = (hop == (int)(qh * g.stride)) &&
+              (wop == (int)(qw * g.stride)) && (int)qh < g.H && (int)qw < g.W;
+      hi = (int)qh;
+      wi = (int)qw;
+    }
+  }
+  return img + (((long)n * g.H + hi) * g.W + wi) * g.C + c;
+}
+
 // ---------------------------------------------------------------------------
 // TN kernel
 // ---------------------------------------------------------------------------
@@ -98,10 +126,10 @@ DEV_INLINE void tn_stage_gather(const unsigned short* __restrict__ img,
       int r, s, c, n, ho, wo;
       k_decode(g, (unsigned)k, r, s, c);
       np_decode(g, (unsigned)np, n, ho, wo);
-      int hi = ho * g.stride - g.pad + r;
-      int wi = wo * g.stride - g.pad + s;
-      if (hi >= 0 && hi < g.H && wi >= 0 && wi < g.W)
-        src = img + (((long)n * g.H + hi) * g.W + wi) * g.C + c;
+      bool valid;
+      const unsigned short* p = gather_addr(g, img, n, ho, wo, r, s, c,
+                                            valid);
+      if (valid) src = p;
     }
     char* dst = lds + (i * 256 + wid * 64) * 16;
     GLDS16(src, dst);
@@ -316,10 +344,10 @@ DEV_INLINE void nt_stage_gather(const unsigned short* __restrict__ img,
       int r, s, cc, n, ho, wo;
       k_decode(g, (unsigned)kk, r, s, cc);
       np_decode(g, (unsigned)np, n, ho, wo);
-      int hi = ho * g.stride - g.pad + r;
-      int wi = wo * g.stride - g.pad + s;
-      if (hi >= 0 && hi < g.H && wi >= 0 && wi < g.W)
-        src = img + (((long)n * g.H + hi) * g.W + wi) * g.C + cc;
+      bool valid;
+      const unsigned short* p = gather_addr(g, img, n, ho, wo, r, s, cc,
+                                            valid);
+      if (valid) src = p;
     }
     char* dst = lds + (i * 512 + wid * 64) * 16;
     GLDS16(src, dst);
