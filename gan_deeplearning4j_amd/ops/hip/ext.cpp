@@ -180,7 +180,7 @@ torch::Tensor conv_fwd_implicit(torch::Tensor x, torch::Tensor Wp,
                                 int64_t H, int64_t W, int64_t C, int64_t Ho,
                                 int64_t Wo, int64_t R, int64_t S,
                                 int64_t stride, int64_t pad, int64_t act,
-                                double slope) {
+                                double slope, int64_t mode) {
   check_bf16(x, "x");
   check_bf16(Wp, "Wp");
   TORCH_CHECK(C % 8 == 0, "implicit conv needs C % 8 == 0");
