@@ -249,30 +249,49 @@ class _Conv2d(torch.autograd.Function):
         dpre = ext.act_bwd(dy2d, y2d, ctx.act, ctx.slope) if ctx.act else dy2d
 
         dx = dw = db = None
+        Ko8 = (Kout + 7) // 8 * 8
+        dpre8 = _pad8(dpre) if Ko8 != Kout else dpre
         if ctx.needs_input_grad[1]:
             # wgrad: implicit gathered-B NT (rows = N*Ho*Wo)
             npq = dpre.shape[0]
-            sk = _splitk_for((Kout + 127) // 128, (kpad + 127) // 128,
+            sk = _splitk_for((Ko8 + 127) // 128, (kpad + 127) // 128,
                              (npq + 63) // 64)
             dw = ext.gemm_nt_implicit(
-                dpre, xh, 2, Kout, kpad, npq,
+                dpre8, xh, 2, Ko8, kpad, npq,
                 _dims(N, H, W, C8, Ho, Wo, R, S, stride, pad), sk,
                 _zp(xh.device))
-            dw = (dw[:, :R * S * C8].reshape(Kout, R, S, C8)[..., :C]
+            dw = (dw[:Kout, :R * S * C8].reshape(Kout, R, S, C8)[..., :C]
                   .permute(0, 3, 1, 2).contiguous().to(ctx.dtypes[1]))
         if ctx.needs_input_grad[0]:
-            # dgrad as ONE transposed-gather GEMM (no dcol, no col2im):
-            # dx[npix][cin] = sum_{r,s,kout} dpre[(hi+pad-r)/stride..][kout]
-            #                 * W[kout][r][s][cin]
             w = ctx.wref
-            wd = _packed(w, "dgrad_w", lambda: _pad_k(
-                _bf(w.detach().permute(1, 2, 3, 0))
-                .reshape(C, R * S * Kout)))          # [C][rs*Kout] pad64
-            dpre_img = dpre.view(N, Ho, Wo, Kout)
-            dx2d = ext.conv_fwd_implicit(
-                dpre_img, wd, None, _zp(dpre.device), N, Ho, Wo, Kout,
-                H, W, R, S, stride, pad, 0, 0.0, 1)   # mode 1
-            dx = _as_nchw_view(dx2d.view(N, H, W, C)).to(ctx.dtypes[0])
+            if stride == 1:
+                # dgrad as ONE transposed-gather GEMM (no dcol/col2im;
+                # every tap valid at stride 1)
+                def build_wd():
+                    wc = _bf(w.detach().permute(1, 2, 3, 0))  # [C,R,S,Kout]
+                    wc = _pad_channels(wc.reshape(C * R * S, 1, 1, Kout),
+                                       Ko8).reshape(C, R * S * Ko8)
+                    return _pad_k(wc)
+
+                wd = _packed(w, "dgrad_w", build_wd)
+                dpre_img = dpre8.view(N, Ho, Wo, Ko8)
+                dx2d = ext.conv_fwd_implicit(
+                    dpre_img, wd, None, _zp(dpre.device), N, Ho, Wo, Ko8,
+                    H, W, R, S, stride, pad, 0, 0.0, 1)   # mode 1
+                dx = _as_nchw_view(dx2d.view(N, H, W, C)).to(ctx.dtypes[0])
+            else:
+                # strided: 3/4 of transposed taps are invalid, so the
+                # dcol + col2im route has 4x fewer MFMA FLOPs
+                rsc8 = R * S * C8
+                wt = _packed(wp, "wt", lambda: _pad_k(
+                    wp[:, :rsc8].t().contiguous()))   # [rsc8, kout_pad]
+                dprep = _pad_k(dpre)
+                dcol = ext.gemm_tn(dprep, wt, None, 0, 0.0, False)
+                dxh = ext.col2im(dcol, N, H, W, C8, Ho, Wo, R, S, stride,
+                                 pad, rsc8, None, 0, 0.0)
+                if C8 != C:
+                    dxh = dxh[..., :C].contiguous()
+                dx = _as_nchw_view(dxh).to(ctx.dtypes[0])
         if ctx.has_bias and ctx.needs_input_grad[2]:
             db = ext.col_sum(dpre).to(ctx.dtypes[2])
         return dx, dw, db, None, None, None, None
@@ -294,16 +313,27 @@ class _ConvTranspose2d(torch.autograd.Function):
         xh = _nhwc(x)                       # [N,Hi,Wi,Cin]; Cin % 8 == 0
         bias = (_packed(b, "f32", lambda: b.detach().float().contiguous())
                 if b is not None else None)
-        # ONE transposed-gather GEMM with fused bias+activation:
-        # y[ho,wo,cout] = act( sum_{r,s,cin}[valid hi=(ho+pad-r)/stride]
-        #                      x[hi,wi,cin] * W[cin][cout][r][s] + b )
-        wt = _packed(w, "convt_fwd_w", lambda: _pad_k(
-            _bf(w.detach().permute(1, 2, 3, 0))      # [Cout,R,S,Cin]
-            .reshape(Cout, R * S * Cin)))
-        y2d = ext.conv_fwd_implicit(xh, wt, bias, _zp(x.device), N, Hi, Wi,
-                                    Cin, Ho, Wo, R, S, stride, pad, act,
-                                    slope, 1)         # mode 1
-        yh = y2d.view(N, Ho, Wo, Cout)
+        if stride == 1:
+            # ONE transposed-gather GEMM with fused bias+activation:
+            # y[ho,wo,cout] = act( sum_{r,s,cin}[valid hi=ho+pad-r]
+            #                      x[hi,wi,cin] * W[cin][cout][r][s] + b )
+            wt = _packed(w, "convt_fwd_w", lambda: _pad_k(
+                _bf(w.detach().permute(1, 2, 3, 0))      # [Cout,R,S,Cin]
+                .reshape(Cout, R * S * Cin)))
+            y2d = ext.conv_fwd_implicit(xh, wt, bias, _zp(x.device), N, Hi,
+                                        Wi, Cin, Ho, Wo, R, S, stride, pad,
+                                        act, slope, 1)   # mode 1
+            yh = y2d.view(N, Ho, Wo, Cout)
+        else:
+            # strided: GEMM over Cin then scatter-free col2im gather with
+            # fused bias+activation (transposed taps 3/4-invalid at s=2)
+            x2d = _pad_k(xh.reshape(-1, Cin))
+            w2a = _packed(w, "w2a", lambda: _pad_k(
+                _bf(w.detach().permute(2, 3, 1, 0))
+                .reshape(R * S * Cout, Cin)))
+            col = ext.gemm_tn(x2d, w2a, None, 0, 0.0, False)  # [NPin,RSCout]
+            yh = ext.col2im(col, N, Ho, Wo, Cout, Hi, Wi, R, S, stride, pad,
+                            R * S * Cout, bias, act, slope)
         ctx.save_for_backward(xh, yh)
         ctx.geom = (N, Cin, Hi, Wi, Cout, R, S, Ho, Wo, stride, pad)
         ctx.act, ctx.slope = act, slope
