@@ -168,13 +168,17 @@ class ComputationGraph(nn.Module):
         return self
 
     # ---------------------------------------------------------- forward
+    @staticmethod
+    def _shape_input(t: Optional[InputType], x: torch.Tensor) -> torch.Tensor:
+        if (t is not None and x.dim() == 2
+                and t.kind in ("convolutional_flat", "convolutional")):
+            return x.reshape(x.shape[0], t.channels, t.height, t.width)
+        return x
+
     def forward(self, *inputs: torch.Tensor) -> torch.Tensor:
         acts: dict[str, torch.Tensor] = {}
         for name, x, in zip(self.input_names, inputs):
-            t = self.input_types.get(name)
-            if t is not None and t.kind == "convolutional_flat" and x.dim() == 2:
-                x = x.reshape(x.shape[0], t.channels, t.height, t.width)
-            acts[name] = x
+            acts[name] = self._shape_input(self.input_types.get(name), x)
         for name in self._topo:
             layer = self.layers[name]
             srcs = [acts[s] for s in self._vertex_inputs[name]]
@@ -194,10 +198,7 @@ class ComputationGraph(nn.Module):
         try:
             acts: dict[str, torch.Tensor] = {}
             for name, x in zip(self.input_names, inputs):
-                t = self.input_types.get(name)
-                if t is not None and t.kind == "convolutional_flat" and x.dim() == 2:
-                    x = x.reshape(x.shape[0], t.channels, t.height, t.width)
-                acts[name] = x
+                acts[name] = self._shape_input(self.input_types.get(name), x)
             for name in self._topo:
                 layer = self.layers[name]
                 srcs = [acts[s] for s in self._vertex_inputs[name]]

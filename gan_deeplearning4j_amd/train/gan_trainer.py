@@ -366,18 +366,24 @@ class ReferenceProtocolTrainer:
 
     # ---------------------------------------------------------- driver
     def run(self, train_iter, test_iter=None) -> dict:
-        """The while-loop (Java:408): numIterations batches."""
+        """The while-loop (Java:408): numIterations batches (re-iterating
+        the data epoch-wise until the budget is met)."""
         cfg = self.cfg
         last = {}
-        for ds in train_iter:
-            if self.batch_counter >= cfg.train.num_iterations:
+        while self.batch_counter < cfg.train.num_iterations:
+            progressed = False
+            for ds in train_iter:
+                if self.batch_counter >= cfg.train.num_iterations:
+                    break
+                progressed = True
+                last = self.train_iteration(ds)
+                i = self.batch_counter
+                if i % cfg.train.print_every == 0:
+                    self.dump_sample_grid(i)
+                    if test_iter is not None and cfg.train.train_classifier:
+                        self.dump_test_predictions(i, test_iter)
+                if i % cfg.train.save_every == 0:
+                    self.save_checkpoints()
+            if not progressed:
                 break
-            last = self.train_iteration(ds)
-            i = self.batch_counter
-            if i % cfg.train.print_every == 0:
-                self.dump_sample_grid(i)
-                if test_iter is not None and cfg.train.train_classifier:
-                    self.dump_test_predictions(i, test_iter)
-            if i % cfg.train.save_every == 0:
-                self.save_checkpoints()
         return last
