@@ -2,16 +2,195 @@
 // Stats, affine params and gradients are fp32 (SURVEY hard-part #3:
 // "BN in bf16 - keep stats in fp32").
 //
-// Training forward: batch mean/var (biased, like torch) + running-stat
-// update (running_var uses the unbiased estimator, matching torch).
-// Backward: the standard two-reduction formulation.
+// All kernels are 8-channel vectorized (s16x8 = 16B/lane loads, G13) when
+// C % 8 == 0, with a scalar fallback for narrow feature dims (C < 8, e.g.
+// the z=2 generator input BN). Reductions use the row-lane scheme: a block
+// covers a channel window, blockDim/width row-lanes walk rows coalesced,
+// LDS-reduce, one atomic per channel.
 
 #include "common.h"
 
-// pass 1: per-channel sum and sum-of-squares.
-// Row-major coalesced with row-lanes (same scheme as col_sum_bf16):
-// a block covers a <=256-wide channel window; blockDim.x/colsW row-lanes
-// walk rows in parallel, LDS-reduce, one atomic pair per channel.
+// ------------------------------------------------------------- vector path
+// pass 1: per-channel sum / sumsq. Channel groups of 8; colsW8 = number of
+// 8-channel groups covered by one block (<= 32 so lanes >= 8).
+__global__ void bn_stats_v8(const s16x8* __restrict__ x, long m, int c,
+                            float* __restrict__ sum,
+                            float* __restrict__ sumsq) {
+  // each thread accumulates 8 channel-partials in registers; row-lanes are
+  // reduced channel-slot by channel-slot through one LDS array
+  int c8 = c / 8;
+  int g0 = blockIdx.y * 32;                 // first 8-group of this block
+  int groups = min(32, c8 - g0);            // groups covered (<=32)
+  int lanes = (int)blockDim.x / groups;     // row-lanes
+  int sub = (int)threadIdx.x / groups;
+  int g = g0 + (int)threadIdx.x % groups;
+  float s[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+  float ss[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+  if (sub < lanes) {
+    for (long r = (long)blockIdx.x * lanes + sub; r < m;
+         r += (long)gridDim.x * lanes) {
+      s16x8 v = x[r * c8 + g];
+      #pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float f = bf2f((unsigned short)v[j]);
+        s[j] += f;
+        ss[j] += f * f;
+      }
+    }
+  }
+  // reduce row-lanes channel-slot by channel-slot through LDS
+  __shared__ float red[256];
+  #pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    red[threadIdx.x] = s[j];
+    __syncthreads();
+    if (sub == 0) {
+      float acc = 0.f;
+      for (int q = 0; q < lanes; ++q) acc += red[q * groups + g - g0];
+      atomicAdd(&sum[g * 8 + j], acc);
+    }
+    __syncthreads();
+    red[threadIdx.x] = ss[j];
+    __syncthreads();
+    if (sub == 0) {
+      float acc = 0.f;
+      for (int q = 0; q < lanes; ++q) acc += red[q * groups + g - g0];
+      atomicAdd(&sumsq[g * 8 + j], acc);
+    }
+    __syncthreads();
+  }
+}
+
+__global__ void bn_apply_v8(const s16x8* __restrict__ x,
+                            s16x8* __restrict__ y, long m, int c,
+                            const float* __restrict__ mean,
+                            const float* __restrict__ istd,
+                            const float* __restrict__ gamma,
+                            const float* __restrict__ beta) {
+  int c8 = c / 8;
+  long total = m * c8;
+  long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  for (; i < total; i += (long)gridDim.x * blockDim.x) {
+    int g = (int)(i % c8) * 8;
+    s16x8 v = x[i];
+    s16x8 o;
+    #pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float f = (bf2f((unsigned short)v[j]) - mean[g + j]) * istd[g + j];
+      o[j] = (short)f2bf(gamma[g + j] * f + beta[g + j]);
+    }
+    y[i] = o;
+  }
+}
+
+__global__ void bn_apply_eval_v8(const s16x8* __restrict__ x,
+                                 s16x8* __restrict__ y, long m, int c,
+                                 const float* __restrict__ rm,
+                                 const float* __restrict__ rv,
+                                 const float* __restrict__ gamma,
+                                 const float* __restrict__ beta, float eps) {
+  int c8 = c / 8;
+  long total = m * c8;
+  long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  for (; i < total; i += (long)gridDim.x * blockDim.x) {
+    int g = (int)(i % c8) * 8;
+    s16x8 v = x[i];
+    s16x8 o;
+    #pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float is = rsqrtf(rv[g + j] + eps);
+      float f = (bf2f((unsigned short)v[j]) - rm[g + j]) * is;
+      o[j] = (short)f2bf(gamma[g + j] * f + beta[g + j]);
+    }
+    y[i] = o;
+  }
+}
+
+__global__ void bn_bwd_reduce_v8(const s16x8* __restrict__ x,
+                                 const s16x8* __restrict__ dy, long m, int c,
+                                 const float* __restrict__ mean,
+                                 const float* __restrict__ istd,
+                                 float* __restrict__ dgamma,
+                                 float* __restrict__ dbeta) {
+  int c8 = c / 8;
+  int g0 = blockIdx.y * 32;
+  int groups = min(32, c8 - g0);
+  int lanes = (int)blockDim.x / groups;
+  int sub = (int)threadIdx.x / groups;
+  int g = g0 + (int)threadIdx.x % groups;
+  float dg[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+  float db[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+  if (sub < lanes) {
+    float mu[8], is[8];
+    #pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      mu[j] = mean[g * 8 + j];
+      is[j] = istd[g * 8 + j];
+    }
+    for (long r = (long)blockIdx.x * lanes + sub; r < m;
+         r += (long)gridDim.x * lanes) {
+      s16x8 vx = x[r * c8 + g];
+      s16x8 vg = dy[r * c8 + g];
+      #pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float gg = bf2f((unsigned short)vg[j]);
+        float xh = (bf2f((unsigned short)vx[j]) - mu[j]) * is[j];
+        dg[j] += gg * xh;
+        db[j] += gg;
+      }
+    }
+  }
+  __shared__ float red[256];
+  #pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    red[threadIdx.x] = dg[j];
+    __syncthreads();
+    if (sub == 0) {
+      float acc = 0.f;
+      for (int q = 0; q < lanes; ++q) acc += red[q * groups + g - g0];
+      atomicAdd(&dgamma[g * 8 + j], acc);
+    }
+    __syncthreads();
+    red[threadIdx.x] = db[j];
+    __syncthreads();
+    if (sub == 0) {
+      float acc = 0.f;
+      for (int q = 0; q < lanes; ++q) acc += red[q * groups + g - g0];
+      atomicAdd(&dbeta[g * 8 + j], acc);
+    }
+    __syncthreads();
+  }
+}
+
+__global__ void bn_bwd_apply_v8(const s16x8* __restrict__ x,
+                                const s16x8* __restrict__ dy,
+                                s16x8* __restrict__ dx, long m, int c,
+                                const float* __restrict__ mean,
+                                const float* __restrict__ istd,
+                                const float* __restrict__ gamma,
+                                const float* __restrict__ dgamma,
+                                const float* __restrict__ dbeta) {
+  int c8 = c / 8;
+  long total = m * c8;
+  float inv_m = 1.f / (float)m;
+  long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  for (; i < total; i += (long)gridDim.x * blockDim.x) {
+    int g = (int)(i % c8) * 8;
+    s16x8 vx = x[i], vg = dy[i];
+    s16x8 o;
+    #pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float xh = (bf2f((unsigned short)vx[j]) - mean[g + j]) * istd[g + j];
+      float gg = bf2f((unsigned short)vg[j]);
+      o[j] = (short)f2bf(gamma[g + j] * istd[g + j] *
+                         (gg - dbeta[g + j] * inv_m -
+                          xh * dgamma[g + j] * inv_m));
+    }
+    dx[i] = o;
+  }
+}
+
+// ------------------------------------------------------------- scalar path
 __global__ void bn_stats(const unsigned short* __restrict__ x, long m, int c,
                          float* __restrict__ sum, float* __restrict__ sumsq) {
   __shared__ float ls[512];
@@ -42,7 +221,6 @@ __global__ void bn_stats(const unsigned short* __restrict__ x, long m, int c,
   }
 }
 
-// finalize: mean/istd from sums; update running stats in-place (fp32)
 __global__ void bn_finalize(const float* __restrict__ sum,
                             const float* __restrict__ sumsq, long m, int c,
                             float eps, float momentum,
@@ -63,7 +241,6 @@ __global__ void bn_finalize(const float* __restrict__ sum,
   }
 }
 
-// y = gamma * (x - mean) * istd + beta
 __global__ void bn_apply(const unsigned short* __restrict__ x,
                          unsigned short* __restrict__ y, long m, int c,
                          const float* __restrict__ mean,
@@ -79,7 +256,6 @@ __global__ void bn_apply(const unsigned short* __restrict__ x,
   }
 }
 
-// eval-mode apply from running stats
 __global__ void bn_apply_eval(const unsigned short* __restrict__ x,
                               unsigned short* __restrict__ y, long m, int c,
                               const float* __restrict__ running_mean,
@@ -96,7 +272,6 @@ __global__ void bn_apply_eval(const unsigned short* __restrict__ x,
   }
 }
 
-// backward pass 1: dgamma = sum dy*xhat, dbeta = sum dy (coalesced scheme)
 __global__ void bn_bwd_reduce(const unsigned short* __restrict__ x,
                               const unsigned short* __restrict__ dy, long m,
                               int c, const float* __restrict__ mean,
@@ -133,8 +308,6 @@ __global__ void bn_bwd_reduce(const unsigned short* __restrict__ x,
   }
 }
 
-// backward pass 2 (training):
-// dx = gamma*istd * (dy - dbeta/m - xhat * dgamma/m)
 __global__ void bn_bwd_apply(const unsigned short* __restrict__ x,
                              const unsigned short* __restrict__ dy,
                              unsigned short* __restrict__ dx, long m, int c,
@@ -165,8 +338,26 @@ static dim3 _colgrid(long m, int c) {
               (unsigned)((c + 255) / 256));
 }
 
+static dim3 _colgrid_v8(long m, int c) {
+  int c8 = c / 8;
+  int groups = c8 < 32 ? c8 : 32;
+  int lanes = 256 / groups;
+  long chunks = (m + lanes - 1) / lanes;
+  return dim3((unsigned)min((long)1024, max((long)1, chunks)),
+              (unsigned)((c8 + 31) / 32));
+}
+
+static int _egrid(long total) {
+  return (int)min((long)2048, total / 256 + 1);
+}
+
 void launch_bn_stats(const void* x, long m, int c, float* sum, float* sumsq,
                      hipStream_t s) {
+  if (c % 8 == 0) {
+    hipLaunchKernelGGL(bn_stats_v8, _colgrid_v8(m, c), dim3(256), 0, s,
+                       (const s16x8*)x, m, c, sum, sumsq);
+    return;
+  }
   hipLaunchKernelGGL(bn_stats, _colgrid(m, c), dim3(256), 0, s,
                      (const unsigned short*)x, m, c, sum, sumsq);
 }
@@ -183,9 +374,13 @@ void launch_bn_finalize(const float* sum, const float* sumsq, long m, int c,
 void launch_bn_apply(const void* x, void* y, long m, int c, const float* mean,
                      const float* istd, const float* gamma, const float* beta,
                      hipStream_t s) {
-  long total = m * c;
-  int grid = (int)min((long)2048, (total + 255) / 256 + 1);
-  hipLaunchKernelGGL(bn_apply, dim3(grid), dim3(256), 0, s,
+  if (c % 8 == 0) {
+    hipLaunchKernelGGL(bn_apply_v8, dim3(_egrid(m * c / 8)), dim3(256), 0, s,
+                       (const s16x8*)x, (s16x8*)y, m, c, mean, istd, gamma,
+                       beta);
+    return;
+  }
+  hipLaunchKernelGGL(bn_apply, dim3(_egrid(m * c)), dim3(256), 0, s,
                      (const unsigned short*)x, (unsigned short*)y, m, c, mean,
                      istd, gamma, beta);
 }
@@ -193,9 +388,13 @@ void launch_bn_apply(const void* x, void* y, long m, int c, const float* mean,
 void launch_bn_apply_eval(const void* x, void* y, long m, int c,
                           const float* rm, const float* rv, const float* gamma,
                           const float* beta, float eps, hipStream_t s) {
-  long total = m * c;
-  int grid = (int)min((long)2048, (total + 255) / 256 + 1);
-  hipLaunchKernelGGL(bn_apply_eval, dim3(grid), dim3(256), 0, s,
+  if (c % 8 == 0) {
+    hipLaunchKernelGGL(bn_apply_eval_v8, dim3(_egrid(m * c / 8)), dim3(256),
+                       0, s, (const s16x8*)x, (s16x8*)y, m, c, rm, rv, gamma,
+                       beta, eps);
+    return;
+  }
+  hipLaunchKernelGGL(bn_apply_eval, dim3(_egrid(m * c)), dim3(256), 0, s,
                      (const unsigned short*)x, (unsigned short*)y, m, c, rm,
                      rv, gamma, beta, eps);
 }
@@ -203,6 +402,12 @@ void launch_bn_apply_eval(const void* x, void* y, long m, int c,
 void launch_bn_bwd_reduce(const void* x, const void* dy, long m, int c,
                           const float* mean, const float* istd, float* dgamma,
                           float* dbeta, hipStream_t s) {
+  if (c % 8 == 0) {
+    hipLaunchKernelGGL(bn_bwd_reduce_v8, _colgrid_v8(m, c), dim3(256), 0, s,
+                       (const s16x8*)x, (const s16x8*)dy, m, c, mean, istd,
+                       dgamma, dbeta);
+    return;
+  }
   hipLaunchKernelGGL(bn_bwd_reduce, _colgrid(m, c), dim3(256), 0, s,
                      (const unsigned short*)x, (const unsigned short*)dy, m, c,
                      mean, istd, dgamma, dbeta);
@@ -212,9 +417,13 @@ void launch_bn_bwd_apply(const void* x, const void* dy, void* dx, long m,
                          int c, const float* mean, const float* istd,
                          const float* gamma, const float* dgamma,
                          const float* dbeta, hipStream_t s) {
-  long total = m * c;
-  int grid = (int)min((long)2048, (total + 255) / 256 + 1);
-  hipLaunchKernelGGL(bn_bwd_apply, dim3(grid), dim3(256), 0, s,
+  if (c % 8 == 0) {
+    hipLaunchKernelGGL(bn_bwd_apply_v8, dim3(_egrid(m * c / 8)), dim3(256), 0,
+                       s, (const s16x8*)x, (const s16x8*)dy, (s16x8*)dx, m, c,
+                       mean, istd, gamma, dgamma, dbeta);
+    return;
+  }
+  hipLaunchKernelGGL(bn_bwd_apply, dim3(_egrid(m * c)), dim3(256), 0, s,
                      (const unsigned short*)x, (const unsigned short*)dy,
                      (unsigned short*)dx, m, c, mean, istd, gamma, dgamma,
                      dbeta);
