@@ -38,7 +38,8 @@ __global__ void bn_stats_v8(const s16x8* __restrict__ x, long m, int c,
       }
     }
   }
-  // reduce row-lanes channel-slot by channel-slot through LDS
+  // reduce row-lanes through LDS; ONE partial row per block, no atomics
+  // (global-atomic contention across 1024 blocks was 4x slower)
   __shared__ float red[256];
   #pragma unroll
   for (int j = 0; j < 8; ++j) {
@@ -47,7 +48,7 @@ __global__ void bn_stats_v8(const s16x8* __restrict__ x, long m, int c,
     if (sub == 0) {
       float acc = 0.f;
       for (int q = 0; q < lanes; ++q) acc += red[q * groups + g - g0];
-      atomicAdd(&sum[g * 8 + j], acc);
+      sum[(long)blockIdx.x * 2 * c + g * 8 + j] = acc;   // partials [gx][2c]
     }
     __syncthreads();
     red[threadIdx.x] = ss[j];
@@ -55,10 +56,25 @@ __global__ void bn_stats_v8(const s16x8* __restrict__ x, long m, int c,
     if (sub == 0) {
       float acc = 0.f;
       for (int q = 0; q < lanes; ++q) acc += red[q * groups + g - g0];
-      atomicAdd(&sumsq[g * 8 + j], acc);
+      sum[(long)blockIdx.x * 2 * c + c + g * 8 + j] = acc;
     }
     __syncthreads();
   }
+}
+
+// second pass: partials [gx][2c] -> sum[c], sumsq[c]
+__global__ void bn_stats_sum2(const float* __restrict__ part, int gx, int c,
+                              float* __restrict__ sum,
+                              float* __restrict__ sumsq) {
+  int col = blockIdx.x * blockDim.x + threadIdx.x;
+  if (col >= c) return;
+  float a = 0.f, b = 0.f;
+  for (int r = 0; r < gx; ++r) {
+    a += part[(long)r * 2 * c + col];
+    b += part[(long)r * 2 * c + c + col];
+  }
+  sum[col] = a;
+  sumsq[col] = b;
 }
 
 __global__ void bn_apply_v8(const s16x8* __restrict__ x,
@@ -148,7 +164,7 @@ __global__ void bn_bwd_reduce_v8(const s16x8* __restrict__ x,
     if (sub == 0) {
       float acc = 0.f;
       for (int q = 0; q < lanes; ++q) acc += red[q * groups + g - g0];
-      atomicAdd(&dgamma[g * 8 + j], acc);
+      dgamma[(long)blockIdx.x * 2 * c + g * 8 + j] = acc;  // partials
     }
     __syncthreads();
     red[threadIdx.x] = db[j];
@@ -156,10 +172,24 @@ __global__ void bn_bwd_reduce_v8(const s16x8* __restrict__ x,
     if (sub == 0) {
       float acc = 0.f;
       for (int q = 0; q < lanes; ++q) acc += red[q * groups + g - g0];
-      atomicAdd(&dbeta[g * 8 + j], acc);
+      dgamma[(long)blockIdx.x * 2 * c + c + g * 8 + j] = acc;
     }
     __syncthreads();
   }
+}
+
+__global__ void bn_bwd_sum2(const float* __restrict__ part, int gx, int c,
+                            float* __restrict__ dgamma,
+                            float* __restrict__ dbeta) {
+  int col = blockIdx.x * blockDim.x + threadIdx.x;
+  if (col >= c) return;
+  float a = 0.f, b = 0.f;
+  for (int r = 0; r < gx; ++r) {
+    a += part[(long)r * 2 * c + col];
+    b += part[(long)r * 2 * c + c + col];
+  }
+  dgamma[col] = a;
+  dbeta[col] = b;
 }
 
 __global__ void bn_bwd_apply_v8(const s16x8* __restrict__ x,
@@ -351,13 +381,24 @@ static int _egrid(long total) {
   return (int)min((long)2048, total / 256 + 1);
 }
 
+// v8 two-pass (partials in scratch [gx][2c], no atomics); returns gx.
+int launch_bn_stats_part(const void* x, long m, int c, float* scratch,
+                         hipStream_t s) {
+  dim3 g = _colgrid_v8(m, c);
+  if (g.x > 256) g.x = 256;
+  hipLaunchKernelGGL(bn_stats_v8, g, dim3(256), 0, s, (const s16x8*)x, m, c,
+                     scratch, nullptr);
+  return (int)g.x;
+}
+
+void launch_bn_stats_sum2(const float* scratch, int gx, int c, float* sum,
+                          float* sumsq, hipStream_t s) {
+  hipLaunchKernelGGL(bn_stats_sum2, dim3((c + 255) / 256), dim3(256), 0, s,
+                     scratch, gx, c, sum, sumsq);
+}
+
 void launch_bn_stats(const void* x, long m, int c, float* sum, float* sumsq,
                      hipStream_t s) {
-  if (c % 8 == 0) {
-    hipLaunchKernelGGL(bn_stats_v8, _colgrid_v8(m, c), dim3(256), 0, s,
-                       (const s16x8*)x, m, c, sum, sumsq);
-    return;
-  }
   hipLaunchKernelGGL(bn_stats, _colgrid(m, c), dim3(256), 0, s,
                      (const unsigned short*)x, m, c, sum, sumsq);
 }
@@ -399,15 +440,25 @@ void launch_bn_apply_eval(const void* x, void* y, long m, int c,
                      rv, gamma, beta, eps);
 }
 
+int launch_bn_bwd_reduce_part(const void* x, const void* dy, long m, int c,
+                              const float* mean, const float* istd,
+                              float* scratch, hipStream_t s) {
+  dim3 g = _colgrid_v8(m, c);
+  if (g.x > 256) g.x = 256;
+  hipLaunchKernelGGL(bn_bwd_reduce_v8, g, dim3(256), 0, s, (const s16x8*)x,
+                     (const s16x8*)dy, m, c, mean, istd, scratch, nullptr);
+  return (int)g.x;
+}
+
+void launch_bn_bwd_sum2(const float* scratch, int gx, int c, float* dgamma,
+                        float* dbeta, hipStream_t s) {
+  hipLaunchKernelGGL(bn_bwd_sum2, dim3((c + 255) / 256), dim3(256), 0, s,
+                     scratch, gx, c, dgamma, dbeta);
+}
+
 void launch_bn_bwd_reduce(const void* x, const void* dy, long m, int c,
                           const float* mean, const float* istd, float* dgamma,
                           float* dbeta, hipStream_t s) {
-  if (c % 8 == 0) {
-    hipLaunchKernelGGL(bn_bwd_reduce_v8, _colgrid_v8(m, c), dim3(256), 0, s,
-                       (const s16x8*)x, (const s16x8*)dy, m, c, mean, istd,
-                       dgamma, dbeta);
-    return;
-  }
   hipLaunchKernelGGL(bn_bwd_reduce, _colgrid(m, c), dim3(256), 0, s,
                      (const unsigned short*)x, (const unsigned short*)dy, m, c,
                      mean, istd, dgamma, dbeta);

@@ -88,10 +88,19 @@ __global__ void col_sum_v8(const s16x8* __restrict__ a,
     if (sub == 0) {
       float acc = 0.f;
       for (int q = 0; q < lanes; ++q) acc += red[q * groups + g - g0];
-      atomicAdd(&out[g * 8 + j], acc);
+      out[(long)blockIdx.x * n + g * 8 + j] = acc;   // partials [gx][n]
     }
     __syncthreads();
   }
+}
+
+__global__ void col_sum_sum2(const float* __restrict__ part, int gx, int n,
+                             float* __restrict__ out) {
+  int col = blockIdx.x * blockDim.x + threadIdx.x;
+  if (col >= n) return;
+  float a = 0.f;
+  for (int r = 0; r < gx; ++r) a += part[(long)r * n + col];
+  out[col] = a;
 }
 
 __global__ void col_sum_bf16(const unsigned short* __restrict__ a,
@@ -225,18 +234,26 @@ void launch_act_bwd(const void* dy, const void* y, void* dx, long n, int act,
                        (unsigned short*)dx, n8 * 8, n, act, slope);
 }
 
+int launch_col_sum_part(const void* a, float* scratch, long m, int n,
+                        hipStream_t s) {
+  int n8 = n / 8;
+  int groups = n8 < 32 ? n8 : 32;
+  int lanes = 256 / groups;
+  long chunks = (m + lanes - 1) / lanes;
+  dim3 grid((unsigned)min((long)256, max((long)1, chunks)),
+            (unsigned)ceil_div(n8, 32));
+  hipLaunchKernelGGL(col_sum_v8, grid, dim3(256), 0, s, (const s16x8*)a,
+                     scratch, m, n);
+  return (int)grid.x;
+}
+
+void launch_col_sum_sum2(const float* scratch, int gx, int n, float* out,
+                         hipStream_t s) {
+  hipLaunchKernelGGL(col_sum_sum2, dim3(ceil_div(n, 256)), dim3(256), 0, s,
+                     scratch, gx, n, out);
+}
+
 void launch_col_sum(const void* a, float* out, long m, int n, hipStream_t s) {
-  if (n % 8 == 0) {
-    int n8 = n / 8;
-    int groups = n8 < 32 ? n8 : 32;
-    int lanes = 256 / groups;
-    long chunks = (m + lanes - 1) / lanes;
-    dim3 grid((unsigned)min((long)1024, max((long)1, chunks)),
-              (unsigned)ceil_div(n8, 32));
-    hipLaunchKernelGGL(col_sum_v8, grid, dim3(256), 0, s, (const s16x8*)a,
-                       out, m, n);
-    return;
-  }
   int colsW = min(256, n);
   int lanes = 256 / colsW;
   long chunks = (m + lanes - 1) / lanes;

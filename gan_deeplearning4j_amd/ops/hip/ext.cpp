@@ -80,6 +80,15 @@ void launch_softmax_xent_fwd(const void*, const void*, float*, void*, int,
 void launch_softmax_xent_bwd(const void*, const void*, void*, float, long,
                              hipStream_t);
 void launch_bn_stats(const void*, long, int, float*, float*, hipStream_t);
+int launch_bn_stats_part(const void*, long, int, float*, hipStream_t);
+void launch_bn_stats_sum2(const float*, int, int, float*, float*,
+                          hipStream_t);
+int launch_bn_bwd_reduce_part(const void*, const void*, long, int,
+                              const float*, const float*, float*,
+                              hipStream_t);
+void launch_bn_bwd_sum2(const float*, int, int, float*, float*, hipStream_t);
+int launch_col_sum_part(const void*, float*, long, int, hipStream_t);
+void launch_col_sum_sum2(const float*, int, int, float*, hipStream_t);
 void launch_bn_finalize(const float*, const float*, long, int, float, float,
                         float*, float*, float*, float*, hipStream_t);
 void launch_bn_apply(const void*, void*, long, int, const float*,
@@ -327,7 +336,17 @@ torch::Tensor act_bwd(torch::Tensor dy, torch::Tensor y, int64_t act,
 torch::Tensor col_sum(torch::Tensor a) {
   check_bf16(a, "a");
   int64_t m = a.size(0), n = a.size(1);
-  torch::Tensor out = torch::zeros({n}, a.options().dtype(torch::kFloat32));
+  auto f32 = a.options().dtype(torch::kFloat32);
+  if (n % 8 == 0) {
+    torch::Tensor out = torch::empty({n}, f32);
+    torch::Tensor scratch = torch::empty({256, n}, f32);
+    int gx = launch_col_sum_part(a.data_ptr(), scratch.data_ptr<float>(),
+                                 (long)m, (int)n, cur_stream());
+    launch_col_sum_sum2(scratch.data_ptr<float>(), gx, (int)n,
+                        out.data_ptr<float>(), cur_stream());
+    return out;
+  }
+  torch::Tensor out = torch::zeros({n}, f32);
   launch_col_sum(a.data_ptr(), out.data_ptr<float>(), (long)m, (int)n,
                  cur_stream());
   return out;
@@ -385,12 +404,24 @@ std::vector<torch::Tensor> bn_fwd_train(torch::Tensor x, torch::Tensor gamma,
   int64_t c = x.size(-1);
   long m = x.numel() / c;
   auto f32 = x.options().dtype(torch::kFloat32);
-  torch::Tensor sum = torch::zeros({c}, f32), sumsq = torch::zeros({c}, f32);
   torch::Tensor mean = torch::empty({c}, f32), istd = torch::empty({c}, f32);
   torch::Tensor y = torch::empty_like(x);
   auto s = cur_stream();
-  launch_bn_stats(x.data_ptr(), m, (int)c, sum.data_ptr<float>(),
-                  sumsq.data_ptr<float>(), s);
+  torch::Tensor sum, sumsq;
+  if (c % 8 == 0) {
+    sum = torch::empty({c}, f32);
+    sumsq = torch::empty({c}, f32);
+    torch::Tensor scratch = torch::empty({256, 2 * c}, f32);
+    int gx = launch_bn_stats_part(x.data_ptr(), m, (int)c,
+                                  scratch.data_ptr<float>(), s);
+    launch_bn_stats_sum2(scratch.data_ptr<float>(), gx, (int)c,
+                         sum.data_ptr<float>(), sumsq.data_ptr<float>(), s);
+  } else {
+    sum = torch::zeros({c}, f32);
+    sumsq = torch::zeros({c}, f32);
+    launch_bn_stats(x.data_ptr(), m, (int)c, sum.data_ptr<float>(),
+                    sumsq.data_ptr<float>(), s);
+  }
   launch_bn_finalize(sum.data_ptr<float>(), sumsq.data_ptr<float>(), m,
                      (int)c, (float)eps, (float)momentum,
                      mean.data_ptr<float>(), istd.data_ptr<float>(),
@@ -429,13 +460,27 @@ std::vector<torch::Tensor> bn_bwd(torch::Tensor x, torch::Tensor dy,
   int64_t c = x.size(-1);
   long m = x.numel() / c;
   auto f32 = x.options().dtype(torch::kFloat32);
-  torch::Tensor dgamma = torch::zeros({c}, f32);
-  torch::Tensor dbeta = torch::zeros({c}, f32);
   torch::Tensor dx = torch::empty_like(x);
   auto s = cur_stream();
-  launch_bn_bwd_reduce(x.data_ptr(), dy.data_ptr(), m, (int)c,
-                       mean.data_ptr<float>(), istd.data_ptr<float>(),
+  torch::Tensor dgamma, dbeta;
+  if (c % 8 == 0) {
+    dgamma = torch::empty({c}, f32);
+    dbeta = torch::empty({c}, f32);
+    torch::Tensor scratch = torch::empty({256, 2 * c}, f32);
+    int gx = launch_bn_bwd_reduce_part(x.data_ptr(), dy.data_ptr(), m,
+                                       (int)c, mean.data_ptr<float>(),
+                                       istd.data_ptr<float>(),
+                                       scratch.data_ptr<float>(), s);
+    launch_bn_bwd_sum2(scratch.data_ptr<float>(), gx, (int)c,
                        dgamma.data_ptr<float>(), dbeta.data_ptr<float>(), s);
+  } else {
+    dgamma = torch::zeros({c}, f32);
+    dbeta = torch::zeros({c}, f32);
+    launch_bn_bwd_reduce(x.data_ptr(), dy.data_ptr(), m, (int)c,
+                         mean.data_ptr<float>(), istd.data_ptr<float>(),
+                         dgamma.data_ptr<float>(), dbeta.data_ptr<float>(),
+                         s);
+  }
   launch_bn_bwd_apply(x.data_ptr(), dy.data_ptr(), dx.data_ptr(), m, (int)c,
                       mean.data_ptr<float>(), istd.data_ptr<float>(),
                       gamma.data_ptr<float>(), dgamma.data_ptr<float>(),
