@@ -62,19 +62,25 @@ __global__ void bn_stats_v8(const s16x8* __restrict__ x, long m, int c,
   }
 }
 
-// second pass: partials [gx][2c] -> sum[c], sumsq[c]
+// second pass: partials [gx][2c] -> sum[c], sumsq[c]. One wave per
+// channel (a single 256-thread block serializes on load latency).
 __global__ void bn_stats_sum2(const float* __restrict__ part, int gx, int c,
                               float* __restrict__ sum,
                               float* __restrict__ sumsq) {
-  int col = blockIdx.x * blockDim.x + threadIdx.x;
+  int col = blockIdx.x * (blockDim.x >> 6) + ((int)threadIdx.x >> 6);
+  int lane = (int)threadIdx.x & 63;
   if (col >= c) return;
   float a = 0.f, b = 0.f;
-  for (int r = 0; r < gx; ++r) {
+  for (int r = lane; r < gx; r += 64) {
     a += part[(long)r * 2 * c + col];
     b += part[(long)r * 2 * c + c + col];
   }
-  sum[col] = a;
-  sumsq[col] = b;
+  a = wave_reduce_sum(a);
+  b = wave_reduce_sum(b);
+  if (lane == 0) {
+    sum[col] = a;
+    sumsq[col] = b;
+  }
 }
 
 __global__ void bn_apply_v8(const s16x8* __restrict__ x,
@@ -181,15 +187,20 @@ __global__ void bn_bwd_reduce_v8(const s16x8* __restrict__ x,
 __global__ void bn_bwd_sum2(const float* __restrict__ part, int gx, int c,
                             float* __restrict__ dgamma,
                             float* __restrict__ dbeta) {
-  int col = blockIdx.x * blockDim.x + threadIdx.x;
+  int col = blockIdx.x * (blockDim.x >> 6) + ((int)threadIdx.x >> 6);
+  int lane = (int)threadIdx.x & 63;
   if (col >= c) return;
   float a = 0.f, b = 0.f;
-  for (int r = 0; r < gx; ++r) {
+  for (int r = lane; r < gx; r += 64) {
     a += part[(long)r * 2 * c + col];
     b += part[(long)r * 2 * c + c + col];
   }
-  dgamma[col] = a;
-  dbeta[col] = b;
+  a = wave_reduce_sum(a);
+  b = wave_reduce_sum(b);
+  if (lane == 0) {
+    dgamma[col] = a;
+    dbeta[col] = b;
+  }
 }
 
 __global__ void bn_bwd_apply_v8(const s16x8* __restrict__ x,
@@ -393,7 +404,7 @@ int launch_bn_stats_part(const void* x, long m, int c, float* scratch,
 
 void launch_bn_stats_sum2(const float* scratch, int gx, int c, float* sum,
                           float* sumsq, hipStream_t s) {
-  hipLaunchKernelGGL(bn_stats_sum2, dim3((c + 255) / 256), dim3(256), 0, s,
+  hipLaunchKernelGGL(bn_stats_sum2, dim3((c + 3) / 4), dim3(256), 0, s,
                      scratch, gx, c, sum, sumsq);
 }
 
@@ -452,7 +463,7 @@ int launch_bn_bwd_reduce_part(const void* x, const void* dy, long m, int c,
 
 void launch_bn_bwd_sum2(const float* scratch, int gx, int c, float* dgamma,
                         float* dbeta, hipStream_t s) {
-  hipLaunchKernelGGL(bn_bwd_sum2, dim3((c + 255) / 256), dim3(256), 0, s,
+  hipLaunchKernelGGL(bn_bwd_sum2, dim3((c + 3) / 4), dim3(256), 0, s,
                      scratch, gx, c, dgamma, dbeta);
 }
 

@@ -96,11 +96,13 @@ __global__ void col_sum_v8(const s16x8* __restrict__ a,
 
 __global__ void col_sum_sum2(const float* __restrict__ part, int gx, int n,
                              float* __restrict__ out) {
-  int col = blockIdx.x * blockDim.x + threadIdx.x;
+  int col = blockIdx.x * (blockDim.x >> 6) + ((int)threadIdx.x >> 6);
+  int lane = (int)threadIdx.x & 63;
   if (col >= n) return;
   float a = 0.f;
-  for (int r = 0; r < gx; ++r) a += part[(long)r * n + col];
-  out[col] = a;
+  for (int r = lane; r < gx; r += 64) a += part[(long)r * n + col];
+  a = wave_reduce_sum(a);
+  if (lane == 0) out[col] = a;
 }
 
 __global__ void col_sum_bf16(const unsigned short* __restrict__ a,
@@ -249,7 +251,7 @@ int launch_col_sum_part(const void* a, float* scratch, long m, int n,
 
 void launch_col_sum_sum2(const float* scratch, int gx, int n, float* out,
                          hipStream_t s) {
-  hipLaunchKernelGGL(col_sum_sum2, dim3(ceil_div(n, 256)), dim3(256), 0, s,
+  hipLaunchKernelGGL(col_sum_sum2, dim3(ceil_div(n, 4)), dim3(256), 0, s,
                      scratch, gx, n, out);
 }
 
