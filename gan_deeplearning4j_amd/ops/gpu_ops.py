@@ -280,18 +280,29 @@ class _Conv2d(torch.autograd.Function):
                     H, W, R, S, stride, pad, 0, 0.0, 1)   # mode 1
                 dx = _as_nchw_view(dx2d.view(N, H, W, C)).to(ctx.dtypes[0])
             else:
-                # strided: 3/4 of transposed taps are invalid, so the
-                # dcol + col2im route has 4x fewer MFMA FLOPs
-                rsc8 = R * S * C8
-                wt = _packed(wp, "wt", lambda: _pad_k(
-                    wp[:, :rsc8].t().contiguous()))   # [rsc8, kout_pad]
-                dprep = _pad_k(dpre)
-                dcol = ext.gemm_tn(dprep, wt, None, 0, 0.0, False)
-                dxh = ext.col2im(dcol, N, H, W, C8, Ho, Wo, R, S, stride,
-                                 pad, rsc8, None, 0, 0.0)
-                if C8 != C:
-                    dxh = dxh[..., :C].contiguous()
-                dx = _as_nchw_view(dxh).to(ctx.dtypes[0])
+                # strided: parity-decomposed gather (each output parity
+                # class is a dense GEMM over its valid taps only)
+                def build_wd_cls():
+                    wc = _bf(w.detach().permute(1, 2, 3, 0))  # [C,R,S,Kout]
+                    wc = _pad_channels(
+                        wc.reshape(C * R * S, 1, 1, Kout), Ko8
+                    ).reshape(C, R, S, Ko8)
+                    packs = []
+                    for qh in range(stride):
+                        for qw in range(stride):
+                            prh = (qh + pad) % stride
+                            prw = (qw + pad) % stride
+                            sub = wc[:, prh::stride, prw::stride, :]
+                            packs.append(_pad_k(
+                                sub.reshape(C, -1).contiguous()))
+                    return packs
+
+                wd_cls = _packed(w, "dgrad_wcls", build_wd_cls)
+                dpre_img = dpre8.view(N, Ho, Wo, Ko8)
+                dx2d = ext.conv_parity_implicit(
+                    dpre_img, wd_cls, None, _zp(dpre.device), N, Ho, Wo,
+                    Ko8, H, W, C, R, S, stride, pad, 0, 0.0)
+                dx = _as_nchw_view(dx2d.view(N, H, W, C)).to(ctx.dtypes[0])
         if ctx.has_bias and ctx.needs_input_grad[2]:
             db = ext.col_sum(dpre).to(ctx.dtypes[2])
         return dx, dw, db, None, None, None, None
@@ -325,15 +336,26 @@ class _ConvTranspose2d(torch.autograd.Function):
                                         act, slope, 1)   # mode 1
             yh = y2d.view(N, Ho, Wo, Cout)
         else:
-            # strided: GEMM over Cin then scatter-free col2im gather with
-            # fused bias+activation (transposed taps 3/4-invalid at s=2)
-            x2d = _pad_k(xh.reshape(-1, Cin))
-            w2a = _packed(w, "w2a", lambda: _pad_k(
-                _bf(w.detach().permute(2, 3, 1, 0))
-                .reshape(R * S * Cout, Cin)))
-            col = ext.gemm_tn(x2d, w2a, None, 0, 0.0, False)  # [NPin,RSCout]
-            yh = ext.col2im(col, N, Ho, Wo, Cout, Hi, Wi, R, S, stride, pad,
-                            R * S * Cout, bias, act, slope)
+            # strided: parity-decomposed gathered GEMMs with fused
+            # bias+activation (each output parity class reads only its
+            # valid taps; no col buffer / col2im pass)
+            def build_wcls():
+                wc = _bf(w.detach().permute(1, 2, 3, 0))  # [Cout,R,S,Cin]
+                packs = []
+                for qh in range(stride):
+                    for qw in range(stride):
+                        prh = (qh + pad) % stride
+                        prw = (qw + pad) % stride
+                        sub = wc[:, prh::stride, prw::stride, :]
+                        packs.append(_pad_k(sub.reshape(Cout, -1)
+                                            .contiguous()))
+                return packs
+
+            wcls = _packed(w, "convt_fwd_wcls", build_wcls)
+            y2d = ext.conv_parity_implicit(
+                xh, wcls, bias, _zp(x.device), N, Hi, Wi, Cin, Ho, Wo,
+                Cout, R, S, stride, pad, act, slope)
+            yh = y2d.view(N, Ho, Wo, Cout)
         ctx.save_for_backward(xh, yh)
         ctx.geom = (N, Cin, Hi, Wi, Cout, R, S, Ho, Wo, stride, pad)
         ctx.act, ctx.slope = act, slope

@@ -106,12 +106,19 @@ DEV_INLINE unsigned fmod_(unsigned n, FastDiv f, unsigned q) {
 // mode 0 (forward):    src pixel = grid_pos * stride - pad + (r,s)
 // mode 1 (transposed): src pixel = (grid_pos + pad - (r,s)) / stride,
 //                      valid only when divisible (conv dgrad / convT fwd)
+// mode 2 (parity class of a stride-2 transposed conv): the GEMM covers one
+//   (qh,qw) output-parity class; R,S are the per-class tap counts and
+//   (r,s) taps map to src pixel (grid_pos + off - tap); output rows
+//   scatter back to the full image via (oH,oW,oqh,oqw) in the epilogue.
+//   Every tap is valid (no 4x zero-padding like mode 1 at stride 2).
 struct ConvGather {
   int N, H, W, C;        // source image dims (NHWC)
-  int Ho, Wo;            // patch grid
+  int Ho, Wo;            // patch grid (mode 2: the class grid)
   int R, S, stride, pad;
   int rsc;               // R*S*C (valid k range; >= rsc is zero padding)
   int mode;
+  int off_h, off_w;      // mode 2: ho = hi2 + off - r2
+  int oH, oW, oqh, oqw;  // mode 2: output scatter (full dims + class)
   FastDiv fC, fS, fWo, fHo, fStride;
 };
 

@@ -43,6 +43,8 @@ static ConvGather make_gather(int N, int H, int W, int C, int Ho, int Wo,
   g.R = R; g.S = S; g.stride = stride; g.pad = pad;
   g.rsc = R * S * C;
   g.mode = mode;
+  g.off_h = g.off_w = 0;
+  g.oH = g.oW = g.oqh = g.oqw = 0;
   g.fC = make_fastdiv_h(C);
   g.fS = make_fastdiv_h(S);
   g.fWo = make_fastdiv_h(Wo);
@@ -236,6 +238,58 @@ torch::Tensor gemm_nt_implicit(torch::Tensor A, torch::Tensor B,
                  (int)Ndim, (int)Kdim, lda, ldb, (int)splitk, (int)gmode, g,
                  zero_page.data_ptr(), cur_stream());
   return C;
+}
+
+// Parity-decomposed transposed conv (stride s): the output splits into
+// s*s parity classes; each class is a dense gathered GEMM over its valid
+// (r2,s2) taps (no 4x zero-tap inflation, no dcol/col2im round trip).
+// Covers conv data-grad AND convT forward. wcls: one [Nout][kpad_class]
+// weight pack per class, ordered qh-major.
+torch::Tensor conv_parity_implicit(
+    torch::Tensor img, std::vector<torch::Tensor> wcls,
+    c10::optional<torch::Tensor> bias, torch::Tensor zero_page, int64_t Nb,
+    int64_t Hs, int64_t Ws, int64_t Cs, int64_t oHd, int64_t oWd,
+    int64_t Nout, int64_t R, int64_t S, int64_t stride, int64_t pad,
+    int64_t act, double slope) {
+  check_bf16(img, "img");
+  TORCH_CHECK(Cs % 8 == 0, "parity conv needs source C % 8 == 0");
+  TORCH_CHECK((int64_t)wcls.size() == stride * stride, "one pack per class");
+  const float* bias_p = nullptr;
+  if (bias.has_value() && bias->defined() && bias->numel() > 0) {
+    check_f32(*bias, "bias");
+    bias_p = bias->data_ptr<float>();
+  }
+  torch::Tensor y = torch::empty({Nb * oHd * oWd, Nout}, img.options());
+  int s_ = (int)stride;
+  for (int qh = 0; qh < s_; ++qh) {
+    for (int qw = 0; qw < s_; ++qw) {
+      torch::Tensor& wq = wcls[qh * s_ + qw];
+      check_bf16(wq, "wcls");
+      int prh = (int)((qh + pad) % stride);
+      int prw = (int)((qw + pad) % stride);
+      int R2 = (int)((R - prh + stride - 1) / stride);
+      int S2 = (int)((S - prw + stride - 1) / stride);
+      int Hq = (int)((oHd - qh + stride - 1) / stride);
+      int Wq = (int)((oWd - qw + stride - 1) / stride);
+      if (Hq <= 0 || Wq <= 0 || R2 <= 0 || S2 <= 0) continue;
+      TORCH_CHECK(wq.size(0) == Nout, "pack Nout");
+      ConvGather g = make_gather((int)Nb, (int)Hs, (int)Ws, (int)Cs, Hq, Wq,
+                                 R2, S2, (int)stride, (int)pad, 2);
+      g.off_h = (int)((qh + pad - prh) / stride);
+      g.off_w = (int)((qw + pad - prw) / stride);
+      g.oH = (int)oHd;
+      g.oW = (int)oWd;
+      g.oqh = qh;
+      g.oqw = qw;
+      int M = (int)(Nb * Hq * Wq);
+      int K = (int)wq.size(1);
+      launch_gemm_tn_gather(img.data_ptr(), wq.data_ptr(), y.data_ptr(),
+                            bias_p, M, (int)Nout, K, K, (int)act,
+                            (float)slope, g, zero_page.data_ptr(),
+                            cur_stream());
+    }
+  }
+  return y;
 }
 
 // ------------------------------------------------------------------ conv
@@ -585,6 +639,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("gemm_nt", &gemm_nt, "C = A^T.B (contraction over rows) fp32 out");
   mod.def("conv_fwd_implicit", &conv_fwd_implicit,
           "implicit-GEMM conv forward (gathered im2col A)");
+  mod.def("conv_parity_implicit", &conv_parity_implicit,
+          "parity-decomposed strided transposed conv / conv dgrad");
   mod.def("gemm_nt_implicit", &gemm_nt_implicit,
           "weight-grad GEMM with one operand gathered as im2col");
   mod.def("fp8_quantize", &fp8_quantize,

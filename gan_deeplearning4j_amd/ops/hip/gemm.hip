@@ -64,6 +64,10 @@ DEV_INLINE const unsigned short* gather_addr(
     hi = ho * g.stride - g.pad + r;
     wi = wo * g.stride - g.pad + s;
     valid = hi >= 0 && hi < g.H && wi >= 0 && wi < g.W;
+  } else if (g.mode == 2) {
+    hi = ho + g.off_h - r;
+    wi = wo + g.off_w - s;
+    valid = hi >= 0 && hi < g.H && wi >= 0 && wi < g.W;
   } else {
     int hop = ho + g.pad - r;
     int wop = wo + g.pad - s;
@@ -244,12 +248,19 @@ __global__ __launch_bounds__(256, 2) void gemm_tn_core(
       int grow = m0 + row;
       int gcol = n0 + seg * 8;
       if (grow < M && gcol < N) {
+        long crow = grow;
+        if (GATHER_A && ga.mode == 2) {
+          int n, h2, w2;
+          np_decode(ga, (unsigned)grow, n, h2, w2);
+          crow = ((long)n * ga.oH + h2 * ga.stride + ga.oqh) * ga.oW +
+                 w2 * ga.stride + ga.oqw;
+        }
         s16x8 v = *(const s16x8*)(ctile + row * 128 + seg * 8);
         if (gcol + 8 <= N) {
-          *(s16x8*)(&C[(long)grow * N + gcol]) = v;
+          *(s16x8*)(&C[crow * N + gcol]) = v;
         } else {
           for (int j = 0; j < 8 && gcol + j < N; ++j)
-            C[(long)grow * N + gcol + j] = (unsigned short)v[j];
+            C[crow * N + gcol + j] = (unsigned short)v[j];
         }
       }
     }
@@ -267,11 +278,18 @@ __global__ __launch_bounds__(256, 2) void gemm_tn_core(
       for (int r = 0; r < 4; ++r) {
         int row = m0 + wr * 64 + mi * 16 + fq * 4 + r;
         if (row >= M) continue;
+        long crow = row;
+        if (GATHER_A && ga.mode == 2) {
+          int n, h2, w2;
+          np_decode(ga, (unsigned)row, n, h2, w2);
+          crow = ((long)n * ga.oH + h2 * ga.stride + ga.oqh) * ga.oW +
+                 w2 * ga.stride + ga.oqw;
+        }
         float v = act_fwd(acc[mi][ni][r] + bv, act, slope);
         if (C != nullptr)
-          C[(long)row * N + col] = f2bf(v);
+          C[crow * N + col] = f2bf(v);
         else
-          Cf[(long)row * N + col] = v;
+          Cf[crow * N + col] = v;
       }
     }
   }
