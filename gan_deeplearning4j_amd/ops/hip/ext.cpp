@@ -21,6 +21,7 @@ struct FastDiv {
 };
 struct ConvGather {
   int N, H, W, C, Ho, Wo, R, S, stride, pad, rsc, mode;
+  int off_h, off_w, oH, oW, oqh, oqw;
   FastDiv fC, fS, fWo, fHo, fStride;
 };
 
