@@ -264,6 +264,27 @@ class ReferenceProtocolTrainer:
         self.soft_real = (1 + std * torch.randn(b, 1, generator=self._cpu_gen))
         self.batch_counter = 0
 
+    # ----------------------------------------------------------- resume
+    def resume(self) -> bool:
+        """Restore all four graphs (incl. updater state) from the zip
+        checkpoints in out_dir, if present. The reference is save-only
+         — resume is this framework's extension."""
+        paths = {name: self.out_dir / f"mnist_{name}_model.zip"
+                 for name in ("dis", "gan", "gen", "CV")}
+        if not all(p.exists() for p in paths.values()):
+            return False
+        dt = torch.bfloat16 if self.device.type == "cuda" else None
+        self.dis = ModelSerializer.restore_computation_graph(
+            paths["dis"]).to_device(self.device, dt)
+        self.gan = ModelSerializer.restore_computation_graph(
+            paths["gan"]).to_device(self.device, dt)
+        self.gen = ModelSerializer.restore_computation_graph(
+            paths["gen"]).to_device(self.device, dt)
+        self.cv = ModelSerializer.restore_computation_graph(
+            paths["CV"]).to_device(self.device, dt)
+        log.info("resumed 4 graphs from %s", self.out_dir)
+        return True
+
     # ------------------------------------------------------------------
     def _uniform_z(self, n: int) -> torch.Tensor:
         # Nd4j.rand.muli(2).subi(1): U(-1,1) (Java:420, 465)
