@@ -156,8 +156,13 @@ class _Linear(torch.autograd.Function):
         ext = hip_ext()
         xp, wp, y = ctx.saved_tensors
         dy = _bf(dy)
-        dpre = ext.act_bwd(dy, y, ctx.act, ctx.slope) if ctx.act else dy
         dx = dw = db = None
+        want_bias = ctx.has_bias and ctx.needs_input_grad[2]
+        if ctx.act and want_bias and dy.shape[1] % 8 == 0:
+            dpre, db_f = ext.act_bwd_bias(dy, y, ctx.act, ctx.slope)
+            db = db_f.to(ctx.dtypes[2])
+        else:
+            dpre = ext.act_bwd(dy, y, ctx.act, ctx.slope) if ctx.act else dy
         if ctx.needs_input_grad[0]:
             # dgrad: dx = dpre @ w ;  B = w^T padded over nout
             nin = ctx.nin
@@ -176,7 +181,7 @@ class _Linear(torch.autograd.Function):
             dpre8 = _pad8(dpre)
             dw = ext.gemm_nt(dpre8, xp, sk, _zp(xp.device))
             dw = dw[: nout, : ctx.nin].contiguous().to(ctx.dtypes[1])
-        if ctx.has_bias and ctx.needs_input_grad[2]:
+        if ctx.has_bias and ctx.needs_input_grad[2] and db is None:
             db = ext.col_sum(dpre).to(ctx.dtypes[2])
         if dx is not None:
             dx = dx.to(ctx.dtypes[0])
@@ -255,9 +260,14 @@ class _Conv2d(torch.autograd.Function):
         xh, wp, y2d = ctx.saved_tensors
         N, C, H, W, Kout, R, S, Ho, Wo, stride, pad, kpad, C8 = ctx.geom
         dy2d = _bf(dy.permute(0, 2, 3, 1)).reshape(-1, Kout)
-        dpre = ext.act_bwd(dy2d, y2d, ctx.act, ctx.slope) if ctx.act else dy2d
-
         dx = dw = db = None
+        want_bias = ctx.has_bias and ctx.needs_input_grad[2]
+        if ctx.act and want_bias and Kout % 8 == 0:
+            dpre, db_f = ext.act_bwd_bias(dy2d, y2d, ctx.act, ctx.slope)
+            db = db_f.to(ctx.dtypes[2])
+        else:
+            dpre = (ext.act_bwd(dy2d, y2d, ctx.act, ctx.slope)
+                    if ctx.act else dy2d)
         Ko8 = (Kout + 7) // 8 * 8
         dpre8 = _pad8(dpre) if Ko8 != Kout else dpre
         if ctx.needs_input_grad[1]:
@@ -324,7 +334,7 @@ class _Conv2d(torch.autograd.Function):
                 if C8 != C:
                     dxh = dxh[..., :C].contiguous()
                 dx = _as_nchw_view(dxh).to(ctx.dtypes[0])
-        if ctx.has_bias and ctx.needs_input_grad[2]:
+        if ctx.has_bias and ctx.needs_input_grad[2] and db is None:
             db = ext.col_sum(dpre).to(ctx.dtypes[2])
         return dx, dw, db, None, None, None, None
 
@@ -403,7 +413,15 @@ class _ConvTranspose2d(torch.autograd.Function):
         rsco = R * S * Cout
         rscop = _rup64(rsco)
         dyh = _nhwc(dy)                       # [N,Ho,Wo,Cout]
-        if ctx.act:
+        db = None
+        want_bias = ctx.has_bias and ctx.needs_input_grad[2]
+        if ctx.act and want_bias and Cout % 8 == 0:
+            dpre, db_f = ext.act_bwd_bias(dyh.reshape(-1, Cout),
+                                          yh.reshape(-1, Cout), ctx.act,
+                                          ctx.slope)
+            db = db_f.to(ctx.dtypes[2])
+            dpre_img = dpre.view(N, Ho, Wo, Cout)
+        elif ctx.act:
             dpre = ext.act_bwd(dyh.reshape(-1, Cout),
                                yh.reshape(-1, Cout), ctx.act, ctx.slope)
             dpre_img = dpre.view(N, Ho, Wo, Cout)
@@ -415,7 +433,7 @@ class _ConvTranspose2d(torch.autograd.Function):
         Co8 = (Cout + 7) // 8 * 8
         dpre8 = _pad_channels(dpre_img, Co8)
         npq = N * Hi * Wi
-        dx = dw = db = None
+        dx = dw = None
         if ctx.needs_input_grad[0]:
             # dx[np_in][cin] = sum_{r,s,cout} dpre[ho=hi*s-p+r..][cout]
             #                  * W[cin][cout][r][s]  (forward-gather, mode 0)
@@ -445,7 +463,7 @@ class _ConvTranspose2d(torch.autograd.Function):
                 sk, _zp(x2d.device))
             dw = (dw2a[:rsco8, :Cin].reshape(R, S, Co8, Cin)[:, :, :Cout]
                   .permute(3, 2, 0, 1).contiguous().to(ctx.dtypes[1]))
-        if ctx.has_bias and ctx.needs_input_grad[2]:
+        if ctx.has_bias and ctx.needs_input_grad[2] and db is None:
             db = ext.col_sum(dpre_img.reshape(-1, Cout)).to(ctx.dtypes[2])
         return dx, dw, db, None, None, None, None
 

@@ -105,6 +105,61 @@ __global__ void col_sum_sum2(const float* __restrict__ part, int gx, int n,
   if (lane == 0) out[col] = a;
 }
 
+// fused activation-backward + bias-grad partials over [M][N] (N % 8 == 0):
+// dx = dy * act'(y) is written AND per-column partial sums of dx land in
+// partials [gx][n] (same second pass as col_sum). Saves a full re-read of
+// dpre for the bias gradient.
+__global__ void act_bwd_bias_v8(const s16x8* __restrict__ dy,
+                                const s16x8* __restrict__ y,
+                                s16x8* __restrict__ dx,
+                                float* __restrict__ partials, long m, int n,
+                                int act, float slope) {
+  int n8 = n / 8;
+  int g0 = blockIdx.y * 32;
+  int groups = min(32, n8 - g0);
+  int lanes = (int)blockDim.x / groups;
+  int sub = (int)threadIdx.x / groups;
+  int g = g0 + (int)threadIdx.x % groups;
+  float s[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+  if (sub < lanes) {
+    for (long r = (long)blockIdx.x * lanes + sub; r < m;
+         r += (long)gridDim.x * lanes) {
+      s16x8 vg = dy[r * n8 + g];
+      s16x8 o;
+      if (act != 0) {
+        s16x8 vy = y[r * n8 + g];
+        #pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          float d = bf2f((unsigned short)vg[j]) *
+                    act_bwd_from_y(bf2f((unsigned short)vy[j]), act, slope);
+          o[j] = (short)f2bf(d);
+          s[j] += d;
+        }
+      } else {
+        #pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          float d = bf2f((unsigned short)vg[j]);
+          o[j] = (short)f2bf(d);
+          s[j] += d;
+        }
+      }
+      dx[r * n8 + g] = o;
+    }
+  }
+  __shared__ float red[256];
+  #pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    red[threadIdx.x] = s[j];
+    __syncthreads();
+    if (sub == 0) {
+      float acc = 0.f;
+      for (int q = 0; q < lanes; ++q) acc += red[q * groups + g - g0];
+      partials[(long)blockIdx.x * n + g * 8 + j] = acc;
+    }
+    __syncthreads();
+  }
+}
+
 __global__ void col_sum_bf16(const unsigned short* __restrict__ a,
                              float* __restrict__ out, long m, int n) {
   __shared__ float ls[256];
@@ -234,6 +289,21 @@ void launch_act_bwd(const void* dy, const void* y, void* dx, long n, int act,
     hipLaunchKernelGGL(act_bwd_tail, dim3(1), dim3(64), 0, s,
                        (const unsigned short*)dy, (const unsigned short*)y,
                        (unsigned short*)dx, n8 * 8, n, act, slope);
+}
+
+int launch_act_bwd_bias(const void* dy, const void* y, void* dx,
+                        float* scratch, long m, int n, int act, float slope,
+                        hipStream_t s) {
+  int n8 = n / 8;
+  int groups = n8 < 32 ? n8 : 32;
+  int lanes = 256 / groups;
+  long chunks = (m + lanes - 1) / lanes;
+  dim3 grid((unsigned)min((long)256, max((long)1, chunks)),
+            (unsigned)ceil_div(n8, 32));
+  hipLaunchKernelGGL(act_bwd_bias_v8, grid, dim3(256), 0, s,
+                     (const s16x8*)dy, (const s16x8*)y, (s16x8*)dx, scratch,
+                     m, n, act, slope);
+  return (int)grid.x;
 }
 
 int launch_col_sum_part(const void* a, float* scratch, long m, int n,

@@ -92,6 +92,8 @@ int launch_bn_bwd_reduce_part(const void*, const void*, long, int,
 void launch_bn_bwd_sum2(const float*, int, int, float*, float*, hipStream_t);
 int launch_col_sum_part(const void*, float*, long, int, hipStream_t);
 void launch_col_sum_sum2(const float*, int, int, float*, hipStream_t);
+int launch_act_bwd_bias(const void*, const void*, void*, float*, long, int,
+                        int, float, hipStream_t);
 void launch_bn_finalize(const float*, const float*, long, int, float, float,
                         float*, float*, float*, float*, hipStream_t);
 void launch_bn_apply(const void*, void*, long, int, const float*,
@@ -378,6 +380,32 @@ torch::Tensor act_fwd(torch::Tensor x, int64_t act, double slope) {
   return y;
 }
 
+// fused act-backward + bias grad: returns (dpre, dbias_f32)
+std::vector<torch::Tensor> act_bwd_bias(torch::Tensor dy,
+                                        c10::optional<torch::Tensor> y,
+                                        int64_t act, double slope) {
+  check_bf16(dy, "dy");
+  int64_t m = dy.size(0), n = dy.size(1);
+  TORCH_CHECK(n % 8 == 0, "act_bwd_bias needs N % 8 == 0");
+  auto f32 = dy.options().dtype(torch::kFloat32);
+  torch::Tensor dx = torch::empty_like(dy);
+  torch::Tensor scratch = torch::empty({256, n}, f32);
+  torch::Tensor db = torch::empty({n}, f32);
+  const void* y_p = nullptr;
+  if (act != 0) {
+    TORCH_CHECK(y.has_value(), "non-identity act needs y");
+    check_bf16(*y, "y");
+    y_p = y->data_ptr();
+  }
+  auto s = cur_stream();
+  int gx = launch_act_bwd_bias(dy.data_ptr(), y_p, dx.data_ptr(),
+                               scratch.data_ptr<float>(), (long)m, (int)n,
+                               (int)act, (float)slope, s);
+  launch_col_sum_sum2(scratch.data_ptr<float>(), gx, (int)n,
+                      db.data_ptr<float>(), s);
+  return {dx, db};
+}
+
 torch::Tensor act_bwd(torch::Tensor dy, torch::Tensor y, int64_t act,
                       double slope) {
   check_bf16(dy, "dy");
@@ -656,6 +684,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("upsample_bwd", &upsample_bwd);
   mod.def("act_fwd", &act_fwd);
   mod.def("act_bwd", &act_bwd);
+  mod.def("act_bwd_bias", &act_bwd_bias,
+          "fused activation backward + bias gradient");
   mod.def("col_sum", &col_sum);
   mod.def("bce_fwd", &bce_fwd);
   mod.def("bce_bwd", &bce_bwd);
