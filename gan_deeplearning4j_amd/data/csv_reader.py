@@ -52,6 +52,19 @@ class CSVRecordReader:
 
     def initialize(self, path: str | Path) -> "CSVRecordReader":
         path = Path(path)
+        # native multithreaded parser from the _C extension when available
+        # (~50x numpy.loadtxt); numpy fallback keeps the reader usable
+        # without the built extension (and for non-comma delimiters)
+        if self.delimiter == ",":
+            try:
+                from ..ops.backend import hip_ext
+
+                t = hip_ext().csv_load(str(path), self.skip_lines)
+                if t.numel():
+                    self._data = t.numpy()
+                    return self
+            except (RuntimeError, ImportError):
+                pass
         self._data = np.loadtxt(
             path, delimiter=self.delimiter, skiprows=self.skip_lines,
             dtype=np.float32, ndmin=2,

@@ -6,6 +6,9 @@
 #include <torch/extension.h>
 #include <ATen/cuda/CUDAContext.h>
 
+// csv_loader.cpp (native data loader)
+torch::Tensor csv_load(const std::string& path, int64_t skip_rows);
+
 namespace {
 
 struct ConvGeom {
@@ -696,5 +699,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("bn_bwd", &bn_bwd);
   mod.def("fused_adam", &fused_adam);
   mod.def("fused_rmsprop", &fused_rmsprop);
+  mod.def("csv_load", &csv_load, "multithreaded CSV -> fp32 tensor");
   mod.attr("arch") = "gfx950";
 }

@@ -50,7 +50,7 @@ def compile_hip_objects() -> list[str]:
 
 ext = cpp_extension.CUDAExtension(
     name="gan_deeplearning4j_amd._C",
-    sources=[str(HIP_DIR / "ext.cpp")],
+    sources=[str(HIP_DIR / "ext.cpp"), str(HIP_DIR / "csv_loader.cpp")],
     extra_objects=compile_hip_objects(),
     extra_compile_args={"cxx": ["-O2"], "nvcc": ["-O2"]},
 )

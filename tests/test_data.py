@@ -48,3 +48,19 @@ def test_iterator_shuffle_determinism(tmp_path):
     a = list(RecordReaderDataSetIterator(r, 16, 8, 2, shuffle=True, seed=3))
     b = list(RecordReaderDataSetIterator(r, 16, 8, 2, shuffle=True, seed=3))
     assert torch.equal(a[0].features, b[0].features)
+
+
+def test_native_csv_loader_matches_numpy(tmp_path):
+    """The _C extension's multithreaded CSV parser must agree with numpy."""
+    from gan_deeplearning4j_amd.ops.backend import has_hip_ext, hip_ext
+
+    if not has_hip_ext():
+        import pytest
+
+        pytest.skip("extension not built")
+    p = write_synthetic_csv(tmp_path / "n.csv", "transactions", n=123,
+                            num_features=17)
+    a = hip_ext().csv_load(str(p), 0).numpy()
+    b = np.loadtxt(p, delimiter=",", dtype=np.float32)
+    assert a.shape == b.shape
+    assert np.allclose(a, b, atol=1e-5)
