@@ -122,7 +122,9 @@ __global__ void col2im_nhwc(const unsigned short* __restrict__ dcol,
   }
 }
 
-// fast col2im: C % 8 == 0
+// fast col2im: C % 8 == 0. SS: compile-time stride (1, 2) or 0 = generic
+// (the stride div/mod chain was a measurable cost in the gather loop).
+template <int SS>
 __global__ void col2im_nhwc_v8(const unsigned short* __restrict__ dcol,
                                s16x8* __restrict__ din, ConvGeom g,
                                const float* __restrict__ bias, int act,
@@ -141,13 +143,16 @@ __global__ void col2im_nhwc_v8(const unsigned short* __restrict__ dcol,
     float acc[8] = {0, 0, 0, 0, 0, 0, 0, 0};
     for (int r = 0; r < g.R; ++r) {
       int hop = h + g.pad - r;
-      if (hop < 0 || hop % g.stride) continue;
-      int ho = hop / g.stride;
+      int st = SS ? SS : g.stride;
+      if (hop < 0 || (SS == 2 ? (hop & 1) : (SS == 1 ? 0 : hop % st)))
+        continue;
+      int ho = SS == 2 ? (hop >> 1) : (SS == 1 ? hop : hop / st);
       if (ho >= g.Ho) continue;
       for (int s_ = 0; s_ < g.S; ++s_) {
         int wop = w + g.pad - s_;
-        if (wop < 0 || wop % g.stride) continue;
-        int wo = wop / g.stride;
+        if (wop < 0 || (SS == 2 ? (wop & 1) : (SS == 1 ? 0 : wop % st)))
+          continue;
+        int wo = SS == 2 ? (wop >> 1) : (SS == 1 ? wop : wop / st);
         if (wo >= g.Wo) continue;
         long np = ((long)n * g.Ho + ho) * g.Wo + wo;
         s16x8 v = *(const s16x8*)(&dcol[np * g.kpad +
@@ -189,9 +194,18 @@ void launch_col2im(const void* dcol, void* din, ConvGeom g, const float* bias,
   if (g.C % 8 == 0 && g.kpad % 8 == 0) {
     long total8 = (long)g.N * g.H * g.W * (g.C / 8);
     int grid = (int)min((long)2048, (total8 + 255) / 256 + 1);
-    hipLaunchKernelGGL(col2im_nhwc_v8, dim3(grid), dim3(256), 0, s,
-                       (const unsigned short*)dcol, (s16x8*)din, g, bias, act,
-                       slope);
+    if (g.stride == 2)
+      hipLaunchKernelGGL((col2im_nhwc_v8<2>), dim3(grid), dim3(256), 0, s,
+                         (const unsigned short*)dcol, (s16x8*)din, g, bias,
+                         act, slope);
+    else if (g.stride == 1)
+      hipLaunchKernelGGL((col2im_nhwc_v8<1>), dim3(grid), dim3(256), 0, s,
+                         (const unsigned short*)dcol, (s16x8*)din, g, bias,
+                         act, slope);
+    else
+      hipLaunchKernelGGL((col2im_nhwc_v8<0>), dim3(grid), dim3(256), 0, s,
+                         (const unsigned short*)dcol, (s16x8*)din, g, bias,
+                         act, slope);
     return;
   }
   long total = (long)g.N * g.H * g.W * g.C;
