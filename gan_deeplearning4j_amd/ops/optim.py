@@ -99,7 +99,10 @@ class Updater:
             if self._t_dev is None:
                 dev = next(s.param.device for s in self.slots
                            if s.param.is_cuda)
-                self._t_dev = torch.zeros(1, dtype=torch.int32, device=dev)
+                # seed from the host count so bias correction is right
+                # after a checkpoint resume (self.t was already bumped)
+                self._t_dev = torch.full((1,), self.t - 1,
+                                         dtype=torch.int32, device=dev)
             self._t_dev += 1
         for s in self.slots:
             g = s.param.grad

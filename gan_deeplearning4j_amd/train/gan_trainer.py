@@ -200,6 +200,9 @@ class GanTrainer:
                 self._graph_failed = True
                 self._graph = None
                 return self._step_eager(real)
+        if real.shape != self._static_real.shape:
+            # shape changed after capture: run this batch eagerly
+            return self._step_eager(real)
         self._static_real.copy_(real.to(self.device, self.dtype),
                                 non_blocking=True)
         self._graph.replay()
