@@ -135,6 +135,37 @@ class ComputationGraph(nn.Module):
         self.preprocessors = nn.ModuleDict(preprocessors)
         self._updater: Optional[Updater] = None
         self._initialized = False
+        self._bn_fusion_pass()
+
+    def _bn_fusion_pass(self) -> None:
+        """Producer->BatchNorm statistics fusion: when a Conv/ConvT/Dense
+        vertex's sole consumer is a BatchNorm over the same feature axis
+        (no preprocessor in between), the producer's GPU epilogue also
+        emits the BN batch statistics, skipping BN's own stats pass."""
+        from .layers import (BatchNormLayer, Conv2dLayer,
+                             ConvTranspose2dLayer, DenseLayer, OutputLayer)
+
+        consumers: dict[str, list[str]] = {}
+        for name in self._topo:
+            for src in self._vertex_inputs[name]:
+                consumers.setdefault(src, []).append(name)
+        for name in self._topo:
+            layer = self.layers[name]
+            if not isinstance(layer, BatchNormLayer):
+                continue
+            if name in self.preprocessors:
+                continue  # reshape between producer and BN: axes differ
+            srcs = self._vertex_inputs[name]
+            if len(srcs) != 1 or srcs[0] not in self.layers:
+                continue
+            prod = self.layers[srcs[0]]
+            if len(consumers.get(srcs[0], [])) != 1:
+                continue
+            if isinstance(prod, OutputLayer):
+                continue
+            if isinstance(prod, (Conv2dLayer, ConvTranspose2dLayer,
+                                 DenseLayer)):
+                prod.emit_bn_stats = True
 
     # ------------------------------------------------------------ build
     def _toposort(self, vertices) -> list[str]:

@@ -33,6 +33,10 @@ class BaseLayer(nn.Module):
         self.lr = lr
         self.frozen = frozen
         self.name: str = ""
+        # set by the graph's fusion pass when this layer's sole consumer is
+        # a BatchNorm over the same feature axis: the producer's GPU kernel
+        # then also emits the BN batch statistics (skips one full pass)
+        self.emit_bn_stats: bool = False
 
     # -- DL4J-style param access (getParam/setParam, reference Java:431-459)
     def get_param(self, key: str) -> torch.Tensor:
@@ -105,7 +109,9 @@ class DenseLayer(BaseLayer):
                 self.bias.zero_()
 
     def forward(self, x):
-        return OF.linear(x, self.weight, self.bias, self.activation, self.slope)
+        return OF.linear(x, self.weight, self.bias, self.activation,
+                         self.slope,
+                         emit_stats=self.emit_bn_stats and self.training)
 
     def out_shape(self, in_shape):
         return (*in_shape[:-1], self.n_out)
@@ -151,7 +157,8 @@ class Conv2dLayer(BaseLayer):
 
     def forward(self, x):
         return OF.conv2d(x, self.weight, self.bias, self.stride, self.padding,
-                         self.activation, self.slope)
+                         self.activation, self.slope,
+                         emit_stats=self.emit_bn_stats and self.training)
 
     def out_shape(self, in_shape):
         n, c, h, w = in_shape
@@ -183,7 +190,9 @@ class ConvTranspose2dLayer(BaseLayer):
 
     def forward(self, x):
         return OF.conv_transpose2d(x, self.weight, self.bias, self.stride,
-                                   self.padding, self.activation, self.slope)
+                                   self.padding, self.activation, self.slope,
+                                   emit_stats=self.emit_bn_stats and
+                                   self.training)
 
     def out_shape(self, in_shape):
         n, c, h, w = in_shape

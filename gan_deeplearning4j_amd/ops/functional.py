@@ -38,12 +38,15 @@ def linear(
     b: Optional[torch.Tensor] = None,
     act: Optional[str] = None,
     slope: float = 0.2,
+    emit_stats: bool = False,
 ) -> torch.Tensor:
-    """y = act(x @ w.T + b); w is [out, in] (torch convention)."""
+    """y = act(x @ w.T + b); w is [out, in] (torch convention).
+    emit_stats: fuse the consumer BatchNorm's batch statistics into the
+    GEMM epilogue (GPU; no-op on CPU)."""
     if x.is_cuda:
         from . import gpu_ops
 
-        return gpu_ops.linear(x, w, b, act or "identity", slope)
+        return gpu_ops.linear(x, w, b, act or "identity", slope, emit_stats)
     return _apply_act(F.linear(x, w, b), act, slope)
 
 
@@ -55,12 +58,14 @@ def conv2d(
     padding: int = 0,
     act: Optional[str] = None,
     slope: float = 0.2,
+    emit_stats: bool = False,
 ) -> torch.Tensor:
     """NCHW-logical conv; on GPU runs NHWC im2col-MFMA-GEMM kernels."""
     if x.is_cuda:
         from . import gpu_ops
 
-        return gpu_ops.conv2d(x, w, b, stride, padding, act or "identity", slope)
+        return gpu_ops.conv2d(x, w, b, stride, padding, act or "identity",
+                              slope, emit_stats)
     return _apply_act(F.conv2d(x, w, b, stride=stride, padding=padding), act, slope)
 
 
@@ -72,6 +77,7 @@ def conv_transpose2d(
     padding: int = 0,
     act: Optional[str] = None,
     slope: float = 0.2,
+    emit_stats: bool = False,
 ) -> torch.Tensor:
     """True transposed conv (the reference emulates it as upsample+conv,
     Java:201-219; the north-star names a real transposed-conv kernel)."""
@@ -79,7 +85,7 @@ def conv_transpose2d(
         from . import gpu_ops
 
         return gpu_ops.conv_transpose2d(
-            x, w, b, stride, padding, act or "identity", slope
+            x, w, b, stride, padding, act or "identity", slope, emit_stats
         )
     return _apply_act(
         F.conv_transpose2d(x, w, b, stride=stride, padding=padding), act, slope

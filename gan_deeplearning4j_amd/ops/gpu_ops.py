@@ -79,6 +79,20 @@ def _splitk_for(m_tiles: int, n_tiles: int, kchunks: int) -> int:
 _zero_pages: dict = {}
 _dims_cache: dict = {}
 
+# side-channel for producer-fused BN statistics: the producing Function
+# deposits (sum, sumsq) here keyed by the output tensor's data_ptr; the
+# BatchNorm wrapper consumes (and clears) them. Entries are overwritten
+# every producer call, so the dict stays O(#fused layers).
+_bn_stats_chan: dict = {}
+
+
+def deposit_bn_stats(t: torch.Tensor, stats) -> None:
+    _bn_stats_chan[t.data_ptr()] = stats
+
+
+def take_bn_stats(t: torch.Tensor):
+    return _bn_stats_chan.pop(t.data_ptr(), None)
+
 # Strided dgrad/convT-fwd algorithm choice: parity-decomposed gathered
 # GEMMs vs dcol+col2im. Measured on DCGAN-64: dcol wins by ~4% (the
 # parity gather re-reads the source once per tap, so traffic is a wash
@@ -137,13 +151,17 @@ def _packed(w: torch.Tensor, key: str, builder):
 # ===================================================================== linear
 class _Linear(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, x, w, b, act: int, slope: float):
+    def forward(ctx, x, w, b, act: int, slope: float, emit_stats: bool):
         ext = hip_ext()
         xp = _pad_k(_bf(x))
         wp = _packed(w, "wp", lambda: _pad_k(_bf(w.detach())))
         bias = (_packed(b, "f32", lambda: b.detach().float().contiguous())
                 if b is not None else None)
-        y = ext.gemm_tn(xp, wp, bias, act, slope, False)
+        if emit_stats and w.shape[0] % 8 == 0:
+            y, ssum, ssq = ext.gemm_tn_stats(xp, wp, bias, act, slope)
+            deposit_bn_stats(y, (ssum, ssq))
+        else:
+            y = ext.gemm_tn(xp, wp, bias, act, slope, False)
         ctx.save_for_backward(xp, wp, y)
         ctx.act, ctx.slope = act, slope
         ctx.nin = x.shape[-1]
@@ -185,12 +203,12 @@ class _Linear(torch.autograd.Function):
             db = ext.col_sum(dpre).to(ctx.dtypes[2])
         if dx is not None:
             dx = dx.to(ctx.dtypes[0])
-        return dx, dw, db, None, None
+        return dx, dw, db, None, None, None
 
 
-def linear(x, w, b=None, act="identity", slope=0.2):
+def linear(x, w, b=None, act="identity", slope=0.2, emit_stats=False):
     x2 = x.reshape(-1, x.shape[-1]) if x.dim() > 2 else x
-    y = _Linear.apply(_bf(x2), w, b, ACT_CODES[act], slope)
+    y = _Linear.apply(_bf(x2), w, b, ACT_CODES[act], slope, emit_stats)
     if x.dim() > 2:
         y = y.reshape(*x.shape[:-1], y.shape[-1])
     return y
@@ -213,7 +231,8 @@ def _pad_channels(xh: torch.Tensor, c8: int) -> torch.Tensor:
 
 class _Conv2d(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, x, w, b, stride: int, pad: int, act: int, slope: float):
+    def forward(ctx, x, w, b, stride: int, pad: int, act: int, slope: float,
+                emit_stats: bool):
         ext = hip_ext()
         N, C, H, W = x.shape
         Kout, _, R, S = w.shape
@@ -232,6 +251,7 @@ class _Conv2d(torch.autograd.Function):
         wp = _packed(w, "conv_wp", build_wp)
         bias = (_packed(b, "f32", lambda: b.detach().float().contiguous())
                 if b is not None else None)
+        stats = None
         if FP8_CONV and C8 % 16 == 0:
             xq, _, ix = ext.fp8_quantize(xh)
             wq, _, iw = _packed(w, "fp8", lambda: tuple(
@@ -243,16 +263,23 @@ class _Conv2d(torch.autograd.Function):
                                             act, slope)
         else:
             # implicit GEMM: im2col gather fused into the MFMA staging
-            y2d = ext.conv_fwd_implicit(xh, wp, bias, _zp(x.device), N, H, W,
+            # (optionally also emitting the consumer BN's batch statistics)
+            res = ext.conv_fwd_implicit(xh, wp, bias, _zp(x.device), N, H, W,
                                         C8, Ho, Wo, R, S, stride, pad, act,
-                                        slope, 0)
+                                        slope, 0, 1 if emit_stats else 0)
+            y2d = res[0]
+            if len(res) == 3:
+                stats = (res[1], res[2])
         ctx.save_for_backward(xh, wp, y2d)
         ctx.geom = (N, C, H, W, Kout, R, S, Ho, Wo, stride, pad, kpad, C8)
         ctx.act, ctx.slope = act, slope
         ctx.has_bias = b is not None
         ctx.dtypes = (x.dtype, w.dtype, b.dtype if b is not None else None)
         ctx.wref = w
-        return _as_nchw_view(y2d.view(N, Ho, Wo, Kout))
+        out = _as_nchw_view(y2d.view(N, Ho, Wo, Kout))
+        if stats is not None:
+            deposit_bn_stats(out, stats)
+        return out
 
     @staticmethod
     def backward(ctx, dy):
@@ -296,7 +323,7 @@ class _Conv2d(torch.autograd.Function):
                 dpre_img = dpre8.view(N, Ho, Wo, Ko8)
                 dx2d = ext.conv_fwd_implicit(
                     dpre_img, wd, None, _zp(dpre.device), N, Ho, Wo, Ko8,
-                    H, W, R, S, stride, pad, 0, 0.0, 1)   # mode 1
+                    H, W, R, S, stride, pad, 0, 0.0, 1, 0)[0]   # mode 1
                 dx = _as_nchw_view(dx2d.view(N, H, W, C)).to(ctx.dtypes[0])
             elif PARITY_STRIDED:
                 # strided: parity-decomposed gather (each output parity
@@ -336,17 +363,20 @@ class _Conv2d(torch.autograd.Function):
                 dx = _as_nchw_view(dxh).to(ctx.dtypes[0])
         if ctx.has_bias and ctx.needs_input_grad[2] and db is None:
             db = ext.col_sum(dpre).to(ctx.dtypes[2])
-        return dx, dw, db, None, None, None, None
+        return dx, dw, db, None, None, None, None, None
 
 
-def conv2d(x, w, b=None, stride=1, padding=0, act="identity", slope=0.2):
-    return _Conv2d.apply(x, w, b, stride, padding, ACT_CODES[act], slope)
+def conv2d(x, w, b=None, stride=1, padding=0, act="identity", slope=0.2,
+           emit_stats=False):
+    return _Conv2d.apply(x, w, b, stride, padding, ACT_CODES[act], slope,
+                         emit_stats)
 
 
 # ============================================================ conv transpose
 class _ConvTranspose2d(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, x, w, b, stride: int, pad: int, act: int, slope: float):
+    def forward(ctx, x, w, b, stride: int, pad: int, act: int, slope: float,
+                emit_stats: bool):
         ext = hip_ext()
         N, Cin, Hi, Wi = x.shape
         _, Cout, R, S = w.shape
@@ -355,6 +385,7 @@ class _ConvTranspose2d(torch.autograd.Function):
         xh = _nhwc(x)                       # [N,Hi,Wi,Cin]; Cin % 8 == 0
         bias = (_packed(b, "f32", lambda: b.detach().float().contiguous())
                 if b is not None else None)
+        stats = None
         if stride == 1:
             # ONE transposed-gather GEMM with fused bias+activation:
             # y[ho,wo,cout] = act( sum_{r,s,cin}[valid hi=ho+pad-r]
@@ -362,9 +393,13 @@ class _ConvTranspose2d(torch.autograd.Function):
             wt = _packed(w, "convt_fwd_w", lambda: _pad_k(
                 _bf(w.detach().permute(1, 2, 3, 0))      # [Cout,R,S,Cin]
                 .reshape(Cout, R * S * Cin)))
-            y2d = ext.conv_fwd_implicit(xh, wt, bias, _zp(x.device), N, Hi,
+            res = ext.conv_fwd_implicit(xh, wt, bias, _zp(x.device), N, Hi,
                                         Wi, Cin, Ho, Wo, R, S, stride, pad,
-                                        act, slope, 1)   # mode 1
+                                        act, slope, 1,
+                                        1 if emit_stats else 0)   # mode 1
+            y2d = res[0]
+            if len(res) == 3:
+                stats = (res[1], res[2])
             yh = y2d.view(N, Ho, Wo, Cout)
         elif PARITY_STRIDED:
             # strided: parity-decomposed gathered GEMMs with fused
@@ -389,20 +424,30 @@ class _ConvTranspose2d(torch.autograd.Function):
             yh = y2d.view(N, Ho, Wo, Cout)
         else:
             # strided default: GEMM over Cin + col2im with fused bias+act
+            # (and optionally the consumer BN's batch statistics)
             x2d = _pad_k(xh.reshape(-1, Cin))
             w2a = _packed(w, "w2a", lambda: _pad_k(
                 _bf(w.detach().permute(2, 3, 1, 0))
                 .reshape(R * S * Cout, Cin)))
             col = ext.gemm_tn(x2d, w2a, None, 0, 0.0, False)  # [NPin,RSCout]
-            yh = ext.col2im(col, N, Ho, Wo, Cout, Hi, Wi, R, S, stride, pad,
-                            R * S * Cout, bias, act, slope)
+            if emit_stats and Cout % 8 == 0:
+                yh, ssum, ssq = ext.col2im_stats(
+                    col, N, Ho, Wo, Cout, Hi, Wi, R, S, stride, pad,
+                    R * S * Cout, bias, act, slope)
+                stats = (ssum, ssq)
+            else:
+                yh = ext.col2im(col, N, Ho, Wo, Cout, Hi, Wi, R, S, stride,
+                                pad, R * S * Cout, bias, act, slope)
         ctx.save_for_backward(xh, yh)
         ctx.geom = (N, Cin, Hi, Wi, Cout, R, S, Ho, Wo, stride, pad)
         ctx.act, ctx.slope = act, slope
         ctx.has_bias = b is not None
         ctx.dtypes = (x.dtype, w.dtype, b.dtype if b is not None else None)
         ctx.wref = w
-        return _as_nchw_view(yh)
+        out = _as_nchw_view(yh)
+        if stats is not None:
+            deposit_bn_stats(out, stats)
+        return out
 
     @staticmethod
     def backward(ctx, dy):
@@ -447,7 +492,7 @@ class _ConvTranspose2d(torch.autograd.Function):
             dx2d = ext.conv_fwd_implicit(dpre8, w2b, None,
                                          _zp(dpre8.device), N, Ho, Wo,
                                          Co8, Hi, Wi, R, S, stride, pad,
-                                         0, 0.0, 0)
+                                         0, 0.0, 0, 0)[0]
             dx = _as_nchw_view(dx2d.view(N, Hi, Wi, Cin)).to(ctx.dtypes[0])
         if ctx.needs_input_grad[1]:
             # wgrad: dW[(r,s,cout)][cin] = sum_np im2col(dpre)[np][rs*cout]
@@ -465,23 +510,30 @@ class _ConvTranspose2d(torch.autograd.Function):
                   .permute(3, 2, 0, 1).contiguous().to(ctx.dtypes[1]))
         if ctx.has_bias and ctx.needs_input_grad[2] and db is None:
             db = ext.col_sum(dpre_img.reshape(-1, Cout)).to(ctx.dtypes[2])
-        return dx, dw, db, None, None, None, None
+        return dx, dw, db, None, None, None, None, None
 
 
 def conv_transpose2d(x, w, b=None, stride=1, padding=0, act="identity",
-                     slope=0.2):
+                     slope=0.2, emit_stats=False):
     return _ConvTranspose2d.apply(x, w, b, stride, padding, ACT_CODES[act],
-                                  slope)
+                                  slope, emit_stats)
 
 
 # ================================================================ batch norm
 class _BatchNorm(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, x2, gamma, beta, rm, rv, momentum, eps):
+    def forward(ctx, x2, gamma, beta, rm, rv, momentum, eps, pre_stats):
         ext = hip_ext()
-        y, mean, istd = ext.bn_fwd_train(x2, gamma.detach().float(),
-                                         beta.detach().float(), rm, rv,
-                                         momentum, eps)
+        if pre_stats is not None:
+            # producer-fused statistics (conv/dense epilogue): skip the
+            # stats pass entirely
+            y, mean, istd = ext.bn_fwd_train_pre(
+                x2, pre_stats[0], pre_stats[1], gamma.detach().float(),
+                beta.detach().float(), rm, rv, momentum, eps)
+        else:
+            y, mean, istd = ext.bn_fwd_train(x2, gamma.detach().float(),
+                                             beta.detach().float(), rm, rv,
+                                             momentum, eps)
         ctx.save_for_backward(x2, mean, istd, gamma)
         return y
 
@@ -492,12 +544,13 @@ class _BatchNorm(torch.autograd.Function):
         dx, dgamma, dbeta = ext.bn_bwd(x2, _bf(dy), mean, istd,
                                        gamma.detach().float())
         return dx, dgamma.to(gamma.dtype), dbeta.to(gamma.dtype), None, \
-            None, None, None
+            None, None, None, None
 
 
 def batch_norm(x, weight, bias, running_mean, running_var, training,
                momentum=0.1, eps=1e-5):
     is4d = x.dim() == 4
+    pre_stats = take_bn_stats(x)
     if is4d:
         xh = _nhwc(x)
         x2 = xh.reshape(-1, xh.shape[-1])
@@ -505,7 +558,7 @@ def batch_norm(x, weight, bias, running_mean, running_var, training,
         x2 = _bf(x)
     if training:
         y2 = _BatchNorm.apply(x2, weight, bias, running_mean, running_var,
-                              momentum, eps)
+                              momentum, eps, pre_stats)
     else:
         y2 = hip_ext().bn_fwd_eval(x2, weight.detach().float(),
                                    bias.detach().float(), running_mean,

@@ -58,11 +58,14 @@ static ConvGather make_gather(int N, int H, int W, int C, int Ho, int Wo,
 }
 
 extern "C" {
-void launch_gemm_tn(const void*, const void*, void*, float*, const float*,
-                    int, int, int, long, long, int, float, hipStream_t);
-void launch_gemm_tn_gather(const void*, const void*, void*, const float*,
-                           int, int, int, long, int, float, ConvGather,
-                           const void*, hipStream_t);
+int launch_gemm_tn(const void*, const void*, void*, float*, const float*,
+                   int, int, int, long, long, int, float, float*,
+                   hipStream_t);
+int launch_gemm_tn_gather(const void*, const void*, void*, const float*,
+                          int, int, int, long, int, float, ConvGather,
+                          const void*, float*, hipStream_t);
+int launch_col2im_stats(const void*, void*, ConvGeom, const float*, int,
+                        float, float*, hipStream_t);
 void launch_gemm_nt(const void*, const void*, float*, int, int, int, long,
                     long, int, int, ConvGather, const void*, hipStream_t);
 void launch_im2col(const void*, void*, ConvGeom, hipStream_t);
@@ -160,8 +163,36 @@ torch::Tensor gemm_tn(torch::Tensor A, torch::Tensor B,
   launch_gemm_tn(A.data_ptr(), B.data_ptr(),
                  out_f32 ? nullptr : C.data_ptr(),
                  out_f32 ? C.data_ptr<float>() : nullptr, bias_p, (int)M,
-                 (int)N, (int)K, K, K, (int)act, (float)slope, cur_stream());
+                 (int)N, (int)K, K, K, (int)act, (float)slope, nullptr,
+                 cur_stream());
   return C;
+}
+
+// TN GEMM with fused BN statistics of the output: returns {y, sum, sumsq}
+std::vector<torch::Tensor> gemm_tn_stats(torch::Tensor A, torch::Tensor B,
+                                         c10::optional<torch::Tensor> bias,
+                                         int64_t act, double slope) {
+  check_bf16(A, "A");
+  check_bf16(B, "B");
+  int64_t M = A.size(0), K = A.size(1), N = B.size(0);
+  TORCH_CHECK(B.size(1) == K && K % 64 == 0 && N % 8 == 0);
+  const float* bias_p = nullptr;
+  if (bias.has_value() && bias->defined() && bias->numel() > 0) {
+    check_f32(*bias, "bias");
+    bias_p = bias->data_ptr<float>();
+  }
+  auto f32 = A.options().dtype(torch::kFloat32);
+  torch::Tensor C = torch::empty({M, N}, A.options());
+  int64_t gx = (M + 127) / 128;
+  torch::Tensor part = torch::empty({gx, 2 * N}, f32);
+  torch::Tensor sum = torch::empty({N}, f32), sumsq = torch::empty({N}, f32);
+  auto st = cur_stream();
+  launch_gemm_tn(A.data_ptr(), B.data_ptr(), C.data_ptr(), nullptr, bias_p,
+                 (int)M, (int)N, (int)K, K, K, (int)act, (float)slope,
+                 part.data_ptr<float>(), st);
+  launch_bn_stats_sum2(part.data_ptr<float>(), (int)gx, (int)N,
+                       sum.data_ptr<float>(), sumsq.data_ptr<float>(), st);
+  return {C, sum, sumsq};
 }
 
 torch::Tensor gemm_nt(torch::Tensor A, torch::Tensor B, int64_t splitk,
@@ -191,13 +222,14 @@ torch::Tensor gemm_nt(torch::Tensor A, torch::Tensor B, int64_t splitk,
 
 // Implicit-GEMM conv forward: y2d[NP][Kout] = act(im2col(x).Wp^T + bias)
 // without materializing col. x NHWC [Nb,H,W,C] (C % 8 == 0), Wp [Kout][kpad].
-torch::Tensor conv_fwd_implicit(torch::Tensor x, torch::Tensor Wp,
+std::vector<torch::Tensor> conv_fwd_implicit(torch::Tensor x, torch::Tensor Wp,
                                 c10::optional<torch::Tensor> bias,
                                 torch::Tensor zero_page, int64_t Nb,
                                 int64_t H, int64_t W, int64_t C, int64_t Ho,
                                 int64_t Wo, int64_t R, int64_t S,
                                 int64_t stride, int64_t pad, int64_t act,
-                                double slope, int64_t mode) {
+                                double slope, int64_t mode,
+                                int64_t want_stats) {
   check_bf16(x, "x");
   check_bf16(Wp, "Wp");
   TORCH_CHECK(C % 8 == 0, "implicit conv needs C % 8 == 0");
@@ -212,10 +244,26 @@ torch::Tensor conv_fwd_implicit(torch::Tensor x, torch::Tensor Wp,
   ConvGather g = make_gather((int)Nb, (int)H, (int)W, (int)C, (int)Ho,
                              (int)Wo, (int)R, (int)S, (int)stride, (int)pad,
                              (int)mode);
+  auto st = cur_stream();
+  if (want_stats && Kout % 8 == 0) {
+    auto f32 = x.options().dtype(torch::kFloat32);
+    int64_t gx = (M + 127) / 128;
+    torch::Tensor part = torch::empty({gx, 2 * Kout}, f32);
+    torch::Tensor sum = torch::empty({Kout}, f32);
+    torch::Tensor sumsq = torch::empty({Kout}, f32);
+    launch_gemm_tn_gather(x.data_ptr(), Wp.data_ptr(), y.data_ptr(), bias_p,
+                          (int)M, (int)Kout, (int)kpad, kpad, (int)act,
+                          (float)slope, g, zero_page.data_ptr(),
+                          part.data_ptr<float>(), st);
+    launch_bn_stats_sum2(part.data_ptr<float>(), (int)((M + 127) / 128),
+                         (int)Kout, sum.data_ptr<float>(),
+                         sumsq.data_ptr<float>(), st);
+    return std::vector<torch::Tensor>{y, sum, sumsq};
+  }
   launch_gemm_tn_gather(x.data_ptr(), Wp.data_ptr(), y.data_ptr(), bias_p,
                         (int)M, (int)Kout, (int)kpad, kpad, (int)act,
-                        (float)slope, g, zero_page.data_ptr(), cur_stream());
-  return y;
+                        (float)slope, g, zero_page.data_ptr(), nullptr, st);
+  return std::vector<torch::Tensor>{y};
 }
 
 // Implicit weight-grad: C[M][N] += sum_np A'[np][M] * B'[np][N] where the
@@ -291,7 +339,7 @@ torch::Tensor conv_parity_implicit(
       int K = (int)wq.size(1);
       launch_gemm_tn_gather(img.data_ptr(), wq.data_ptr(), y.data_ptr(),
                             bias_p, M, (int)Nout, K, K, (int)act,
-                            (float)slope, g, zero_page.data_ptr(),
+                            (float)slope, g, zero_page.data_ptr(), nullptr,
                             cur_stream());
     }
   }
@@ -327,6 +375,34 @@ torch::Tensor col2im(torch::Tensor dcol, int64_t N, int64_t H, int64_t W,
   launch_col2im(dcol.data_ptr(), out.data_ptr(), g, bias_p, (int)act,
                 (float)slope, cur_stream());
   return out;
+}
+
+// col2im with fused BN stats: returns {y_nhwc, sum, sumsq}
+std::vector<torch::Tensor> col2im_stats(
+    torch::Tensor dcol, int64_t N, int64_t H, int64_t W, int64_t C,
+    int64_t Ho, int64_t Wo, int64_t R, int64_t S, int64_t stride,
+    int64_t pad, int64_t kpad, c10::optional<torch::Tensor> bias,
+    int64_t act, double slope) {
+  check_bf16(dcol, "dcol");
+  TORCH_CHECK(C % 8 == 0 && kpad % 8 == 0);
+  ConvGeom g{(int)N, (int)H, (int)W, (int)C, (int)Ho, (int)Wo,
+             (int)R, (int)S, (int)stride, (int)pad, (int)kpad};
+  const float* bias_p = nullptr;
+  if (bias.has_value() && bias->defined() && bias->numel() > 0) {
+    check_f32(*bias, "bias");
+    bias_p = bias->data_ptr<float>();
+  }
+  auto f32 = dcol.options().dtype(torch::kFloat32);
+  torch::Tensor out = torch::empty({N, H, W, C}, dcol.options());
+  torch::Tensor part = torch::empty({256, 2 * C}, f32);
+  torch::Tensor sum = torch::empty({C}, f32), sumsq = torch::empty({C}, f32);
+  auto st = cur_stream();
+  int gx = launch_col2im_stats(dcol.data_ptr(), out.data_ptr(), g, bias_p,
+                               (int)act, (float)slope,
+                               part.data_ptr<float>(), st);
+  launch_bn_stats_sum2(part.data_ptr<float>(), gx, (int)C,
+                       sum.data_ptr<float>(), sumsq.data_ptr<float>(), st);
+  return {out, sum, sumsq};
 }
 
 // ------------------------------------------------------------------ pool
@@ -524,6 +600,36 @@ std::vector<torch::Tensor> bn_fwd_train(torch::Tensor x, torch::Tensor gamma,
   return {y, mean, istd};
 }
 
+// training forward from PRE-COMPUTED sums (fused producer statistics)
+std::vector<torch::Tensor> bn_fwd_train_pre(
+    torch::Tensor x, torch::Tensor sum, torch::Tensor sumsq,
+    torch::Tensor gamma, torch::Tensor beta, torch::Tensor running_mean,
+    torch::Tensor running_var, double momentum, double eps) {
+  check_bf16(x, "x");
+  check_f32(sum, "sum");
+  check_f32(sumsq, "sumsq");
+  int64_t c = x.size(-1);
+  long m = x.numel() / c;
+  auto f32 = x.options().dtype(torch::kFloat32);
+  torch::Tensor mean = torch::empty({c}, f32), istd = torch::empty({c}, f32);
+  torch::Tensor y = torch::empty_like(x);
+  auto s = cur_stream();
+  launch_bn_finalize(sum.data_ptr<float>(), sumsq.data_ptr<float>(), m,
+                     (int)c, (float)eps, (float)momentum,
+                     mean.data_ptr<float>(), istd.data_ptr<float>(),
+                     running_mean.defined() && running_mean.numel() > 0
+                         ? running_mean.data_ptr<float>()
+                         : nullptr,
+                     running_var.defined() && running_var.numel() > 0
+                         ? running_var.data_ptr<float>()
+                         : nullptr,
+                     s);
+  launch_bn_apply(x.data_ptr(), y.data_ptr(), m, (int)c,
+                  mean.data_ptr<float>(), istd.data_ptr<float>(),
+                  gamma.data_ptr<float>(), beta.data_ptr<float>(), s);
+  return {y, mean, istd};
+}
+
 torch::Tensor bn_fwd_eval(torch::Tensor x, torch::Tensor gamma,
                           torch::Tensor beta, torch::Tensor rm,
                           torch::Tensor rv, double eps) {
@@ -681,6 +787,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
           "implicit-GEMM conv forward on fp8 MFMA");
   mod.def("im2col", &im2col);
   mod.def("col2im", &col2im);
+  mod.def("col2im_stats", &col2im_stats,
+          "col2im with fused bias+act+BN statistics");
+  mod.def("gemm_tn_stats", &gemm_tn_stats,
+          "TN GEMM with fused output BN statistics");
   mod.def("maxpool_fwd", &maxpool_fwd);
   mod.def("maxpool_bwd", &maxpool_bwd);
   mod.def("upsample_fwd", &upsample_fwd);
@@ -695,6 +805,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("softmax_xent_fwd", &softmax_xent_fwd);
   mod.def("softmax_xent_bwd", &softmax_xent_bwd);
   mod.def("bn_fwd_train", &bn_fwd_train);
+  mod.def("bn_fwd_train_pre", &bn_fwd_train_pre,
+          "BN training forward from producer-fused statistics");
   mod.def("bn_fwd_eval", &bn_fwd_eval);
   mod.def("bn_bwd", &bn_bwd);
   mod.def("fused_adam", &fused_adam);

@@ -151,7 +151,8 @@ __global__ __launch_bounds__(256, 2) void gemm_tn_core(
     unsigned short* __restrict__ C, float* __restrict__ Cf,
     const float* __restrict__ bias, int M, int N, int K, long lda, long ldb,
     int act, float slope, ConvGather ga,
-    const unsigned short* __restrict__ zp) {
+    const unsigned short* __restrict__ zp,
+    float* __restrict__ bn_part) {
   __shared__ __attribute__((aligned(128))) char lds[4 * TN_TILE_B];
   auto abuf = [&](int i) -> char* { return lds + (i ? 2 * TN_TILE_B : 0); };
   auto bbuf = [&](int i) -> char* {
@@ -262,6 +263,31 @@ __global__ __launch_bounds__(256, 2) void gemm_tn_core(
           for (int j = 0; j < 8 && gcol + j < N; ++j)
             C[crow * N + gcol + j] = (unsigned short)v[j];
         }
+      }
+    }
+    // optional fused BN-statistics: per-channel (tile-col) sum/sumsq of
+    // the post-activation tile -> partials [gridDim.x][2N] (second pass =
+    // bn_stats_sum2). Rows beyond M are masked out.
+    if (bn_part != nullptr) {
+      float* red = (float*)(lds + 2 * TN_TILE_B);  // past the 32KB ctile
+      const int col = t & 127;
+      const int rl = t >> 7;  // 2 row-lanes x 64 rows
+      float s = 0.f, ss = 0.f;
+      for (int row = rl; row < 128; row += 2) {
+        if (m0 + row < M) {
+          float v = bf2f(ctile[row * 128 + col]);
+          s += v;
+          ss += v * v;
+        }
+      }
+      red[t] = s;
+      red[256 + t] = ss;
+      __syncthreads();
+      if (rl == 0 && n0 + col < N) {
+        s += red[128 + col];
+        ss += red[256 + 128 + col];
+        bn_part[(long)blockIdx.x * 2 * N + n0 + col] = s;
+        bn_part[(long)blockIdx.x * 2 * N + N + n0 + col] = ss;
       }
     }
     return;
@@ -492,28 +518,32 @@ __global__ __launch_bounds__(NT_THREADS, 2) void gemm_nt_core(
 // ---------------------------------------------------------------------------
 extern "C" {
 
-void launch_gemm_tn(const void* A, const void* B, void* C_bf16, float* C_f32,
-                    const float* bias, int M, int N, int K, long lda, long ldb,
-                    int act, float slope, hipStream_t s) {
+int launch_gemm_tn(const void* A, const void* B, void* C_bf16, float* C_f32,
+                   const float* bias, int M, int N, int K, long lda, long ldb,
+                   int act, float slope, float* bn_part, hipStream_t s) {
   dim3 grid(ceil_div(M, TN_BM), ceil_div(N, TN_BN));
   ConvGather dummy{};
   hipLaunchKernelGGL((gemm_tn_core<false>), grid, dim3(256), 0, s,
                      (const unsigned short*)A, (const unsigned short*)B,
                      (unsigned short*)C_bf16, C_f32, bias, M, N, K, lda, ldb,
-                     act, slope, dummy, nullptr);
+                     act, slope, dummy, nullptr, bn_part);
+  return (int)grid.x;
 }
 
 // implicit-GEMM conv forward / gathered-A TN: A is an NHWC image; M = the
 // number of patch positions; K = kpad (zero page covers k >= rsc).
-void launch_gemm_tn_gather(const void* img, const void* B, void* C_bf16,
-                           const float* bias, int M, int N, int K, long ldb,
-                           int act, float slope, ConvGather ga,
-                           const void* zero_page, hipStream_t s) {
+int launch_gemm_tn_gather(const void* img, const void* B, void* C_bf16,
+                          const float* bias, int M, int N, int K, long ldb,
+                          int act, float slope, ConvGather ga,
+                          const void* zero_page, float* bn_part,
+                          hipStream_t s) {
   dim3 grid(ceil_div(M, TN_BM), ceil_div(N, TN_BN));
   hipLaunchKernelGGL((gemm_tn_core<true>), grid, dim3(256), 0, s,
                      (const unsigned short*)img, (const unsigned short*)B,
                      (unsigned short*)C_bf16, nullptr, bias, M, N, K, 0, ldb,
-                     act, slope, ga, (const unsigned short*)zero_page);
+                     act, slope, ga, (const unsigned short*)zero_page,
+                     bn_part);
+  return (int)grid.x;
 }
 
 void launch_gemm_nt(const void* A, const void* B, float* C, int M, int N,

@@ -129,3 +129,40 @@ def test_weight_sync_propagates_on_gpu():
     sync_params(dis, gan, DIS_TO_GAN_SYNC)
     after = gan.output(z).float()
     assert not torch.allclose(before, after, atol=1e-4)
+
+
+def test_bn_stats_fusion_equivalence():
+    """Producer-fused BN statistics must equal the standalone stats pass."""
+    import torch
+    from gan_deeplearning4j_amd.config import preset
+    from gan_deeplearning4j_amd.models import build_dcgan
+
+    cfg = preset("dcgan64")
+    gen, dis = build_dcgan(cfg)
+    # the fusion pass marks conv->BN producers
+    fused = [n for n in dis.layer_names() if dis.layers[n].emit_bn_stats]
+    assert len(fused) >= 3, fused
+    dis.to_device(torch.device("cuda:0"), torch.bfloat16)
+    gen.to_device(torch.device("cuda:0"), torch.bfloat16)
+    x = (torch.rand(16, 3, 64, 64, device="cuda:0",
+                    dtype=torch.bfloat16) * 2 - 1)
+    dis.train()
+    y_fused = dis(x).float().clone()
+    # disable fusion and compare the full forward
+    for n in fused:
+        dis.layers[n].emit_bn_stats = False
+    y_plain = dis(x).float()
+    err = (y_fused - y_plain).abs().max() / y_plain.abs().max().clamp_min(1e-5)
+    assert float(err) < 0.03, float(err)
+    # generator side too (convT producers)
+    gfused = [n for n in gen.layer_names() if gen.layers[n].emit_bn_stats]
+    assert len(gfused) >= 2, gfused
+    z = torch.randn(16, cfg.model.z_size, device="cuda:0",
+                    dtype=torch.bfloat16)
+    gen.train()
+    o1 = gen(z).float().clone()
+    for n in gfused:
+        gen.layers[n].emit_bn_stats = False
+    o2 = gen(z).float()
+    err = (o1 - o2).abs().max() / o2.abs().max().clamp_min(1e-5)
+    assert float(err) < 0.03, float(err)
