@@ -142,6 +142,10 @@ class ComputationGraph(nn.Module):
         vertex's sole consumer is a BatchNorm over the same feature axis
         (no preprocessor in between), the producer's GPU epilogue also
         emits the BN batch statistics, skipping BN's own stats pass."""
+        import os
+
+        if os.environ.get("GDLJ_BN_FUSE", "0") != "1":
+            return  # measured ~neutral-to-negative on DCGAN-64; opt-in
         from .layers import (BatchNormLayer, Conv2dLayer,
                              ConvTranspose2dLayer, DenseLayer, OutputLayer)
 
