@@ -131,12 +131,15 @@ def test_weight_sync_propagates_on_gpu():
     assert not torch.allclose(before, after, atol=1e-4)
 
 
-def test_bn_stats_fusion_equivalence():
-    """Producer-fused BN statistics must equal the standalone stats pass."""
+def test_bn_stats_fusion_equivalence(monkeypatch):
+    """Producer-fused BN statistics must equal the standalone stats pass.
+    (Fusion is opt-in via GDLJ_BN_FUSE: measured -8% on DCGAN-64, kept
+    for shapes where the stats pass dominates.)"""
     import torch
     from gan_deeplearning4j_amd.config import preset
     from gan_deeplearning4j_amd.models import build_dcgan
 
+    monkeypatch.setenv("GDLJ_BN_FUSE", "1")
     cfg = preset("dcgan64")
     gen, dis = build_dcgan(cfg)
     # the fusion pass marks conv->BN producers
