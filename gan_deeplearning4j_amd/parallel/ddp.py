@@ -22,9 +22,10 @@ import torch.distributed as dist
 
 
 class _Bucket:
-    def __init__(self, params: list[torch.Tensor]):
+    def __init__(self, params: list[torch.Tensor], comm_dtype: torch.dtype):
         self.params = params
         self.numel = sum(p.numel() for p in params)
+        self.comm_dtype = comm_dtype
         self.flat: Optional[torch.Tensor] = None
         self.ready = 0
         self.work = None
@@ -33,7 +34,7 @@ class _Bucket:
         if self.flat is None:
             p0 = self.params[0]
             self.flat = torch.zeros(
-                self.numel, dtype=torch.float32, device=p0.device
+                self.numel, dtype=self.comm_dtype, device=p0.device
             )
         return self.flat
 
@@ -68,12 +69,22 @@ class GradReducer:
         self.buckets: list[_Bucket] = []
         self._param_bucket: dict[int, _Bucket] = {}
         if self.enabled:
+            # communicate in the gradients' own dtype over RCCL (bf16
+            # halves xGMI bytes; all-reduce of tens-of-MB GAN gradients is
+            # latency/bandwidth-bound); gloo lacks bf16 -> fp32 there
+            self.comm_dtype = (
+                torch.bfloat16
+                if dist.get_backend() == "nccl"
+                and all(p.dtype == torch.bfloat16 for p in self.params)
+                else torch.float32
+            )
             self._build_buckets(bucket_cap_mb)
             for p in self.params:
                 p.register_post_accumulate_grad_hook(self._hook)
 
     def _build_buckets(self, cap_mb: int):
-        cap = cap_mb * (1 << 20) // 4  # fp32 elements
+        elt = 2 if self.comm_dtype == torch.bfloat16 else 4
+        cap = cap_mb * (1 << 20) // elt
         cur: list[torch.Tensor] = []
         size = 0
         # reverse order ~ backward completion order (last layers first)
@@ -81,10 +92,10 @@ class GradReducer:
             cur.append(p)
             size += p.numel()
             if size >= cap:
-                self.buckets.append(_Bucket(cur))
+                self.buckets.append(_Bucket(cur, self.comm_dtype))
                 cur, size = [], 0
         if cur:
-            self.buckets.append(_Bucket(cur))
+            self.buckets.append(_Bucket(cur, self.comm_dtype))
         for b in self.buckets:
             for p in b.params:
                 self._param_bucket[id(p)] = b
@@ -103,7 +114,9 @@ class GradReducer:
         off = 0
         for p in b.params:
             n = p.numel()
-            flat[off : off + n].copy_(p.grad.detach().reshape(-1).float())
+            flat[off : off + n].copy_(
+                p.grad.detach().reshape(-1).to(b.comm_dtype)
+            )
             off += n
         b.work = dist.all_reduce(flat, op=dist.ReduceOp.SUM, async_op=True)
 

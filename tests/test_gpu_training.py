@@ -169,3 +169,29 @@ def test_bn_stats_fusion_equivalence(monkeypatch):
     o2 = gen(z).float()
     err = (o1 - o2).abs().max() / o2.abs().max().clamp_min(1e-5)
     assert float(err) < 0.03, float(err)
+
+
+def test_reference_protocol_cpu_gpu_consistency(tmp_path):
+    """Same seed, one reference iteration: GPU bf16 losses must track the
+    CPU fp32 reference within bf16 tolerance."""
+    from gan_deeplearning4j_amd.config import GanConfig
+    from gan_deeplearning4j_amd.data.csv_reader import DataSet
+    from gan_deeplearning4j_amd.train import ReferenceProtocolTrainer
+
+    def one_iter(device):
+        cfg = GanConfig()
+        cfg.data.batch_size_per_worker = 64
+        torch.manual_seed(0)
+        tr = ReferenceProtocolTrainer(cfg, device=torch.device(device),
+                                      out_dir=str(tmp_path / device.replace(
+                                          ":", "_")))
+        g = torch.Generator().manual_seed(42)
+        feats = torch.rand(64, 784, generator=g)
+        labels = torch.eye(10)[torch.randint(0, 10, (64,), generator=g)]
+        return tr.train_iteration(DataSet(feats, labels))
+
+    cpu = one_iter("cpu")
+    gpu = one_iter("cuda:0")
+    for k in ("loss_d", "loss_g", "loss_cv"):
+        rel = abs(gpu[k] - cpu[k]) / (abs(cpu[k]) + 1.0)
+        assert rel < 0.15, (k, cpu[k], gpu[k])
