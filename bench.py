@@ -112,7 +112,7 @@ def main():
         print(json.dumps({
             "metric": metric,
             "value": round(images_per_sec, 2),
-            "unit": "images/s",
+            "unit": "images/s" if args.arch != "mlp" else "samples/s",
             "n_gpus": n_gpus,
             "steps": args.steps,
             "warmup": args.warmup,

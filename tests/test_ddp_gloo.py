@@ -97,6 +97,57 @@ def _body_param_averaging(rank, world):
     return m.weight.detach().clone()
 
 
+def _body_grad_reducer_exact(rank, world):
+    """Reduced grads must equal the mean of the per-rank local grads."""
+    from gan_deeplearning4j_amd.parallel.ddp import GradReducer
+
+    torch.manual_seed(0)
+    m = torch.nn.Linear(8, 4)
+    torch.manual_seed(100 + rank)
+    x = torch.randn(4, 8)
+    # local gradient, no reducer
+    m(x).sum().backward()
+    local = [p.grad.clone() for p in m.parameters()]
+    for p in m.parameters():
+        p.grad = None
+    red = GradReducer([m], bucket_cap_mb=1)
+    red.prepare()
+    m(x).sum().backward()
+    red.finish()
+    return [local, [p.grad.clone() for p in m.parameters()]]
+
+
+def _body_local_steps(rank, world):
+    """local_steps=2 (the averaging_frequency analog): step 1 keeps grads
+    local, step 2 reduces."""
+    from gan_deeplearning4j_amd.parallel.ddp import GradReducer
+
+    torch.manual_seed(0)
+    m = torch.nn.Linear(4, 1, bias=False)
+    red = GradReducer([m], bucket_cap_mb=1, local_steps=2)
+    torch.manual_seed(100 + rank)
+    x = torch.randn(2, 4)
+    red.prepare()
+    m(x).sum().backward()
+    red.finish()
+    g1 = m.weight.grad.clone()
+    m.weight.grad = None
+    red.prepare()
+    m(x).sum().backward()
+    red.finish()
+    return [g1, m.weight.grad.clone()]
+
+
+def _body_broadcast(rank, world):
+    from gan_deeplearning4j_amd.parallel.ddp import broadcast_parameters
+
+    m = torch.nn.Linear(4, 1, bias=False)
+    with torch.no_grad():
+        m.weight.fill_(float(rank + 1))
+    broadcast_parameters(m, src=0)
+    return m.weight.detach().clone()
+
+
 # ------------------------------------------------------------------- tests
 def test_grad_reducer_averages_across_ranks():
     out = _run_mp("_body_grad_reducer")
@@ -114,3 +165,31 @@ def test_parameter_averaging():
     # ranks filled 1.0 and 2.0 -> average 1.5 on both
     assert torch.allclose(torch.tensor(out[0]), torch.full((1, 4), 1.5))
     assert torch.allclose(torch.tensor(out[1]), torch.full((1, 4), 1.5))
+
+
+def test_grad_reducer_exact_mean():
+    out = _run_mp("_body_grad_reducer_exact")
+    local0, _ = out[0]
+    local1, red0 = out[1][0], out[0][1]
+    red1 = out[1][1]
+    for l0, l1, r0, r1 in zip(local0, local1, red0, red1):
+        want = (torch.tensor(l0) + torch.tensor(l1)) / 2
+        assert torch.allclose(torch.tensor(r0), want, atol=1e-6)
+        assert torch.allclose(torch.tensor(r1), want, atol=1e-6)
+
+
+def test_local_steps_gates_reduction():
+    out = _run_mp("_body_local_steps")
+    g1_r0, g2_r0 = (torch.tensor(t) for t in out[0])
+    g1_r1, g2_r1 = (torch.tensor(t) for t in out[1])
+    # step 1 (local): per-rank data differs -> grads differ
+    assert not torch.allclose(g1_r0, g1_r1, atol=1e-6)
+    # step 2 (reduced): grads identical and equal to the mean of the locals
+    assert torch.allclose(g2_r0, g2_r1, atol=1e-6)
+    assert torch.allclose(g2_r0, (g1_r0 + g1_r1) / 2, atol=1e-6)
+
+
+def test_broadcast_parameters():
+    out = _run_mp("_body_broadcast")
+    assert torch.allclose(torch.tensor(out[0]), torch.full((1, 4), 1.0))
+    assert torch.allclose(torch.tensor(out[1]), torch.full((1, 4), 1.0))
