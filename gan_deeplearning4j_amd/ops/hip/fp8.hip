@@ -109,38 +109,55 @@ DEV_INLINE void k_decode_f8(const ConvGather& g, unsigned k, int& r, int& s,
 
 // NOTE: 16-fp8 chunks must not straddle (r,s) boundaries -> requires
 // C % 16 == 0 (enforced by the binding).
-DEV_INLINE void f8_stage_gather(const unsigned char* __restrict__ img,
-                                const ConvGather& g,
-                                const unsigned char* __restrict__ zp,
-                                int row0, int nrows, int k0, char* lds) {
-  const int t = threadIdx.x;
-  const int wid = t >> 6;
-  #pragma unroll
-  for (int i = 0; i < 2; ++i) {
-    int chunk = i * 256 + t;
-    int row = chunk >> 2;
-    int slot = chunk & 3;
-    int gslot = slot ^ (row & 3);
-    int np = min(row0 + row, nrows - 1);
-    int k = k0 + gslot * 16;       // 16 fp8 channels per chunk
-    const unsigned char* src = zp;
-    if (k < g.rsc) {
-      int r, s, c, n, ho, wo;
-      k_decode_f8(g, (unsigned)k, r, s, c);
+// Row decode hoisted out of the K-loop (same scheme as TnGatherStager in
+// gemm.hip): each thread stages the same 2 rows every K-step.
+struct F8GatherStager {
+  long base[2];      // (long)n * H*W*C element offset per chunk
+  int h0[2], w0[2];  // ho*stride-pad / wo*stride-pad per chunk
+
+  DEV_INLINE void init(const ConvGather& g, int row0, int nrows) {
+    const int t = threadIdx.x;
+    #pragma unroll
+    for (int i = 0; i < 2; ++i) {
+      int row = (i * 256 + t) >> 2;
+      int np = min(row0 + row, nrows - 1);
       unsigned q1 = fdiv((unsigned)np, g.fWo);
-      wo = (int)((unsigned)np - q1 * g.Wo);
+      int wo = (int)((unsigned)np - q1 * g.Wo);
       unsigned q2 = fdiv(q1, g.fHo);
-      ho = (int)(q1 - q2 * g.Ho);
-      n = (int)q2;
-      int hi = ho * g.stride - g.pad + r;
-      int wi = wo * g.stride - g.pad + s;
-      if (hi >= 0 && hi < g.H && wi >= 0 && wi < g.W)
-        src = img + (((long)n * g.H + hi) * g.W + wi) * g.C + c;
+      int ho = (int)(q1 - q2 * g.Ho);
+      base[i] = (long)(int)q2 * g.H * g.W * g.C;
+      h0[i] = ho * g.stride - g.pad;
+      w0[i] = wo * g.stride - g.pad;
     }
-    char* dst = lds + (i * 256 + wid * 64) * 16;
-    GLDS16(src, dst);
   }
-}
+
+  DEV_INLINE void stage(const unsigned char* __restrict__ img,
+                        const ConvGather& g,
+                        const unsigned char* __restrict__ zp, int k0,
+                        char* lds) const {
+    const int t = threadIdx.x;
+    const int wid = t >> 6;
+    #pragma unroll
+    for (int i = 0; i < 2; ++i) {
+      int chunk = i * 256 + t;
+      int row = chunk >> 2;
+      int slot = chunk & 3;
+      int gslot = slot ^ (row & 3);
+      int k = k0 + gslot * 16;  // 16 fp8 channels per chunk
+      const unsigned char* src = zp;
+      if (k < g.rsc) {
+        int r, s, c;
+        k_decode_f8(g, (unsigned)k, r, s, c);
+        int hi = h0[i] + r;
+        int wi = w0[i] + s;
+        if (hi >= 0 && hi < g.H && wi >= 0 && wi < g.W)
+          src = img + base[i] + (long)(hi * g.W + wi) * g.C + c;
+      }
+      char* dst = lds + (i * 256 + wid * 64) * 16;
+      GLDS16(src, dst);
+    }
+  }
+};
 
 DEV_INLINE fp8x8 f8_frag(const char* lds, int row, int kslot8) {
   // fragment = 8 fp8 at k-octet kslot8; 16B swizzle on slot16 = kslot8>>1
@@ -183,10 +200,13 @@ __global__ __launch_bounds__(256, 2) void gemm_tn_fp8_core(
     for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
 
   const int ntiles = K / F8_BK;
-  if (GATHER_A)
-    f8_stage_gather(A, ga, zp, m0, M, 0, abuf(0));
-  else
+  F8GatherStager gs;
+  if (GATHER_A) {
+    gs.init(ga, m0, M);
+    gs.stage(A, ga, zp, 0, abuf(0));
+  } else {
     f8_stage(A, m0, M, lda, 0, abuf(0));
+  }
   f8_stage(B, n0, N, ldb, 0, bbuf(0));
   asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
   __syncthreads();
@@ -195,7 +215,7 @@ __global__ __launch_bounds__(256, 2) void gemm_tn_fp8_core(
     int cur = t & 1;
     if (t + 1 < ntiles) {
       if (GATHER_A)
-        f8_stage_gather(A, ga, zp, m0, M, (t + 1) * F8_BK, abuf(cur ^ 1));
+        gs.stage(A, ga, zp, (t + 1) * F8_BK, abuf(cur ^ 1));
       else
         f8_stage(A, m0, M, lda, (t + 1) * F8_BK, abuf(cur ^ 1));
       f8_stage(B, n0, N, ldb, (t + 1) * F8_BK, bbuf(cur ^ 1));

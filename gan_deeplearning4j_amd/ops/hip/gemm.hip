@@ -110,35 +110,87 @@ DEV_INLINE void tn_stage(const unsigned short* __restrict__ g, int row0,
   }
 }
 
-// implicit-GEMM stage: A rows are im2col rows gathered from the NHWC image
-DEV_INLINE void tn_stage_gather(const unsigned short* __restrict__ img,
-                                const ConvGather& g,
-                                const unsigned short* __restrict__ zp,
-                                int row0, int nrows, int k0, char* lds) {
-  const int t = threadIdx.x;
-  const int wid = t >> 6;
-  #pragma unroll
-  for (int i = 0; i < 4; ++i) {
-    int chunk = i * 256 + t;
-    int row = chunk >> 3;
-    int slot = chunk & 7;
-    int gslot = slot ^ (row & 7);
-    int np = min(row0 + row, nrows - 1);
-    int k = k0 + gslot * 8;
-    const unsigned short* src = zp;
-    if (k < g.rsc) {
-      int r, s, c, n, ho, wo;
-      k_decode(g, (unsigned)k, r, s, c);
+// implicit-GEMM stage: A rows are im2col rows gathered from the NHWC image.
+// The (np -> n,ho,wo) row decode and the n-plane base offset are invariant
+// across the whole K-loop (each thread stages the same 4 rows every step),
+// so they are hoisted into registers once (init) and only the cheap k part
+// (tap + channel) is decoded per chunk per step (stage).
+struct TnGatherStager {
+  long base[4];        // (long)n * H*W*C element offset per chunk
+  int h0[4], w0[4];    // mode-adjusted spatial bases per chunk
+
+  DEV_INLINE void init(const ConvGather& g, int row0, int nrows) {
+    const int t = threadIdx.x;
+    #pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      int row = (i * 256 + t) >> 3;
+      int np = min(row0 + row, nrows - 1);
+      int n, ho, wo;
       np_decode(g, (unsigned)np, n, ho, wo);
-      bool valid;
-      const unsigned short* p = gather_addr(g, img, n, ho, wo, r, s, c,
-                                            valid);
-      if (valid) src = p;
+      base[i] = (long)n * g.H * g.W * g.C;
+      if (g.mode == 0) {
+        h0[i] = ho * g.stride - g.pad;
+        w0[i] = wo * g.stride - g.pad;
+      } else if (g.mode == 2) {
+        h0[i] = ho + g.off_h;
+        w0[i] = wo + g.off_w;
+      } else {
+        h0[i] = ho + g.pad;
+        w0[i] = wo + g.pad;
+      }
     }
-    char* dst = lds + (i * 256 + wid * 64) * 16;
-    GLDS16(src, dst);
   }
-}
+
+  DEV_INLINE void stage(const unsigned short* __restrict__ img,
+                        const ConvGather& g,
+                        const unsigned short* __restrict__ zp, int k0,
+                        char* lds) const {
+    const int t = threadIdx.x;
+    const int wid = t >> 6;
+    #pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      int chunk = i * 256 + t;
+      int row = chunk >> 3;
+      int slot = chunk & 7;
+      int gslot = slot ^ (row & 7);
+      int k = k0 + gslot * 8;
+      const unsigned short* src = zp;
+      if (k < g.rsc) {
+        int r, s, c;
+        k_decode(g, (unsigned)k, r, s, c);
+        int hi, wi;
+        bool valid;
+        if (g.mode == 0) {
+          hi = h0[i] + r;
+          wi = w0[i] + s;
+          valid = hi >= 0 && hi < g.H && wi >= 0 && wi < g.W;
+        } else if (g.mode == 2) {
+          hi = h0[i] - r;
+          wi = w0[i] - s;
+          valid = hi >= 0 && hi < g.H && wi >= 0 && wi < g.W;
+        } else {
+          int hop = h0[i] - r;
+          int wop = w0[i] - s;
+          if (hop < 0 || wop < 0) {
+            valid = false;
+            hi = wi = 0;
+          } else {
+            unsigned qh = fdiv((unsigned)hop, g.fStride);
+            unsigned qw = fdiv((unsigned)wop, g.fStride);
+            valid = (hop == (int)(qh * g.stride)) &&
+                    (wop == (int)(qw * g.stride)) && (int)qh < g.H &&
+                    (int)qw < g.W;
+            hi = (int)qh;
+            wi = (int)qw;
+          }
+        }
+        if (valid) src = img + base[i] + (long)(hi * g.W + wi) * g.C + c;
+      }
+      char* dst = lds + (i * 256 + wid * 64) * 16;
+      GLDS16(src, dst);
+    }
+  }
+};
 
 DEV_INLINE bf16x8 tn_frag(const char* lds, int row, int kslot) {
   int byte = row * 128 + ((kslot ^ (row & 7)) * 16);
@@ -182,10 +234,13 @@ __global__ __launch_bounds__(256, 2) void gemm_tn_core(
     for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
 
   const int ntiles = K / TN_BK;
-  if (GATHER_A)
-    tn_stage_gather(A, ga, zp, m0, M, 0, abuf(0));
-  else
+  TnGatherStager gs;
+  if (GATHER_A) {
+    gs.init(ga, m0, M);
+    gs.stage(A, ga, zp, 0, abuf(0));
+  } else {
     tn_stage(A, m0, M, lda, 0, abuf(0));
+  }
   tn_stage(B, n0, N, ldb, 0, bbuf(0));
   asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
   __syncthreads();
@@ -194,7 +249,7 @@ __global__ __launch_bounds__(256, 2) void gemm_tn_core(
     int cur = t & 1;
     if (t + 1 < ntiles) {
       if (GATHER_A)
-        tn_stage_gather(A, ga, zp, m0, M, (t + 1) * TN_BK, abuf(cur ^ 1));
+        gs.stage(A, ga, zp, (t + 1) * TN_BK, abuf(cur ^ 1));
       else
         tn_stage(A, m0, M, lda, (t + 1) * TN_BK, abuf(cur ^ 1));
       tn_stage(B, n0, N, ldb, (t + 1) * TN_BK, bbuf(cur ^ 1));
