@@ -28,6 +28,9 @@ for _ in range(10):
         n=rng.choice([2, 3, 5]),
         cin=rng.choice([3, 8, 16, 24, 32]),
         cout=rng.choice([8, 16, 32, 48]),
+        # convT output channels incl. non-multiples of 8 (generator heads
+        # are Cout=3: exercises the padded vectorized col2im path)
+        cout_t=rng.choice([3, 5, 8, 12, 16, 48]),
         h=rng.choice([7, 9, 12, 16, 17]),
         r=rng.choice([3, 4, 5]),
         stride=rng.choice([1, 2]),
@@ -72,7 +75,7 @@ def test_conv_transpose2d_fuzz(case):
     from gan_deeplearning4j_amd.ops import gpu_ops
 
     n, cin, cout, h, r, stride, pad = (case[k] for k in
-                                       ("n", "cin", "cout", "h", "r",
+                                       ("n", "cin", "cout_t", "h", "r",
                                         "stride", "pad"))
     if cin % 8 != 0:
         cin = 8  # convT gather requires Cin%8 (model contract)
