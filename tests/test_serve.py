@@ -1,0 +1,77 @@
+"""Serving API tests (CPU eager path; on GPU the same app replays a
+captured hipGraph — covered by tests/test_gpu_training.py's
+compile_inference test)."""
+
+import base64
+
+import pytest
+import torch
+
+fastapi = pytest.importorskip("fastapi")
+from fastapi.testclient import TestClient  # noqa: E402
+
+from gan_deeplearning4j_amd.config import preset
+from gan_deeplearning4j_amd.models import build_dcgan
+from gan_deeplearning4j_amd.serve import create_app, load_graph
+
+
+@pytest.fixture(scope="module")
+def client():
+    cfg = preset("dcgan28")
+    gen, dis = build_dcgan(cfg)
+    app = create_app(generator=gen, discriminator=dis,
+                     device=torch.device("cpu"), max_batch=8)
+    return TestClient(app)
+
+
+def test_healthz_and_info(client):
+    r = client.get("/healthz")
+    assert r.status_code == 200
+    assert r.json()["generator"] and r.json()["discriminator"]
+    info = client.get("/info").json()
+    assert info["generator"]["n_params"] > 0
+    assert info["discriminator"]["inputs"] == ["d_input"]
+
+
+def test_generate_array(client):
+    r = client.post("/generate", json={"n": 3, "seed": 1})
+    assert r.status_code == 200
+    s = torch.tensor(r.json()["samples"])
+    assert s.shape[0] == 3 and s.ndim == 4
+    # seeded requests are reproducible
+    r2 = client.post("/generate", json={"n": 3, "seed": 1})
+    assert torch.allclose(s, torch.tensor(r2.json()["samples"]), atol=1e-5)
+    r3 = client.post("/generate", json={"n": 3, "seed": 2})
+    assert not torch.allclose(s, torch.tensor(r3.json()["samples"]))
+
+
+def test_generate_png(client):
+    r = client.post("/generate", json={"n": 4, "format": "png_base64"})
+    assert r.status_code == 200
+    png = base64.b64decode(r.json()["png_base64"])
+    assert png[:8] == b"\x89PNG\r\n\x1a\n"
+
+
+def test_generate_over_max_batch(client):
+    r = client.post("/generate", json={"n": 99})
+    assert r.status_code == 400
+
+
+def test_discriminate(client):
+    x = torch.rand(2, 784).tolist()
+    r = client.post("/discriminate", json={"inputs": x})
+    assert r.status_code == 200
+    scores = r.json()["scores"]
+    assert len(scores) == 2
+    assert all(0.0 <= v <= 1.0 for v in scores)
+
+
+def test_load_graph_roundtrip(tmp_path):
+    from gan_deeplearning4j_amd.graph.serialization import ModelSerializer
+
+    cfg = preset("dcgan28")
+    gen, _ = build_dcgan(cfg)
+    p = ModelSerializer.write_model(gen, tmp_path / "gen.zip")
+    g2 = load_graph(p)
+    z = torch.rand(2, cfg.model.z_size)
+    assert torch.allclose(gen.output(z), g2.output(z), atol=1e-6)
