@@ -152,7 +152,10 @@ class GradReducer:
 @torch.no_grad()
 def average_parameters(module: torch.nn.Module):
     """Synchronous parameter averaging (exact DL4J TrainingMaster semantics:
-    broadcast/average the PARAMETER vector, not gradients)."""
+    broadcast/average the PARAMETER vector, not gradients). BatchNorm
+    running statistics are averaged too — they are part of DL4J's flat
+    params() vector (the reference syncs mean/var by name, Java:445-455),
+    and leaving them per-rank would diverge the replicas' eval paths."""
     if not (dist.is_initialized() and dist.get_world_size() > 1):
         return
     world = dist.get_world_size()
@@ -162,6 +165,12 @@ def average_parameters(module: torch.nn.Module):
         p.data.copy_((t / world).to(p.dtype))
         if hasattr(p, "_gdlj_cache"):
             del p._gdlj_cache
+    for b in module.buffers():
+        if not b.dtype.is_floating_point:
+            continue
+        t = b.data.float()
+        dist.all_reduce(t, op=dist.ReduceOp.SUM)
+        b.data.copy_((t / world).to(b.dtype))
 
 
 @torch.no_grad()

@@ -90,11 +90,14 @@ def _body_trainer_sync(rank, world):
 def _body_param_averaging(rank, world):
     from gan_deeplearning4j_amd.parallel.ddp import average_parameters
 
-    m = torch.nn.Linear(4, 1, bias=False)
+    m = torch.nn.Sequential(torch.nn.Linear(4, 4, bias=False),
+                            torch.nn.BatchNorm1d(4))
     with torch.no_grad():
-        m.weight.fill_(float(rank + 1))
+        m[0].weight.fill_(float(rank + 1))
+        m[1].running_mean.fill_(float(rank))  # 0.0 / 1.0 -> 0.5
     average_parameters(m)
-    return m.weight.detach().clone()
+    return [m[0].weight.detach().clone(),
+            m[1].running_mean.detach().clone()]
 
 
 def _body_grad_reducer_exact(rank, world):
@@ -162,9 +165,12 @@ def test_trainer_replicas_stay_synced():
 
 def test_parameter_averaging():
     out = _run_mp("_body_param_averaging")
-    # ranks filled 1.0 and 2.0 -> average 1.5 on both
-    assert torch.allclose(torch.tensor(out[0]), torch.full((1, 4), 1.5))
-    assert torch.allclose(torch.tensor(out[1]), torch.full((1, 4), 1.5))
+    # ranks filled 1.0 and 2.0 -> average 1.5 on both; BN running stats
+    # (buffers) average too (0.0/1.0 -> 0.5)
+    for r in (0, 1):
+        w, mean = (torch.tensor(t) for t in out[r])
+        assert torch.allclose(w, torch.full((4, 4), 1.5))
+        assert torch.allclose(mean, torch.full((4,), 0.5))
 
 
 def test_grad_reducer_exact_mean():
