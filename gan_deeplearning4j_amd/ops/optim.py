@@ -180,6 +180,10 @@ class Updater:
 
     def load_state_dict(self, sd: dict):
         self.t = sd["t"]
+        # drop any device-side step counter: it reseeds from the restored
+        # host count on the next step (a stale one would skew Adam bias
+        # correction after an in-process resume)
+        self._t_dev = None
         for s, ss in zip(self.slots, sd["slots"]):
             s.lr = ss["lr"]
             dev = s.param.device
