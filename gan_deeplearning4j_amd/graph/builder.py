@@ -138,14 +138,21 @@ class ComputationGraph(nn.Module):
         self._bn_fusion_pass()
 
     def _bn_fusion_pass(self) -> None:
-        """Producer->BatchNorm statistics fusion: when a Conv/ConvT/Dense
-        vertex's sole consumer is a BatchNorm over the same feature axis
-        (no preprocessor in between), the producer's GPU epilogue also
-        emits the BN batch statistics, skipping BN's own stats pass."""
+        """Producer<->BatchNorm fusions for Conv/ConvT/Dense vertices whose
+        SOLE consumer is a BatchNorm over the same feature axis (no
+        preprocessor in between):
+
+        - backward act-fusion (always on; kill switch GDLJ_NO_ACT_FUSE):
+          BN's backward kernel applies the producer's activation backward
+          and reduces the producer's bias gradient in the same pass,
+          removing the standalone act_bwd_bias tensor streams.
+        - statistics fusion (opt-in via GDLJ_BN_FUSE, measured ~neutral-
+          to-negative on DCGAN-64): the producer's GPU epilogue also emits
+          the BN batch statistics, skipping BN's own stats pass."""
         import os
 
-        if os.environ.get("GDLJ_BN_FUSE", "0") != "1":
-            return  # measured ~neutral-to-negative on DCGAN-64; opt-in
+        stats_fuse = os.environ.get("GDLJ_BN_FUSE", "0") == "1"
+        from ..ops.gpu_ops import ACT_CODES
         from .layers import (BatchNormLayer, Conv2dLayer,
                              ConvTranspose2dLayer, DenseLayer, OutputLayer)
 
@@ -169,7 +176,13 @@ class ComputationGraph(nn.Module):
                 continue
             if isinstance(prod, (Conv2dLayer, ConvTranspose2dLayer,
                                  DenseLayer)):
-                prod.emit_bn_stats = True
+                if stats_fuse:
+                    prod.emit_bn_stats = True
+                if prod.activation in ACT_CODES and \
+                        ACT_CODES[prod.activation] != 0:
+                    layer.bwd_act = (ACT_CODES[prod.activation],
+                                     getattr(prod, "slope", 0.2),
+                                     getattr(prod, "bias", None) is not None)
 
     # ------------------------------------------------------------ build
     def _toposort(self, vertices) -> list[str]:

@@ -232,7 +232,7 @@ class BatchNormLayer(BaseLayer):
     def forward(self, x):
         return OF.batch_norm(x, self.weight, self.bias, self.running_mean,
                              self.running_var, self.training, self.momentum,
-                             self.eps)
+                             self.eps, getattr(self, "bwd_act", None))
 
 
 class MaxPool2dLayer(BaseLayer):

@@ -101,13 +101,20 @@ def batch_norm(
     training: bool,
     momentum: float = 0.1,
     eps: float = 1e-5,
+    bwd_act=None,
 ) -> torch.Tensor:
-    """BatchNorm over dim 1 (2D) or channel dim (4D). Stats kept fp32."""
+    """BatchNorm over dim 1 (2D) or channel dim (4D). Stats kept fp32.
+
+    bwd_act: optional (act_code, slope, want_bias) describing the sole
+    producer's fused output activation — GPU path folds that activation's
+    backward into the BN backward kernel (see gpu_ops deposit_act_fused).
+    """
     if x.is_cuda:
         from . import gpu_ops
 
         return gpu_ops.batch_norm(
-            x, weight, bias, running_mean, running_var, training, momentum, eps
+            x, weight, bias, running_mean, running_var, training, momentum,
+            eps, bwd_act
         )
     return F.batch_norm(
         x, running_mean, running_var, weight, bias, training, momentum, eps

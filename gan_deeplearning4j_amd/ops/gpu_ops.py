@@ -93,6 +93,26 @@ def deposit_bn_stats(t: torch.Tensor, stats) -> None:
 def take_bn_stats(t: torch.Tensor):
     return _bn_stats_chan.pop(t.data_ptr(), None)
 
+
+# Backward act-fusion side channel (consumer BatchNorm -> producer conv/
+# dense): when BN is the sole consumer of a producer's activation output,
+# BN's backward applies the activation backward INSIDE its apply kernel
+# (the activation output is already in registers there) and reduces the
+# producer's bias gradient on the way out, then deposits
+# (act_code, n_cols, db) keyed by the grad tensor it returns. The
+# producer's backward pops the entry by data_ptr (grads flow through
+# view-only nodes, so the storage pointer is preserved) and skips its own
+# act_bwd_bias pass — removing 3 full tensor streams per fused layer.
+_act_fused_chan: dict = {}
+
+
+def deposit_act_fused(t: torch.Tensor, payload) -> None:
+    _act_fused_chan[t.data_ptr()] = payload
+
+
+def take_act_fused(t: torch.Tensor):
+    return _act_fused_chan.pop(t.data_ptr(), None)
+
 # Strided dgrad/convT-fwd algorithm choice: parity-decomposed gathered
 # GEMMs vs dcol+col2im. Measured on DCGAN-64: dcol wins by ~4% (the
 # parity gather re-reads the source once per tap, so traffic is a wash
@@ -173,10 +193,17 @@ class _Linear(torch.autograd.Function):
     def backward(ctx, dy):
         ext = hip_ext()
         xp, wp, y = ctx.saved_tensors
+        fused = take_act_fused(dy)
         dy = _bf(dy)
         dx = dw = db = None
         want_bias = ctx.has_bias and ctx.needs_input_grad[2]
-        if ctx.act and want_bias and dy.shape[1] % 8 == 0:
+        if fused is not None and fused[0] == ctx.act and \
+                fused[1] == dy.shape[1]:
+            # consumer BN already applied act backward + bias reduction
+            dpre = dy
+            if want_bias and fused[2] is not None:
+                db = fused[2].to(ctx.dtypes[2])
+        elif ctx.act and want_bias and dy.shape[1] % 8 == 0:
             dpre, db_f = ext.act_bwd_bias(dy, y, ctx.act, ctx.slope)
             db = db_f.to(ctx.dtypes[2])
         else:
@@ -286,10 +313,16 @@ class _Conv2d(torch.autograd.Function):
         ext = hip_ext()
         xh, wp, y2d = ctx.saved_tensors
         N, C, H, W, Kout, R, S, Ho, Wo, stride, pad, kpad, C8 = ctx.geom
+        fused = take_act_fused(dy)
         dy2d = _bf(dy.permute(0, 2, 3, 1)).reshape(-1, Kout)
         dx = dw = db = None
         want_bias = ctx.has_bias and ctx.needs_input_grad[2]
-        if ctx.act and want_bias and Kout % 8 == 0:
+        if fused is not None and fused[0] == ctx.act and fused[1] == Kout:
+            # consumer BN already applied act backward + bias reduction
+            dpre = dy2d
+            if want_bias and fused[2] is not None:
+                db = fused[2].to(ctx.dtypes[2])
+        elif ctx.act and want_bias and Kout % 8 == 0:
             dpre, db_f = ext.act_bwd_bias(dy2d, y2d, ctx.act, ctx.slope)
             db = db_f.to(ctx.dtypes[2])
         else:
@@ -474,10 +507,16 @@ class _ConvTranspose2d(torch.autograd.Function):
         N, Cin, Hi, Wi, Cout, R, S, Ho, Wo, stride, pad = ctx.geom
         rsco = R * S * Cout
         rscop = _rup64(rsco)
+        fused = take_act_fused(dy)
         dyh = _nhwc(dy)                       # [N,Ho,Wo,Cout]
         db = None
         want_bias = ctx.has_bias and ctx.needs_input_grad[2]
-        if ctx.act and want_bias and Cout % 8 == 0:
+        if fused is not None and fused[0] == ctx.act and fused[1] == Cout:
+            # consumer BN already applied act backward + bias reduction
+            dpre_img = dyh
+            if want_bias and fused[2] is not None:
+                db = fused[2].to(ctx.dtypes[2])
+        elif ctx.act and want_bias and Cout % 8 == 0:
             dpre, db_f = ext.act_bwd_bias(dyh.reshape(-1, Cout),
                                           yh.reshape(-1, Cout), ctx.act,
                                           ctx.slope)
@@ -539,7 +578,8 @@ def conv_transpose2d(x, w, b=None, stride=1, padding=0, act="identity",
 # ================================================================ batch norm
 class _BatchNorm(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, x2, gamma, beta, rm, rv, momentum, eps, pre_stats):
+    def forward(ctx, x2, gamma, beta, rm, rv, momentum, eps, pre_stats,
+                bwd_act):
         ext = hip_ext()
         if pre_stats is not None:
             # producer-fused statistics (conv/dense epilogue): skip the
@@ -552,20 +592,32 @@ class _BatchNorm(torch.autograd.Function):
                                              beta.detach().float(), rm, rv,
                                              momentum, eps)
         ctx.save_for_backward(x2, mean, istd, gamma)
+        ctx.bwd_act = bwd_act
         return y
 
     @staticmethod
     def backward(ctx, dy):
         ext = hip_ext()
         x2, mean, istd, gamma = ctx.saved_tensors
-        dx, dgamma, dbeta = ext.bn_bwd(x2, _bf(dy), mean, istd,
-                                       gamma.detach().float())
+        if ctx.bwd_act is not None:
+            # x2 is the producer's activation output: fold the activation
+            # backward + producer bias-grad into the BN apply kernel and
+            # hand the result to the producer via the side channel
+            act, slope, want_bias = ctx.bwd_act
+            dx, dgamma, dbeta, db = ext.bn_bwd_act(
+                x2, _bf(dy), mean, istd, gamma.detach().float(), act, slope,
+                want_bias)
+            deposit_act_fused(
+                dx, (act, x2.shape[-1], db if want_bias else None))
+        else:
+            dx, dgamma, dbeta = ext.bn_bwd(x2, _bf(dy), mean, istd,
+                                           gamma.detach().float())
         return dx, dgamma.to(gamma.dtype), dbeta.to(gamma.dtype), None, \
-            None, None, None, None
+            None, None, None, None, None
 
 
 def batch_norm(x, weight, bias, running_mean, running_var, training,
-               momentum=0.1, eps=1e-5):
+               momentum=0.1, eps=1e-5, bwd_act=None):
     is4d = x.dim() == 4
     pre_stats = take_bn_stats(x)
     if is4d:
@@ -573,9 +625,12 @@ def batch_norm(x, weight, bias, running_mean, running_var, training,
         x2 = xh.reshape(-1, xh.shape[-1])
     else:
         x2 = _bf(x)
+    if bwd_act is not None and (x2.shape[-1] % 8 != 0
+                                or os.environ.get("GDLJ_NO_ACT_FUSE") == "1"):
+        bwd_act = None
     if training:
         y2 = _BatchNorm.apply(x2, weight, bias, running_mean, running_var,
-                              momentum, eps, pre_stats)
+                              momentum, eps, pre_stats, bwd_act)
     else:
         y2 = hip_ext().bn_fwd_eval(x2, weight.detach().float(),
                                    bias.detach().float(), running_mean,

@@ -231,6 +231,69 @@ __global__ void bn_bwd_apply_v8(const s16x8* __restrict__ x,
   }
 }
 
+// BN backward apply with the PRODUCER activation's backward fused in.
+// x (the BatchNorm input) IS the producing conv's activation output, so
+// dpre = dxbn * act'(x) costs zero extra memory traffic, and the conv's
+// bias gradient (column sums of dpre) accumulates into per-block
+// partials on the way out — this removes the conv's standalone
+// act_bwd_bias pass (3 full tensor streams) from the backward.
+// Col-group structure (fixed channels per thread) like bn_bwd_reduce_v8.
+__global__ void bn_bwd_apply_act_v8(
+    const s16x8* __restrict__ x, const s16x8* __restrict__ dy,
+    s16x8* __restrict__ dx, long m, int c, const float* __restrict__ mean,
+    const float* __restrict__ istd, const float* __restrict__ gamma,
+    const float* __restrict__ dgamma, const float* __restrict__ dbeta,
+    int act, float slope, float* __restrict__ bias_part) {
+  int c8 = c / 8;
+  int g0 = blockIdx.y * 32;
+  int groups = min(32, c8 - g0);
+  int lanes = (int)blockDim.x / groups;
+  int sub = (int)threadIdx.x / groups;
+  int g = g0 + (int)threadIdx.x % groups;
+  float inv_m = 1.f / (float)m;
+  float db[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+  if (sub < lanes) {
+    float mu[8], is[8], gis[8], dgs[8], dbs[8];
+    #pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      mu[j] = mean[g * 8 + j];
+      is[j] = istd[g * 8 + j];
+      gis[j] = gamma[g * 8 + j] * is[j];
+      dgs[j] = dgamma[g * 8 + j] * inv_m;
+      dbs[j] = dbeta[g * 8 + j] * inv_m;
+    }
+    for (long r = (long)blockIdx.x * lanes + sub; r < m;
+         r += (long)gridDim.x * lanes) {
+      s16x8 vx = x[r * c8 + g], vg = dy[r * c8 + g];
+      s16x8 o;
+      #pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float y = bf2f((unsigned short)vx[j]);
+        float xh = (y - mu[j]) * is[j];
+        float gg = bf2f((unsigned short)vg[j]);
+        float dbn = gis[j] * (gg - dbs[j] - xh * dgs[j]);
+        float dpre = dbn * act_bwd_from_y(y, act, slope);
+        o[j] = (short)f2bf(dpre);
+        db[j] += dpre;
+      }
+      dx[r * c8 + g] = o;
+    }
+  }
+  if (bias_part == nullptr) return;
+  __shared__ float red[256];
+  #pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    red[threadIdx.x] = db[j];
+    __syncthreads();
+    if (sub == 0) {
+      float acc = 0.f;
+      for (int q = 0; q < lanes; ++q) acc += red[q * groups + g - g0];
+      bias_part[(long)blockIdx.x * c + g * 8 + j] = acc;
+    }
+    __syncthreads();
+  }
+}
+
 // ------------------------------------------------------------- scalar path
 __global__ void bn_stats(const unsigned short* __restrict__ x, long m, int c,
                          float* __restrict__ sum, float* __restrict__ sumsq) {
@@ -473,6 +536,19 @@ void launch_bn_bwd_reduce(const void* x, const void* dy, long m, int c,
   hipLaunchKernelGGL(bn_bwd_reduce, _colgrid(m, c), dim3(256), 0, s,
                      (const unsigned short*)x, (const unsigned short*)dy, m, c,
                      mean, istd, dgamma, dbeta);
+}
+
+int launch_bn_bwd_apply_act(const void* x, const void* dy, void* dx, long m,
+                            int c, const float* mean, const float* istd,
+                            const float* gamma, const float* dgamma,
+                            const float* dbeta, int act, float slope,
+                            float* bias_part, hipStream_t s) {
+  dim3 g = _colgrid_v8(m, c);
+  if (g.x > 256) g.x = 256;
+  hipLaunchKernelGGL(bn_bwd_apply_act_v8, g, dim3(256), 0, s, (const s16x8*)x,
+                     (const s16x8*)dy, (s16x8*)dx, m, c, mean, istd, gamma,
+                     dgamma, dbeta, act, slope, bias_part);
+  return (int)g.x;
 }
 
 void launch_bn_bwd_apply(const void* x, const void* dy, void* dx, long m,

@@ -109,6 +109,10 @@ void launch_bn_apply_eval(const void*, void*, long, int, const float*,
                           hipStream_t);
 void launch_bn_bwd_reduce(const void*, const void*, long, int, const float*,
                           const float*, float*, float*, hipStream_t);
+int launch_bn_bwd_apply_act(const void*, const void*, void*, long, int,
+                            const float*, const float*, const float*,
+                            const float*, const float*, int, float, float*,
+                            hipStream_t);
 void launch_bn_bwd_apply(const void*, const void*, void*, long, int,
                          const float*, const float*, const float*,
                          const float*, const float*, hipStream_t);
@@ -680,6 +684,50 @@ std::vector<torch::Tensor> bn_bwd(torch::Tensor x, torch::Tensor dy,
   return {dx, dgamma, dbeta};
 }
 
+// BN backward with the producer activation's backward fused into the
+// apply pass: returns {dpre, dgamma, dbeta, db_bias}. x must be the
+// producer's ACTIVATION OUTPUT (= BN input); dpre is the gradient at the
+// producer's pre-activation, db_bias its bias gradient (column sums of
+// dpre). Requires C % 8 == 0.
+std::vector<torch::Tensor> bn_bwd_act(torch::Tensor x, torch::Tensor dy,
+                                      torch::Tensor mean, torch::Tensor istd,
+                                      torch::Tensor gamma, int64_t act,
+                                      double slope, bool want_bias) {
+  check_bf16(x, "x");
+  check_bf16(dy, "dy");
+  int64_t c = x.size(-1);
+  long m = x.numel() / c;
+  TORCH_CHECK(c % 8 == 0, "bn_bwd_act needs C % 8 == 0");
+  auto f32 = x.options().dtype(torch::kFloat32);
+  torch::Tensor dx = torch::empty_like(x);
+  auto s = cur_stream();
+  torch::Tensor dgamma = torch::empty({c}, f32);
+  torch::Tensor dbeta = torch::empty({c}, f32);
+  torch::Tensor scratch = torch::empty({256, 2 * c}, f32);
+  int gx = launch_bn_bwd_reduce_part(x.data_ptr(), dy.data_ptr(), m, (int)c,
+                                     mean.data_ptr<float>(),
+                                     istd.data_ptr<float>(),
+                                     scratch.data_ptr<float>(), s);
+  launch_bn_bwd_sum2(scratch.data_ptr<float>(), gx, (int)c,
+                     dgamma.data_ptr<float>(), dbeta.data_ptr<float>(), s);
+  torch::Tensor db = torch::empty({want_bias ? c : 0}, f32);
+  torch::Tensor bpart;
+  float* bpart_p = nullptr;
+  if (want_bias) {
+    bpart = torch::empty({256, c}, f32);
+    bpart_p = bpart.data_ptr<float>();
+  }
+  int gx2 = launch_bn_bwd_apply_act(
+      x.data_ptr(), dy.data_ptr(), dx.data_ptr(), m, (int)c,
+      mean.data_ptr<float>(), istd.data_ptr<float>(),
+      gamma.data_ptr<float>(), dgamma.data_ptr<float>(),
+      dbeta.data_ptr<float>(), (int)act, (float)slope, bpart_p, s);
+  if (want_bias)
+    launch_col_sum_sum2(bpart.data_ptr<float>(), gx2, (int)c,
+                        db.data_ptr<float>(), s);
+  return {dx, dgamma, dbeta, db};
+}
+
 // -------------------------------------------------------------------- fp8
 std::vector<torch::Tensor> fp8_quantize(torch::Tensor x) {
   // bf16 -> e4m3 with per-tensor dynamic amax scaling.
@@ -809,6 +857,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
           "BN training forward from producer-fused statistics");
   mod.def("bn_fwd_eval", &bn_fwd_eval);
   mod.def("bn_bwd", &bn_bwd);
+  mod.def("bn_bwd_act", &bn_bwd_act);
   mod.def("fused_adam", &fused_adam);
   mod.def("fused_rmsprop", &fused_rmsprop);
   mod.def("csv_load", &csv_load, "multithreaded CSV -> fp32 tensor");

@@ -171,6 +171,53 @@ def test_bn_stats_fusion_equivalence(monkeypatch):
     assert float(err) < 0.03, float(err)
 
 
+def test_act_bwd_fusion_equivalence(monkeypatch):
+    """BN-consumer act-backward fusion (default ON) must produce the same
+    gradients as the unfused path (GDLJ_NO_ACT_FUSE=1) for every
+    parameter of a conv->BN and convT->BN model."""
+    from gan_deeplearning4j_amd.config import preset
+    from gan_deeplearning4j_amd.models import build_dcgan
+
+    def grads(no_fuse):
+        if no_fuse:
+            monkeypatch.setenv("GDLJ_NO_ACT_FUSE", "1")
+        else:
+            monkeypatch.delenv("GDLJ_NO_ACT_FUSE", raising=False)
+        torch.manual_seed(0)
+        cfg = preset("dcgan64")
+        gen, dis = build_dcgan(cfg)
+        # fusion pass must have marked BN layers
+        marked = [n for n in dis.layer_names()
+                  if getattr(dis.layers[n], "bwd_act", None)]
+        assert len(marked) >= 1, marked
+        dis.to_device(torch.device("cuda:0"), torch.bfloat16)
+        gen.to_device(torch.device("cuda:0"), torch.bfloat16)
+        dis.train()
+        gen.train()
+        g = torch.Generator().manual_seed(7)
+        x = (torch.rand(16, 3, 64, 64, generator=g) * 2 - 1).to(
+            "cuda:0", torch.bfloat16)
+        z = torch.randn(16, cfg.model.z_size, generator=g).to(
+            "cuda:0", torch.bfloat16)
+        (dis(gen(z)).float() ** 2).mean().backward()
+        (dis(x).float() ** 2).mean().backward()
+        out = {}
+        for graph, tag in ((dis, "d"), (gen, "g")):
+            for i, p in enumerate(graph.parameters()):
+                if p.grad is not None:
+                    out[f"{tag}{i}"] = p.grad.float().cpu()
+        return out
+
+    fused = grads(no_fuse=False)
+    plain = grads(no_fuse=True)
+    assert fused.keys() == plain.keys() and len(fused) > 10
+    for k in fused:
+        a, b = fused[k], plain[k]
+        denom = b.abs().max().clamp_min(1e-6)
+        err = (a - b).abs().max() / denom
+        assert float(err) < 0.05, (k, float(err))
+
+
 def test_reference_protocol_cpu_gpu_consistency(tmp_path):
     """Same seed, one reference iteration: GPU bf16 losses must track the
     CPU fp32 reference within bf16 tolerance."""
