@@ -162,27 +162,33 @@ class ComputationGraph(nn.Module):
                 consumers.setdefault(src, []).append(name)
         for name in self._topo:
             layer = self.layers[name]
-            if not isinstance(layer, BatchNormLayer):
-                continue
             if name in self.preprocessors:
-                continue  # reshape between producer and BN: axes differ
+                continue  # reshape between producer and consumer: axes differ
             srcs = self._vertex_inputs[name]
             if len(srcs) != 1 or srcs[0] not in self.layers:
                 continue
             prod = self.layers[srcs[0]]
             if len(consumers.get(srcs[0], [])) != 1:
                 continue
-            if isinstance(prod, OutputLayer):
+            if isinstance(prod, OutputLayer) or not isinstance(
+                    prod, (Conv2dLayer, ConvTranspose2dLayer, DenseLayer)):
                 continue
-            if isinstance(prod, (Conv2dLayer, ConvTranspose2dLayer,
-                                 DenseLayer)):
+            prod_act = (prod.activation in ACT_CODES
+                        and ACT_CODES[prod.activation] != 0)
+            if isinstance(layer, BatchNormLayer):
                 if stats_fuse:
                     prod.emit_bn_stats = True
-                if prod.activation in ACT_CODES and \
-                        ACT_CODES[prod.activation] != 0:
+                if prod_act:
                     layer.bwd_act = (ACT_CODES[prod.activation],
                                      getattr(prod, "slope", 0.2),
                                      getattr(prod, "bias", None) is not None)
+            elif isinstance(layer, Conv2dLayer) and prod_act and \
+                    not isinstance(prod, DenseLayer):
+                # strided-conv consumer: its dgrad col2im folds the
+                # producer's activation backward + bias grad in
+                layer.prev_act = (ACT_CODES[prod.activation],
+                                  getattr(prod, "slope", 0.2),
+                                  getattr(prod, "bias", None) is not None)
 
     # ------------------------------------------------------------ build
     def _toposort(self, vertices) -> list[str]:

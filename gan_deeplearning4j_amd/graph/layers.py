@@ -158,7 +158,8 @@ class Conv2dLayer(BaseLayer):
     def forward(self, x):
         return OF.conv2d(x, self.weight, self.bias, self.stride, self.padding,
                          self.activation, self.slope,
-                         emit_stats=self.emit_bn_stats and self.training)
+                         emit_stats=self.emit_bn_stats and self.training,
+                         prev_act=getattr(self, "prev_act", None))
 
     def out_shape(self, in_shape):
         n, c, h, w = in_shape

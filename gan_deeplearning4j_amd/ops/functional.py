@@ -59,13 +59,19 @@ def conv2d(
     act: Optional[str] = None,
     slope: float = 0.2,
     emit_stats: bool = False,
+    prev_act=None,
 ) -> torch.Tensor:
-    """NCHW-logical conv; on GPU runs NHWC im2col-MFMA-GEMM kernels."""
+    """NCHW-logical conv; on GPU runs NHWC im2col-MFMA-GEMM kernels.
+
+    prev_act: optional (act_code, slope, has_bias) of the sole producer
+    feeding x — the strided dgrad folds that activation's backward into
+    its col2im pass (see gpu_ops deposit_act_fused).
+    """
     if x.is_cuda:
         from . import gpu_ops
 
         return gpu_ops.conv2d(x, w, b, stride, padding, act or "identity",
-                              slope, emit_stats)
+                              slope, emit_stats, prev_act)
     return _apply_act(F.conv2d(x, w, b, stride=stride, padding=padding), act, slope)
 
 

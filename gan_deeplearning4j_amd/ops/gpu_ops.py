@@ -259,7 +259,7 @@ def _pad_channels(xh: torch.Tensor, c8: int) -> torch.Tensor:
 class _Conv2d(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, w, b, stride: int, pad: int, act: int, slope: float,
-                emit_stats: bool):
+                emit_stats: bool, prev_act):
         ext = hip_ext()
         N, C, H, W = x.shape
         Kout, _, R, S = w.shape
@@ -303,6 +303,7 @@ class _Conv2d(torch.autograd.Function):
         ctx.has_bias = b is not None
         ctx.dtypes = (x.dtype, w.dtype, b.dtype if b is not None else None)
         ctx.wref = w
+        ctx.prev_act = prev_act
         out = _as_nchw_view(y2d.view(N, Ho, Wo, Kout))
         if stats is not None:
             deposit_bn_stats(out, stats)
@@ -389,20 +390,38 @@ class _Conv2d(torch.autograd.Function):
                     wp[:, :rsc8].t().contiguous()))   # [rsc8, kout_pad]
                 dprep = _pad_k(dpre)
                 dcol = ext.gemm_tn(dprep, wt, None, 0, 0.0, False)
-                dxh = ext.col2im(dcol, N, H, W, C8, Ho, Wo, R, S, stride,
-                                 pad, rsc8, None, 0, 0.0)
-                if C8 != C:
-                    dxh = dxh[..., :C].contiguous()
-                dx = _as_nchw_view(dxh).to(ctx.dtypes[0])
+                pact = ctx.prev_act
+                if pact is not None and (
+                        C8 != C
+                        or os.environ.get("GDLJ_NO_ACT_FUSE") == "1"):
+                    pact = None
+                if pact is not None:
+                    # the producer's act backward folds into the col2im
+                    # (xh IS the producer's activation output) along with
+                    # its bias-grad column sums; hand both over via the
+                    # side channel so the producer skips its own pass
+                    p_code, p_slope, p_bias = pact
+                    dxh, pdb = ext.col2im_dact(
+                        dcol, N, H, W, C8, Ho, Wo, R, S, stride, pad,
+                        rsc8, xh, p_code, p_slope, p_bias)
+                    dx = _as_nchw_view(dxh).to(ctx.dtypes[0])
+                    deposit_act_fused(
+                        dx, (p_code, C, pdb if p_bias else None))
+                else:
+                    dxh = ext.col2im(dcol, N, H, W, C8, Ho, Wo, R, S,
+                                     stride, pad, rsc8, None, 0, 0.0)
+                    if C8 != C:
+                        dxh = dxh[..., :C].contiguous()
+                    dx = _as_nchw_view(dxh).to(ctx.dtypes[0])
         if ctx.has_bias and ctx.needs_input_grad[2] and db is None:
             db = ext.col_sum(dpre).to(ctx.dtypes[2])
-        return dx, dw, db, None, None, None, None, None
+        return dx, dw, db, None, None, None, None, None, None
 
 
 def conv2d(x, w, b=None, stride=1, padding=0, act="identity", slope=0.2,
-           emit_stats=False):
+           emit_stats=False, prev_act=None):
     return _Conv2d.apply(x, w, b, stride, padding, ACT_CODES[act], slope,
-                         emit_stats)
+                         emit_stats, prev_act)
 
 
 # ============================================================ conv transpose

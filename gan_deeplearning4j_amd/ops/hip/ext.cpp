@@ -64,6 +64,8 @@ int launch_gemm_tn(const void*, const void*, void*, float*, const float*,
 int launch_gemm_tn_gather(const void*, const void*, void*, const float*,
                           int, int, int, long, int, float, ConvGather,
                           const void*, float*, hipStream_t);
+int launch_col2im_dact(const void*, void*, ConvGeom, const void*, int,
+                       float, float*, hipStream_t);
 int launch_col2im_stats(const void*, void*, ConvGeom, const float*, int,
                         float, float*, hipStream_t);
 void launch_gemm_nt(const void*, const void*, float*, int, int, int, long,
@@ -379,6 +381,39 @@ torch::Tensor col2im(torch::Tensor dcol, int64_t N, int64_t H, int64_t W,
   launch_col2im(dcol.data_ptr(), out.data_ptr(), g, bias_p, (int)act,
                 (float)slope, cur_stream());
   return out;
+}
+
+// col2im with the producer conv's activation backward fused (strided
+// dgrad): y0 is the producer's activation output [N,H,W,C]; returns
+// {dpre_nhwc, db} where db is the producer's bias gradient.
+std::vector<torch::Tensor> col2im_dact(
+    torch::Tensor dcol, int64_t N, int64_t H, int64_t W, int64_t C,
+    int64_t Ho, int64_t Wo, int64_t R, int64_t S, int64_t stride,
+    int64_t pad, int64_t kpad, torch::Tensor y0, int64_t act, double slope,
+    bool want_bias) {
+  check_bf16(dcol, "dcol");
+  check_bf16(y0, "y0");
+  TORCH_CHECK(C % 8 == 0 && kpad % 8 == 0, "col2im_dact needs C,kpad %8==0");
+  TORCH_CHECK(y0.numel() == N * H * W * C, "y0 shape");
+  ConvGeom g{(int)N, (int)H, (int)W, (int)C, (int)Ho, (int)Wo,
+             (int)R, (int)S, (int)stride, (int)pad, (int)kpad};
+  auto f32 = dcol.options().dtype(torch::kFloat32);
+  torch::Tensor out = torch::empty({N, H, W, C}, dcol.options());
+  torch::Tensor db = torch::empty({want_bias ? C : 0}, f32);
+  torch::Tensor part;
+  float* part_p = nullptr;
+  auto s = cur_stream();
+  if (want_bias) {
+    part = torch::empty({256, C}, f32);
+    part_p = part.data_ptr<float>();
+  }
+  int gx = launch_col2im_dact(dcol.data_ptr(), out.data_ptr(), g,
+                              y0.data_ptr(), (int)act, (float)slope, part_p,
+                              s);
+  if (want_bias)
+    launch_col_sum_sum2(part.data_ptr<float>(), gx, (int)C,
+                        db.data_ptr<float>(), s);
+  return {out, db};
 }
 
 // col2im with fused BN stats: returns {y_nhwc, sum, sumsq}
@@ -858,6 +893,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("bn_fwd_eval", &bn_fwd_eval);
   mod.def("bn_bwd", &bn_bwd);
   mod.def("bn_bwd_act", &bn_bwd_act);
+  mod.def("col2im_dact", &col2im_dact);
   mod.def("fused_adam", &fused_adam);
   mod.def("fused_rmsprop", &fused_rmsprop);
   mod.def("csv_load", &csv_load, "multithreaded CSV -> fp32 tensor");

@@ -251,6 +251,82 @@ __global__ void col2im_stats_v8(const unsigned short* __restrict__ dcol,
   }
 }
 
+// col2im with the PRODUCER's activation backward fused: din here is the
+// gradient at the producer conv's activation output, and y0 is that
+// activation output itself ([N,H,W,C], elementwise-aligned) — so
+// dpre = col_accum * act'(y0) costs one extra read instead of the
+// producer's whole standalone act_bwd_bias pass, and the producer's
+// bias gradient (column sums of dpre) reduces into partials [gx][C] on
+// the way out. Col-group structure like col2im_stats_v8.
+template <int SS>
+__global__ void col2im_dact_v8(const unsigned short* __restrict__ dcol,
+                               s16x8* __restrict__ din, ConvGeom g,
+                               const s16x8* __restrict__ y0, int act,
+                               float slope, float* __restrict__ part) {
+  int c8 = g.C / 8;
+  long rows = (long)g.N * g.H * g.W;          // pixels
+  int g0 = blockIdx.y * 32;
+  int groups = min(32, c8 - g0);
+  int lanes = (int)blockDim.x / groups;
+  int sub = (int)threadIdx.x / groups;
+  int cg = g0 + (int)threadIdx.x % groups;
+  float db[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+  if (sub < lanes) {
+    for (long px = (long)blockIdx.x * lanes + sub; px < rows;
+         px += (long)gridDim.x * lanes) {
+      long t = px;
+      int w = (int)(t % g.W);
+      t /= g.W;
+      int h = (int)(t % g.H);
+      int n = (int)(t / g.H);
+      float acc[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+      for (int r = 0; r < g.R; ++r) {
+        int hop = h + g.pad - r;
+        int st = SS ? SS : g.stride;
+        if (hop < 0 || (SS == 2 ? (hop & 1) : (SS == 1 ? 0 : hop % st)))
+          continue;
+        int ho = SS == 2 ? (hop >> 1) : (SS == 1 ? hop : hop / st);
+        if (ho >= g.Ho) continue;
+        for (int s_ = 0; s_ < g.S; ++s_) {
+          int wop = w + g.pad - s_;
+          if (wop < 0 || (SS == 2 ? (wop & 1) : (SS == 1 ? 0 : wop % st)))
+            continue;
+          int wo = SS == 2 ? (wop >> 1) : (SS == 1 ? wop : wop / st);
+          if (wo >= g.Wo) continue;
+          long np = ((long)n * g.Ho + ho) * g.Wo + wo;
+          s16x8 v = *(const s16x8*)(&dcol[np * g.kpad +
+                                          (r * g.S + s_) * g.C + cg * 8]);
+          #pragma unroll
+          for (int j = 0; j < 8; ++j) acc[j] += bf2f((unsigned short)v[j]);
+        }
+      }
+      s16x8 vy = y0[px * c8 + cg];
+      s16x8 o;
+      #pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float y = bf2f((unsigned short)vy[j]);
+        float dpre = acc[j] * act_bwd_from_y(y, act, slope);
+        o[j] = (short)f2bf(dpre);
+        db[j] += dpre;
+      }
+      din[px * c8 + cg] = o;
+    }
+  }
+  if (part == nullptr) return;
+  __shared__ float red[256];
+  #pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    red[threadIdx.x] = db[j];
+    __syncthreads();
+    if (sub == 0) {
+      float acc = 0.f;
+      for (int q = 0; q < lanes; ++q) acc += red[q * groups + cg - g0];
+      part[(long)blockIdx.x * g.C + cg * 8 + j] = acc;
+    }
+    __syncthreads();
+  }
+}
+
 extern "C" {
 
 // returns gx (partials rows)
@@ -276,6 +352,32 @@ int launch_col2im_stats(const void* dcol, void* din, ConvGeom g,
     hipLaunchKernelGGL((col2im_stats_v8<0>), grid, dim3(256), 0, s,
                        (const unsigned short*)dcol, (s16x8*)din, g, bias,
                        act, slope, part);
+  return (int)grid.x;
+}
+
+// returns gx (bias-partials rows)
+int launch_col2im_dact(const void* dcol, void* din, ConvGeom g,
+                       const void* y0, int act, float slope, float* part,
+                       hipStream_t s) {
+  int c8 = g.C / 8;
+  int groups = c8 < 32 ? c8 : 32;
+  int lanes = 256 / groups;
+  long rows = (long)g.N * g.H * g.W;
+  long chunks = (rows + lanes - 1) / lanes;
+  dim3 grid((unsigned)min((long)256, max((long)1, chunks)),
+            (unsigned)ceil_div(c8, 32));
+  if (g.stride == 2)
+    hipLaunchKernelGGL((col2im_dact_v8<2>), grid, dim3(256), 0, s,
+                       (const unsigned short*)dcol, (s16x8*)din, g,
+                       (const s16x8*)y0, act, slope, part);
+  else if (g.stride == 1)
+    hipLaunchKernelGGL((col2im_dact_v8<1>), grid, dim3(256), 0, s,
+                       (const unsigned short*)dcol, (s16x8*)din, g,
+                       (const s16x8*)y0, act, slope, part);
+  else
+    hipLaunchKernelGGL((col2im_dact_v8<0>), grid, dim3(256), 0, s,
+                       (const unsigned short*)dcol, (s16x8*)din, g,
+                       (const s16x8*)y0, act, slope, part);
   return (int)grid.x;
 }
 

@@ -190,6 +190,9 @@ def test_act_bwd_fusion_equivalence(monkeypatch):
         marked = [n for n in dis.layer_names()
                   if getattr(dis.layers[n], "bwd_act", None)]
         assert len(marked) >= 1, marked
+        # conv->conv chain: consumer fuses the producer's act backward
+        # into its dgrad col2im
+        assert getattr(dis.layers["d_conv_1"], "prev_act", None) is not None
         dis.to_device(torch.device("cuda:0"), torch.bfloat16)
         gen.to_device(torch.device("cuda:0"), torch.bfloat16)
         dis.train()
