@@ -404,7 +404,7 @@ std::vector<torch::Tensor> col2im_dact(
   float* part_p = nullptr;
   auto s = cur_stream();
   if (want_bias) {
-    part = torch::empty({256, C}, f32);
+    part = torch::empty({2048, C}, f32);
     part_p = part.data_ptr<float>();
   }
   int gx = launch_col2im_dact(dcol.data_ptr(), out.data_ptr(), g,

@@ -364,7 +364,10 @@ int launch_col2im_dact(const void* dcol, void* din, ConvGeom g,
   int lanes = 256 / groups;
   long rows = (long)g.N * g.H * g.W;
   long chunks = (rows + lanes - 1) / lanes;
-  dim3 grid((unsigned)min((long)256, max((long)1, chunks)),
+  // col2im gathers R*S taps per pixel: it needs a near-full grid (a
+  // 256-block cap measured 4x slower at DCGAN-64 dgrad sizes); partials
+  // rows scale with grid.x (buffer sized 2048 in the binding)
+  dim3 grid((unsigned)min((long)2048, max((long)1, chunks)),
             (unsigned)ceil_div(c8, 32));
   if (g.stride == 2)
     hipLaunchKernelGGL((col2im_dact_v8<2>), grid, dim3(256), 0, s,
