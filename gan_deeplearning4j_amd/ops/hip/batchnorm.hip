@@ -459,7 +459,7 @@ static int _egrid(long total) {
 int launch_bn_stats_part(const void* x, long m, int c, float* scratch,
                          hipStream_t s) {
   dim3 g = _colgrid_v8(m, c);
-  if (g.x > 256) g.x = 256;
+  if (g.x > 2048) g.x = 2048;
   hipLaunchKernelGGL(bn_stats_v8, g, dim3(256), 0, s, (const s16x8*)x, m, c,
                      scratch, nullptr);
   return (int)g.x;
@@ -518,7 +518,7 @@ int launch_bn_bwd_reduce_part(const void* x, const void* dy, long m, int c,
                               const float* mean, const float* istd,
                               float* scratch, hipStream_t s) {
   dim3 g = _colgrid_v8(m, c);
-  if (g.x > 256) g.x = 256;
+  if (g.x > 2048) g.x = 2048;
   hipLaunchKernelGGL(bn_bwd_reduce_v8, g, dim3(256), 0, s, (const s16x8*)x,
                      (const s16x8*)dy, m, c, mean, istd, scratch, nullptr);
   return (int)g.x;
@@ -544,7 +544,7 @@ int launch_bn_bwd_apply_act(const void* x, const void* dy, void* dx, long m,
                             const float* dbeta, int act, float slope,
                             float* bias_part, hipStream_t s) {
   dim3 g = _colgrid_v8(m, c);
-  if (g.x > 256) g.x = 256;
+  if (g.x > 2048) g.x = 2048;
   hipLaunchKernelGGL(bn_bwd_apply_act_v8, g, dim3(256), 0, s, (const s16x8*)x,
                      (const s16x8*)dy, (s16x8*)dx, m, c, mean, istd, gamma,
                      dgamma, dbeta, act, slope, bias_part);

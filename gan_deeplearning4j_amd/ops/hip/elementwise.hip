@@ -298,7 +298,7 @@ int launch_act_bwd_bias(const void* dy, const void* y, void* dx,
   int groups = n8 < 32 ? n8 : 32;
   int lanes = 256 / groups;
   long chunks = (m + lanes - 1) / lanes;
-  dim3 grid((unsigned)min((long)256, max((long)1, chunks)),
+  dim3 grid((unsigned)min((long)2048, max((long)1, chunks)),
             (unsigned)ceil_div(n8, 32));
   hipLaunchKernelGGL(act_bwd_bias_v8, grid, dim3(256), 0, s,
                      (const s16x8*)dy, (const s16x8*)y, (s16x8*)dx, scratch,
@@ -312,7 +312,7 @@ int launch_col_sum_part(const void* a, float* scratch, long m, int n,
   int groups = n8 < 32 ? n8 : 32;
   int lanes = 256 / groups;
   long chunks = (m + lanes - 1) / lanes;
-  dim3 grid((unsigned)min((long)256, max((long)1, chunks)),
+  dim3 grid((unsigned)min((long)2048, max((long)1, chunks)),
             (unsigned)ceil_div(n8, 32));
   hipLaunchKernelGGL(col_sum_v8, grid, dim3(256), 0, s, (const s16x8*)a,
                      scratch, m, n);

@@ -433,7 +433,7 @@ std::vector<torch::Tensor> col2im_stats(
   }
   auto f32 = dcol.options().dtype(torch::kFloat32);
   torch::Tensor out = torch::empty({N, H, W, C}, dcol.options());
-  torch::Tensor part = torch::empty({256, 2 * C}, f32);
+  torch::Tensor part = torch::empty({2048, 2 * C}, f32);
   torch::Tensor sum = torch::empty({C}, f32), sumsq = torch::empty({C}, f32);
   auto st = cur_stream();
   int gx = launch_col2im_stats(dcol.data_ptr(), out.data_ptr(), g, bias_p,
@@ -507,7 +507,7 @@ std::vector<torch::Tensor> act_bwd_bias(torch::Tensor dy,
   TORCH_CHECK(n % 8 == 0, "act_bwd_bias needs N % 8 == 0");
   auto f32 = dy.options().dtype(torch::kFloat32);
   torch::Tensor dx = torch::empty_like(dy);
-  torch::Tensor scratch = torch::empty({256, n}, f32);
+  torch::Tensor scratch = torch::empty({2048, n}, f32);
   torch::Tensor db = torch::empty({n}, f32);
   const void* y_p = nullptr;
   if (act != 0) {
@@ -540,7 +540,7 @@ torch::Tensor col_sum(torch::Tensor a) {
   auto f32 = a.options().dtype(torch::kFloat32);
   if (n % 8 == 0) {
     torch::Tensor out = torch::empty({n}, f32);
-    torch::Tensor scratch = torch::empty({256, n}, f32);
+    torch::Tensor scratch = torch::empty({2048, n}, f32);
     int gx = launch_col_sum_part(a.data_ptr(), scratch.data_ptr<float>(),
                                  (long)m, (int)n, cur_stream());
     launch_col_sum_sum2(scratch.data_ptr<float>(), gx, (int)n,
@@ -612,7 +612,7 @@ std::vector<torch::Tensor> bn_fwd_train(torch::Tensor x, torch::Tensor gamma,
   if (c % 8 == 0) {
     sum = torch::empty({c}, f32);
     sumsq = torch::empty({c}, f32);
-    torch::Tensor scratch = torch::empty({256, 2 * c}, f32);
+    torch::Tensor scratch = torch::empty({2048, 2 * c}, f32);
     int gx = launch_bn_stats_part(x.data_ptr(), m, (int)c,
                                   scratch.data_ptr<float>(), s);
     launch_bn_stats_sum2(scratch.data_ptr<float>(), gx, (int)c,
@@ -697,7 +697,7 @@ std::vector<torch::Tensor> bn_bwd(torch::Tensor x, torch::Tensor dy,
   if (c % 8 == 0) {
     dgamma = torch::empty({c}, f32);
     dbeta = torch::empty({c}, f32);
-    torch::Tensor scratch = torch::empty({256, 2 * c}, f32);
+    torch::Tensor scratch = torch::empty({2048, 2 * c}, f32);
     int gx = launch_bn_bwd_reduce_part(x.data_ptr(), dy.data_ptr(), m,
                                        (int)c, mean.data_ptr<float>(),
                                        istd.data_ptr<float>(),
@@ -738,7 +738,7 @@ std::vector<torch::Tensor> bn_bwd_act(torch::Tensor x, torch::Tensor dy,
   auto s = cur_stream();
   torch::Tensor dgamma = torch::empty({c}, f32);
   torch::Tensor dbeta = torch::empty({c}, f32);
-  torch::Tensor scratch = torch::empty({256, 2 * c}, f32);
+  torch::Tensor scratch = torch::empty({2048, 2 * c}, f32);
   int gx = launch_bn_bwd_reduce_part(x.data_ptr(), dy.data_ptr(), m, (int)c,
                                      mean.data_ptr<float>(),
                                      istd.data_ptr<float>(),
@@ -749,7 +749,7 @@ std::vector<torch::Tensor> bn_bwd_act(torch::Tensor x, torch::Tensor dy,
   torch::Tensor bpart;
   float* bpart_p = nullptr;
   if (want_bias) {
-    bpart = torch::empty({256, c}, f32);
+    bpart = torch::empty({2048, c}, f32);
     bpart_p = bpart.data_ptr<float>();
   }
   int gx2 = launch_bn_bwd_apply_act(

@@ -338,7 +338,7 @@ int launch_col2im_stats(const void* dcol, void* din, ConvGeom g,
   int lanes = 256 / groups;
   long rows = (long)g.N * g.H * g.W;
   long chunks = (rows + lanes - 1) / lanes;
-  dim3 grid((unsigned)min((long)256, max((long)1, chunks)),
+  dim3 grid((unsigned)min((long)2048, max((long)1, chunks)),
             (unsigned)ceil_div(c8, 32));
   if (g.stride == 2)
     hipLaunchKernelGGL((col2im_stats_v8<2>), grid, dim3(256), 0, s,
