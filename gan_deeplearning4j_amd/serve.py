@@ -82,10 +82,94 @@ class _Endpoint:
         return out[:n].float().cpu()
 
 
+class _MicroBatcher:
+    """Accumulates concurrent requests into one fixed-shape graph replay.
+
+    Requests arriving within `window_ms` of each other (up to the
+    endpoint's max_batch rows) share a single padded replay — the
+    serving idiom for a captured hipGraph, where launch count, not
+    per-row work, is the cost at small batches.
+    """
+
+    def __init__(self, endpoint: _Endpoint, window_ms: float = 2.0):
+        self.ep = endpoint
+        self.window = window_ms / 1000.0
+        self.queue = None
+        self._task = None
+        self._loop = None
+
+    def ensure_running(self):
+        import asyncio
+
+        loop = asyncio.get_running_loop()
+        if self._loop is not loop:
+            # fresh event loop (first request, or a per-request test
+            # client): rebind the queue and worker to it
+            self.queue = asyncio.Queue()
+            self._loop = loop
+            self._task = loop.create_task(self._run())
+        elif self._task is None or self._task.done():
+            self._task = loop.create_task(self._run())
+
+    async def submit(self, x: torch.Tensor) -> torch.Tensor:
+        import asyncio
+
+        self.ensure_running()
+        fut = asyncio.get_running_loop().create_future()
+        await self.queue.put((x, fut))
+        return await fut
+
+    async def _run(self):
+        import asyncio
+
+        loop = asyncio.get_running_loop()
+        while True:
+            x, fut = await self.queue.get()
+            items = [(x, fut)]
+            rows = x.shape[0]
+            deadline = loop.time() + self.window
+            while rows < self.ep.max_batch:
+                timeout = deadline - loop.time()
+                if timeout <= 0:
+                    break
+                try:
+                    nx, nfut = await asyncio.wait_for(self.queue.get(),
+                                                      timeout)
+                except asyncio.TimeoutError:
+                    break
+                if rows + nx.shape[0] > self.ep.max_batch:
+                    # flush current batch; start the next with this item
+                    self._dispatch(items)
+                    items, rows = [], 0
+                    deadline = loop.time() + self.window
+                items.append((nx, nfut))
+                rows += nx.shape[0]
+            self._dispatch(items)
+
+    def _dispatch(self, items):
+        if not items:
+            return
+        xs = torch.cat([x for x, _ in items], dim=0)
+        try:
+            out = self.ep.run(xs)
+        except Exception as e:  # propagate to every waiter
+            for _, fut in items:
+                if not fut.done():
+                    fut.set_exception(e)
+            return
+        off = 0
+        for x, fut in items:
+            n = x.shape[0]
+            if not fut.done():
+                fut.set_result(out[off:off + n])
+            off += n
+
+
 def create_app(generator: Optional[ComputationGraph] = None,
                discriminator: Optional[ComputationGraph] = None,
                device: Optional[torch.device] = None,
-               max_batch: int = 64):
+               max_batch: int = 64,
+               batch_window_ms: float = 2.0):
     """Build the FastAPI app. Models may be None (their routes 404)."""
     from fastapi import FastAPI, HTTPException
 
@@ -97,6 +181,7 @@ def create_app(generator: Optional[ComputationGraph] = None,
               if generator is not None else None)
     dis_ep = (_Endpoint(discriminator, device, dtype, max_batch)
               if discriminator is not None else None)
+    gen_mb = _MicroBatcher(gen_ep, batch_window_ms) if gen_ep else None
 
     app = FastAPI(title="gan_deeplearning4j_amd serving")
 
@@ -121,7 +206,7 @@ def create_app(generator: Optional[ComputationGraph] = None,
         return out
 
     @app.post("/generate")
-    def generate(req: GenerateReq):
+    async def generate(req: GenerateReq):
         if gen_ep is None:
             raise HTTPException(404, "no generator loaded")
         if req.n > max_batch:
@@ -131,7 +216,8 @@ def create_app(generator: Optional[ComputationGraph] = None,
         g = torch.Generator().manual_seed(req.seed) if req.seed is not None \
             else None
         z = torch.randn(req.n, z_size, generator=g)
-        samples = gen_ep.run(z)
+        # concurrent requests coalesce into one padded graph replay
+        samples = await gen_mb.submit(z)
         if req.format == "png_base64":
             from .utils.imaging import save_image_grid
 

@@ -66,6 +66,56 @@ def test_discriminate(client):
     assert all(0.0 <= v <= 1.0 for v in scores)
 
 
+def test_microbatcher_coalesces_concurrent_requests():
+    """Concurrent submits within the window share ONE padded replay."""
+    import asyncio
+
+    from gan_deeplearning4j_amd.serve import _Endpoint, _MicroBatcher
+
+    cfg = preset("dcgan28")
+    gen, _ = build_dcgan(cfg)
+    ep = _Endpoint(gen, torch.device("cpu"), torch.float32, max_batch=16)
+    calls = []
+    real_run = ep.run
+    ep.run = lambda x: (calls.append(x.shape[0]), real_run(x))[1]
+    mb = _MicroBatcher(ep, window_ms=50.0)
+
+    async def go():
+        zs = [torch.randn(2, cfg.model.z_size) for _ in range(4)]
+        outs = await asyncio.gather(*(mb.submit(z) for z in zs))
+        return zs, outs
+
+    zs, outs = asyncio.run(go())
+    assert calls == [8], calls  # 4 x 2 rows coalesced into one run
+    for z, o in zip(zs, outs):
+        assert torch.allclose(o, gen.output(z), atol=1e-5)
+
+
+def test_microbatcher_splits_over_max_batch():
+    """A flood larger than max_batch splits into multiple replays, each
+    request still answered with its own rows."""
+    import asyncio
+
+    from gan_deeplearning4j_amd.serve import _Endpoint, _MicroBatcher
+
+    cfg = preset("dcgan28")
+    gen, _ = build_dcgan(cfg)
+    ep = _Endpoint(gen, torch.device("cpu"), torch.float32, max_batch=4)
+    calls = []
+    real_run = ep.run
+    ep.run = lambda x: (calls.append(x.shape[0]), real_run(x))[1]
+    mb = _MicroBatcher(ep, window_ms=50.0)
+
+    async def go():
+        zs = [torch.randn(3, cfg.model.z_size) for _ in range(3)]
+        return zs, await asyncio.gather(*(mb.submit(z) for z in zs))
+
+    zs, outs = asyncio.run(go())
+    assert sum(calls) == 9 and all(c <= 4 for c in calls), calls
+    for z, o in zip(zs, outs):
+        assert torch.allclose(o, gen.output(z), atol=1e-5)
+
+
 @pytest.mark.gpu
 def test_serve_gpu_compiled_path():
     """On MI355X the endpoint must serve through the captured hipGraph
