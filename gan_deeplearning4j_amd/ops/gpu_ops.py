@@ -281,8 +281,16 @@ class _Conv2d(torch.autograd.Function):
         stats = None
         if FP8_CONV and C8 % 16 == 0:
             xq, _, ix = ext.fp8_quantize(xh)
-            wq, _, iw = _packed(w, "fp8", lambda: tuple(
-                ext.fp8_quantize(wp)))
+
+            def build_fp8_pack():
+                # the double-rate fp8 MFMA consumes K in 128-deep tiles
+                wpk = wp
+                if wpk.shape[1] % 128:
+                    wpk = torch.nn.functional.pad(
+                        wpk, (0, 128 - wpk.shape[1] % 128))
+                return tuple(ext.fp8_quantize(wpk))
+
+            wq, _, iw = _packed(w, "fp8", build_fp8_pack)
             zp8 = _packed(wp, "zp8", lambda: torch.zeros(
                 32, dtype=torch.uint8, device=x.device))
             y2d = ext.conv_fwd_implicit_fp8(xq, wq, bias, ix, iw, zp8, N, H,

@@ -792,6 +792,7 @@ torch::Tensor conv_fwd_implicit_fp8(
   TORCH_CHECK(wq.scalar_type() == torch::kUInt8 && wq.is_contiguous());
   TORCH_CHECK(C % 16 == 0, "fp8 implicit conv needs C % 16 == 0");
   int64_t Kout = wq.size(0), kpad = wq.size(1);
+  TORCH_CHECK(kpad % 128 == 0, "fp8 pack K must be padded to 128");
   int64_t M = Nb * Ho * Wo;
   const float* bias_p = nullptr;
   if (bias.has_value() && bias->defined() && bias->numel() > 0) {

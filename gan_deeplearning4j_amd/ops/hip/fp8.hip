@@ -12,7 +12,6 @@
 #include "common.h"
 
 typedef __attribute__((ext_vector_type(8))) float f32x8;
-typedef long fp8x8;  // 8 e4m3 bytes = 2 VGPRs (MFMA operand)
 
 #define GLDS16(gsrc, ldst)                                                    \
   __builtin_amdgcn_global_load_lds(                                          \
@@ -77,8 +76,12 @@ __global__ void quant_fp8(const s16x8* __restrict__ x,
 
 // ------------------------------------------------------------ fp8 TN GEMM
 // A gathered from an fp8 NHWC image (implicit conv) or plain [M][K];
-// B plain fp8 [N][K]. LDS tiles [128 rows][64 k] fp8 = 8 KiB each.
-constexpr int F8_BM = 128, F8_BN = 128, F8_BK = 64;
+// B plain fp8 [N][K]. LDS tiles [128 rows][128 k] fp8 = 16 KiB each.
+// K depth is 128 to feed gfx950's DOUBLE-RATE fp8 MFMA
+// (mfma_scale_f32_16x16x128_f8f6f4 with unit e8m0 scales — there is no
+// non-scaled large-K fp8 form; the CDNA3-era 16x16x32_fp8 runs at the
+// bf16 FLOP rate, i.e. half the 5 PF fp8 peak).
+constexpr int F8_BM = 128, F8_BN = 128, F8_BK = 128;
 constexpr int F8_TILE_B = F8_BM * F8_BK;  // bytes (1B/elem)
 
 DEV_INLINE void f8_stage(const unsigned char* __restrict__ g, int row0,
@@ -86,11 +89,11 @@ DEV_INLINE void f8_stage(const unsigned char* __restrict__ g, int row0,
   const int t = threadIdx.x;
   const int wid = t >> 6;
   #pragma unroll
-  for (int i = 0; i < 2; ++i) {
-    int chunk = i * 256 + t;       // 512 chunks = row*4 + slot16
-    int row = chunk >> 2;
-    int slot = chunk & 3;
-    int gslot = slot ^ (row & 3);
+  for (int i = 0; i < 4; ++i) {
+    int chunk = i * 256 + t;       // 1024 chunks = row*8 + slot16
+    int row = chunk >> 3;
+    int slot = chunk & 7;
+    int gslot = slot ^ (row & 7);
     int grow = min(row0 + row, nrows - 1);
     const unsigned char* src = g + (long)grow * ldk + k0 + gslot * 16;
     char* dst = lds + (i * 256 + wid * 64) * 16;
@@ -110,16 +113,16 @@ DEV_INLINE void k_decode_f8(const ConvGather& g, unsigned k, int& r, int& s,
 // NOTE: 16-fp8 chunks must not straddle (r,s) boundaries -> requires
 // C % 16 == 0 (enforced by the binding).
 // Row decode hoisted out of the K-loop (same scheme as TnGatherStager in
-// gemm.hip): each thread stages the same 2 rows every K-step.
+// gemm.hip): each thread stages the same 4 rows every K-step.
 struct F8GatherStager {
-  long base[2];      // (long)n * H*W*C element offset per chunk
-  int h0[2], w0[2];  // ho*stride-pad / wo*stride-pad per chunk
+  long base[4];      // (long)n * H*W*C element offset per chunk
+  int h0[4], w0[4];  // ho*stride-pad / wo*stride-pad per chunk
 
   DEV_INLINE void init(const ConvGather& g, int row0, int nrows) {
     const int t = threadIdx.x;
     #pragma unroll
-    for (int i = 0; i < 2; ++i) {
-      int row = (i * 256 + t) >> 2;
+    for (int i = 0; i < 4; ++i) {
+      int row = (i * 256 + t) >> 3;
       int np = min(row0 + row, nrows - 1);
       unsigned q1 = fdiv((unsigned)np, g.fWo);
       int wo = (int)((unsigned)np - q1 * g.Wo);
@@ -138,11 +141,11 @@ struct F8GatherStager {
     const int t = threadIdx.x;
     const int wid = t >> 6;
     #pragma unroll
-    for (int i = 0; i < 2; ++i) {
+    for (int i = 0; i < 4; ++i) {
       int chunk = i * 256 + t;
-      int row = chunk >> 2;
-      int slot = chunk & 3;
-      int gslot = slot ^ (row & 3);
+      int row = chunk >> 3;
+      int slot = chunk & 7;
+      int gslot = slot ^ (row & 7);
       int k = k0 + gslot * 16;  // 16 fp8 channels per chunk
       const unsigned char* src = zp;
       if (k < g.rsc) {
@@ -159,11 +162,20 @@ struct F8GatherStager {
   }
 };
 
-DEV_INLINE fp8x8 f8_frag(const char* lds, int row, int kslot8) {
-  // fragment = 8 fp8 at k-octet kslot8; 16B swizzle on slot16 = kslot8>>1
-  int s16 = (kslot8 >> 1) ^ (row & 3);
-  int byte = row * 64 + s16 * 16 + (kslot8 & 1) * 8;
-  return *(const fp8x8*)(lds + byte);
+typedef __attribute__((ext_vector_type(8))) int i32x8;
+typedef __attribute__((ext_vector_type(4))) int i32x4;
+
+// fragment = 32 consecutive fp8 (lane's K window kq*32) as two XOR-swizzled
+// 16B LDS reads; rows are 128B (8 slots of 16B)
+DEV_INLINE i32x8 f8_frag32(const char* lds, int row, int kq) {
+  int s0 = (2 * kq) ^ (row & 7);
+  int s1 = (2 * kq + 1) ^ (row & 7);
+  i32x4 lo = *(const i32x4*)(lds + row * 128 + s0 * 16);
+  i32x4 hi = *(const i32x4*)(lds + row * 128 + s1 * 16);
+  i32x8 r;
+  r[0] = lo[0]; r[1] = lo[1]; r[2] = lo[2]; r[3] = lo[3];
+  r[4] = hi[0]; r[5] = hi[1]; r[6] = hi[2]; r[7] = hi[3];
+  return r;
 }
 
 template <bool GATHER_A>
@@ -222,21 +234,22 @@ __global__ __launch_bounds__(256, 2) void gemm_tn_fp8_core(
     }
     const char* Al = abuf(cur);
     const char* Bl = bbuf(cur);
-    #pragma unroll
-    for (int kc = 0; kc < 2; ++kc) {
-      fp8x8 a[4], b[4];
+    {
+      i32x8 a[4], b[4];
       #pragma unroll
       for (int mi = 0; mi < 4; ++mi)
-        a[mi] = f8_frag(Al, wr * 64 + mi * 16 + fr, kc * 4 + fq);
+        a[mi] = f8_frag32(Al, wr * 64 + mi * 16 + fr, fq);
       #pragma unroll
       for (int ni = 0; ni < 4; ++ni)
-        b[ni] = f8_frag(Bl, wc * 64 + ni * 16 + fr, kc * 4 + fq);
+        b[ni] = f8_frag32(Bl, wc * 64 + ni * 16 + fr, fq);
       #pragma unroll
       for (int mi = 0; mi < 4; ++mi)
         #pragma unroll
         for (int ni = 0; ni < 4; ++ni)
-          acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8(
-              a[mi], b[ni], acc[mi][ni], 0, 0, 0);
+          // cbsz=0/blgp=0: both operands e4m3 fp8; e8m0 scale byte 127
+          // = x1.0 (per-tensor scaling stays in the epilogue de-scale)
+          acc[mi][ni] = __builtin_amdgcn_mfma_scale_f32_16x16x128_f8f6f4(
+              a[mi], b[ni], acc[mi][ni], 0, 0, 0, 127, 0, 127);
     }
     asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
     __syncthreads();
