@@ -70,6 +70,34 @@ def test_conv2d_fuzz(case):
     assert relerr(b.grad, br.grad) < 0.06, case
 
 
+def test_parity_strided_path_equivalence(monkeypatch):
+    """The selectable parity-decomposed strided dgrad/convT-fwd
+    (GDLJ_PARITY=1) must match the default dcol+col2im path."""
+    from gan_deeplearning4j_amd.ops import gpu_ops
+
+    def run(parity):
+        monkeypatch.setattr(gpu_ops, "PARITY_STRIDED", parity)
+        torch.manual_seed(3)
+        x = (torch.randn(3, 16, 12, 12) * 0.5).to(DEV, torch.bfloat16)
+        w = (torch.randn(24, 16, 4, 4) * 0.2).to(DEV, torch.bfloat16)
+        b = torch.randn(24).to(DEV, torch.bfloat16)
+        x.requires_grad_(True)
+        w.requires_grad_(True)
+        y = gpu_ops.conv2d(x, w, b, 2, 1, "lrelu", 0.2)
+        y.backward(torch.ones_like(y))
+        xt = (torch.randn(3, 16, 6, 6) * 0.5).to(DEV, torch.bfloat16)
+        wt = (torch.randn(16, 24, 4, 4) * 0.2).to(DEV, torch.bfloat16)
+        yt = gpu_ops.conv_transpose2d(xt, wt, b, 2, 1, "tanh")
+        return y.detach(), x.grad.clone(), w.grad.clone(), yt.detach()
+
+    base = run(False)
+    par = run(True)
+    for a, p in zip(base, par):
+        err = (a.float() - p.float()).abs().max() / \
+            a.float().abs().max().clamp_min(1e-5)
+        assert float(err) < 0.02, float(err)
+
+
 @pytest.mark.parametrize("case", CASES)
 def test_conv_transpose2d_fuzz(case):
     from gan_deeplearning4j_amd.ops import gpu_ops
