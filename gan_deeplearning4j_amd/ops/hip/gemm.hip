@@ -26,6 +26,7 @@
 // zero-page); gemm_nt handles arbitrary K by zero-fill.
 
 #include "common.h"
+#include <stdlib.h>
 
 typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
 
@@ -115,7 +116,8 @@ DEV_INLINE void tn_stage(const unsigned short* __restrict__ g, int row0,
 // across the whole K-loop (each thread stages the same 4 rows every step),
 // so they are hoisted into registers once (init) and only the cheap k part
 // (tap + channel) is decoded per chunk per step (stage).
-struct TnGatherStager {
+template <int NTH>  // block thread count: 4 chunks staged per thread
+struct TnGatherStagerT {
   long base[4];        // (long)n * H*W*C element offset per chunk
   int h0[4], w0[4];    // mode-adjusted spatial bases per chunk
 
@@ -123,7 +125,7 @@ struct TnGatherStager {
     const int t = threadIdx.x;
     #pragma unroll
     for (int i = 0; i < 4; ++i) {
-      int row = (i * 256 + t) >> 3;
+      int row = (i * NTH + t) >> 3;
       int np = min(row0 + row, nrows - 1);
       int n, ho, wo;
       np_decode(g, (unsigned)np, n, ho, wo);
@@ -149,7 +151,7 @@ struct TnGatherStager {
     const int wid = t >> 6;
     #pragma unroll
     for (int i = 0; i < 4; ++i) {
-      int chunk = i * 256 + t;
+      int chunk = i * NTH + t;
       int row = chunk >> 3;
       int slot = chunk & 7;
       int gslot = slot ^ (row & 7);
@@ -186,11 +188,12 @@ struct TnGatherStager {
         }
         if (valid) src = img + base[i] + (long)(hi * g.W + wi) * g.C + c;
       }
-      char* dst = lds + (i * 256 + wid * 64) * 16;
+      char* dst = lds + (i * NTH + wid * 64) * 16;
       GLDS16(src, dst);
     }
   }
 };
+using TnGatherStager = TnGatherStagerT<256>;
 
 DEV_INLINE bf16x8 tn_frag(const char* lds, int row, int kslot) {
   int byte = row * 128 + ((kslot ^ (row & 7)) * 16);
@@ -371,6 +374,187 @@ __global__ __launch_bounds__(256, 2) void gemm_tn_core(
           C[crow * N + col] = f2bf(v);
         else
           Cf[crow * N + col] = v;
+      }
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// 256x128x64 TN variant (GDLJ_TN256): +37% arithmetic intensity per staged
+// byte (2M MACs per 48 KiB vs 1M per 32 KiB) at the SAME 8-wave/CU
+// occupancy — one 512-thread block per CU with 96 KiB dynamic LDS
+// (2x32 KiB A + 2x16 KiB B double buffers; the 64 KiB epilogue ctile
+// reuses the A region). No mode-2 scatter / fp32-out / bn_part here —
+// the launcher falls back to the 128 tile for those.
+// ---------------------------------------------------------------------------
+constexpr int T2_BM = 256, T2_BN = 128, T2_BK = 64;
+constexpr int T2_AT = T2_BM * T2_BK * 2;  // 32 KiB
+constexpr int T2_BT = T2_BN * T2_BK * 2;  // 16 KiB
+
+DEV_INLINE void t2_stage_a(const unsigned short* __restrict__ g, int row0,
+                           int nrows, long ldk, int k0, char* lds) {
+  const int t = threadIdx.x;
+  const int wid = t >> 6;
+  #pragma unroll
+  for (int i = 0; i < 4; ++i) {           // 2048 chunks
+    int chunk = i * 512 + t;
+    int row = chunk >> 3;
+    int slot = chunk & 7;
+    int gslot = slot ^ (row & 7);
+    int grow = min(row0 + row, nrows - 1);
+    const unsigned short* src = g + (long)grow * ldk + k0 + gslot * 8;
+    char* dst = lds + (i * 512 + wid * 64) * 16;
+    GLDS16(src, dst);
+  }
+}
+
+DEV_INLINE void t2_stage_b(const unsigned short* __restrict__ g, int row0,
+                           int nrows, long ldk, int k0, char* lds) {
+  const int t = threadIdx.x;
+  const int wid = t >> 6;
+  #pragma unroll
+  for (int i = 0; i < 2; ++i) {           // 1024 chunks
+    int chunk = i * 512 + t;
+    int row = chunk >> 3;
+    int slot = chunk & 7;
+    int gslot = slot ^ (row & 7);
+    int grow = min(row0 + row, nrows - 1);
+    const unsigned short* src = g + (long)grow * ldk + k0 + gslot * 8;
+    char* dst = lds + (i * 512 + wid * 64) * 16;
+    GLDS16(src, dst);
+  }
+}
+
+template <bool GATHER_A>
+__global__ __launch_bounds__(512, 1) void gemm_tn_core256(
+    const unsigned short* __restrict__ A, const unsigned short* __restrict__ B,
+    unsigned short* __restrict__ C, const float* __restrict__ bias, int M,
+    int N, int K, long lda, long ldb, int act, float slope, ConvGather ga,
+    const unsigned short* __restrict__ zp) {
+  extern __shared__ __attribute__((aligned(128))) char lds[];
+  auto abuf = [&](int i) -> char* { return lds + (i ? T2_AT : 0); };
+  auto bbuf = [&](int i) -> char* {
+    return lds + 2 * T2_AT + (i ? T2_BT : 0);
+  };
+
+  int nwgx = gridDim.x;
+  int bidx = blockIdx.x;
+  if (nwgx >= 8) {
+    int q = nwgx / 8, r = nwgx % 8;
+    int xcd = bidx % 8, idx = bidx / 8;
+    bidx = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + idx;
+  }
+  const int m0 = bidx * T2_BM;
+  const int n0 = blockIdx.y * T2_BN;
+
+  const int wid = threadIdx.x >> 6;
+  const int lane = threadIdx.x & 63;
+  const int wr = wid >> 1, wc = wid & 1;  // 4(m) x 2(n) wave grid
+  const int fr = lane & 15, fq = lane >> 4;
+
+  f32x4 acc[4][4];
+  #pragma unroll
+  for (int i = 0; i < 4; ++i)
+    #pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  const int ntiles = K / T2_BK;
+  TnGatherStagerT<512> gs;
+  if (GATHER_A) {
+    gs.init(ga, m0, M);
+    gs.stage(A, ga, zp, 0, abuf(0));
+  } else {
+    t2_stage_a(A, m0, M, lda, 0, abuf(0));
+  }
+  t2_stage_b(B, n0, N, ldb, 0, bbuf(0));
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __syncthreads();
+
+  for (int t = 0; t < ntiles; ++t) {
+    int cur = t & 1;
+    if (t + 1 < ntiles) {
+      if (GATHER_A)
+        gs.stage(A, ga, zp, (t + 1) * T2_BK, abuf(cur ^ 1));
+      else
+        t2_stage_a(A, m0, M, lda, (t + 1) * T2_BK, abuf(cur ^ 1));
+      t2_stage_b(B, n0, N, ldb, (t + 1) * T2_BK, bbuf(cur ^ 1));
+    }
+    const char* Al = abuf(cur);
+    const char* Bl = bbuf(cur);
+    #pragma unroll
+    for (int kc = 0; kc < 2; ++kc) {
+      bf16x8 a[4], b[4];
+      #pragma unroll
+      for (int mi = 0; mi < 4; ++mi)
+        a[mi] = tn_frag(Al, wr * 64 + mi * 16 + fr, kc * 4 + fq);
+      #pragma unroll
+      for (int ni = 0; ni < 4; ++ni)
+        b[ni] = tn_frag(Bl, wc * 64 + ni * 16 + fr, kc * 4 + fq);
+      #pragma unroll
+      for (int mi = 0; mi < 4; ++mi)
+        #pragma unroll
+        for (int ni = 0; ni < 4; ++ni)
+          acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a[mi], b[ni], acc[mi][ni], 0, 0, 0);
+    }
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __syncthreads();
+  }
+
+  // epilogue: bias + act, LDS-staged coalesced stores (ctile 256x128 bf16
+  // = 64 KiB in the A-buffer region)
+  if ((N & 7) == 0) {
+    unsigned short* ctile = (unsigned short*)lds;
+    __syncthreads();
+    #pragma unroll
+    for (int mi = 0; mi < 4; ++mi) {
+      #pragma unroll
+      for (int ni = 0; ni < 4; ++ni) {
+        int lc = wc * 64 + ni * 16 + fr;
+        float bv = bias != nullptr ? bias[min(n0 + lc, N - 1)] : 0.f;
+        #pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          int lr = wr * 64 + mi * 16 + fq * 4 + r;
+          ctile[lr * 128 + lc] = f2bf(act_fwd(acc[mi][ni][r] + bv, act,
+                                              slope));
+        }
+      }
+    }
+    __syncthreads();
+    const int t = threadIdx.x;
+    #pragma unroll
+    for (int i = 0; i < 8; ++i) {
+      int piece = i * 512 + t;       // 4096 16B pieces = 256 rows x 16
+      int row = piece >> 4;
+      int seg = piece & 15;
+      int grow = m0 + row;
+      int gcol = n0 + seg * 8;
+      if (grow < M && gcol < N) {
+        s16x8 v = *(const s16x8*)(ctile + row * 128 + seg * 8);
+        if (gcol + 8 <= N) {
+          *(s16x8*)(&C[(long)grow * N + gcol]) = v;
+        } else {
+          for (int j = 0; j < 8 && gcol + j < N; ++j)
+            C[(long)grow * N + gcol + j] = (unsigned short)v[j];
+        }
+      }
+    }
+    return;
+  }
+  // N % 8 != 0 fallback: direct predicated stores
+  #pragma unroll
+  for (int mi = 0; mi < 4; ++mi) {
+    #pragma unroll
+    for (int ni = 0; ni < 4; ++ni) {
+      int col = n0 + wc * 64 + ni * 16 + fr;
+      if (col >= N) continue;
+      float bv = bias != nullptr ? bias[col] : 0.f;
+      #pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int row = m0 + wr * 64 + mi * 16 + fq * 4 + r;
+        if (row >= M) continue;
+        C[(long)row * N + col] = f2bf(act_fwd(acc[mi][ni][r] + bv, act,
+                                              slope));
       }
     }
   }
@@ -573,9 +757,39 @@ __global__ __launch_bounds__(NT_THREADS, 2) void gemm_nt_core(
 // ---------------------------------------------------------------------------
 extern "C" {
 
+// env-gated 256-row TN tile (see gemm_tn_core256). Opted in with
+// GDLJ_TN256=1; requires bf16 output, no bn_part, gather mode != 2,
+// and enough rows to fill the chip at 1 block/CU.
+static int t2_enabled() {
+  static int v = -1;
+  if (v < 0) {
+    const char* e = getenv("GDLJ_TN256");
+    v = (e != nullptr && e[0] == '1') ? 1 : 0;
+    if (v) {
+      (void)hipFuncSetAttribute(
+          reinterpret_cast<const void*>(&gemm_tn_core256<false>),
+          hipFuncAttributeMaxDynamicSharedMemorySize, 2 * T2_AT + 2 * T2_BT);
+      (void)hipFuncSetAttribute(
+          reinterpret_cast<const void*>(&gemm_tn_core256<true>),
+          hipFuncAttributeMaxDynamicSharedMemorySize, 2 * T2_AT + 2 * T2_BT);
+    }
+  }
+  return v;
+}
+
 int launch_gemm_tn(const void* A, const void* B, void* C_bf16, float* C_f32,
                    const float* bias, int M, int N, int K, long lda, long ldb,
                    int act, float slope, float* bn_part, hipStream_t s) {
+  if (t2_enabled() && C_bf16 != nullptr && bn_part == nullptr &&
+      M >= 8 * T2_BM) {
+    dim3 grid(ceil_div(M, T2_BM), ceil_div(N, T2_BN));
+    ConvGather dummy{};
+    hipLaunchKernelGGL((gemm_tn_core256<false>), grid, dim3(512),
+                       2 * T2_AT + 2 * T2_BT, s, (const unsigned short*)A,
+                       (const unsigned short*)B, (unsigned short*)C_bf16,
+                       bias, M, N, K, lda, ldb, act, slope, dummy, nullptr);
+    return (int)grid.x;
+  }
   dim3 grid(ceil_div(M, TN_BM), ceil_div(N, TN_BN));
   ConvGather dummy{};
   hipLaunchKernelGGL((gemm_tn_core<false>), grid, dim3(256), 0, s,
@@ -592,6 +806,15 @@ int launch_gemm_tn_gather(const void* img, const void* B, void* C_bf16,
                           int act, float slope, ConvGather ga,
                           const void* zero_page, float* bn_part,
                           hipStream_t s) {
+  if (t2_enabled() && bn_part == nullptr && ga.mode != 2 && M >= 8 * T2_BM) {
+    dim3 grid(ceil_div(M, T2_BM), ceil_div(N, T2_BN));
+    hipLaunchKernelGGL((gemm_tn_core256<true>), grid, dim3(512),
+                       2 * T2_AT + 2 * T2_BT, s, (const unsigned short*)img,
+                       (const unsigned short*)B, (unsigned short*)C_bf16,
+                       bias, M, N, K, 0, ldb, act, slope, ga,
+                       (const unsigned short*)zero_page);
+    return (int)grid.x;
+  }
   dim3 grid(ceil_div(M, TN_BM), ceil_div(N, TN_BN));
   hipLaunchKernelGGL((gemm_tn_core<true>), grid, dim3(256), 0, s,
                      (const unsigned short*)img, (const unsigned short*)B,
