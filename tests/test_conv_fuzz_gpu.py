@@ -98,6 +98,44 @@ def test_parity_strided_path_equivalence(monkeypatch):
         assert float(err) < 0.02, float(err)
 
 
+def test_tn256_path_equivalence():
+    """The env-gated 256-row TN tile (GDLJ_TN256=1) must match PyTorch.
+    The flag is latched per-process at first launch, so this runs in a
+    fresh subprocess."""
+    import os
+    import subprocess
+    import sys
+
+    script = (
+        "import torch; import torch.nn.functional as F;\n"
+        "from gan_deeplearning4j_amd.ops import gpu_ops\n"
+        "torch.manual_seed(5)\n"
+        "x = (torch.randn(32, 16, 40, 40) * 0.5).to('cuda', torch.bfloat16)\n"
+        "# M = 32*20*20 = 12800 >= the 256-tile's 8-tile row gate\n"
+        "w = (torch.randn(32, 16, 4, 4) * 0.2).to('cuda', torch.bfloat16)\n"
+        "b = torch.randn(32).to('cuda', torch.bfloat16)\n"
+        "x.requires_grad_(True); w.requires_grad_(True)\n"
+        "y = gpu_ops.conv2d(x, w, b, 2, 1, 'lrelu', 0.2)\n"
+        "y.backward(torch.ones_like(y))\n"
+        "xr = x.detach().float().cpu().requires_grad_(True)\n"
+        "wr = w.detach().float().cpu().requires_grad_(True)\n"
+        "yr = F.leaky_relu(F.conv2d(xr, wr, b.float().cpu(), 2, 1), 0.2)\n"
+        "yr.backward(torch.ones_like(yr))\n"
+        "def rel(a, r): return ((a.float().cpu()-r).abs().max()"
+        "/r.abs().max().clamp_min(1e-5)).item()\n"
+        "assert rel(y, yr.detach()) < 0.05, rel(y, yr.detach())\n"
+        "assert rel(x.grad, xr.grad) < 0.06\n"
+        "assert rel(w.grad, wr.grad) < 0.06\n"
+        "print('TN256 OK')\n"
+    )
+    env = dict(os.environ)
+    env["GDLJ_TN256"] = "1"
+    r = subprocess.run([sys.executable, "-c", script], env=env,
+                       capture_output=True, text=True, timeout=240)
+    assert r.returncode == 0 and "TN256 OK" in r.stdout, (
+        r.stdout[-2000:], r.stderr[-2000:])
+
+
 @pytest.mark.parametrize("case", CASES)
 def test_conv_transpose2d_fuzz(case):
     from gan_deeplearning4j_amd.ops import gpu_ops
