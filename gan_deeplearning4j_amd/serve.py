@@ -35,6 +35,7 @@ try:  # web deps are optional for the core framework
         n: int = Field(default=1, ge=1)
         seed: Optional[int] = None
         format: str = Field(default="array", pattern="^(array|png_base64)$")
+        model: Optional[str] = None  # named generator (default: "default")
 
     class DiscriminateReq(BaseModel):
         inputs: list
@@ -169,19 +170,32 @@ def create_app(generator: Optional[ComputationGraph] = None,
                discriminator: Optional[ComputationGraph] = None,
                device: Optional[torch.device] = None,
                max_batch: int = 64,
-               batch_window_ms: float = 2.0):
-    """Build the FastAPI app. Models may be None (their routes 404)."""
+               batch_window_ms: float = 2.0,
+               generators: Optional[dict] = None):
+    """Build the FastAPI app. Models may be None (their routes 404).
+
+    generators: optional {name: ComputationGraph} registry of ADDITIONAL
+    generators (multi-model residency — 288 GB HBM holds thousands of
+    GAN-scale models; each gets its own captured graph + micro-batcher).
+    `generator` is registered as "default".
+    """
     from fastapi import FastAPI, HTTPException
 
     if device is None:
         device = torch.device("cuda:0" if torch.cuda.is_available() else "cpu")
     dtype = torch.bfloat16 if device.type == "cuda" else torch.float32
 
-    gen_ep = (_Endpoint(generator, device, dtype, max_batch)
-              if generator is not None else None)
+    gens: dict = {}
+    if generator is not None:
+        gens["default"] = generator
+    for name, g in (generators or {}).items():
+        gens[name] = g
+    gen_eps = {name: _Endpoint(g, device, dtype, max_batch)
+               for name, g in gens.items()}
+    gen_mbs = {name: _MicroBatcher(ep, batch_window_ms)
+               for name, ep in gen_eps.items()}
     dis_ep = (_Endpoint(discriminator, device, dtype, max_batch)
               if discriminator is not None else None)
-    gen_mb = _MicroBatcher(gen_ep, batch_window_ms) if gen_ep else None
 
     app = FastAPI(title="gan_deeplearning4j_amd serving")
 
@@ -190,16 +204,22 @@ def create_app(generator: Optional[ComputationGraph] = None,
         return {
             "status": "ok",
             "device": str(device),
-            "generator": gen_ep is not None,
+            "generator": len(gen_eps) > 0,
             "discriminator": dis_ep is not None,
         }
+
+    @app.get("/models")
+    def models():
+        return {"generators": sorted(gen_eps.keys()),
+                "discriminator": dis_ep is not None}
 
     @app.get("/info")
     def info():
         out = {}
-        if gen_ep is not None:
-            out["generator"] = {"n_params": gen_ep.graph.n_params(),
-                                "inputs": gen_ep.graph.input_names}
+        for name, ep in gen_eps.items():
+            key = "generator" if name == "default" else f"generator:{name}"
+            out[key] = {"n_params": ep.graph.n_params(),
+                        "inputs": ep.graph.input_names}
         if dis_ep is not None:
             out["discriminator"] = {"n_params": dis_ep.graph.n_params(),
                                     "inputs": dis_ep.graph.input_names}
@@ -207,8 +227,11 @@ def create_app(generator: Optional[ComputationGraph] = None,
 
     @app.post("/generate")
     async def generate(req: GenerateReq):
+        name = req.model or "default"
+        gen_ep = gen_eps.get(name)
         if gen_ep is None:
-            raise HTTPException(404, "no generator loaded")
+            raise HTTPException(404, f"no generator {name!r} loaded")
+        gen_mb = gen_mbs[name]
         if req.n > max_batch:
             raise HTTPException(400, f"n > max_batch ({max_batch})")
         it = gen_ep.graph.input_types[gen_ep.graph.input_names[0]]

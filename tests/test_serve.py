@@ -66,6 +66,27 @@ def test_discriminate(client):
     assert all(0.0 <= v <= 1.0 for v in scores)
 
 
+def test_multi_model_registry():
+    """Named generators serve side by side (multi-model residency)."""
+    from gan_deeplearning4j_amd.serve import create_app as mk
+
+    cfg28 = preset("dcgan28")
+    g28, _ = build_dcgan(cfg28)
+    cfg64 = preset("dcgan64")
+    g64, _ = build_dcgan(cfg64)
+    app = mk(generator=g28, generators={"hires": g64},
+             device=torch.device("cpu"), max_batch=4)
+    c = TestClient(app)
+    assert c.get("/models").json()["generators"] == ["default", "hires"]
+    s28 = torch.tensor(c.post("/generate", json={"n": 2}).json()["samples"])
+    assert s28.shape == (2, 1, 28, 28)
+    s64 = torch.tensor(c.post("/generate", json={
+        "n": 2, "model": "hires"}).json()["samples"])
+    assert s64.shape == (2, 3, 64, 64)
+    assert c.post("/generate", json={"n": 1, "model": "nope"}
+                  ).status_code == 404
+
+
 def test_microbatcher_coalesces_concurrent_requests():
     """Concurrent submits within the window share ONE padded replay."""
     import asyncio
