@@ -273,12 +273,23 @@ def main(argv: Optional[list[str]] = None):
     ap = argparse.ArgumentParser(description=__doc__)
     ap.add_argument("--generator", type=str, default=None)
     ap.add_argument("--discriminator", type=str, default=None)
+    ap.add_argument("--add-generator", action="append", default=[],
+                    metavar="NAME=PATH",
+                    help="register an additional named generator")
     ap.add_argument("--host", type=str, default="127.0.0.1")
     ap.add_argument("--port", type=int, default=8000)
     ap.add_argument("--max-batch", type=int, default=64)
     args = ap.parse_args(argv)
-    if args.generator is None and args.discriminator is None:
-        ap.error("load at least one of --generator/--discriminator")
+    if args.generator is None and args.discriminator is None \
+            and not args.add_generator:
+        ap.error("load at least one model (--generator / --discriminator "
+                 "/ --add-generator)")
+    extra = {}
+    for spec in args.add_generator:
+        name, _, path = spec.partition("=")
+        if not name or not path:
+            ap.error(f"--add-generator wants NAME=PATH, got {spec!r}")
+        extra[name] = load_graph(path)
     import uvicorn
 
     app = create_app(
@@ -286,6 +297,7 @@ def main(argv: Optional[list[str]] = None):
         discriminator=(load_graph(args.discriminator)
                        if args.discriminator else None),
         max_batch=args.max_batch,
+        generators=extra,
     )
     uvicorn.run(app, host=args.host, port=args.port)
 
