@@ -55,12 +55,28 @@ class GradReducer:
         modules: Iterable[torch.nn.Module],
         bucket_cap_mb: int = 40,
         local_steps: int = 1,
+        sync_buffers: bool = True,
+        buffer_modules: Optional[Iterable[torch.nn.Module]] = None,
     ):
+        modules = list(modules)
         self.params = [
             p
             for m in modules
             for p in m.parameters()
             if p.requires_grad
+        ]
+        # BatchNorm running statistics update from PER-RANK batch stats in
+        # forward; without syncing them the replicas' eval paths (and
+        # checkpoints — DL4J's flat params() vector includes mean/var)
+        # drift apart. They are a few KB per model: average them every
+        # reduction round. buffer_modules widens the set beyond `modules`
+        # (e.g. the step's LAST reducer also owns the other model's stats,
+        # which keep updating in later phases of the same step).
+        self.sync_buffers = sync_buffers
+        self._buffers = [
+            b for m in (list(buffer_modules) if buffer_modules is not None
+                        else modules)
+            for b in m.buffers() if b.dtype.is_floating_point
         ]
         self.local_steps = max(1, local_steps)
         self._step = 0
@@ -146,6 +162,12 @@ class GradReducer:
                         (b.flat[off : off + n] / world).to(p.grad.dtype)
                     )
                     off += n
+        if self.sync_buffers and self._buffers:
+            with torch.no_grad():
+                for buf in self._buffers:
+                    t = buf.detach().float()
+                    dist.all_reduce(t, op=dist.ReduceOp.SUM)
+                    buf.copy_((t / world).to(buf.dtype))
         self._active = False
 
 

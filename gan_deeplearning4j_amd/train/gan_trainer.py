@@ -85,8 +85,13 @@ class GanTrainer:
         broadcast_parameters(self.gen)
         broadcast_parameters(self.dis)
         cap = bucket_cap_mb or cfg.parallel.bucket_cap_mb
-        self.g_reducer = GradReducer([self.gen], cap, cfg.parallel.local_steps)
-        self.d_reducer = GradReducer([self.dis], cap, cfg.parallel.local_steps)
+        # the G-step reducer is the LAST to finish each step, so it also
+        # averages BOTH models' BN running stats (D's update again during
+        # the G-step forward, after d_reducer.finish)
+        self.g_reducer = GradReducer([self.gen], cap, cfg.parallel.local_steps,
+                                     buffer_modules=[self.gen, self.dis])
+        self.d_reducer = GradReducer([self.dis], cap, cfg.parallel.local_steps,
+                                     sync_buffers=False)
         self.z_size = cfg.model.z_size
         from ..parallel.launch import get_rank
 

@@ -39,20 +39,22 @@ def _worker(rank, world, fn_name, tmpdir, q):
 
 def _run_mp(fn_name, world=2, tmpdir=""):
     ctx = mp.get_context("spawn")
-    q = ctx.SimpleQueue()
+    q = ctx.Queue()
     procs = [
         ctx.Process(target=_worker, args=(r, world, fn_name, tmpdir, q))
         for r in range(world)
     ]
     for p in procs:
         p.start()
-    for p in procs:
-        p.join(180)
-        assert p.exitcode == 0, f"worker failed: exit {p.exitcode}"
+    # drain BEFORE join: large results (full param vectors) overflow the
+    # pipe buffer, and a child blocked on q.put never exits
     out = {}
-    while not q.empty():
-        r, v = q.get()
+    for _ in procs:
+        r, v = q.get(timeout=240)
         out[r] = v
+    for p in procs:
+        p.join(60)
+        assert p.exitcode == 0, f"worker failed: exit {p.exitcode}"
     return out
 
 
@@ -85,6 +87,24 @@ def _body_trainer_sync(rank, world):
         tr.step(torch.rand(8, cfg.data.num_features))
     # replicas must stay bit-identical after synced updates
     return tr.dis.params_flat()
+
+
+def _body_trainer_sync_conv(rank, world):
+    """Conv-model (dcgan28) replicas stay synced — the same path the
+    driver's multi-GPU scale bench exercises, minus RCCL."""
+    from gan_deeplearning4j_amd.config import preset
+    from gan_deeplearning4j_amd.models import build_dcgan
+    from gan_deeplearning4j_amd.train import GanTrainer
+
+    cfg = preset("dcgan28")
+    cfg.train.use_gpu = False
+    gen, dis = build_dcgan(cfg)
+    tr = GanTrainer(gen, dis, cfg, device=torch.device("cpu"))
+    torch.manual_seed(700 + rank)  # different data per rank
+    for _ in range(2):
+        real = torch.rand(4, 1, 28, 28) * 2 - 1
+        tr.step(real)
+    return [tr.dis.params_flat(), tr.gen.params_flat()]
 
 
 def _body_param_averaging(rank, world):
@@ -161,6 +181,13 @@ def test_grad_reducer_averages_across_ranks():
 def test_trainer_replicas_stay_synced():
     out = _run_mp("_body_trainer_sync")
     assert torch.allclose(torch.tensor(out[0]), torch.tensor(out[1]), atol=1e-5)
+
+
+def test_conv_trainer_replicas_stay_synced():
+    out = _run_mp("_body_trainer_sync_conv")
+    for i in range(2):
+        a, b = torch.tensor(out[0][i]), torch.tensor(out[1][i])
+        assert torch.allclose(a, b, atol=1e-5)
 
 
 def test_parameter_averaging():
