@@ -163,11 +163,18 @@ class GradReducer:
                     )
                     off += n
         if self.sync_buffers and self._buffers:
+            # one flat collective (a dozen KB-sized all-reduces would be
+            # pure latency on xGMI)
             with torch.no_grad():
-                for buf in self._buffers:
-                    t = buf.detach().float()
-                    dist.all_reduce(t, op=dist.ReduceOp.SUM)
-                    buf.copy_((t / world).to(buf.dtype))
+                flat = torch.cat([b.detach().reshape(-1).float()
+                                  for b in self._buffers])
+                dist.all_reduce(flat, op=dist.ReduceOp.SUM)
+                flat /= world
+                off = 0
+                for b in self._buffers:
+                    n = b.numel()
+                    b.copy_(flat[off:off + n].view_as(b).to(b.dtype))
+                    off += n
         self._active = False
 
 
