@@ -3,7 +3,6 @@ from .reference_protocol import (
     build_frozen_generator,
     build_stacked_gan,
     build_transfer_classifier,
-    DIS_SYNC_KEYS,
     GAN_TO_GEN_SYNC,
     DIS_TO_GAN_SYNC,
     DIS_TO_CV_SYNC,
@@ -18,7 +17,6 @@ __all__ = [
     "build_transfer_classifier",
     "build_dcgan",
     "build_mlp_gan",
-    "DIS_SYNC_KEYS",
     "GAN_TO_GEN_SYNC",
     "DIS_TO_GAN_SYNC",
     "DIS_TO_CV_SYNC",

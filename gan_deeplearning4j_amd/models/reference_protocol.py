@@ -163,14 +163,6 @@ def build_transfer_classifier(dis: ComputationGraph,
 
 # ---------------------------------------------------------------------------
 # Weight-sync tables for the manual per-tensor copies of the reference loop.
-# (dis layer name, param keys) in copy order.
-DIS_SYNC_KEYS = [
-    ("batch_layer", ("gamma", "beta", "mean", "var")),
-    ("conv2d_layer", ("W", "b")),
-    ("conv2d_layer2", ("W", "b")),
-    ("dense_layer", ("W", "b")),
-    ("output_layer", ("W", "b")),
-]
 
 # dis -> gan frozen-D twin (Java:429-460): 13 tensor-bearing layer pairs
 DIS_TO_GAN_SYNC = [
