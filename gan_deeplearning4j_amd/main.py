@@ -1,7 +1,7 @@
 """CLI entry point (the reference's `main` / `GAN()`, Java:94-101).
 
     python -m gan_deeplearning4j_amd.main [--config cfg.yaml] [--protocol
-        reference|fast] [--data-dir DIR] [section.key=value ...]
+        reference|fast] [section.key=value ...]
 
 - protocol=reference: the exact DL4J alternating protocol on CSV data
   (3 graphs + transfer classifier + CSV artifact dumps + 4 zip
@@ -28,7 +28,12 @@ from .train import GanTrainer, MetricsLogger, ReferenceProtocolTrainer
 log = logging.getLogger("gan_deeplearning4j_amd")
 
 
-def _iterators(cfg: GanConfig):
+def _iterators(cfg: GanConfig, rank: int = 0):
+    """Train/test DataSet iterators. rank decorrelates the training
+    shuffle order across data-parallel workers (every rank still sees
+    the full dataset and steps in lockstep — no tail-imbalance deadlock
+    on the gradient collectives; the reference's TrainingMaster
+    partitions exported batches per worker, Java:325-330)."""
     d = cfg.data
     data_dir = Path(d.data_dir)
     train_csv = data_dir / f"{d.dataset_name}_train.csv"
@@ -51,7 +56,8 @@ def _iterators(cfg: GanConfig):
                             seed=cfg.train.seed + 1, **kw)
     train_it = RecordReaderDataSetIterator(
         CSVRecordReader().initialize(train_csv), d.batch_size_per_worker,
-        d.label_index, d.num_classes, shuffle=True, seed=cfg.train.seed)
+        d.label_index, d.num_classes, shuffle=True,
+        seed=cfg.train.seed + 1000 * rank)
     test_it = RecordReaderDataSetIterator(
         CSVRecordReader().initialize(test_csv), d.batch_size_pred,
         d.label_index, d.num_classes)
@@ -73,7 +79,7 @@ def run_fast(cfg: GanConfig):
     else:
         gen, dis = build_dcgan(cfg)
     tr = GanTrainer(gen, dis, cfg, device=device)
-    train_it, _ = _iterators(cfg)
+    train_it, _ = _iterators(cfg, rank)
     metrics = MetricsLogger(cfg.train.out_dir, cfg.train.print_every,
                             is_main())
     m = cfg.model
