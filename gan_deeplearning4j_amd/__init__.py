@@ -16,7 +16,7 @@ for AMD Instinct MI355X (gfx950, CDNA4):
   semantics (reference Java:408-621).
 """
 
-__version__ = "0.1.0"
+__version__ = "1.0.0"
 
 from . import config  # noqa: F401
 
