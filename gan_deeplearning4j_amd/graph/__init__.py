@@ -5,6 +5,7 @@ from .layers import (
     ConvTranspose2dLayer,
     DenseLayer,
     MaxPool2dLayer,
+    MergeVertex,
     OutputLayer,
     Upsampling2dLayer,
     CnnToFeedForwardPreProcessor,
@@ -24,6 +25,7 @@ __all__ = [
     "ConvTranspose2dLayer",
     "BatchNormLayer",
     "MaxPool2dLayer",
+    "MergeVertex",
     "Upsampling2dLayer",
     "ActivationLayer",
     "OutputLayer",

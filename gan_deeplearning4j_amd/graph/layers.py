@@ -318,6 +318,21 @@ class CnnToFeedForwardPreProcessor(BaseLayer):
         return (n, size)
 
 
+class MergeVertex(BaseLayer):
+    """Concatenate inputs along the feature/channel axis (dim 1) — the
+    DL4J MergeVertex. The reference graphs are single-input chains and
+    never use it; included for ComputationGraph API parity and
+    multi-input models (e.g. conditional GANs, models/cgan.py)."""
+
+    def forward(self, *xs):
+        return torch.cat(xs, dim=1)
+
+    def out_shape(self, *in_shapes):
+        first = in_shapes[0]
+        total = sum(s[1] for s in in_shapes)
+        return (first[0], total, *first[2:])
+
+
 class ReshapeVertex(BaseLayer):
     def __init__(self, *shape: int):
         super().__init__()

@@ -40,6 +40,7 @@ _LAYER_FIELDS = {
     "FeedForwardToCnnPreProcessor": ("height", "width", "channels"),
     "CnnToFeedForwardPreProcessor": ("channels_last",),
     "ReshapeVertex": ("shape",),
+    "MergeVertex": (),
 }
 
 _CTOR_ARGS = {
@@ -70,6 +71,7 @@ _CTOR_ARGS = {
     "CnnToFeedForwardPreProcessor": lambda d: L.CnnToFeedForwardPreProcessor(
         d.get("channels_last", False)),
     "ReshapeVertex": lambda d: L.ReshapeVertex(*d["shape"]),
+    "MergeVertex": lambda d: L.MergeVertex(),
 }
 
 
