@@ -56,6 +56,39 @@ def test_cgan_adversarial_step_reduces_loss():
     assert float(loss_d) < first  # D learned something
 
 
+import pytest  # noqa: E402
+
+
+@pytest.mark.gpu
+def test_cgan_gpu_step():
+    """Multi-input graphs run the HIP kernel path end to end (bf16)."""
+    from gan_deeplearning4j_amd.ops.functional import bce_with_logits_loss
+
+    torch.manual_seed(0)
+    cfg, gen, dis = _models()
+    dev = torch.device("cuda:0")
+    gen.to_device(dev, torch.bfloat16)
+    dis.to_device(dev, torch.bfloat16)
+    n, ncls = 64, cfg.data.num_classes
+    real = torch.rand(n, 1, 28, 28, device=dev, dtype=torch.bfloat16)
+    y = torch.eye(ncls)[torch.randint(0, ncls, (n,))].to(dev, torch.bfloat16)
+    z = torch.randn(n, cfg.model.z_size, device=dev, dtype=torch.bfloat16)
+    fake = gen(z, y)
+    assert fake.shape == (n, 1, 28, 28)
+    dis.updater.zero_grad()
+    loss = bce_with_logits_loss(dis(real, y), torch.ones(n, 1, device=dev)) \
+        + bce_with_logits_loss(dis(fake.detach(), y),
+                               torch.zeros(n, 1, device=dev))
+    loss.backward()
+    dis.updater.step()
+    gen.updater.zero_grad()
+    bce_with_logits_loss(dis(fake, y),
+                         torch.ones(n, 1, device=dev)).backward()
+    gen.updater.step()
+    torch.cuda.synchronize()
+    assert torch.isfinite(loss.float())
+
+
 def test_cgan_serialization_roundtrip(tmp_path):
     from gan_deeplearning4j_amd.graph.serialization import ModelSerializer
 
