@@ -757,6 +757,15 @@ __global__ __launch_bounds__(NT_THREADS, 2) void gemm_nt_core(
 // ---------------------------------------------------------------------------
 extern "C" {
 
+// 8-phase deep-pipelined 256x256 TN core (gemm8p.hip) — the default for
+// large bf16 TN shapes; this file's 128-tile kernels remain the fallback
+// (fp32 out, bn_part fusion, small N/K/M).
+int gemm_tn_8p_eligible(int M, int N, int K);
+int launch_gemm_tn_8p(const void* A, const void* B, void* C,
+                      const float* bias, int M, int N, int K, long lda,
+                      long ldb, int act, float slope, int gather,
+                      ConvGather ga, const void* zp, hipStream_t s);
+
 // env-gated 256-row TN tile (see gemm_tn_core256). Opted in with
 // GDLJ_TN256=1; requires bf16 output, no bn_part, gather mode != 2,
 // and enough rows to fill the chip at 1 block/CU.
@@ -780,6 +789,12 @@ static int t2_enabled() {
 int launch_gemm_tn(const void* A, const void* B, void* C_bf16, float* C_f32,
                    const float* bias, int M, int N, int K, long lda, long ldb,
                    int act, float slope, float* bn_part, hipStream_t s) {
+  if (C_bf16 != nullptr && bn_part == nullptr &&
+      gemm_tn_8p_eligible(M, N, K)) {
+    ConvGather dummy{};
+    return launch_gemm_tn_8p(A, B, C_bf16, bias, M, N, K, lda, ldb, act,
+                             slope, 0, dummy, nullptr, s);
+  }
   if (t2_enabled() && C_bf16 != nullptr && bn_part == nullptr &&
       M >= 8 * T2_BM) {
     dim3 grid(ceil_div(M, T2_BM), ceil_div(N, T2_BN));
@@ -806,6 +821,10 @@ int launch_gemm_tn_gather(const void* img, const void* B, void* C_bf16,
                           int act, float slope, ConvGather ga,
                           const void* zero_page, float* bn_part,
                           hipStream_t s) {
+  if (bn_part == nullptr && gemm_tn_8p_eligible(M, N, K)) {
+    return launch_gemm_tn_8p(img, B, C_bf16, bias, M, N, K, 0, ldb, act,
+                             slope, 1, ga, zero_page, s);
+  }
   if (t2_enabled() && bn_part == nullptr && ga.mode != 2 && M >= 8 * T2_BM) {
     dim3 grid(ceil_div(M, T2_BM), ceil_div(N, T2_BN));
     hipLaunchKernelGGL((gemm_tn_core256<true>), grid, dim3(512),

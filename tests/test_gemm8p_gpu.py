@@ -1,0 +1,122 @@
+"""Numerics for the 8-phase 256x256 deep-pipelined TN GEMM (gemm8p.hip).
+
+Shapes here satisfy gemm_tn_8p_eligible (N >= 192, K >= 128 and a multiple
+of 64, >= 128 blocks), so the launcher routes them to the 8p core; every
+result is compared against a plain fp32 PyTorch reference on the same
+bf16-rounded inputs.  Tail cases (M/N not multiples of the 256 tile,
+minimum K = 2 tiles) exercise the clamped idempotent re-staging path.
+"""
+
+import pytest
+import torch
+import torch.nn.functional as F
+
+pytestmark = pytest.mark.gpu
+
+DEV = "cuda:0"
+
+
+def ext():
+    from gan_deeplearning4j_amd.ops.backend import hip_ext
+
+    return hip_ext()
+
+
+def mk(shape, seed=0, scale=1.0):
+    g = torch.Generator().manual_seed(seed)
+    t = torch.randn(*shape, generator=g) * scale
+    return t.to(DEV, torch.bfloat16)
+
+
+def relerr(a, b):
+    a = a.float().cpu()
+    b = b.float().cpu()
+    denom = b.abs().max().clamp_min(1e-6)
+    return ((a - b).abs().max() / denom).item()
+
+
+@pytest.mark.parametrize(
+    "m,n,k",
+    [
+        (32768, 256, 256),   # exact tiles
+        (32700, 320, 128),   # M tail, N tail, minimum K (2 tiles)
+        (32768, 200, 192),   # N < tile with non-multiple-of-8... (200 % 8 == 0)
+        (33000, 512, 448),   # M tail, odd K-tile count
+    ],
+)
+def test_gemm8p_plain_shapes(m, n, k):
+    e = ext()
+    A, B = mk((m, k), 1, 0.5), mk((n, k), 2, 0.5)
+    C = e.gemm_tn(A, B, None, 0, 0.0, False)
+    ref = A.float() @ B.float().t()
+    assert relerr(C, ref.cpu()) < 0.02
+
+
+def test_gemm8p_identity_asymmetric():
+    # A = I with an asymmetric B catches operand/output transposes (the
+    # fragment C-write maps are easy to get silently wrong).
+    e = ext()
+    m, n, k = 32768, 256, 256
+    A = torch.zeros(m, k, device=DEV, dtype=torch.bfloat16)
+    A[:k].copy_(torch.eye(k, device=DEV, dtype=torch.bfloat16))
+    B = torch.arange(n * k, device=DEV, dtype=torch.float32)
+    B = ((B % 251) / 251.0 - 0.5).reshape(n, k).to(torch.bfloat16)
+    C = e.gemm_tn(A, B, None, 0, 0.0, False)
+    assert torch.allclose(C[:k].float(), B.t().float(), atol=1e-2)
+    assert C[k:].abs().max().item() == 0
+
+
+def test_gemm8p_bias_act():
+    e = ext()
+    A, B = mk((32768, 256), 3, 0.3), mk((256, 256), 4, 0.3)
+    bias = torch.randn(256, device=DEV)
+    C = e.gemm_tn(A, B, bias, 1, 0.0, False)  # tanh epilogue
+    ref = torch.tanh(A.float() @ B.float().t() + bias.float())
+    assert relerr(C, ref.cpu()) < 0.03
+
+
+def test_gemm8p_conv_fwd_gather():
+    # implicit-GEMM conv forward at an 8p-eligible shape (mode-0 gather)
+    from gan_deeplearning4j_amd.ops import gpu_ops
+
+    N, Cin, H, Cout, R, stride, pad = 32, 64, 32, 256, 5, 1, 2
+    x = mk((N, Cin, H, H), 10, 0.4)
+    w = mk((Cout, Cin, R, R), 11, 0.1)
+    b = torch.randn(Cout, device=DEV, dtype=torch.bfloat16)
+    y = gpu_ops.conv2d(x, w, b, stride, pad, "identity", 0.0)
+    yr = F.conv2d(x.float().cpu(), w.float().cpu(), b.float().cpu(),
+                  stride=stride, padding=pad)
+    assert relerr(y, yr) < 0.03
+
+
+def test_gemm8p_conv_bwd_gather():
+    # strided conv fwd+bwd: dgrad runs the mode-1 transposed gather
+    from gan_deeplearning4j_amd.ops import gpu_ops
+
+    N, Cin, H, Cout, R, stride, pad = 32, 64, 32, 256, 5, 2, 2
+    x = mk((N, Cin, H, H), 12, 0.4).requires_grad_(True)
+    w = mk((Cout, Cin, R, R), 13, 0.1).requires_grad_(True)
+    y = gpu_ops.conv2d(x, w, None, stride, pad, "identity", 0.0)
+    gout = mk(y.shape, 14, 0.3)
+    y.backward(gout)
+
+    xr = x.detach().float().cpu().requires_grad_(True)
+    wr = w.detach().float().cpu().requires_grad_(True)
+    yr = F.conv2d(xr, wr, None, stride=stride, padding=pad)
+    yr.backward(gout.float().cpu())
+    assert relerr(y, yr) < 0.03
+    assert relerr(x.grad, xr.grad) < 0.05
+    assert relerr(w.grad, wr.grad) < 0.05
+
+
+def test_gemm8p_conv_transpose_fwd():
+    # stride-2 transposed conv: parity-class mode-2 gather + output scatter
+    from gan_deeplearning4j_amd.ops import gpu_ops
+
+    N, Cin, H, Cout, R, stride, pad = 32, 512, 16, 256, 4, 2, 1
+    x = mk((N, Cin, H, H), 15, 0.3)
+    w = mk((Cin, Cout, R, R), 16, 0.05)
+    y = gpu_ops.conv_transpose2d(x, w, None, stride, pad, "identity", 0.0)
+    yr = F.conv_transpose2d(x.float().cpu(), w.float().cpu(), None,
+                            stride=stride, padding=pad)
+    assert relerr(y, yr) < 0.03
