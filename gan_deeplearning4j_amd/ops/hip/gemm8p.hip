@@ -261,6 +261,9 @@ __global__ __launch_bounds__(512, 2) void gemm_tn_8p(
   stage_b(0, 1);
   stage_a(0, 2);
   stage_b(0, 3);
+  // ordering fence: tile-0 glds must be OLDER than tile-1's so the
+  // counted vmcnt below guards exactly the tile-1 pair
+  asm volatile("" ::: "memory");
   stage_a(1, 0);
   stage_b(1, 1);
   asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
@@ -355,9 +358,9 @@ __global__ __launch_bounds__(512, 2) void gemm_tn_8p(
   const int t = threadIdx.x;
   #pragma unroll
   for (int i = 0; i < 16; ++i) {
-    int piece = i * 512 + t;  // 8192 16B pieces = 256 rows x 16 segs
-    int row = piece >> 4;
-    int seg = piece & 15;
+    int piece = i * 512 + t;  // 8192 16B pieces = 256 rows x 32 segs
+    int row = piece >> 5;
+    int seg = piece & 31;
     int grow = m0 + row;
     int gcol = n0 + seg * 8;
     if (grow < M && gcol < N) {
