@@ -422,15 +422,16 @@ static int p8_enabled() {
   return v;
 }
 
-// Eligibility: bf16 out, K a multiple of 64 with >= 2 K-tiles, N wide
-// enough that the 256-col tile isn't mostly padding, and enough blocks to
-// fill 256 CUs at one block/CU.
+// Eligibility: bf16 out, K deep enough to amortize the pipeline prologue
+// (>= 4 K-tiles; at K = 128 the 2-block/CU 128-tile kernel wins, measured
+// profiles/gemm8p_ab.md), N wide enough that the 256-col tile isn't mostly
+// padding, and >= 256 blocks so every CU gets work at one block/CU.
 int gemm_tn_8p_eligible(int M, int N, int K) {
   if (!p8_enabled()) return 0;
-  if (K % p8::BK != 0 || K < 2 * p8::BK) return 0;
+  if (K % p8::BK != 0 || K < 4 * p8::BK) return 0;
   if (N < 192) return 0;
   long blocks = (long)ceil_div(M, p8::BM) * ceil_div(N, p8::BN);
-  if (blocks < 128) return 0;
+  if (blocks < 256) return 0;
   return 1;
 }
 

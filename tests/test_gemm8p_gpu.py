@@ -1,10 +1,10 @@
 """Numerics for the 8-phase 256x256 deep-pipelined TN GEMM (gemm8p.hip).
 
-Shapes here satisfy gemm_tn_8p_eligible (N >= 192, K >= 128 and a multiple
-of 64, >= 128 blocks), so the launcher routes them to the 8p core; every
+Shapes here satisfy gemm_tn_8p_eligible (N >= 192, K >= 256 and a multiple
+of 64, >= 256 blocks), so the launcher routes them to the 8p core; every
 result is compared against a plain fp32 PyTorch reference on the same
 bf16-rounded inputs.  Tail cases (M/N not multiples of the 256 tile,
-minimum K = 2 tiles) exercise the clamped idempotent re-staging path.
+minimum eligible K) exercise the clamped idempotent re-staging path.
 """
 
 import pytest
@@ -38,9 +38,9 @@ def relerr(a, b):
 @pytest.mark.parametrize(
     "m,n,k",
     [
-        (32768, 256, 256),   # exact tiles
-        (32700, 320, 128),   # M tail, N tail, minimum K (2 tiles)
-        (32768, 200, 192),   # N < tile with non-multiple-of-8... (200 % 8 == 0)
+        (65536, 256, 256),   # exact tiles, minimum eligible K (4 tiles)
+        (65400, 320, 256),   # M tail + N tail
+        (65536, 200, 256),   # N < one tile
         (33000, 512, 448),   # M tail, odd K-tile count
     ],
 )
@@ -56,7 +56,7 @@ def test_gemm8p_identity_asymmetric():
     # A = I with an asymmetric B catches operand/output transposes (the
     # fragment C-write maps are easy to get silently wrong).
     e = ext()
-    m, n, k = 32768, 256, 256
+    m, n, k = 65536, 256, 256
     A = torch.zeros(m, k, device=DEV, dtype=torch.bfloat16)
     A[:k].copy_(torch.eye(k, device=DEV, dtype=torch.bfloat16))
     B = torch.arange(n * k, device=DEV, dtype=torch.float32)
@@ -68,7 +68,7 @@ def test_gemm8p_identity_asymmetric():
 
 def test_gemm8p_bias_act():
     e = ext()
-    A, B = mk((32768, 256), 3, 0.3), mk((256, 256), 4, 0.3)
+    A, B = mk((65536, 256), 3, 0.3), mk((256, 256), 4, 0.3)
     bias = torch.randn(256, device=DEV)
     C = e.gemm_tn(A, B, bias, 1, 0.0, False)  # tanh epilogue
     ref = torch.tanh(A.float() @ B.float().t() + bias.float())
@@ -79,7 +79,7 @@ def test_gemm8p_conv_fwd_gather():
     # implicit-GEMM conv forward at an 8p-eligible shape (mode-0 gather)
     from gan_deeplearning4j_amd.ops import gpu_ops
 
-    N, Cin, H, Cout, R, stride, pad = 32, 64, 32, 256, 5, 1, 2
+    N, Cin, H, Cout, R, stride, pad = 64, 64, 32, 256, 5, 1, 2
     x = mk((N, Cin, H, H), 10, 0.4)
     w = mk((Cout, Cin, R, R), 11, 0.1)
     b = torch.randn(Cout, device=DEV, dtype=torch.bfloat16)
@@ -113,7 +113,7 @@ def test_gemm8p_conv_transpose_fwd():
     # stride-2 transposed conv: parity-class mode-2 gather + output scatter
     from gan_deeplearning4j_amd.ops import gpu_ops
 
-    N, Cin, H, Cout, R, stride, pad = 32, 512, 16, 256, 4, 2, 1
+    N, Cin, H, Cout, R, stride, pad = 256, 512, 16, 256, 4, 2, 1
     x = mk((N, Cin, H, H), 15, 0.3)
     w = mk((Cin, Cout, R, R), 16, 0.05)
     y = gpu_ops.conv_transpose2d(x, w, None, stride, pad, "identity", 0.0)
