@@ -57,6 +57,7 @@ class GradReducer:
         local_steps: int = 1,
         sync_buffers: bool = True,
         buffer_modules: Optional[Iterable[torch.nn.Module]] = None,
+        comm: str = "fp32",
     ):
         modules = list(modules)
         self.params = [
@@ -85,12 +86,20 @@ class GradReducer:
         self.buckets: list[_Bucket] = []
         self._param_bucket: dict[int, _Bucket] = {}
         if self.enabled:
-            # communicate in the gradients' own dtype over RCCL (bf16
-            # halves xGMI bytes; all-reduce of tens-of-MB GAN gradients is
-            # latency/bandwidth-bound); gloo lacks bf16 -> fp32 there
+            # Gradient sums communicate in fp32 by default: an 8-rank bf16
+            # ring-SUM loses ~log2(8) mantissa bits on every gradient, and
+            # at these sizes (GAN grads total 16-60 MB) the fp32 all-reduce
+            # costs ~0.4 ms overlapped with backward — noise at 70 ms
+            # steps. bf16 comms remain an explicit opt-in for
+            # bandwidth-bound regimes (comm="bf16" or GDLJ_COMM_DTYPE=bf16;
+            # gloo lacks bf16 so it always uses fp32).
+            import os as _os
+
+            comm = _os.environ.get("GDLJ_COMM_DTYPE", comm)
             self.comm_dtype = (
                 torch.bfloat16
-                if dist.get_backend() == "nccl"
+                if comm == "bf16"
+                and dist.get_backend() == "nccl"
                 and all(p.dtype == torch.bfloat16 for p in self.params)
                 else torch.float32
             )
