@@ -270,14 +270,6 @@ __global__ __launch_bounds__(512, 2) void gemm_tn_8p(
   __builtin_amdgcn_s_barrier();
 
   bf16x8 a[4][2], blo[2][2], bhi[2][2];
-  // tile 0's B(ni0-1) fragments load here so every steady-state phase
-  // issues <= 8 ds_reads (phase 4 reloads blo for the NEXT tile in place:
-  // the registers are dead after phase 3's quadrant)
-  #pragma unroll
-  for (int ni = 0; ni < 2; ++ni) {
-    blo[ni][0] = frag<SWZ>(bbuf(0), wc * 4 + ni, 0, fr, fq);
-    blo[ni][1] = frag<SWZ>(bbuf(0), wc * 4 + ni, 1, fr, fq);
-  }
 
 #define P8_QUAD(MIH, NIH, BREG)                                              \
   _Pragma("unroll") for (int kc = 0; kc < 2; ++kc)                           \
@@ -299,12 +291,16 @@ __global__ __launch_bounds__(512, 2) void gemm_tn_8p(
   for (int t = 0; t < ntiles; ++t) {
     const char* Ab = abuf(t);
     const char* Bb = bbuf(t);
-    // phase 1: A(mi0-3) reads (blo was loaded in the previous phase 4);
-    // stage R2(t+1)
+    // phase 1: A(mi0-3) + B(ni0-1) reads; stage R2(t+1)
     #pragma unroll
     for (int mi = 0; mi < 4; ++mi) {
       a[mi][0] = frag<SWZ>(Ab, wr * 8 + mi, 0, fr, fq);
       a[mi][1] = frag<SWZ>(Ab, wr * 8 + mi, 1, fr, fq);
+    }
+    #pragma unroll
+    for (int ni = 0; ni < 2; ++ni) {
+      blo[ni][0] = frag<SWZ>(Bb, wc * 4 + ni, 0, fr, fq);
+      blo[ni][1] = frag<SWZ>(Bb, wc * 4 + ni, 1, fr, fq);
     }
     stage_a(t + 1, 2);
     P8_BAR_MFMA(0, 0, blo);
@@ -327,21 +323,10 @@ __global__ __launch_bounds__(512, 2) void gemm_tn_8p(
     stage_a(t + 2, 0);
     P8_BAR_MFMA(1, 0, blo);
 
-    // phase 4: stage R1(t+2); the tile's single counted wait; after the
-    // barrier (tile t+1 fully landed) reload blo with t+1's B(ni0-1) —
-    // those reads' latency hides under this phase's MFMAs and phase 1
-    // starts with its operands in flight
+    // phase 4: no reads; stage R1(t+2); the tile's single counted wait
     stage_b(t + 2, 1);
     asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
     __builtin_amdgcn_s_barrier();
-    if (t + 1 < ntiles) {
-      const char* Bn = bbuf(t + 1);
-      #pragma unroll
-      for (int ni = 0; ni < 2; ++ni) {
-        blo[ni][0] = frag<SWZ>(Bn, wc * 4 + ni, 0, fr, fq);
-        blo[ni][1] = frag<SWZ>(Bn, wc * 4 + ni, 1, fr, fq);
-      }
-    }
     __builtin_amdgcn_s_setprio(1);
     P8_QUAD(1, 1, bhi);
     __builtin_amdgcn_s_setprio(0);
