@@ -205,7 +205,11 @@ DEV_INLINE bf16x8 frag(const char* opb, int rowblk, int kc, int fr, int fq) {
 
 }  // namespace p8
 
-template <bool GATHER_A, int SWZ>
+// DEEP: two counted waits per K-tile (phase-1 tail + phase 4) at depth
+// vmcnt(6) = 3 regions in flight, instead of one vmcnt(4) wait at depth 2.
+// The extra wait is where each next-needed region's landing is forced at
+// the latest legal point, so 50% more staging latency is hidden.
+template <bool GATHER_A, int SWZ, bool DEEP>
 __global__ __launch_bounds__(512, 2) void gemm_tn_8p(
     const unsigned short* __restrict__ A, const unsigned short* __restrict__ B,
     unsigned short* __restrict__ C, const float* __restrict__ bias, int M,
@@ -266,7 +270,10 @@ __global__ __launch_bounds__(512, 2) void gemm_tn_8p(
   asm volatile("" ::: "memory");
   stage_a(1, 0);
   stage_b(1, 1);
-  asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+  if (DEEP)
+    asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
+  else
+    asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
   __builtin_amdgcn_s_barrier();
 
   bf16x8 a[4][2], blo[2][2], bhi[2][2];
@@ -279,13 +286,14 @@ __global__ __launch_bounds__(512, 2) void gemm_tn_8p(
       __builtin_amdgcn_mfma_f32_16x16x32_bf16(                               \
           a[mi][kc], BREG[ni][kc], acc[(MIH)*4 + mi][(NIH)*2 + ni], 0, 0, 0)
 
-#define P8_BAR_MFMA(MIH, NIH, BREG)                                          \
+#define P8_BAR_MFMA(MIH, NIH, BREG, TAILWAIT)                                \
   __builtin_amdgcn_s_barrier();                                              \
   asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");                         \
   __builtin_amdgcn_sched_barrier(0);                                         \
   __builtin_amdgcn_s_setprio(1);                                             \
   P8_QUAD(MIH, NIH, BREG);                                                   \
   __builtin_amdgcn_s_setprio(0);                                             \
+  TAILWAIT;                                                                  \
   __builtin_amdgcn_s_barrier()
 
   for (int t = 0; t < ntiles; ++t) {
@@ -303,7 +311,9 @@ __global__ __launch_bounds__(512, 2) void gemm_tn_8p(
       blo[ni][1] = frag<SWZ>(Bb, wc * 4 + ni, 1, fr, fq);
     }
     stage_a(t + 1, 2);
-    P8_BAR_MFMA(0, 0, blo);
+    // DEEP: force R3(t) (phase-2's bhi source) landed at this phase's tail
+    P8_BAR_MFMA(0, 0, blo,
+                if (DEEP) asm volatile("s_waitcnt vmcnt(6)" ::: "memory"));
 
     // phase 2: B(ni2-3) reads; stage R3(t+1)
     #pragma unroll
@@ -312,7 +322,7 @@ __global__ __launch_bounds__(512, 2) void gemm_tn_8p(
       bhi[ni][1] = frag<SWZ>(Bb, wc * 4 + 2 + ni, 1, fr, fq);
     }
     stage_b(t + 1, 3);
-    P8_BAR_MFMA(0, 1, bhi);
+    P8_BAR_MFMA(0, 1, bhi, );
 
     // phase 3: A(mi4-7) reads; stage R0(t+2)
     #pragma unroll
@@ -321,11 +331,14 @@ __global__ __launch_bounds__(512, 2) void gemm_tn_8p(
       a[mi][1] = frag<SWZ>(Ab, wr * 8 + 4 + mi, 1, fr, fq);
     }
     stage_a(t + 2, 0);
-    P8_BAR_MFMA(1, 0, blo);
+    P8_BAR_MFMA(1, 0, blo, );
 
     // phase 4: no reads; stage R1(t+2); the tile's single counted wait
     stage_b(t + 2, 1);
-    asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+    if (DEEP)
+      asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
+    else
+      asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
     __builtin_amdgcn_s_barrier();
     __builtin_amdgcn_s_setprio(1);
     P8_QUAD(1, 1, bhi);
@@ -399,6 +412,15 @@ static int p8_swz_mode() {
   return v;
 }
 
+static int p8_deep() {
+  static int v = -1;
+  if (v < 0) {
+    const char* e = getenv("GDLJ_8P_DEEP");
+    v = (e != nullptr && e[0] == '1') ? 1 : 0;
+  }
+  return v;
+}
+
 static int p8_enabled() {
   static int v = -1;
   if (v < 0) {
@@ -410,12 +432,14 @@ static int p8_enabled() {
         (void)hipFuncSetAttribute(reinterpret_cast<const void*>(&KF),        \
                                   hipFuncAttributeMaxDynamicSharedMemorySize,\
                                   p8::LDS_B)
-      P8_SETATTR((gemm_tn_8p<false, 0>));
-      P8_SETATTR((gemm_tn_8p<false, 1>));
-      P8_SETATTR((gemm_tn_8p<false, 2>));
-      P8_SETATTR((gemm_tn_8p<true, 0>));
-      P8_SETATTR((gemm_tn_8p<true, 1>));
-      P8_SETATTR((gemm_tn_8p<true, 2>));
+      P8_SETATTR((gemm_tn_8p<false, 0, false>));
+      P8_SETATTR((gemm_tn_8p<false, 1, false>));
+      P8_SETATTR((gemm_tn_8p<false, 2, false>));
+      P8_SETATTR((gemm_tn_8p<true, 0, false>));
+      P8_SETATTR((gemm_tn_8p<true, 1, false>));
+      P8_SETATTR((gemm_tn_8p<true, 2, false>));
+      P8_SETATTR((gemm_tn_8p<false, 2, true>));
+      P8_SETATTR((gemm_tn_8p<true, 2, true>));
       #undef P8_SETATTR
     }
   }
@@ -442,19 +466,22 @@ int launch_gemm_tn_8p(const void* A, const void* B, void* C,
   dim3 grid(ceil_div(M, p8::BM), ceil_div(N, p8::BN));
   dim3 blk(512);
   int swzm = p8_swz_mode();
-  #define P8_LAUNCH(G, S)                                                    \
-    hipLaunchKernelGGL((gemm_tn_8p<G, S>), grid, blk, p8::LDS_B, s,          \
+  #define P8_LAUNCH(G, S, D)                                                 \
+    hipLaunchKernelGGL((gemm_tn_8p<G, S, D>), grid, blk, p8::LDS_B, s,       \
                        (const unsigned short*)A, (const unsigned short*)B,   \
                        (unsigned short*)C, bias, M, N, K, lda, ldb, act,     \
                        slope, ga, (const unsigned short*)zp)
-  if (gather) {
-    if (swzm == 0) P8_LAUNCH(true, 0);
-    else if (swzm == 1) P8_LAUNCH(true, 1);
-    else P8_LAUNCH(true, 2);
+  if (p8_deep()) {
+    if (gather) P8_LAUNCH(true, 2, true);
+    else P8_LAUNCH(false, 2, true);
+  } else if (gather) {
+    if (swzm == 0) P8_LAUNCH(true, 0, false);
+    else if (swzm == 1) P8_LAUNCH(true, 1, false);
+    else P8_LAUNCH(true, 2, false);
   } else {
-    if (swzm == 0) P8_LAUNCH(false, 0);
-    else if (swzm == 1) P8_LAUNCH(false, 1);
-    else P8_LAUNCH(false, 2);
+    if (swzm == 0) P8_LAUNCH(false, 0, false);
+    else if (swzm == 1) P8_LAUNCH(false, 1, false);
+    else P8_LAUNCH(false, 2, false);
   }
   #undef P8_LAUNCH
   return (int)grid.x;
