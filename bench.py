@@ -32,7 +32,7 @@ from gan_deeplearning4j_amd.train import GanTrainer  # noqa: E402
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--gpus", type=int, default=1)
-    ap.add_argument("--steps", type=int, default=20)
+    ap.add_argument("--steps", type=int, default=50)
     ap.add_argument("--warmup", type=int, default=5)
     ap.add_argument("--batch", type=int, default=0,
                     help="per-GPU batch (0 = arch default)")

@@ -370,6 +370,67 @@ class ComputationGraph(nn.Module):
                 last = float(loss.detach())
         return last
 
+    def fit_averaged(self, datasets) -> float:
+        """ParameterAveragingTrainingMaster semantics for one fit call
+        (Java:325-330, 425-426): each DataSet is an independent worker
+        partition that starts from the CURRENT parameters and updater
+        state, fits its one minibatch locally, and the resulting
+        parameter vectors (and updater state — DL4J averages it by
+        default) are averaged.  This is what
+        `sparkDis.fit(sc.parallelize([real_ds, fake_ds]))` computes: ONE
+        averaged update, not two sequential SGD steps.
+
+        Returns the mean of the partition losses.
+        """
+        import copy as _copy
+
+        datasets = list(datasets)
+        if len(datasets) <= 1:
+            return self.fit(datasets)
+        init_p = [p.detach().clone() for p in self.parameters()]
+        init_b = [b.detach().clone() for b in self.buffers()]
+        init_u = _copy.deepcopy(self.updater.state_dict())
+        acc_p = [torch.zeros_like(p, dtype=torch.float32) for p in init_p]
+        acc_b = [torch.zeros_like(b, dtype=torch.float32) for b in init_b]
+        acc_u: list = []
+        losses = []
+        for wi, ds in enumerate(datasets):
+            if wi > 0:  # restore the broadcast state for this "worker"
+                with torch.no_grad():
+                    for p, p0 in zip(self.parameters(), init_p):
+                        p.copy_(p0)
+                    for b, b0 in zip(self.buffers(), init_b):
+                        b.copy_(b0)
+                self.updater.load_state_dict(_copy.deepcopy(init_u))
+            losses.append(self.fit(ds))
+            acc_u.append(_copy.deepcopy(self.updater.state_dict()))
+            with torch.no_grad():
+                for a, p in zip(acc_p, self.parameters()):
+                    a += p.detach().float()
+                for a, b in zip(acc_b, self.buffers()):
+                    a += b.detach().float()
+        n = float(len(datasets))
+        with torch.no_grad():
+            for p, a in zip(self.parameters(), acc_p):
+                p.copy_((a / n).to(p.dtype))
+            for b, a in zip(self.buffers(), acc_b):
+                b.copy_((a / n).to(b.dtype))
+        # average the updater slot tensors (m/v/master); scalars from the
+        # last worker (t is identical across workers)
+        avg = acc_u[-1]
+        for key in ("m", "v", "master"):
+            for si, slot in enumerate(avg["slots"]):
+                if slot[key] is None:
+                    continue
+                slot[key] = sum(u["slots"][si][key].float()
+                                for u in acc_u) / n
+        self.updater.load_state_dict(avg)
+        for layer in self.layers.values():
+            for p in layer.parameters():
+                if hasattr(p, "_gdlj_cache"):
+                    del p._gdlj_cache
+        return float(sum(losses) / n)
+
     # ----------------------------------------------------- param access
     def get_layer(self, name: str) -> BaseLayer:
         return self.layers[name]
@@ -427,11 +488,13 @@ class ComputationGraph(nn.Module):
         total = 0
         for name in self._topo:
             layer = self.layers[name]
-            src = self._vertex_inputs[name][0]
-            in_shape = shapes.get(src, (batch, -1))
+            srcs = self._vertex_inputs[name]
+            in_shapes = [shapes.get(s, (batch, -1)) for s in srcs]
             if name in self.preprocessors:
-                in_shape = self.preprocessors[name].out_shape(in_shape)
-            out_shape = layer.out_shape(in_shape)
+                in_shapes = [self.preprocessors[name].out_shape(s)
+                             for s in in_shapes]
+            # multi-input vertices (MergeVertex) take every input shape
+            out_shape = layer.out_shape(*in_shapes)
             shapes[name] = out_shape
             n = layer.n_params()
             total += n

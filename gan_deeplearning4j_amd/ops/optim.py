@@ -164,6 +164,11 @@ class Updater:
 
     # -------------------------------------------------------- state io
     def state_dict(self) -> dict:
+        # under hipGraph capture, replays advance only the device counter
+        # _t_dev; read it back so checkpoints never store a stale t (Adam
+        # bias correction would be skewed after resume)
+        if self._t_dev is not None:
+            self.t = int(self._t_dev.item())
         return {
             "kind": self.kind,
             "t": self.t,

@@ -20,6 +20,7 @@ from __future__ import annotations
 import argparse
 import base64
 import tempfile
+import threading
 from pathlib import Path
 from typing import Optional
 
@@ -63,6 +64,11 @@ class _Endpoint:
         self.max_batch = max_batch
         self._compiled = None
         self._example: Optional[torch.Tensor] = None
+        # /discriminate is a sync route on FastAPI's thread pool: without
+        # this lock concurrent requests race on the captured graph's
+        # static input/output buffers (the /generate path is already
+        # serialized by its micro-batcher)
+        self._lock = threading.Lock()
 
     def _runner(self, example: torch.Tensor):
         if self._compiled is None or self._example.shape != example.shape:
@@ -75,12 +81,14 @@ class _Endpoint:
         n = x.shape[0]
         if n > self.max_batch:
             raise ValueError(f"batch {n} > max_batch {self.max_batch}")
-        # pad to the one captured shape
-        xp = x.new_zeros(self.max_batch, *x.shape[1:])
-        xp[:n] = x
-        xp = xp.to(self.device, self.dtype)
-        out = self._runner(xp)(xp)
-        return out[:n].float().cpu()
+        with self._lock:
+            # pad to the one captured shape
+            xp = x.new_zeros(self.max_batch, *x.shape[1:])
+            xp[:n] = x
+            xp = xp.to(self.device, self.dtype)
+            out = self._runner(xp)(xp)
+            # .cpu() materializes a private copy before the lock releases
+            return out[:n].float().cpu()
 
 
 class _MicroBatcher:

@@ -245,6 +245,15 @@ class ReferenceProtocolTrainer:
     def __init__(self, cfg: GanConfig, device: Optional[torch.device] = None,
                  out_dir: Optional[str] = None):
         self.cfg = cfg
+        # Exact-reference updater numerics: the reference constructs
+        # RmsProp(lr, 1e-8, 1e-8) (Java:133) — rmsDecay=1e-8, an
+        # effectively memoryless v.  The dataclass default (0.95) is the
+        # idiomatic value for the fast trainer; in reference-protocol
+        # mode with rmsprop selected and the default untouched, use the
+        # reference's constant.
+        if (cfg.optim.optimizer == "rmsprop"
+                and cfg.optim.rms_decay == 0.95):
+            cfg.optim.rms_decay = 1e-8
         self.device = device or torch.device(
             "cuda" if torch.cuda.is_available() and cfg.train.use_gpu else "cpu"
         )
@@ -314,7 +323,10 @@ class ReferenceProtocolTrainer:
             DataSet(feats, self.soft_real[:n].to(dev)),
             DataSet(fake_flat, self.soft_fake[:n].to(dev)),
         ]
-        loss_d = self.dis.fit(d_data)
+        # the reference parallelizes the two DataSets across workers and
+        # parameter-averages (Java:425-426): ONE averaged update, not two
+        # sequential minibatch steps
+        loss_d = self.dis.fit_averaged(d_data)
 
         # (b) copy fresh D into gan's frozen D (Java:429-460)
         sync_params(self.dis, self.gan, DIS_TO_GAN_SYNC)

@@ -1,0 +1,99 @@
+"""Golden loss-trajectory assertions for the exact reference protocol.
+
+VERDICT round-1 item 4c: the old 2-iteration e2e test checked shapes and
+finiteness only — nearly impossible to fail on a numerics regression.
+This test pins the full exact-protocol stack to committed golden VALUES:
+
+- RMSProp with the reference's constructor constants (Java:133
+  `new RmsProp(lr, 1e-8, 1e-8)`: rmsDecay = 1e-8, auto-applied in
+  reference-protocol mode);
+- the D step as a parameter-averaged two-partition fit
+  (`ComputationGraph.fit_averaged`, Java:425-426 semantics), not two
+  sequential updates;
+- label softening drawn once, frozen-copy weight sync, the 6-phase order.
+
+Any change to init, updater math, sync tables, fit semantics or the loss
+kernels moves these values.  Goldens were generated on this image
+(torch 2.10 CPU, fp32) and verified identical across two runs.
+"""
+
+import tempfile
+
+import torch
+
+from gan_deeplearning4j_amd.config import preset
+from gan_deeplearning4j_amd.data.csv_reader import DataSet
+from gan_deeplearning4j_amd.train.gan_trainer import ReferenceProtocolTrainer
+
+GOLDEN = [
+    # (loss_d, loss_g, loss_cv) per iteration; seed 666 init, data seed 4242
+    (0.712352, 1.231502, 2.951957),
+    (0.452114, 1.903526, 3.308198),
+    (0.536285, 0.660316, 2.956998),
+]
+
+
+def test_reference_protocol_golden_losses():
+    cfg = preset("dcgan28")
+    cfg.train.use_gpu = False
+    cfg.optim.optimizer = "rmsprop"
+    cfg.data.batch_size_per_worker = 32
+    with tempfile.TemporaryDirectory() as td:
+        tr = ReferenceProtocolTrainer(cfg, device=torch.device("cpu"),
+                                      out_dir=td)
+        # reference-protocol mode applies the reference's RmsProp
+        # constants (Java:133) when rmsprop is selected
+        assert cfg.optim.rms_decay == 1e-8
+        g = torch.Generator().manual_seed(4242)
+        for want_d, want_g, want_cv in GOLDEN:
+            feats = torch.rand(32, 784, generator=g)
+            labels = torch.nn.functional.one_hot(
+                torch.randint(0, 10, (32,), generator=g), 10).float()
+            out = tr.train_iteration(DataSet(feats, labels))
+            for got, want in ((out["loss_d"], want_d),
+                              (out["loss_g"], want_g),
+                              (out["loss_cv"], want_cv)):
+                assert abs(float(got) - want) < 5e-4, (
+                    f"trajectory drift: got {float(got):.6f}, "
+                    f"golden {want:.6f}")
+
+
+def test_fit_averaged_is_one_averaged_update():
+    """fit_averaged([a, b]) == mean of the two single-partition results
+    (ParameterAveragingTrainingMaster), NOT the sequential two-step fit."""
+    import copy
+
+    from gan_deeplearning4j_amd.config import GanConfig
+    from gan_deeplearning4j_amd.models.reference_protocol import (
+        build_discriminator)
+
+    cfg = GanConfig()
+    cfg.train.use_gpu = False
+    cfg.optim.optimizer = "rmsprop"
+
+    def fresh():
+        torch.manual_seed(1)
+        return build_discriminator(cfg)
+
+    g = torch.Generator().manual_seed(7)
+    a = DataSet(torch.rand(8, 784, generator=g),
+                torch.ones(8, 1))
+    b = DataSet(torch.rand(8, 784, generator=g),
+                torch.zeros(8, 1))
+
+    # expected: average of the two independently-updated param vectors
+    ga_, gb_ = fresh(), fresh()
+    ga_.fit(a)
+    gb_.fit(b)
+    want = (ga_.params_flat() + gb_.params_flat()) / 2
+
+    gavg = fresh()
+    gavg.fit_averaged([a, b])
+    got = gavg.params_flat()
+    assert torch.allclose(got, want, atol=1e-6), (
+        (got - want).abs().max().item())
+
+    # and it differs from the sequential two-step fit
+    gseq = fresh()
+    gseq.fit([a, b])
+    assert not torch.allclose(gseq.params_flat(), want, atol=1e-6)
