@@ -1,4 +1,4 @@
-"""Checkpointing: DL4J ModelSerializer-compatible .zip + native format.
+"""Checkpointing: DL4J ModelSerializer-shaped .zip + native format.
 
 The reference writes all four models each iteration via
 ModelSerializer.writeModel(graph, file, saveUpdater=true) (Java:605-618).
@@ -7,6 +7,17 @@ The zip layout:
     coefficients.bin    - flattened fp32 parameter vector in layer order
                           (per layer: W, b, gamma, beta, mean, var)
     updaterState.bin    - flattened fp32 updater state (saveUpdater=true)
+
+DL4J-compat scope (VERDICT round-1 item 5, documented delta):
+configuration.json is emitted in DL4J 1.0.0-beta3's Jackson layout
+(dl4j_json.py) and a DL4J-style zip restores through the tolerant
+parser there, but byte-compat against a real DL4J install is
+UNVALIDATED — this offline environment has no JVM/DL4J to test against.
+The three-file zip shape and the flat fp32 coefficient ordering match
+the DL4J convention; configuration.native.json (this framework's
+schema) is the authoritative restore source and always round-trips
+exactly. updater.json is a framework extension (DL4J encodes the
+updater layout inside its own binary header).
 
 The reference itself never calls restore (save-only); this framework
 implements BOTH save and resume (SURVEY.md §5 checkpoint/resume).
@@ -145,9 +156,28 @@ class ModelSerializer:
         path = Path(path)
         path.parent.mkdir(parents=True, exist_ok=True)
         params = graph.params_flat().numpy().astype("<f4")
+        # configuration.json follows DL4J 1.0.0-beta3's Jackson layout
+        # where the layer vocabulary maps (see dl4j_json.py's scope
+        # statement: structure-compatible, byte-compat unvalidated
+        # offline); configuration.native.json is this framework's
+        # authoritative schema and always restores exactly.
+        try:
+            from .dl4j_json import to_dl4j_json
+
+            dl4j_conf: Optional[dict] = to_dl4j_json(graph)
+        except TypeError:
+            dl4j_conf = None
         with zipfile.ZipFile(path, "w", zipfile.ZIP_DEFLATED) as zf:
             zf.writestr(
                 "configuration.json",
+                json.dumps(
+                    dl4j_conf if dl4j_conf is not None
+                    else graph_config_dict(graph),
+                    indent=1,
+                ),
+            )
+            zf.writestr(
+                "configuration.native.json",
                 json.dumps(graph_config_dict(graph), indent=1),
             )
             zf.writestr("coefficients.bin", params.tobytes())
@@ -184,8 +214,21 @@ class ModelSerializer:
                                   load_updater: bool = True) -> ComputationGraph:
         path = Path(path)
         with zipfile.ZipFile(path) as zf:
-            conf = json.loads(zf.read("configuration.json"))
-            graph = graph_from_config_dict(conf)
+            names = zf.namelist()
+            if "configuration.native.json" in names:
+                conf = json.loads(zf.read("configuration.native.json"))
+                graph = graph_from_config_dict(conf)
+            else:
+                conf = json.loads(zf.read("configuration.json"))
+                if conf.get("format") == \
+                        "gan_deeplearning4j_amd/ComputationGraph":
+                    graph = graph_from_config_dict(conf)
+                else:
+                    # a zip written by DL4J itself (or by this writer's
+                    # DL4J-style emitter alone)
+                    from .dl4j_json import from_dl4j_json
+
+                    graph = from_dl4j_json(conf)
             vec = torch.from_numpy(
                 np.frombuffer(zf.read("coefficients.bin"), dtype="<f4").copy()
             )
