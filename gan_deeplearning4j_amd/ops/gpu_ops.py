@@ -134,6 +134,59 @@ def set_fp8_conv(enabled: bool) -> None:
     FP8_CONV = bool(enabled)
 
 
+# fp8 backward (dgrad family) rides the same switch; GDLJ_FP8_BWD=0 keeps
+# backward in bf16 for A/B comparisons. wgrad stays bf16: the NT
+# contraction reads both operands k(=np)-major, and glds cannot build the
+# transposed fp8 LDS image (lane-linear dest; no tr16-style transpose
+# read exists for 8-bit fragments at the 16x16x128 MFMA's layout) — a
+# register-staged transpose pass measured as the round-1 parity-dgrad
+# wash makes it a documented non-goal this round.
+def _fp8_bwd_on() -> bool:
+    return FP8_CONV and os.environ.get("GDLJ_FP8_BWD") != "0"
+
+
+def _fp8_state(w: torch.Tensor, role: str, device) -> list:
+    """Per-layer persistent delayed-scaling state (scale, inv, amax,
+    first) attached to the layer's weight tensor (stable identity across
+    steps, unlike activations)."""
+    st = getattr(w, "_gdlj_fp8st", None)
+    if st is None:
+        st = {}
+        w._gdlj_fp8st = st
+    if role not in st:
+        st[role] = [
+            torch.ones(1, dtype=torch.float32, device=device),
+            torch.ones(1, dtype=torch.float32, device=device),
+            torch.zeros(1, dtype=torch.int32, device=device),
+            True,
+        ]
+    return st[role]
+
+
+def _quant_delayed(x: torch.Tensor, w: torch.Tensor, role: str):
+    """Quantize x to e4m3 with the previous step's scale (fused amax
+    accumulation; exact two-pass on first use). Returns (q, inv_scale)."""
+    ext = hip_ext()
+    st = _fp8_state(w, role, x.device)
+    y = ext.fp8_quantize_delayed(x.contiguous(), st[0], st[1], st[2], st[3])
+    st[3] = False
+    return y, st[1]
+
+
+def _pad_k128(t: torch.Tensor) -> torch.Tensor:
+    k = t.shape[1]
+    if k % 128:
+        t = torch.nn.functional.pad(t, (0, 128 - k % 128))
+    return t
+
+
+def _zp8(device) -> torch.Tensor:
+    key = "fp8:" + str(device)
+    if key not in _zero_pages:
+        _zero_pages[key] = torch.zeros(32, dtype=torch.uint8, device=device)
+    return _zero_pages[key]
+
+
 def _zp(device) -> torch.Tensor:
     """16B zero page for gathered global_load_lds (OOB/pad redirect)."""
     key = str(device)
@@ -280,7 +333,7 @@ class _Conv2d(torch.autograd.Function):
                 if b is not None else None)
         stats = None
         if FP8_CONV and C8 % 16 == 0:
-            xq, _, ix = ext.fp8_quantize(xh)
+            xq, ix = _quant_delayed(xh, w, "x")
 
             def build_fp8_pack():
                 # the double-rate fp8 MFMA consumes K in 128-deep tiles
@@ -295,7 +348,7 @@ class _Conv2d(torch.autograd.Function):
                 32, dtype=torch.uint8, device=x.device))
             y2d = ext.conv_fwd_implicit_fp8(xq, wq, bias, ix, iw, zp8, N, H,
                                             W, C8, Ho, Wo, R, S, stride, pad,
-                                            act, slope)
+                                            act, slope, 0)
         else:
             # implicit GEMM: im2col gather fused into the MFMA staging
             # (optionally also emitting the consumer BN's batch statistics)
@@ -363,9 +416,17 @@ class _Conv2d(torch.autograd.Function):
 
                 wd = _packed(w, "dgrad_w", build_wd)
                 dpre_img = dpre8.view(N, Ho, Wo, Ko8)
-                dx2d = ext.conv_fwd_implicit(
-                    dpre_img, wd, None, _zp(dpre.device), N, Ho, Wo, Ko8,
-                    H, W, R, S, stride, pad, 0, 0.0, 1, 0)[0]   # mode 1
+                if _fp8_bwd_on() and Ko8 % 16 == 0:
+                    wdq, _, iwd = _packed(w, "dgrad_w_fp8", lambda: tuple(
+                        ext.fp8_quantize(_pad_k128(wd))))
+                    dq, idy = _quant_delayed(dpre_img, w, "dy")
+                    dx2d = ext.conv_fwd_implicit_fp8(
+                        dq, wdq, None, idy, iwd, _zp8(dpre.device), N, Ho,
+                        Wo, Ko8, H, W, R, S, stride, pad, 0, 0.0, 1)
+                else:
+                    dx2d = ext.conv_fwd_implicit(
+                        dpre_img, wd, None, _zp(dpre.device), N, Ho, Wo,
+                        Ko8, H, W, R, S, stride, pad, 0, 0.0, 1, 0)[0]
                 dx = _as_nchw_view(dx2d.view(N, H, W, C)).to(ctx.dtypes[0])
             elif PARITY_STRIDED:
                 # strided: parity-decomposed gather (each output parity
@@ -396,8 +457,15 @@ class _Conv2d(torch.autograd.Function):
                 rsc8 = R * S * C8
                 wt = _packed(wp, "wt", lambda: _pad_k(
                     wp[:, :rsc8].t().contiguous()))   # [rsc8, kout_pad]
-                dprep = _pad_k(dpre)
-                dcol = ext.gemm_tn(dprep, wt, None, 0, 0.0, False)
+                if _fp8_bwd_on():
+                    wtq, _, iwt = _packed(wp, "wt_fp8", lambda: tuple(
+                        ext.fp8_quantize(_pad_k128(wt))))
+                    dq, idy = _quant_delayed(_pad_k128(_pad_k(dpre)), w,
+                                             "dy")
+                    dcol = ext.gemm_tn_fp8(dq, wtq, idy, iwt, None, 0, 0.0)
+                else:
+                    dprep = _pad_k(dpre)
+                    dcol = ext.gemm_tn(dprep, wt, None, 0, 0.0, False)
                 pact = ctx.prev_act
                 if pact is not None and (
                         C8 != C
@@ -453,13 +521,21 @@ class _ConvTranspose2d(torch.autograd.Function):
             wt = _packed(w, "convt_fwd_w", lambda: _pad_k(
                 _bf(w.detach().permute(1, 2, 3, 0))      # [Cout,R,S,Cin]
                 .reshape(Cout, R * S * Cin)))
-            res = ext.conv_fwd_implicit(xh, wt, bias, _zp(x.device), N, Hi,
-                                        Wi, Cin, Ho, Wo, R, S, stride, pad,
-                                        act, slope, 1,
-                                        1 if emit_stats else 0)   # mode 1
-            y2d = res[0]
-            if len(res) == 3:
-                stats = (res[1], res[2])
+            if FP8_CONV and Cin % 16 == 0 and not emit_stats:
+                wtq, _, iw = _packed(w, "convt_fwd_w_fp8", lambda: tuple(
+                    ext.fp8_quantize(_pad_k128(wt))))
+                xq, ix = _quant_delayed(xh, w, "x")
+                y2d = ext.conv_fwd_implicit_fp8(
+                    xq, wtq, bias, ix, iw, _zp8(x.device), N, Hi, Wi, Cin,
+                    Ho, Wo, R, S, stride, pad, act, slope, 1)
+            else:
+                res = ext.conv_fwd_implicit(xh, wt, bias, _zp(x.device), N,
+                                            Hi, Wi, Cin, Ho, Wo, R, S,
+                                            stride, pad, act, slope, 1,
+                                            1 if emit_stats else 0)  # mode 1
+                y2d = res[0]
+                if len(res) == 3:
+                    stats = (res[1], res[2])
             yh = y2d.view(N, Ho, Wo, Cout)
         elif PARITY_STRIDED:
             # strided: parity-decomposed gathered GEMMs with fused
@@ -504,7 +580,13 @@ class _ConvTranspose2d(torch.autograd.Function):
                 w2a = _packed(w, "w2a", lambda: _pad_k(
                     _bf(w.detach().permute(2, 3, 1, 0))
                     .reshape(R * S * Cout, Cin)))
-            col = ext.gemm_tn(x2d, w2a, None, 0, 0.0, False)  # [NPin,RSCo8]
+            if FP8_CONV:
+                w2aq, _, iw2 = _packed(w, "w2a_fp8", lambda: tuple(
+                    ext.fp8_quantize(_pad_k128(w2a))))
+                xq2, ix2 = _quant_delayed(_pad_k128(x2d), w, "xT")
+                col = ext.gemm_tn_fp8(xq2, w2aq, ix2, iw2, None, 0, 0.0)
+            else:
+                col = ext.gemm_tn(x2d, w2a, None, 0, 0.0, False)  # [NPin,RSCo8]
             if emit_stats and Cout % 8 == 0:
                 yh, ssum, ssq = ext.col2im_stats(
                     col, N, Ho, Wo, Cout, Hi, Wi, R, S, stride, pad,
@@ -572,10 +654,20 @@ class _ConvTranspose2d(torch.autograd.Function):
                 return _pad_k(wc)
 
             w2b = _packed(w, "convt_dgrad_w", build_w2b)
-            dx2d = ext.conv_fwd_implicit(dpre8, w2b, None,
-                                         _zp(dpre8.device), N, Ho, Wo,
-                                         Co8, Hi, Wi, R, S, stride, pad,
-                                         0, 0.0, 0, 0)[0]
+            if _fp8_bwd_on() and Co8 % 16 == 0:
+                w2bq, _, iw2b = _packed(w, "convt_dgrad_w_fp8",
+                                        lambda: tuple(
+                                            ext.fp8_quantize(
+                                                _pad_k128(w2b))))
+                dq, idy = _quant_delayed(dpre8, w, "dy")
+                dx2d = ext.conv_fwd_implicit_fp8(
+                    dq, w2bq, None, idy, iw2b, _zp8(dpre8.device), N, Ho,
+                    Wo, Co8, Hi, Wi, R, S, stride, pad, 0, 0.0, 0)
+            else:
+                dx2d = ext.conv_fwd_implicit(dpre8, w2b, None,
+                                             _zp(dpre8.device), N, Ho, Wo,
+                                             Co8, Hi, Wi, R, S, stride,
+                                             pad, 0, 0.0, 0, 0)[0]
             dx = _as_nchw_view(dx2d.view(N, Hi, Wi, Cin)).to(ctx.dtypes[0])
         if ctx.needs_input_grad[1]:
             # wgrad: dW[(r,s,cout)][cin] = sum_np im2col(dpre)[np][rs*cout]

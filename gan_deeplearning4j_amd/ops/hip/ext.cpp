@@ -124,6 +124,9 @@ void launch_fused_adam(void*, const void*, float*, float*, float*, long, int,
 void launch_amax(const void*, long, unsigned*, hipStream_t);
 void launch_fp8_make_scale(const unsigned*, float*, float*, hipStream_t);
 void launch_quant_fp8(const void*, void*, long, const float*, hipStream_t);
+void launch_quant_fp8_delayed(const void*, void*, long, const float*,
+                              unsigned*, hipStream_t);
+void launch_fp8_roll_scale(unsigned*, float*, float*, hipStream_t);
 void launch_gemm_tn_fp8(const void*, const void*, void*, const float*,
                         const float*, const float*, int, int, int, long,
                         long, int, float, int, ConvGather, const void*,
@@ -782,12 +785,67 @@ std::vector<torch::Tensor> fp8_quantize(torch::Tensor x) {
   return {y, scale, inv};
 }
 
+torch::Tensor fp8_quantize_delayed(torch::Tensor x, torch::Tensor scale,
+                                   torch::Tensor inv, torch::Tensor amax,
+                                   bool first) {
+  // Delayed-scaling quantize (TransformerEngine-style): quantize with the
+  // scale rolled from the PREVIOUS step's amax (cvt saturates at +-448 so
+  // a stale scale clips, never overflows) while accumulating this
+  // tensor's amax for the next roll.  first=true bootstraps with an
+  // exact two-pass amax.
+  check_bf16(x, "x");
+  TORCH_CHECK(x.numel() % 8 == 0, "fp8 quantize needs numel % 8 == 0");
+  torch::Tensor y = torch::empty(x.sizes(), x.options().dtype(torch::kUInt8));
+  auto s = cur_stream();
+  if (first) {
+    launch_amax(x.data_ptr(), x.numel(), (unsigned*)amax.data_ptr(), s);
+    launch_fp8_roll_scale((unsigned*)amax.data_ptr(),
+                          scale.data_ptr<float>(), inv.data_ptr<float>(), s);
+    launch_quant_fp8_delayed(x.data_ptr(), y.data_ptr(), x.numel(),
+                             scale.data_ptr<float>(),
+                             (unsigned*)amax.data_ptr(), s);
+  } else {
+    launch_fp8_roll_scale((unsigned*)amax.data_ptr(),
+                          scale.data_ptr<float>(), inv.data_ptr<float>(), s);
+    launch_quant_fp8_delayed(x.data_ptr(), y.data_ptr(), x.numel(),
+                             scale.data_ptr<float>(),
+                             (unsigned*)amax.data_ptr(), s);
+  }
+  return y;
+}
+
+torch::Tensor gemm_tn_fp8(torch::Tensor Aq, torch::Tensor Bq,
+                          torch::Tensor inv_qa, torch::Tensor inv_qb,
+                          c10::optional<torch::Tensor> bias, int64_t act,
+                          double slope) {
+  // C[M][N] bf16 = descale * (Aq . Bq^T) — both operands e4m3, K-major,
+  // K padded to 128 (the double-rate fp8 MFMA's K depth)
+  TORCH_CHECK(Aq.scalar_type() == torch::kUInt8 && Aq.is_contiguous());
+  TORCH_CHECK(Bq.scalar_type() == torch::kUInt8 && Bq.is_contiguous());
+  TORCH_CHECK(Aq.size(1) == Bq.size(1), "K mismatch");
+  TORCH_CHECK(Aq.size(1) % 128 == 0, "fp8 K must be padded to 128");
+  int64_t M = Aq.size(0), N = Bq.size(0), K = Aq.size(1);
+  const float* bias_p = nullptr;
+  if (bias.has_value() && bias->defined() && bias->numel() > 0) {
+    check_f32(*bias, "bias");
+    bias_p = bias->data_ptr<float>();
+  }
+  torch::Tensor C = torch::empty({M, N},
+                                 Aq.options().dtype(torch::kBFloat16));
+  ConvGather dummy{};
+  launch_gemm_tn_fp8(Aq.data_ptr(), Bq.data_ptr(), C.data_ptr(), bias_p,
+                     inv_qa.data_ptr<float>(), inv_qb.data_ptr<float>(),
+                     (int)M, (int)N, (int)K, K, K, (int)act, (float)slope,
+                     0, dummy, nullptr, cur_stream());
+  return C;
+}
+
 torch::Tensor conv_fwd_implicit_fp8(
     torch::Tensor xq, torch::Tensor wq, c10::optional<torch::Tensor> bias,
     torch::Tensor inv_qx, torch::Tensor inv_qw, torch::Tensor zero_page,
     int64_t Nb, int64_t H, int64_t W, int64_t C, int64_t Ho, int64_t Wo,
     int64_t R, int64_t S, int64_t stride, int64_t pad, int64_t act,
-    double slope) {
+    double slope, int64_t mode) {
   TORCH_CHECK(xq.scalar_type() == torch::kUInt8 && xq.is_contiguous());
   TORCH_CHECK(wq.scalar_type() == torch::kUInt8 && wq.is_contiguous());
   TORCH_CHECK(C % 16 == 0, "fp8 implicit conv needs C % 16 == 0");
@@ -802,7 +860,8 @@ torch::Tensor conv_fwd_implicit_fp8(
   torch::Tensor y = torch::empty({M, Kout},
                                  xq.options().dtype(torch::kBFloat16));
   ConvGather g = make_gather((int)Nb, (int)H, (int)W, (int)C, (int)Ho,
-                             (int)Wo, (int)R, (int)S, (int)stride, (int)pad);
+                             (int)Wo, (int)R, (int)S, (int)stride, (int)pad,
+                             (int)mode);
   launch_gemm_tn_fp8(xq.data_ptr(), wq.data_ptr(), y.data_ptr(), bias_p,
                      inv_qx.data_ptr<float>(), inv_qw.data_ptr<float>(),
                      (int)M, (int)Kout, (int)kpad, 0, kpad, (int)act,
@@ -865,6 +924,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
           "parity-decomposed strided transposed conv / conv dgrad");
   mod.def("gemm_nt_implicit", &gemm_nt_implicit,
           "weight-grad GEMM with one operand gathered as im2col");
+  mod.def("fp8_quantize_delayed", &fp8_quantize_delayed,
+          "bf16 -> e4m3 with delayed per-tensor scaling (fused amax)");
+  mod.def("gemm_tn_fp8", &gemm_tn_fp8,
+          "C = descale*(Aq.Bq^T) e4m3 MFMA, bf16 out");
   mod.def("fp8_quantize", &fp8_quantize,
           "bf16 -> e4m3 (OCP) with dynamic per-tensor scale");
   mod.def("conv_fwd_implicit_fp8", &conv_fwd_implicit_fp8,

@@ -74,6 +74,54 @@ __global__ void quant_fp8(const s16x8* __restrict__ x,
   }
 }
 
+// bf16 -> e4m3 with the previous step's scale (delayed scaling), fused
+// with this tensor's amax reduction so the standalone full-read amax
+// pass disappears.  cvt_pk_fp8_f32 saturates at +-448, so a modestly
+// stale scale clips instead of overflowing; the updated scale applies
+// from the next step (TransformerEngine-style recipe).
+__global__ void quant_fp8_delayed(const s16x8* __restrict__ x,
+                                  unsigned long long* __restrict__ y, long n8,
+                                  const float* __restrict__ scale,
+                                  unsigned* __restrict__ amax_bits) {
+  float s = *scale;
+  float m = 0.f;
+  long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  for (; i < n8; i += (long)gridDim.x * blockDim.x) {
+    s16x8 v = x[i];
+    float f[8];
+    #pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      f[j] = bf2f((unsigned short)v[j]);
+      m = fmaxf(m, fabsf(f[j]));
+    }
+    unsigned lo = 0, hi = 0;
+    lo = __builtin_amdgcn_cvt_pk_fp8_f32(f[0] * s, f[1] * s, lo, false);
+    lo = __builtin_amdgcn_cvt_pk_fp8_f32(f[2] * s, f[3] * s, lo, true);
+    hi = __builtin_amdgcn_cvt_pk_fp8_f32(f[4] * s, f[5] * s, hi, false);
+    hi = __builtin_amdgcn_cvt_pk_fp8_f32(f[6] * s, f[7] * s, hi, true);
+    y[i] = ((unsigned long long)hi << 32) | lo;
+  }
+  m = wave_reduce_max(m);
+  if ((threadIdx.x & 63) == 0) {
+    union { float f; unsigned u; } c;
+    c.f = m;
+    atomicMax(amax_bits, c.u);
+  }
+}
+
+// scale/inv for the NEXT step from the accumulated amax; resets amax
+__global__ void fp8_roll_scale(unsigned* __restrict__ amax_bits,
+                               float* __restrict__ scale,
+                               float* __restrict__ inv) {
+  union { float f; unsigned u; } c;
+  c.u = *amax_bits;
+  float a = c.f;
+  float sc = (a > 1e-20f) ? 448.f / a : 1.f;
+  *scale = sc;
+  *inv = 1.f / sc;
+  *amax_bits = 0;
+}
+
 // ------------------------------------------------------------ fp8 TN GEMM
 // A gathered from an fp8 NHWC image (implicit conv) or plain [M][K];
 // B plain fp8 [N][K]. LDS tiles [128 rows][128 k] fp8 = 16 KiB each.
@@ -116,7 +164,7 @@ DEV_INLINE void k_decode_f8(const ConvGather& g, unsigned k, int& r, int& s,
 // gemm.hip): each thread stages the same 4 rows every K-step.
 struct F8GatherStager {
   long base[4];      // (long)n * H*W*C element offset per chunk
-  int h0[4], w0[4];  // ho*stride-pad / wo*stride-pad per chunk
+  int h0[4], w0[4];  // mode-adjusted spatial bases per chunk
 
   DEV_INLINE void init(const ConvGather& g, int row0, int nrows) {
     const int t = threadIdx.x;
@@ -129,8 +177,16 @@ struct F8GatherStager {
       unsigned q2 = fdiv(q1, g.fHo);
       int ho = (int)(q1 - q2 * g.Ho);
       base[i] = (long)(int)q2 * g.H * g.W * g.C;
-      h0[i] = ho * g.stride - g.pad;
-      w0[i] = wo * g.stride - g.pad;
+      if (g.mode == 0) {
+        h0[i] = ho * g.stride - g.pad;
+        w0[i] = wo * g.stride - g.pad;
+      } else if (g.mode == 2) {
+        h0[i] = ho + g.off_h;
+        w0[i] = wo + g.off_w;
+      } else {
+        h0[i] = ho + g.pad;
+        w0[i] = wo + g.pad;
+      }
     }
   }
 
@@ -151,10 +207,33 @@ struct F8GatherStager {
       if (k < g.rsc) {
         int r, s, c;
         k_decode_f8(g, (unsigned)k, r, s, c);
-        int hi = h0[i] + r;
-        int wi = w0[i] + s;
-        if (hi >= 0 && hi < g.H && wi >= 0 && wi < g.W)
-          src = img + base[i] + (long)(hi * g.W + wi) * g.C + c;
+        int hi, wi;
+        bool valid;
+        if (g.mode == 0) {
+          hi = h0[i] + r;
+          wi = w0[i] + s;
+          valid = hi >= 0 && hi < g.H && wi >= 0 && wi < g.W;
+        } else if (g.mode == 2) {
+          hi = h0[i] - r;
+          wi = w0[i] - s;
+          valid = hi >= 0 && hi < g.H && wi >= 0 && wi < g.W;
+        } else {
+          int hop = h0[i] - r;
+          int wop = w0[i] - s;
+          if (hop < 0 || wop < 0) {
+            valid = false;
+            hi = wi = 0;
+          } else {
+            unsigned qh = fdiv((unsigned)hop, g.fStride);
+            unsigned qw = fdiv((unsigned)wop, g.fStride);
+            valid = (hop == (int)(qh * g.stride)) &&
+                    (wop == (int)(qw * g.stride)) && (int)qh < g.H &&
+                    (int)qw < g.W;
+            hi = (int)qh;
+            wi = (int)qw;
+          }
+        }
+        if (valid) src = img + base[i] + (long)(hi * g.W + wi) * g.C + c;
       }
       char* dst = lds + (i * 256 + wid * 64) * 16;
       GLDS16(src, dst);
@@ -284,9 +363,21 @@ __global__ __launch_bounds__(256, 2) void gemm_tn_fp8_core(
       int seg = piece & 15;
       int grow = m0 + row;
       int gcol = n0 + seg * 8;
-      if (grow < M && gcol + 8 <= N)
-        *(s16x8*)(&C[(long)grow * N + gcol]) =
+      if (grow < M && gcol + 8 <= N) {
+        long crow = grow;
+        if (GATHER_A && ga.mode == 2) {
+          int n2, h2, w2;
+          unsigned q1 = fdiv((unsigned)grow, ga.fWo);
+          w2 = (int)((unsigned)grow - q1 * ga.Wo);
+          unsigned q2 = fdiv(q1, ga.fHo);
+          h2 = (int)(q1 - q2 * ga.Ho);
+          n2 = (int)q2;
+          crow = ((long)n2 * ga.oH + h2 * ga.stride + ga.oqh) * ga.oW +
+                 w2 * ga.stride + ga.oqw;
+        }
+        *(s16x8*)(&C[crow * N + gcol]) =
             *(const s16x8*)(ctile + row * 128 + seg * 8);
+      }
     }
     return;
   }
@@ -328,6 +419,22 @@ void launch_quant_fp8(const void* x, void* y, long n, const float* scale,
   int grid = (int)min((long)2048, n8 / 256 + 1);
   hipLaunchKernelGGL(quant_fp8, dim3(grid), dim3(256), 0, s, (const s16x8*)x,
                      (unsigned long long*)y, n8, scale);
+}
+
+void launch_quant_fp8_delayed(const void* x, void* y, long n,
+                              const float* scale, unsigned* amax_bits,
+                              hipStream_t s) {
+  long n8 = n / 8;
+  int grid = (int)min((long)2048, n8 / 256 + 1);
+  hipLaunchKernelGGL(quant_fp8_delayed, dim3(grid), dim3(256), 0, s,
+                     (const s16x8*)x, (unsigned long long*)y, n8, scale,
+                     amax_bits);
+}
+
+void launch_fp8_roll_scale(unsigned* amax_bits, float* scale, float* inv,
+                           hipStream_t s) {
+  hipLaunchKernelGGL(fp8_roll_scale, dim3(1), dim3(1), 0, s, amax_bits,
+                     scale, inv);
 }
 
 void launch_gemm_tn_fp8(const void* A, const void* B, void* C,

@@ -66,7 +66,7 @@ def test_fp8_conv_fwd_vs_fp32():
     wq, _, iw = e.fp8_quantize(wp)
     zp8 = torch.zeros(32, dtype=torch.uint8, device=DEV)
     y = e.conv_fwd_implicit_fp8(xq, wq, None, ix, iw, zp8, N, H, H, C, Ho,
-                                Ho, R, R, stride, pad, 0, 0.0)
+                                Ho, R, R, stride, pad, 0, 0.0, 0)
     ref = F.conv2d(x.float().cpu(), w.float().cpu(), stride=stride,
                    padding=pad)
     ref2d = ref.permute(0, 2, 3, 1).reshape(-1, Kout)
@@ -113,5 +113,91 @@ def test_dcgan128_fp8_step():
         torch.cuda.synchronize()
         assert math.isfinite(float(out["loss_d"]))
         assert math.isfinite(float(out["loss_g"]))
+    finally:
+        gpu_ops.set_fp8_conv(False)
+
+
+def test_fp8_quantize_delayed_rolls_scale():
+    """Delayed scaling: first call bootstraps exactly; the second call's
+    scale comes from the first call's accumulated amax."""
+    e = ext()
+    scale = torch.ones(1, device=DEV)
+    inv = torch.ones(1, device=DEV)
+    amax = torch.zeros(1, dtype=torch.int32, device=DEV)
+    x1 = (torch.arange(4096, device=DEV).float() / 4096 * 10).to(
+        torch.bfloat16)
+    q1 = e.fp8_quantize_delayed(x1, scale, inv, amax, True)
+    torch.cuda.synchronize()
+    s1 = float(scale)
+    assert abs(s1 - 448.0 / float(x1.float().abs().max())) / s1 < 0.02
+    # second tensor with twice the amax: quantized with s1 (stale), and
+    # the roll for call 3 must reflect x2's amax
+    x2 = x1 * 2
+    q2 = e.fp8_quantize_delayed(x2, scale, inv, amax, False)
+    torch.cuda.synchronize()
+    assert abs(float(scale) - s1) / s1 < 0.02  # still the x1-era scale
+    x3 = x1
+    e.fp8_quantize_delayed(x3, scale, inv, amax, False)
+    torch.cuda.synchronize()
+    s3 = float(scale)
+    assert abs(s3 - 448.0 / float(x2.float().abs().max())) / s3 < 0.02
+    assert q1.shape == q2.shape
+
+
+def test_fp8_conv_bwd_through_function():
+    """conv2d fwd+bwd with FP8 on: dgrad runs the fp8 dcol/mode-1 path
+    (wgrad stays bf16). Coarse tolerances: e4m3 is a 3-bit mantissa."""
+    from gan_deeplearning4j_amd.ops import gpu_ops
+
+    gpu_ops.set_fp8_conv(True)
+    try:
+        for stride in (1, 2):
+            g = torch.Generator().manual_seed(10 + stride)
+            x = (torch.randn(8, 32, 16, 16, generator=g) * 0.5).to(
+                DEV, torch.bfloat16).requires_grad_(True)
+            w = (torch.randn(64, 32, 4, 4, generator=g) * 0.2).to(
+                DEV, torch.bfloat16).requires_grad_(True)
+            y = gpu_ops.conv2d(x, w, None, stride, 1, "identity", 0.0)
+            gout = (torch.randn(y.shape, generator=g) * 0.3).to(
+                DEV, torch.bfloat16)
+            y.backward(gout)
+            xr = x.detach().float().cpu().requires_grad_(True)
+            wr = w.detach().float().cpu().requires_grad_(True)
+            yr = F.conv2d(xr, wr, None, stride=stride, padding=1)
+            yr.backward(gout.float().cpu())
+            for got, want, tol in ((y, yr, 0.12), (x.grad, xr.grad, 0.15),
+                                   (w.grad, wr.grad, 0.08)):
+                errn = ((got.detach().float().cpu() - want.detach())
+                        .abs().max() / want.detach().abs().max())
+                assert float(errn) < tol, (stride, float(errn))
+    finally:
+        gpu_ops.set_fp8_conv(False)
+
+
+def test_fp8_convtranspose_through_function():
+    from gan_deeplearning4j_amd.ops import gpu_ops
+
+    gpu_ops.set_fp8_conv(True)
+    try:
+        for stride in (1, 2):
+            g = torch.Generator().manual_seed(20 + stride)
+            x = (torch.randn(8, 64, 8, 8, generator=g) * 0.5).to(
+                DEV, torch.bfloat16).requires_grad_(True)
+            w = (torch.randn(64, 32, 4, 4, generator=g) * 0.2).to(
+                DEV, torch.bfloat16).requires_grad_(True)
+            y = gpu_ops.conv_transpose2d(x, w, None, stride, 1,
+                                         "identity", 0.0)
+            gout = (torch.randn(y.shape, generator=g) * 0.3).to(
+                DEV, torch.bfloat16)
+            y.backward(gout)
+            xr = x.detach().float().cpu().requires_grad_(True)
+            wr = w.detach().float().cpu().requires_grad_(True)
+            yr = F.conv_transpose2d(xr, wr, None, stride=stride, padding=1)
+            yr.backward(gout.float().cpu())
+            for got, want, tol in ((y, yr, 0.12), (x.grad, xr.grad, 0.15),
+                                   (w.grad, wr.grad, 0.08)):
+                errn = ((got.detach().float().cpu() - want.detach())
+                        .abs().max() / want.detach().abs().max())
+                assert float(errn) < tol, (stride, float(errn))
     finally:
         gpu_ops.set_fp8_conv(False)
