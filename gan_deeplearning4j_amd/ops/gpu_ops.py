@@ -165,8 +165,12 @@ def _fp8_state(w: torch.Tensor, role: str, device) -> list:
 
 def _quant_delayed(x: torch.Tensor, w: torch.Tensor, role: str):
     """Quantize x to e4m3 with the previous step's scale (fused amax
-    accumulation; exact two-pass on first use). Returns (q, inv_scale)."""
+    accumulation; exact two-pass on first use). Returns (q, inv_scale).
+    GDLJ_FP8_DELAYED=0 falls back to exact per-call scaling."""
     ext = hip_ext()
+    if os.environ.get("GDLJ_FP8_DELAYED") == "0":
+        q, _, inv = ext.fp8_quantize(x.contiguous())
+        return q, inv
     st = _fp8_state(w, role, x.device)
     y = ext.fp8_quantize_delayed(x.contiguous(), st[0], st[1], st[2], st[3])
     st[3] = False
@@ -521,7 +525,8 @@ class _ConvTranspose2d(torch.autograd.Function):
             wt = _packed(w, "convt_fwd_w", lambda: _pad_k(
                 _bf(w.detach().permute(1, 2, 3, 0))      # [Cout,R,S,Cin]
                 .reshape(Cout, R * S * Cin)))
-            if FP8_CONV and Cin % 16 == 0 and not emit_stats:
+            if (FP8_CONV and Cin % 16 == 0 and not emit_stats
+                    and os.environ.get("GDLJ_FP8_CONVT") != "0"):
                 wtq, _, iw = _packed(w, "convt_fwd_w_fp8", lambda: tuple(
                     ext.fp8_quantize(_pad_k128(wt))))
                 xq, ix = _quant_delayed(xh, w, "x")
@@ -580,7 +585,7 @@ class _ConvTranspose2d(torch.autograd.Function):
                 w2a = _packed(w, "w2a", lambda: _pad_k(
                     _bf(w.detach().permute(2, 3, 1, 0))
                     .reshape(R * S * Cout, Cin)))
-            if FP8_CONV:
+            if FP8_CONV and os.environ.get("GDLJ_FP8_CONVT") != "0":
                 w2aq, _, iw2 = _packed(w, "w2a_fp8", lambda: tuple(
                     ext.fp8_quantize(_pad_k128(w2a))))
                 xq2, ix2 = _quant_delayed(_pad_k128(x2d), w, "xT")
