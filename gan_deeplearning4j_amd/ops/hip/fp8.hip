@@ -76,9 +76,11 @@ __global__ void quant_fp8(const s16x8* __restrict__ x,
 
 // bf16 -> e4m3 with the previous step's scale (delayed scaling), fused
 // with this tensor's amax reduction so the standalone full-read amax
-// pass disappears.  cvt_pk_fp8_f32 saturates at +-448, so a modestly
-// stale scale clips instead of overflowing; the updated scale applies
-// from the next step (TransformerEngine-style recipe).
+// pass disappears.  v_cvt_pk_fp8_f32 does NOT saturate — a scaled value
+// past +-448 encodes as e4m3fn NaN (0x7f) and poisons the GEMM (measured:
+// the second D forward of a GAN step overflowed its first-step scale) —
+// so the scaled input is clamped to +-448 explicitly; the updated scale
+// applies from the next step (TransformerEngine-style recipe).
 __global__ void quant_fp8_delayed(const s16x8* __restrict__ x,
                                   unsigned long long* __restrict__ y, long n8,
                                   const float* __restrict__ scale,
@@ -91,14 +93,15 @@ __global__ void quant_fp8_delayed(const s16x8* __restrict__ x,
     float f[8];
     #pragma unroll
     for (int j = 0; j < 8; ++j) {
-      f[j] = bf2f((unsigned short)v[j]);
-      m = fmaxf(m, fabsf(f[j]));
+      float raw = bf2f((unsigned short)v[j]);
+      m = fmaxf(m, fabsf(raw));
+      f[j] = fminf(fmaxf(raw * s, -448.f), 448.f);  // saturate, never NaN
     }
     unsigned lo = 0, hi = 0;
-    lo = __builtin_amdgcn_cvt_pk_fp8_f32(f[0] * s, f[1] * s, lo, false);
-    lo = __builtin_amdgcn_cvt_pk_fp8_f32(f[2] * s, f[3] * s, lo, true);
-    hi = __builtin_amdgcn_cvt_pk_fp8_f32(f[4] * s, f[5] * s, hi, false);
-    hi = __builtin_amdgcn_cvt_pk_fp8_f32(f[6] * s, f[7] * s, hi, true);
+    lo = __builtin_amdgcn_cvt_pk_fp8_f32(f[0], f[1], lo, false);
+    lo = __builtin_amdgcn_cvt_pk_fp8_f32(f[2], f[3], lo, true);
+    hi = __builtin_amdgcn_cvt_pk_fp8_f32(f[4], f[5], hi, false);
+    hi = __builtin_amdgcn_cvt_pk_fp8_f32(f[6], f[7], hi, true);
     y[i] = ((unsigned long long)hi << 32) | lo;
   }
   m = wave_reduce_max(m);
