@@ -403,6 +403,14 @@ __global__ __launch_bounds__(256, 2) void gemm_tn_fp8_core(
 
 extern "C" {
 
+// 8-phase fp8 pipeline (gemm8p_fp8.hip) — default for large shapes
+int gemm_tn_8p_fp8_eligible(int M, int N, int K);
+int launch_gemm_tn_8p_fp8(const void* A, const void* B, void* C,
+                          const float* bias, const float* inv_qa,
+                          const float* inv_qb, int M, int N, int K, long lda,
+                          long ldb, int act, float slope, int gather,
+                          ConvGather ga, const void* zp, hipStream_t s);
+
 void launch_amax(const void* x, long n, unsigned* amax_bits, hipStream_t s) {
   long n8 = n / 8;
   int grid = (int)min((long)1024, n8 / 256 + 1);
@@ -445,6 +453,11 @@ void launch_gemm_tn_fp8(const void* A, const void* B, void* C,
                         const float* inv_qb, int M, int N, int K, long lda,
                         long ldb, int act, float slope, int gather,
                         ConvGather ga, const void* zp, hipStream_t s) {
+  if (gemm_tn_8p_fp8_eligible(M, N, K)) {
+    launch_gemm_tn_8p_fp8(A, B, C, bias, inv_qa, inv_qb, M, N, K, lda, ldb,
+                          act, slope, gather, ga, zp, s);
+    return;
+  }
   dim3 grid(ceil_div(M, F8_BM), ceil_div(N, F8_BN));
   if (gather)
     hipLaunchKernelGGL((gemm_tn_fp8_core<true>), grid, dim3(256), 0, s,

@@ -201,3 +201,40 @@ def test_fp8_convtranspose_through_function():
                 assert float(errn) < tol, (stride, float(errn))
     finally:
         gpu_ops.set_fp8_conv(False)
+
+
+def test_fp8_8p_plain_gemm():
+    """Eligible shapes route to the 8-phase fp8 pipeline (gemm8p_fp8.hip)."""
+    e = ext()
+    g = torch.Generator().manual_seed(3)
+    for m, n, k in [(65536, 256, 256), (65400, 320, 384)]:
+        A = (torch.randn(m, k, generator=g) * 0.5).to(DEV, torch.bfloat16)
+        B = (torch.randn(n, k, generator=g) * 0.5).to(DEV, torch.bfloat16)
+        qa, _, ia = e.fp8_quantize(A)
+        qb, _, ib = e.fp8_quantize(B)
+        C = e.gemm_tn_fp8(qa, qb, ia, ib, None, 0, 0.0)
+        ref = A.float() @ B.float().t()
+        err = (C.float() - ref).abs().max() / ref.abs().max()
+        assert float(err) < 0.12, (m, n, k, float(err))
+
+
+def test_fp8_8p_conv_fwd_gather():
+    """Eligible conv shape routes the gathered fp8 8p kernel."""
+    from gan_deeplearning4j_amd.ops import gpu_ops
+
+    gpu_ops.set_fp8_conv(True)
+    try:
+        N, Cin, H, Cout, R, stride, pad = 64, 64, 32, 256, 5, 1, 2
+        g = torch.Generator().manual_seed(4)
+        x = (torch.randn(N, Cin, H, H, generator=g) * 0.4).to(
+            DEV, torch.bfloat16)
+        w = (torch.randn(Cout, Cin, R, R, generator=g) * 0.1).to(
+            DEV, torch.bfloat16)
+        y = gpu_ops.conv2d(x, w, None, stride, pad, "identity", 0.0)
+        yr = F.conv2d(x.float().cpu(), w.float().cpu(), None,
+                      stride=stride, padding=pad)
+        err = ((y.detach().float().cpu() - yr).abs().max()
+               / yr.abs().max())
+        assert float(err) < 0.12, float(err)
+    finally:
+        gpu_ops.set_fp8_conv(False)
