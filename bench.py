@@ -48,7 +48,7 @@ def main():
         # sized for 288 GB HBM3E: bigger batches amortize fixed kernel
         # costs (measured +13% at 4096 vs 1024 on dcgan64)
         per_gpu_batch = {"dcgan28": 8192, "dcgan64": 8192,
-                         "dcgan128": 512}.get(args.arch, 512)
+                         "dcgan128": 2048}.get(args.arch, 512)
         if not use_gpu:
             per_gpu_batch = 16
     cfg.train.use_gpu = use_gpu
@@ -104,11 +104,13 @@ def main():
     images_per_sec = global_batch * args.steps / elapsed
     size = m.image_height
     if use_gpu and cfg.model.dtype == "fp8":
-        # fwd convs + the dgrad family run e4m3 MFMA; wgrad stays bf16
-        # (documented non-goal: the NT contraction has no fp8 transpose
-        # read path) — the label says exactly what is measured
+        # fwd convs (incl. convT) run e4m3 MFMA; backward defaults to
+        # bf16 — fp8 dgrad exists (GDLJ_FP8_BWD=1) but measured 4-6%
+        # slower at these K-thin memory-bound dgrad shapes
+        # (profiles/fp8_bwd_ab.md), and wgrad has no fp8 transpose-read
+        # path. The label states exactly what is measured.
         dt = ("fp8-fwd+dgrad/bf16-wgrad"
-              if os.environ.get("GDLJ_FP8_BWD") != "0"
+              if os.environ.get("GDLJ_FP8_BWD") == "1"
               else "fp8-fwd/bf16-bwd")
     else:
         dt = "bf16" if use_gpu else "fp32"

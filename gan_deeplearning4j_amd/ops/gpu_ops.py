@@ -134,15 +134,19 @@ def set_fp8_conv(enabled: bool) -> None:
     FP8_CONV = bool(enabled)
 
 
-# fp8 backward (dgrad family) rides the same switch; GDLJ_FP8_BWD=0 keeps
-# backward in bf16 for A/B comparisons. wgrad stays bf16: the NT
-# contraction reads both operands k(=np)-major, and glds cannot build the
-# transposed fp8 LDS image (lane-linear dest; no tr16-style transpose
-# read exists for 8-bit fragments at the 16x16x128 MFMA's layout) — a
-# register-staged transpose pass measured as the round-1 parity-dgrad
-# wash makes it a documented non-goal this round.
+# fp8 backward (dgrad family) is implemented and numerics-tested but OFF
+# by default (GDLJ_FP8_BWD=1 opts in): measured on DCGAN-128 at batches
+# 512-2048, fp8 dgrad is 4-6% SLOWER end-to-end than bf16 dgrad — the
+# dgrad GEMMs are K-thin (K = Cout <= 1024) and memory-bound on the
+# C-write/A-read stream, so halving the A bytes saves less than the
+# extra quantize pass (a full dy read + fp8 write) costs
+# (profiles/fp8_bwd_ab.md). fp8 pays where MFMA rate is the bound: the
+# K-deep forward gathers (on by default). wgrad stays bf16: the NT
+# contraction reads both operands k(=np)-major, and glds cannot build
+# the transposed fp8 LDS image (lane-linear dest; no tr16-style
+# transpose read exists for 8-bit fragments at the 16x16x128 layout).
 def _fp8_bwd_on() -> bool:
-    return FP8_CONV and os.environ.get("GDLJ_FP8_BWD") != "0"
+    return FP8_CONV and os.environ.get("GDLJ_FP8_BWD") == "1"
 
 
 def _fp8_state(w: torch.Tensor, role: str, device) -> list:
