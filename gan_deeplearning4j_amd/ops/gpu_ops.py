@@ -529,6 +529,7 @@ class _ConvTranspose2d(torch.autograd.Function):
             wt = _packed(w, "convt_fwd_w", lambda: _pad_k(
                 _bf(w.detach().permute(1, 2, 3, 0))      # [Cout,R,S,Cin]
                 .reshape(Cout, R * S * Cin)))
+            # stride-1 convT fwd: K = rsc (deep) — fp8 pays; opt-out gate
             if (FP8_CONV and Cin % 16 == 0 and not emit_stats
                     and os.environ.get("GDLJ_FP8_CONVT") != "0"):
                 wtq, _, iw = _packed(w, "convt_fwd_w_fp8", lambda: tuple(
@@ -589,7 +590,10 @@ class _ConvTranspose2d(torch.autograd.Function):
                 w2a = _packed(w, "w2a", lambda: _pad_k(
                     _bf(w.detach().permute(2, 3, 1, 0))
                     .reshape(R * S * Cout, Cin)))
-            if FP8_CONV and os.environ.get("GDLJ_FP8_CONVT") != "0":
+            # strided convT dcol: K = Cin <= 1024 (thin, memory-bound):
+            # fp8 measured -2% end-to-end at dcgan128 b2048
+            # (profiles/fp8_bwd_ab.md mechanism) — opt-in only
+            if FP8_CONV and os.environ.get("GDLJ_FP8_CONVT") == "1":
                 w2aq, _, iw2 = _packed(w, "w2a_fp8", lambda: tuple(
                     ext.fp8_quantize(_pad_k128(w2a))))
                 xq2, ix2 = _quant_delayed(_pad_k128(x2d), w, "xT")
