@@ -605,9 +605,36 @@ DEV_INLINE void nt_stage(const unsigned short* __restrict__ g, int k0, int K,
 }
 
 // gathered flavor: operand is the im2col of an NHWC image
-// (contraction row = np; tile column = rsc coordinate)
+// (contraction row = np; tile column = rsc coordinate).  The column-side
+// k_decode (tap r,s + channel) is K-loop-invariant per thread — each
+// thread stages the same two rsc coordinates every K-step — so it is
+// hoisted into registers once (the row side np varies with k0 and is
+// decoded per step).
+struct NtGatherCols {
+  int r[2], s[2], cc[2];
+  bool in_rsc[2];
+
+  DEV_INLINE void init(const ConvGather& g, int c0) {
+    const int t = threadIdx.x;
+    #pragma unroll
+    for (int i = 0; i < 2; ++i) {
+      int c = i * 512 + t;
+      int w = c >> 3;
+      int h = c & 1;
+      int mb = w & 7;
+      int kk = c0 + mb * 16 + h * 8;
+      in_rsc[i] = kk < g.rsc;
+      if (in_rsc[i])
+        k_decode(g, (unsigned)kk, r[i], s[i], cc[i]);
+      else
+        r[i] = s[i] = cc[i] = 0;
+    }
+  }
+};
+
 DEV_INLINE void nt_stage_gather(const unsigned short* __restrict__ img,
-                                const ConvGather& g, int k0, int K, int c0,
+                                const ConvGather& g, const NtGatherCols& gc,
+                                int k0, int K,
                                 const unsigned short* __restrict__ zp,
                                 char* lds) {
   const int t = threadIdx.x;
@@ -617,19 +644,15 @@ DEV_INLINE void nt_stage_gather(const unsigned short* __restrict__ img,
     int c = i * 512 + t;
     int w = c >> 3;
     int j = (c >> 1) & 3;
-    int h = c & 1;
     int kb = w >> 3;
-    int mb = w & 7;
     int np = k0 + kb * 4 + j;
-    int kk = c0 + mb * 16 + h * 8;
     const unsigned short* src = zp;
-    if (np < K && kk < g.rsc) {
-      int r, s, cc, n, ho, wo;
-      k_decode(g, (unsigned)kk, r, s, cc);
+    if (np < K && gc.in_rsc[i]) {
+      int n, ho, wo;
       np_decode(g, (unsigned)np, n, ho, wo);
       bool valid;
-      const unsigned short* p = gather_addr(g, img, n, ho, wo, r, s, cc,
-                                            valid);
+      const unsigned short* p = gather_addr(g, img, n, ho, wo, gc.r[i],
+                                            gc.s[i], gc.cc[i], valid);
       if (valid) src = p;
     }
     char* dst = lds + (i * 512 + wid * 64) * 16;
@@ -689,15 +712,18 @@ __global__ __launch_bounds__(NT_THREADS, 2) void gemm_nt_core(
   int kt1 = min(kt0 + kchunks_per_block, kchunks);
   if (kt0 >= kt1) return;
 
+  NtGatherCols gca, gcb;
+  if (GMODE == 1) gca.init(gg, m0);
+  if (GMODE == 2) gcb.init(gg, n0);
   auto stage_a = [&](int kt, char* buf) {
     if (GMODE == 1)
-      nt_stage_gather(A, gg, kt * NT_BK, K, m0, zp, buf);
+      nt_stage_gather(A, gg, gca, kt * NT_BK, K, zp, buf);
     else
       nt_stage(A, kt * NT_BK, K, m0, M, lda, zp, buf);
   };
   auto stage_b = [&](int kt, char* buf) {
     if (GMODE == 2)
-      nt_stage_gather(B, gg, kt * NT_BK, K, n0, zp, buf);
+      nt_stage_gather(B, gg, gcb, kt * NT_BK, K, zp, buf);
     else
       nt_stage(B, kt * NT_BK, K, n0, N, ldb, zp, buf);
   };
