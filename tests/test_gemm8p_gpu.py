@@ -120,3 +120,32 @@ def test_gemm8p_conv_transpose_fwd():
     yr = F.conv_transpose2d(x.float().cpu(), w.float().cpu(), None,
                             stride=stride, padding=pad)
     assert relerr(y, yr) < 0.03
+
+
+@pytest.mark.parametrize(
+    "m,n,k",
+    [
+        (65536, 128, 256),   # BNT=128 tile exact
+        (65400, 120, 320),   # BNT=128 with N tail
+    ],
+)
+def test_gemm8p_n128_shapes(m, n, k):
+    e = ext()
+    A, B = mk((m, k), 21, 0.5), mk((n, k), 22, 0.5)
+    C = e.gemm_tn(A, B, None, 0, 0.0, False)
+    ref = A.float() @ B.float().t()
+    assert relerr(C, ref.cpu()) < 0.02
+
+
+def test_gemm8p_n128_conv_fwd():
+    # Cout=128 conv routes the BNT=128 gathered tile
+    from gan_deeplearning4j_amd.ops import gpu_ops
+
+    N, Cin, H, Cout, R, stride, pad = 64, 64, 32, 128, 5, 1, 2
+    x = mk((N, Cin, H, H), 23, 0.4)
+    w = mk((Cout, Cin, R, R), 24, 0.1)
+    y = gpu_ops.conv2d(x, w, None, stride, pad, "lrelu", 0.2)
+    yr = torch.nn.functional.leaky_relu(
+        F.conv2d(x.float().cpu(), w.float().cpu(), None, stride=stride,
+                 padding=pad), 0.2)
+    assert relerr(y, yr) < 0.03
