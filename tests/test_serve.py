@@ -167,3 +167,25 @@ def test_load_graph_roundtrip(tmp_path):
     g2 = load_graph(p)
     z = torch.rand(2, cfg.model.z_size)
     assert torch.allclose(gen.output(z), g2.output(z), atol=1e-6)
+
+
+def test_discriminate_concurrent_consistency(client):
+    """Concurrent /discriminate requests must not race on the endpoint's
+    static buffers (the _Endpoint.run lock, round-1 advisor finding):
+    every thread's scores must equal the single-thread result for its
+    own input."""
+    import concurrent.futures
+
+    torch.manual_seed(5)
+    inputs = [torch.rand(2, 784).tolist() for _ in range(8)]
+    want = [client.post("/discriminate", json={"inputs": x}).json()["scores"]
+            for x in inputs]
+
+    def hit(i):
+        r = client.post("/discriminate", json={"inputs": inputs[i]})
+        assert r.status_code == 200
+        return i, r.json()["scores"]
+
+    with concurrent.futures.ThreadPoolExecutor(max_workers=8) as ex:
+        for i, scores in ex.map(hit, list(range(8)) * 4):
+            assert scores == pytest.approx(want[i], abs=1e-5), i
