@@ -357,6 +357,21 @@ class _Conv2d(torch.autograd.Function):
             y2d = ext.conv_fwd_implicit_fp8(xq, wq, bias, ix, iw, zp8, N, H,
                                             W, C8, Ho, Wo, R, S, stride, pad,
                                             act, slope, 0)
+        elif C8 <= 16 and os.environ.get("GDLJ_SMALLC_COL") != "0":
+            # small-C convs (conv1-class): a gathered 16B chunk spans 2+
+            # taps, so the implicit stage issues 64 SCATTERED addresses
+            # per glds — TA-address-bound, not bandwidth-bound. An
+            # explicit coalesced im2col (image read once, col written
+            # linear) + plain-TN GEMM restores coalesced staging at the
+            # cost of one col round trip.
+            col = ext.im2col(xh, N, H, W, C8, Ho, Wo, R, S, stride, pad,
+                             kpad)
+            if emit_stats and Kout % 8 == 0:
+                y2d, ssum, ssq = ext.gemm_tn_stats(col, wp, bias, act,
+                                                   slope)
+                stats = (ssum, ssq)
+            else:
+                y2d = ext.gemm_tn(col, wp, bias, act, slope, False)
         else:
             # implicit GEMM: im2col gather fused into the MFMA staging
             # (optionally also emitting the consumer BN's batch statistics)
