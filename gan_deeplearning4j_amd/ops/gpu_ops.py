@@ -357,13 +357,13 @@ class _Conv2d(torch.autograd.Function):
             y2d = ext.conv_fwd_implicit_fp8(xq, wq, bias, ix, iw, zp8, N, H,
                                             W, C8, Ho, Wo, R, S, stride, pad,
                                             act, slope, 0)
-        elif C8 <= 16 and os.environ.get("GDLJ_SMALLC_COL") != "0":
-            # small-C convs (conv1-class): a gathered 16B chunk spans 2+
-            # taps, so the implicit stage issues 64 SCATTERED addresses
-            # per glds — TA-address-bound, not bandwidth-bound. An
-            # explicit coalesced im2col (image read once, col written
-            # linear) + plain-TN GEMM restores coalesced staging at the
-            # cost of one col round trip.
+        elif C8 <= 16 and os.environ.get("GDLJ_SMALLC_COL") == "1":
+            # MEASURED NEGATIVE, opt-in only (dcgan64 117.0k -> 114.4k,
+            # dcgan28 791k -> 765k): the hypothesis was that small-C
+            # gathers are TA-address-bound (a gathered 16B chunk spans
+            # 2+ taps -> 64 scattered addresses per glds) and an explicit
+            # coalesced im2col + plain-TN GEMM would win; the col round
+            # trip costs more than the scattered addressing does.
             col = ext.im2col(xh, N, H, W, C8, Ho, Wo, R, S, stride, pad,
                              kpad)
             if emit_stats and Kout % 8 == 0:
