@@ -510,11 +510,23 @@ static int p8_enabled() {
 
 // Tile-N selection: 0 = not eligible, else the BNT to launch.
 // Both shapes need K deep enough to amortize the pipeline prologue
-// (>= 4 K-tiles; at K = 128 the 2-block/CU 128-tile kernel wins,
-// profiles/gemm8p_ab.md) and >= 256 blocks to fill the chip at one
-// block/CU.
+// (>= 4 K-tiles; at K = 128 the 2-block/CU 128-tile kernel wins),
+// >= 256 blocks to fill the chip at one block/CU, and a large-M regime
+// (dcgan28's M=8192 dense family measured -2.6% under 8p routing:
+// 812.8k -> 791.5k img/s; every shape the 8p wins on has M >= 131k).
+// GDLJ_8P_MINM overrides (0 for microbenches). profiles/gemm8p_ab.md.
+static int p8_min_m() {
+  static int v = -1;
+  if (v < 0) {
+    const char* e = getenv("GDLJ_8P_MINM");
+    v = (e != nullptr) ? atoi(e) : 32768;
+  }
+  return v;
+}
+
 static int p8_pick_bnt(int M, int N, int K) {
   if (!p8_enabled()) return 0;
+  if (M < p8_min_m()) return 0;
   if (K % p8::BK != 0 || K < 4 * p8::BK) return 0;
   if (N >= 192) {
     long blocks = (long)ceil_div(M, p8::BM) * ceil_div(N, 256);

@@ -352,6 +352,7 @@ static int p8f_enabled() {
 
 int gemm_tn_8p_fp8_eligible(int M, int N, int K) {
   if (!p8f_enabled()) return 0;
+  if (M < 32768) return 0;  // large-M regime only (see bf16 8p note)
   if (K % p8f::BK != 0 || K < 2 * p8f::BK) return 0;
   if (N < 192) return 0;
   long blocks = (long)ceil_div(M, p8f::BM) * ceil_div(N, p8f::BN);

@@ -8,6 +8,9 @@ import os
 import sys
 import time
 
+# microbench shapes include M=4096 squares: lift the 8p large-M product gate
+os.environ.setdefault("GDLJ_8P_MINM", "0")
+
 sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 import torch
