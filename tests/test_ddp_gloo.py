@@ -226,3 +226,24 @@ def test_broadcast_parameters():
     out = _run_mp("_body_broadcast")
     assert torch.allclose(torch.tensor(out[0]), torch.full((1, 4), 1.0))
     assert torch.allclose(torch.tensor(out[1]), torch.full((1, 4), 1.0))
+
+
+def _body_comm_dtype_env(rank, world):
+    """gloo never uses bf16 comms even when requested (fp32 fallback)."""
+    import os
+
+    from gan_deeplearning4j_amd.parallel.ddp import GradReducer
+
+    os.environ["GDLJ_COMM_DTYPE"] = "bf16"
+    try:
+        m = torch.nn.Linear(4, 2).to(torch.bfloat16)
+        red = GradReducer([m], bucket_cap_mb=1)
+        return [str(red.comm_dtype)]
+    finally:
+        del os.environ["GDLJ_COMM_DTYPE"]
+
+
+def test_comm_dtype_gloo_fp32_fallback():
+    out = _run_mp("_body_comm_dtype_env")
+    assert out[0][0] == "torch.float32"
+    assert out[1][0] == "torch.float32"

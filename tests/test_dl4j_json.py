@@ -179,3 +179,18 @@ def test_parse_hand_constructed_dl4j_fixture():
     assert g.optim_cfg.rms_decay == 1e-8
     y = g.output(torch.rand(3, 8))
     assert y.shape == (3, 1)
+
+
+def test_dl4j_parser_rejects_unknown_layer():
+    bad = json.loads(json.dumps(KNOWN_GOOD))
+    bad["vertices"]["hidden"]["layerConf"]["layer"]["@class"] = (
+        "org.deeplearning4j.nn.conf.layers.LocallyConnected2D")
+    with pytest.raises(TypeError, match="LocallyConnected2D"):
+        from_dl4j_json(bad)
+
+
+def test_dl4j_parser_rejects_dangling_wiring():
+    bad = json.loads(json.dumps(KNOWN_GOOD))
+    bad["vertexInputs"]["hidden"] = ["missing_vertex"]
+    with pytest.raises(ValueError, match="cyclic or dangling"):
+        from_dl4j_json(bad)

@@ -97,3 +97,22 @@ def test_fit_averaged_is_one_averaged_update():
     gseq = fresh()
     gseq.fit([a, b])
     assert not torch.allclose(gseq.params_flat(), want, atol=1e-6)
+
+
+def test_fit_averaged_single_partition_equals_fit():
+    from gan_deeplearning4j_amd.config import GanConfig
+    from gan_deeplearning4j_amd.models.reference_protocol import (
+        build_discriminator)
+
+    cfg = GanConfig()
+    cfg.train.use_gpu = False
+    torch.manual_seed(3)
+    g1 = build_discriminator(cfg)
+    torch.manual_seed(3)
+    g2 = build_discriminator(cfg)
+    g = torch.Generator().manual_seed(9)
+    ds = DataSet(torch.rand(4, 784, generator=g), torch.ones(4, 1))
+    l1 = g1.fit(ds)
+    l2 = g2.fit_averaged([ds])
+    assert abs(l1 - l2) < 1e-7
+    assert torch.allclose(g1.params_flat(), g2.params_flat(), atol=1e-7)
