@@ -29,6 +29,7 @@ def init_distributed(backend: str = "auto", timeout_s: int = 600):
     device = torch.device("cuda", local_rank % max(1, torch.cuda.device_count())) \
         if cuda else torch.device("cpu")
     if world > 1 and not dist.is_initialized():
+        backend = os.environ.get("GDLJ_BACKEND", backend)
         if backend == "auto":
             backend = "nccl" if cuda else "gloo"
         if cuda and backend == "nccl":
