@@ -104,6 +104,15 @@ class GanTrainer:
         self._soft_fake = None
         self._noise_std = std
         self.it = 0
+        # D-step batching: separate real/fake forwards (default; per-half
+        # BatchNorm statistics, the conventional DCGAN recipe and the
+        # reference's semantics) vs one concatenated 2n forward
+        # (GDLJ_D_CONCAT=1: half the D launches at twice the GEMM M,
+        # mixed-batch BN statistics — a legitimate alternative recipe,
+        # kept opt-in pending a quality + throughput A/B).
+        import os as _os
+
+        self._d_concat = _os.environ.get("GDLJ_D_CONCAT") == "1"
         # hipGraph capture of the whole training step (one replay per step;
         # removes launch/dispatch host overhead). GPU-only; falls back to
         # eager if capture fails.
@@ -158,10 +167,17 @@ class GanTrainer:
         # ---- D step -------------------------------------------------
         self.dis.updater.zero_grad()
         self.d_reducer.prepare()
-        d_real = self.dis(real)
-        d_fake = self.dis(fake.detach())
-        loss_d = OF.bce_with_logits_loss(d_real, soft_real) + \
-            OF.bce_with_logits_loss(d_fake, soft_fake)
+        if self._d_concat:
+            both = torch.cat([real, fake.detach()], dim=0)
+            d_all = self.dis(both)
+            lab = torch.cat([soft_real, soft_fake], dim=0)
+            # x2 keeps the gradient scale of mean(real)+mean(fake)
+            loss_d = OF.bce_with_logits_loss(d_all, lab) * 2
+        else:
+            d_real = self.dis(real)
+            d_fake = self.dis(fake.detach())
+            loss_d = OF.bce_with_logits_loss(d_real, soft_real) + \
+                OF.bce_with_logits_loss(d_fake, soft_fake)
         loss_d.backward()
         self.d_reducer.finish()
         self.dis.updater.step()
