@@ -61,6 +61,8 @@ extern "C" {
 int launch_gemm_tn(const void*, const void*, void*, float*, const float*,
                    int, int, int, long, long, int, float, float*,
                    hipStream_t);
+int launch_gemm_tn_8p_tweak(int, const void*, const void*, void*, int, int,
+                            int, long, long, hipStream_t);
 int launch_gemm_tn_gather(const void*, const void*, void*, const float*,
                           int, int, int, long, int, float, ConvGather,
                           const void*, float*, hipStream_t);
@@ -869,6 +871,19 @@ torch::Tensor conv_fwd_implicit_fp8(
   return y;
 }
 
+torch::Tensor gemm_tn_8p_tweak(torch::Tensor A, torch::Tensor B,
+                               int64_t tweak) {
+  check_bf16(A, "A");
+  check_bf16(B, "B");
+  int64_t M = A.size(0), N = B.size(0), K = A.size(1);
+  TORCH_CHECK(B.size(1) == K && K % 64 == 0);
+  torch::Tensor C = torch::empty({M, N}, A.options());
+  launch_gemm_tn_8p_tweak((int)tweak, A.data_ptr(), B.data_ptr(),
+                          C.data_ptr(), (int)M, (int)N, (int)K, K, K,
+                          cur_stream());
+  return C;
+}
+
 // ------------------------------------------------------------------ optim
 void fused_adam(torch::Tensor param, torch::Tensor grad,
                 c10::optional<torch::Tensor> master, torch::Tensor m,
@@ -917,6 +932,8 @@ void fused_rmsprop(torch::Tensor param, torch::Tensor grad,
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("gemm_tn", &gemm_tn, "C = act(A.B^T + bias) bf16 MFMA");
+  mod.def("gemm_tn_8p_tweak", &gemm_tn_8p_tweak,
+          "within-probe A/B entry for 8p kernel variants");
   mod.def("gemm_nt", &gemm_nt, "C = A^T.B (contraction over rows) fp32 out");
   mod.def("conv_fwd_implicit", &conv_fwd_implicit,
           "implicit-GEMM conv forward (gathered im2col A)");

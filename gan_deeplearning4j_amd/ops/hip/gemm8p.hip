@@ -243,7 +243,11 @@ DEV_INLINE bf16x8 frag(const char* opb, int rowblk, int kc, int fr, int fq) {
 
 }  // namespace p8
 
-template <bool GATHER_A, int SWZ, bool DEEP, int BNT = 256>
+// TWEAK bits (within-probe A/B candidates, tools/ab_gemm8p.py):
+//   1 = partial s_waitcnt lgkmcnt(8) before phase-1's first barrier
+//   2 = static young-half setprio instead of per-cluster flips (T5 static)
+//   4 = n-major XCD grid decomposition (B-panel L2 affinity)
+template <bool GATHER_A, int SWZ, bool DEEP, int BNT = 256, int TWEAK = 0>
 __global__ __launch_bounds__(512, 2) void gemm_tn_8p(
     const unsigned short* __restrict__ A, const unsigned short* __restrict__ B,
     unsigned short* __restrict__ C, const float* __restrict__ bias, int M,
@@ -264,8 +268,18 @@ __global__ __launch_bounds__(512, 2) void gemm_tn_8p(
     int xcd = bid % 8, idx = bid / 8;
     bid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + idx;
   }
-  const int m0 = (bid % gridDim.x) * BM;
-  const int n0 = (bid / gridDim.x) * BNT;
+  const int m0 = (TWEAK & 4) ? (bid / gridDim.y) * BM
+                             : (bid % gridDim.x) * BM;
+  const int n0 = (TWEAK & 4) ? (bid % gridDim.y) * BNT
+                             : (bid / gridDim.x) * BNT;
+
+  if (TWEAK & 2) {
+    // one-time priority for the younger dispatch half (waves 4-7);
+    // condition must be provably wave-uniform (readfirstlane) or the
+    // s_setprio is emitted unconditionally under exec masking
+    if (__builtin_amdgcn_readfirstlane(threadIdx.x) >= 256)
+      __builtin_amdgcn_s_setprio(1);
+  }
 
   const int wid = threadIdx.x >> 6;
   const int lane = threadIdx.x & 63;
@@ -333,9 +347,9 @@ __global__ __launch_bounds__(512, 2) void gemm_tn_8p(
   __builtin_amdgcn_s_barrier();                                              \
   asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");                         \
   __builtin_amdgcn_sched_barrier(0);                                         \
-  __builtin_amdgcn_s_setprio(1);                                             \
+  if (!(TWEAK & 2)) __builtin_amdgcn_s_setprio(1);                           \
   P8_QUAD(MIH, NIH, BREG);                                                   \
-  __builtin_amdgcn_s_setprio(0);                                             \
+  if (!(TWEAK & 2)) __builtin_amdgcn_s_setprio(0);                           \
   TAILWAIT;                                                                  \
   __builtin_amdgcn_s_barrier()
 
@@ -354,6 +368,8 @@ __global__ __launch_bounds__(512, 2) void gemm_tn_8p(
       blo[ni][1] = frag<SWZ>(Bb, wc * 4 + ni, 1, fr, fq);
     }
     stage_a(t + 1, 1);
+    if (TWEAK & 1)  // drain most of the 12-read burst before the barrier
+      asm volatile("s_waitcnt lgkmcnt(8)" ::: "memory");
     // DEEP: force R3(t) (phase-2's bhi source) landed at this phase's tail
     P8_BAR_MFMA(0, 0, blo,
                 if (DEEP && BNT == 256)
@@ -388,9 +404,9 @@ __global__ __launch_bounds__(512, 2) void gemm_tn_8p(
       asm volatile("s_waitcnt vmcnt(3)" ::: "memory");
     }
     __builtin_amdgcn_s_barrier();
-    __builtin_amdgcn_s_setprio(1);
+    if (!(TWEAK & 2)) __builtin_amdgcn_s_setprio(1);
     P8_QUAD(1, 1, bhi);
-    __builtin_amdgcn_s_setprio(0);
+    if (!(TWEAK & 2)) __builtin_amdgcn_s_setprio(0);
     __builtin_amdgcn_s_barrier();
   }
 #undef P8_BAR_MFMA
@@ -578,6 +594,42 @@ int launch_gemm_tn_8p(const void* A, const void* B, void* C,
     else P8_LAUNCH(false, 2, false, 256);
   }
   #undef P8_LAUNCH
+  return (int)grid.x;
+}
+
+// Within-probe A/B entry (tools/ab_gemm8p.py): plain bf16 TN at a fixed
+// variant id; co-compiled instantiations keep codegen context shared so
+// interleaved deltas are reliable (guide rule 19/24).
+int launch_gemm_tn_8p_tweak(int tweak, const void* A, const void* B, void* C,
+                            int M, int N, int K, long lda, long ldb,
+                            hipStream_t s) {
+  (void)p8_enabled();  // ensure LDS attributes are set
+  constexpr int L256 = 2 * (p8::OPTA + 256 * p8::BK * 2);
+  dim3 grid(ceil_div(M, p8::BM), ceil_div(N, 256));
+  dim3 blk(512);
+  ConvGather dummy{};
+  #define P8_TW(T)                                                           \
+    do {                                                                     \
+      (void)hipFuncSetAttribute(                                             \
+          reinterpret_cast<const void*>(&gemm_tn_8p<false, 2, false, 256,    \
+                                                    T>),                     \
+          hipFuncAttributeMaxDynamicSharedMemorySize, L256);                 \
+      hipLaunchKernelGGL((gemm_tn_8p<false, 2, false, 256, T>), grid, blk,   \
+                         L256, s, (const unsigned short*)A,                  \
+                         (const unsigned short*)B, (unsigned short*)C,       \
+                         nullptr, M, N, K, lda, ldb, 0, 0.0f, dummy,         \
+                         nullptr);                                           \
+    } while (0)
+  switch (tweak) {
+    case 1: P8_TW(1); break;
+    case 2: P8_TW(2); break;
+    case 3: P8_TW(3); break;
+    case 4: P8_TW(4); break;
+    case 5: P8_TW(5); break;
+    case 7: P8_TW(7); break;
+    default: P8_TW(0); break;
+  }
+  #undef P8_TW
   return (int)grid.x;
 }
 
