@@ -631,6 +631,16 @@ int launch_gemm_tn_8p_tweak(int tweak, const void* A, const void* B, void* C,
     case 4: P8_TW(4); break;
     case 5: P8_TW(5); break;
     case 7: P8_TW(7); break;
+    case 8:  // DEEP pipeline re-test under the static-priority regime
+      (void)hipFuncSetAttribute(
+          reinterpret_cast<const void*>(&gemm_tn_8p<false, 2, true, 256, 2>),
+          hipFuncAttributeMaxDynamicSharedMemorySize, L256);
+      hipLaunchKernelGGL((gemm_tn_8p<false, 2, true, 256, 2>), grid, blk,
+                         L256, s, (const unsigned short*)A,
+                         (const unsigned short*)B, (unsigned short*)C,
+                         nullptr, M, N, K, lda, ldb, 0, 0.0f, dummy,
+                         nullptr);
+      break;
     default: P8_TW(0); break;
   }
   #undef P8_TW
