@@ -696,6 +696,9 @@ __global__ __launch_bounds__(NT_THREADS, 2) void gemm_nt_core(
 
   const int m0 = blockIdx.x * NT_BM;
   const int n0 = blockIdx.y * NT_BN;
+  // (static young-half wave priority measured NEUTRAL here — at 2
+  // blocks/CU the SIMDs arbitrate across blocks, unlike the 1-block/CU
+  // 8p template where it is +4.9-8.9%; tools/ab_gemm8p.py)
   const int wid = threadIdx.x >> 6;
   const int lane = threadIdx.x & 63;
   const int wr = wid >> 2, wc = wid & 3;
