@@ -380,6 +380,124 @@ __global__ __launch_bounds__(256, 2) void gemm_tn_core(
 }
 
 // ---------------------------------------------------------------------------
+// Single-shot TN for K in {64, 128} (the K-thin dcol/dense family):
+// every K-tile is staged ONCE up front (no double-buffer loop, no
+// per-tile vmcnt(0)+barrier drain pairs) and the block runs
+// stage -> one wait -> MFMA -> epilogue.  Same 128x128 tile / 4 waves /
+// 2 blocks per CU geometry as gemm_tn_core; these shapes are bound by
+// the C-write stream + per-block overhead, so the win is structural
+// overhead removal, not MFMA scheduling.
+// ---------------------------------------------------------------------------
+template <int KT>  // 64 or 128
+__global__ __launch_bounds__(256, 2) void gemm_tn_kshort(
+    const unsigned short* __restrict__ A, const unsigned short* __restrict__ B,
+    unsigned short* __restrict__ C, const float* __restrict__ bias, int M,
+    int N, long lda, long ldb, int act, float slope) {
+  constexpr int NT = KT / TN_BK;  // K-tiles
+  __shared__ __attribute__((aligned(128))) char lds[2 * NT * TN_TILE_B];
+  auto abuf = [&](int t) -> char* { return lds + t * TN_TILE_B; };
+  auto bbuf = [&](int t) -> char* { return lds + (NT + t) * TN_TILE_B; };
+
+  int nwgx = gridDim.x;
+  int bidx = blockIdx.x;
+  if (nwgx >= 8) {
+    int q = nwgx / 8, r = nwgx % 8;
+    int xcd = bidx % 8, idx = bidx / 8;
+    bidx = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + idx;
+  }
+  const int m0 = bidx * TN_BM;
+  const int n0 = blockIdx.y * TN_BN;
+  const int wid = threadIdx.x >> 6;
+  const int lane = threadIdx.x & 63;
+  const int wr = wid >> 1, wc = wid & 1;
+  const int fr = lane & 15, fq = lane >> 4;
+
+  f32x4 acc[4][4];
+  #pragma unroll
+  for (int i = 0; i < 4; ++i)
+    #pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  #pragma unroll
+  for (int t = 0; t < NT; ++t) {
+    tn_stage(A, m0, M, lda, t * TN_BK, abuf(t));
+    tn_stage(B, n0, N, ldb, t * TN_BK, bbuf(t));
+  }
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __syncthreads();
+
+  #pragma unroll
+  for (int t = 0; t < NT; ++t) {
+    const char* Al = abuf(t);
+    const char* Bl = bbuf(t);
+    #pragma unroll
+    for (int kc = 0; kc < 2; ++kc) {
+      bf16x8 a[4], b[4];
+      #pragma unroll
+      for (int mi = 0; mi < 4; ++mi)
+        a[mi] = tn_frag(Al, wr * 64 + mi * 16 + fr, kc * 4 + fq);
+      #pragma unroll
+      for (int ni = 0; ni < 4; ++ni)
+        b[ni] = tn_frag(Bl, wc * 64 + ni * 16 + fr, kc * 4 + fq);
+      #pragma unroll
+      for (int mi = 0; mi < 4; ++mi)
+        #pragma unroll
+        for (int ni = 0; ni < 4; ++ni)
+          acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a[mi], b[ni], acc[mi][ni], 0, 0, 0);
+    }
+  }
+
+  if ((N & 7) == 0) {
+    unsigned short* ctile = (unsigned short*)lds;  // [128][128] bf16
+    __syncthreads();
+    #pragma unroll
+    for (int mi = 0; mi < 4; ++mi) {
+      #pragma unroll
+      for (int ni = 0; ni < 4; ++ni) {
+        int lc = wc * 64 + ni * 16 + fr;
+        float bv = bias != nullptr ? bias[min(n0 + lc, N - 1)] : 0.f;
+        #pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          int lr = wr * 64 + mi * 16 + fq * 4 + r;
+          ctile[lr * 128 + lc] = f2bf(act_fwd(acc[mi][ni][r] + bv, act,
+                                              slope));
+        }
+      }
+    }
+    __syncthreads();
+    const int t = threadIdx.x;
+    #pragma unroll
+    for (int i = 0; i < 8; ++i) {
+      int piece = i * 256 + t;
+      int row = piece >> 4;
+      int seg = piece & 15;
+      int grow = m0 + row;
+      int gcol = n0 + seg * 8;
+      if (grow < M && gcol + 8 <= N)
+        *(s16x8*)(&C[(long)grow * N + gcol]) =
+            *(const s16x8*)(ctile + row * 128 + seg * 8);
+    }
+    return;
+  }
+  #pragma unroll
+  for (int mi = 0; mi < 4; ++mi)
+    #pragma unroll
+    for (int ni = 0; ni < 4; ++ni) {
+      int col = n0 + wc * 64 + ni * 16 + fr;
+      if (col >= N) continue;
+      float bv = bias != nullptr ? bias[col] : 0.f;
+      #pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int row = m0 + wr * 64 + mi * 16 + fq * 4 + r;
+        if (row >= M) continue;
+        C[(long)row * N + col] = f2bf(act_fwd(acc[mi][ni][r] + bv, act,
+                                              slope));
+      }
+    }
+}
+
+// ---------------------------------------------------------------------------
 // 256x128x64 TN variant (GDLJ_TN256): +37% arithmetic intensity per staged
 // byte (2M MACs per 48 KiB vs 1M per 32 KiB) at the SAME 8-wave/CU
 // occupancy — one 512-thread block per CU with 96 KiB dynamic LDS
@@ -815,9 +933,33 @@ static int t2_enabled() {
   return v;
 }
 
+static int kshort_enabled() {
+  static int v = -1;
+  if (v < 0) {
+    const char* e = getenv("GDLJ_KSHORT");
+    v = (e != nullptr && e[0] == '0') ? 0 : 1;
+  }
+  return v;
+}
+
 int launch_gemm_tn(const void* A, const void* B, void* C_bf16, float* C_f32,
                    const float* bias, int M, int N, int K, long lda, long ldb,
                    int act, float slope, float* bn_part, hipStream_t s) {
+  if (C_bf16 != nullptr && bn_part == nullptr && (K == 64 || K == 128) &&
+      kshort_enabled()) {
+    dim3 grid(ceil_div(M, TN_BM), ceil_div(N, TN_BN));
+    if (K == 64)
+      hipLaunchKernelGGL((gemm_tn_kshort<64>), grid, dim3(256), 0, s,
+                         (const unsigned short*)A, (const unsigned short*)B,
+                         (unsigned short*)C_bf16, bias, M, N, lda, ldb, act,
+                         slope);
+    else
+      hipLaunchKernelGGL((gemm_tn_kshort<128>), grid, dim3(256), 0, s,
+                         (const unsigned short*)A, (const unsigned short*)B,
+                         (unsigned short*)C_bf16, bias, M, N, lda, ldb, act,
+                         slope);
+    return (int)grid.x;
+  }
   if (C_bf16 != nullptr && bn_part == nullptr &&
       gemm_tn_8p_eligible(M, N, K)) {
     ConvGather dummy{};
