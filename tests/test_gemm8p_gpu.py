@@ -174,3 +174,16 @@ def test_gemm8p_n128_conv_fwd():
         F.conv2d(x.float().cpu(), w.float().cpu(), None, stride=stride,
                  padding=pad), 0.2)
     assert relerr(y, yr) < 0.03
+
+
+@pytest.mark.parametrize("m,n,k", [(32768, 1024, 128), (32768, 64, 64),
+                                   (5000, 200, 128)])
+def test_gemm_kshort_shapes(m, n, k):
+    """Single-shot K-short TN (K in {64,128}) vs fp32 reference."""
+    e = ext()
+    A, B = mk((m, k), 31, 0.5), mk((n, k), 32, 0.5)
+    bias = torch.randn(n, device=DEV)
+    C = e.gemm_tn(A, B, bias, 3, 0.2, False)  # lrelu epilogue
+    ref = torch.nn.functional.leaky_relu(
+        A.float() @ B.float().t() + bias.float(), 0.2)
+    assert relerr(C, ref.cpu()) < 0.02
