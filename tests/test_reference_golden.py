@@ -116,3 +116,31 @@ def test_fit_averaged_single_partition_equals_fit():
     l2 = g2.fit_averaged([ds])
     assert abs(l1 - l2) < 1e-7
     assert torch.allclose(g1.params_flat(), g2.params_flat(), atol=1e-7)
+
+
+FAST_GOLDEN = [
+    # (loss_d, loss_g) per step; mlp_tabular_cpu, hidden=32, data seed 777
+    (1.471612, 0.736157),
+    (1.433109, 0.679899),
+    (1.466658, 0.652197),
+    (1.486723, 0.627431),
+]
+
+
+def test_fast_trainer_golden_losses():
+    """The flagship GanTrainer pinned to committed loss values (the
+    reference-protocol twin of this gate lives above). Catches numerics
+    regressions in the fast path's loss/update/label-softening stack."""
+    from gan_deeplearning4j_amd.models import build_mlp_gan
+    from gan_deeplearning4j_amd.train import GanTrainer
+
+    cfg = preset("mlp_tabular_cpu")
+    cfg.train.use_gpu = False
+    torch.manual_seed(cfg.train.seed)
+    gen, dis = build_mlp_gan(cfg, hidden=32)
+    tr = GanTrainer(gen, dis, cfg, device=torch.device("cpu"))
+    g = torch.Generator().manual_seed(777)
+    for want_d, want_g in FAST_GOLDEN:
+        out = tr.step(torch.rand(16, cfg.data.num_features, generator=g))
+        assert abs(float(out["loss_d"]) - want_d) < 5e-4, out
+        assert abs(float(out["loss_g"]) - want_g) < 5e-4, out
