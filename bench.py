@@ -72,6 +72,11 @@ def main():
     g = torch.Generator().manual_seed(1234 + rank)
     real = (torch.rand(per_gpu_batch, m.image_channels, m.image_height,
                        m.image_width, generator=g) * 2 - 1).to(device, dtype)
+    if use_gpu and args.arch != "mlp":
+        # channels-last once up front: the NHWC conv path then consumes
+        # the batch copy-free every step (a user pipeline would store
+        # images channels-last for the same reason)
+        real = real.contiguous(memory_format=torch.channels_last)
     if args.arch == "mlp":
         real = torch.rand(per_gpu_batch, cfg.data.num_features,
                           generator=g).to(device, dtype)

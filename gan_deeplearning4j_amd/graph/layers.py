@@ -279,13 +279,27 @@ class ActivationLayer(BaseLayer):
 
 
 class FeedForwardToCnnPreProcessor(BaseLayer):
-    """[N, H*W*C] -> [N, C, H, W] (reference Java:200)."""
+    """[N, H*W*C] -> [N, C, H, W] (reference Java:200).
 
-    def __init__(self, height: int, width: int, channels: int):
+    channels_last=True interprets the flat features in (H, W, C) order and
+    returns a channels-last VIEW — the downstream NHWC conv path then
+    consumes it copy-free (the flat ordering is a fixed permutation of
+    untrained dense columns, so the model family is identical; measured:
+    the C-major reshape forced a slow strided permute-copy of the full
+    activation per step). The reference-protocol graphs keep DL4J's
+    C-major order (default False)."""
+
+    def __init__(self, height: int, width: int, channels: int,
+                 channels_last: bool = False):
         super().__init__()
         self.height, self.width, self.channels = height, width, channels
+        self.channels_last = channels_last
 
     def forward(self, x):
+        if self.channels_last:
+            nhwc = x.reshape(x.shape[0], self.height, self.width,
+                             self.channels)
+            return nhwc.permute(0, 3, 1, 2)
         return x.reshape(x.shape[0], self.channels, self.height, self.width)
 
     def out_shape(self, in_shape):

@@ -48,7 +48,7 @@ _LAYER_FIELDS = {
     "MaxPool2dLayer": ("kernel", "stride"),
     "Upsampling2dLayer": ("scale",),
     "ActivationLayer": ("activation", "slope"),
-    "FeedForwardToCnnPreProcessor": ("height", "width", "channels"),
+    "FeedForwardToCnnPreProcessor": ("height", "width", "channels", "channels_last"),
     "CnnToFeedForwardPreProcessor": ("channels_last",),
     "ReshapeVertex": ("shape",),
     "MergeVertex": (),
@@ -78,7 +78,8 @@ _CTOR_ARGS = {
     "ActivationLayer": lambda d: L.ActivationLayer(d["activation"],
                                                    d.get("slope", 0.2)),
     "FeedForwardToCnnPreProcessor": lambda d: L.FeedForwardToCnnPreProcessor(
-        d["height"], d["width"], d["channels"]),
+        d["height"], d["width"], d["channels"],
+        d.get("channels_last", False)),
     "CnnToFeedForwardPreProcessor": lambda d: L.CnnToFeedForwardPreProcessor(
         d.get("channels_last", False)),
     "ReshapeVertex": lambda d: L.ReshapeVertex(*d["shape"]),

@@ -38,7 +38,8 @@ def _gen_graph(cfg: GanConfig, stages: list[int], s0: int) -> ComputationGraph:
     gb.add_layer("g_dense_0", DenseLayer(m.z_size, s0 * s0 * c0, "identity", lr),
                  "g_input")
     gb.add_layer("g_bn_0", BatchNormLayer(c0, lr=lr), "g_dense_0",
-                 preprocessor=FeedForwardToCnnPreProcessor(s0, s0, c0))
+                 preprocessor=FeedForwardToCnnPreProcessor(s0, s0, c0,
+                                                           channels_last=True))
     prev = "g_bn_0"
     # hidden ConvT stages: stages[i] -> stages[i+1], spatial x2
     for i in range(len(stages) - 1):

@@ -35,7 +35,8 @@ def main():
     pool = max(4096, batch)
     imgs, _ = pixel_lattice_images(pool, m.image_height, m.image_width,
                                    m.image_channels, seed=7)
-    real_all = (imgs * 2 - 1).to("cuda:0", torch.bfloat16)
+    real_all = (imgs * 2 - 1).to("cuda:0", torch.bfloat16).contiguous(
+        memory_format=torch.channels_last)
     t0 = time.time()
     out = {}
     for i in range(steps):
