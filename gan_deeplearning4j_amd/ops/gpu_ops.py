@@ -113,6 +113,32 @@ def deposit_act_fused(t: torch.Tensor, payload) -> None:
 def take_act_fused(t: torch.Tensor):
     return _act_fused_chan.pop(t.data_ptr(), None)
 
+
+# Channel-pad pass-through side channel: NHWC producers whose true channel
+# count C is below the pad C8 (the 3-channel image boundary) return a
+# trimmed [..., :C] VIEW and deposit the PADDED buffer here (pad channels
+# provably zero: zero weight columns + tanh/identity epilogue). The next
+# conv consumer takes the padded buffer directly instead of re-padding —
+# removing a slow strided trim-copy AND a re-pad per boundary crossing.
+# Entries hold the padded tensor (keeps its storage from allocator reuse,
+# so a data_ptr key can never go stale); FIFO-capped so at most a few
+# boundary buffers stay pinned. Peek semantics: the same fake batch is
+# consumed by both the D-step and the G-step forward.
+_chan_pad: dict = {}
+
+
+def deposit_chan_pad(trimmed: torch.Tensor, padded: torch.Tensor) -> None:
+    if len(_chan_pad) >= 4:
+        _chan_pad.pop(next(iter(_chan_pad)))
+    _chan_pad[trimmed.data_ptr()] = padded
+
+
+def take_chan_pad(t: torch.Tensor, c8: int):
+    p = _chan_pad.get(t.data_ptr())
+    if p is not None and p.shape[-1] == c8 and p.shape[0] == t.shape[0]:
+        return p
+    return None
+
 # Strided dgrad/convT-fwd algorithm choice: parity-decomposed gathered
 # GEMMs vs dcol+col2im. Measured on DCGAN-64: dcol wins by ~4% (the
 # parity gather re-reads the source once per tap, so traffic is a wash
@@ -308,12 +334,15 @@ def _conv_out(h, k, stride, pad):
 
 
 def _pad_channels(xh: torch.Tensor, c8: int) -> torch.Tensor:
-    """[N,H,W,C] -> [N,H,W,c8] zero-padded channels (gather needs C%8==0)."""
+    """[N,H,W,C] -> [N,H,W,c8] zero-padded channels (gather needs C%8==0).
+    new_empty + two slice writes beats new_zeros + copy (the full-buffer
+    zero fill re-wrote the data region for nothing)."""
     n, h, w, c = xh.shape
     if c == c8:
         return xh
-    out = xh.new_zeros(n, h, w, c8)
+    out = xh.new_empty(n, h, w, c8)
     out[..., :c] = xh
+    out[..., c:] = 0
     return out
 
 
@@ -327,7 +356,9 @@ class _Conv2d(torch.autograd.Function):
         Ho, Wo = _conv_out(H, R, stride, pad), _conv_out(W, S, stride, pad)
         # channel dim padded to 8 so every conv runs the implicit path
         C8 = (C + 7) // 8 * 8
-        xh = _pad_channels(_nhwc(x), C8)
+        xh = take_chan_pad(x, C8)
+        if xh is None:
+            xh = _pad_channels(_nhwc(x), C8)
         kpad = _rup64(R * S * C8)
 
         def build_wp():
@@ -509,9 +540,14 @@ class _Conv2d(torch.autograd.Function):
                 else:
                     dxh = ext.col2im(dcol, N, H, W, C8, Ho, Wo, R, S,
                                      stride, pad, rsc8, None, 0, 0.0)
-                    if C8 != C:
-                        dxh = dxh[..., :C].contiguous()
-                    dx = _as_nchw_view(dxh).to(ctx.dtypes[0])
+                    if C8 != C and dxh.dtype == ctx.dtypes[0]:
+                        trimmed = dxh[..., :C]
+                        dx = _as_nchw_view(trimmed)
+                        deposit_chan_pad(trimmed, dxh)
+                    else:
+                        if C8 != C:
+                            dxh = dxh[..., :C].contiguous()
+                        dx = _as_nchw_view(dxh).to(ctx.dtypes[0])
         if ctx.has_bias and ctx.needs_input_grad[2] and db is None:
             db = ext.col_sum(dpre).to(ctx.dtypes[2])
         return dx, dw, db, None, None, None, None, None, None
@@ -624,7 +660,12 @@ class _ConvTranspose2d(torch.autograd.Function):
                 yh = ext.col2im(col, N, Ho, Wo, Co8, Hi, Wi, R, S, stride,
                                 pad, R * S * Co8, bias, act, slope)
                 if Co8 != Cout:
-                    yh = yh[..., :Cout].contiguous()
+                    if act in (0, 1):  # identity/tanh: act(0 + 0-bias) == 0
+                        yh_pad = yh
+                        yh = yh_pad[..., :Cout]
+                        deposit_chan_pad(yh, yh_pad)
+                    else:
+                        yh = yh[..., :Cout].contiguous()
         ctx.save_for_backward(xh, yh)
         ctx.geom = (N, Cin, Hi, Wi, Cout, R, S, Ho, Wo, stride, pad)
         ctx.act, ctx.slope = act, slope
@@ -645,31 +686,55 @@ class _ConvTranspose2d(torch.autograd.Function):
         rsco = R * S * Cout
         rscop = _rup64(rsco)
         fused = take_act_fused(dy)
-        dyh = _nhwc(dy)                       # [N,Ho,Wo,Cout]
         db = None
         want_bias = ctx.has_bias and ctx.needs_input_grad[2]
-        if fused is not None and fused[0] == ctx.act and fused[1] == Cout:
-            # consumer BN already applied act backward + bias reduction
-            dpre_img = dyh
-            if want_bias and fused[2] is not None:
-                db = fused[2].to(ctx.dtypes[2])
-        elif ctx.act and want_bias and Cout % 8 == 0:
-            dpre, db_f = ext.act_bwd_bias(dyh.reshape(-1, Cout),
-                                          yh.reshape(-1, Cout), ctx.act,
-                                          ctx.slope)
-            db = db_f.to(ctx.dtypes[2])
-            dpre_img = dpre.view(N, Ho, Wo, Cout)
-        elif ctx.act:
-            dpre = ext.act_bwd(dyh.reshape(-1, Cout),
-                               yh.reshape(-1, Cout), ctx.act, ctx.slope)
-            dpre_img = dpre.view(N, Ho, Wo, Cout)
-        else:
-            dpre_img = dyh
-        dpre_img = dpre_img.contiguous()
-        # forward-gather geometry over dOut (chunks of 8 channels):
-        # pad Cout to 8 if needed (g_out has Cout = image channels = 3)
         Co8 = (Cout + 7) // 8 * 8
-        dpre8 = _pad_channels(dpre_img, Co8)
+        dpre8 = None
+        # 3-channel image boundary pass-through: when the downstream
+        # conv's dgrad deposited the PADDED dy and the forward deposited
+        # the padded y, act backward runs at Co8 width (the pad lanes are
+        # exact zeros through tanh/identity) and no trim/re-pad copies
+        # are paid on either side of the boundary.
+        if fused is None and Co8 != Cout:
+            dy_pad = take_chan_pad(dy, Co8)
+            y_pad = take_chan_pad(yh, Co8) if dy_pad is not None else None
+            if dy_pad is not None and (y_pad is not None or ctx.act == 0):
+                if ctx.act:
+                    if want_bias:
+                        dp, db_f = ext.act_bwd_bias(
+                            dy_pad.reshape(-1, Co8),
+                            y_pad.reshape(-1, Co8), ctx.act, ctx.slope)
+                        db = db_f[:Cout].to(ctx.dtypes[2])
+                    else:
+                        dp = ext.act_bwd(dy_pad.reshape(-1, Co8),
+                                         y_pad.reshape(-1, Co8), ctx.act,
+                                         ctx.slope)
+                    dpre8 = dp.view(N, Ho, Wo, Co8)
+                else:
+                    dpre8 = dy_pad
+        if dpre8 is None:
+            dyh = _nhwc(dy)                   # [N,Ho,Wo,Cout]
+            if fused is not None and fused[0] == ctx.act and \
+                    fused[1] == Cout:
+                # consumer BN already applied act backward + bias grad
+                dpre_img = dyh
+                if want_bias and fused[2] is not None:
+                    db = fused[2].to(ctx.dtypes[2])
+            elif ctx.act and want_bias and Cout % 8 == 0:
+                dpre, db_f = ext.act_bwd_bias(
+                    dyh.reshape(-1, Cout),
+                    yh.reshape(-1, Cout).contiguous(), ctx.act, ctx.slope)
+                db = db_f.to(ctx.dtypes[2])
+                dpre_img = dpre.view(N, Ho, Wo, Cout)
+            elif ctx.act:
+                dpre = ext.act_bwd(dyh.reshape(-1, Cout),
+                                   yh.reshape(-1, Cout).contiguous(),
+                                   ctx.act, ctx.slope)
+                dpre_img = dpre.view(N, Ho, Wo, Cout)
+            else:
+                dpre_img = dyh
+            dpre_img = dpre_img.contiguous()
+            dpre8 = _pad_channels(dpre_img, Co8)
         npq = N * Hi * Wi
         dx = dw = None
         if ctx.needs_input_grad[0]:
@@ -712,7 +777,8 @@ class _ConvTranspose2d(torch.autograd.Function):
             dw = (dw2a[:rsco8, :Cin].reshape(R, S, Co8, Cin)[:, :, :Cout]
                   .permute(3, 2, 0, 1).contiguous().to(ctx.dtypes[1]))
         if ctx.has_bias and ctx.needs_input_grad[2] and db is None:
-            db = ext.col_sum(dpre_img.reshape(-1, Cout)).to(ctx.dtypes[2])
+            db = ext.col_sum(dpre8.reshape(-1, Co8))[:Cout].to(
+                ctx.dtypes[2])
         return dx, dw, db, None, None, None, None, None
 
 

@@ -384,3 +384,56 @@ def test_reference_protocol_iteration_gpu(tmp_path):
     import math
 
     assert math.isfinite(out["loss_d"]) and math.isfinite(out["loss_g"])
+
+
+def test_three_channel_boundary_chain():
+    """convT (Cout=3, tanh, bias) -> conv (Cin=3) fwd+bwd vs torch: the
+    padded-channel pass-through (deposit/take side channel) must be
+    numerically invisible across the image boundary — the exact G-head ->
+    D-conv1 wiring of the GAN step, including the act backward run at
+    padded width."""
+    from gan_deeplearning4j_amd.ops import gpu_ops
+
+    N, Cin, H = 8, 32, 8
+    x = mk((N, Cin, H, H), 40, 0.4).requires_grad_(True)
+    wt = mk((Cin, 3, 4, 4), 41, 0.1).requires_grad_(True)
+    bt = torch.randn(3, device=DEV, dtype=torch.bfloat16,
+                     requires_grad=True)
+    wc = mk((16, 3, 5, 5), 42, 0.1).requires_grad_(True)
+    img = gpu_ops.conv_transpose2d(x, wt, bt, 2, 1, "tanh", 0.0)
+    y = gpu_ops.conv2d(img, wc, None, 2, 2, "lrelu", 0.2)
+    gout = mk(y.shape, 43, 0.3)
+    y.backward(gout)
+
+    xr = x.detach().float().cpu().requires_grad_(True)
+    wtr = wt.detach().float().cpu().requires_grad_(True)
+    btr = bt.detach().float().cpu().requires_grad_(True)
+    wcr = wc.detach().float().cpu().requires_grad_(True)
+    imgr = torch.tanh(F.conv_transpose2d(xr, wtr, btr, stride=2, padding=1))
+    yr = torch.nn.functional.leaky_relu(
+        F.conv2d(imgr, wcr, None, stride=2, padding=2), 0.2)
+    yr.backward(gout.float().cpu())
+    assert relerr(y, yr) < 0.04
+    assert relerr(x.grad, xr.grad) < 0.06
+    assert relerr(wt.grad, wtr.grad) < 0.06
+    assert relerr(bt.grad, btr.grad) < 0.06
+    assert relerr(wc.grad, wcr.grad) < 0.06
+
+
+def test_three_channel_boundary_repeated_consumers():
+    """The same 3-channel producer output consumed by TWO conv forwards
+    (the GAN's fake batch goes through D twice) — peek semantics must
+    serve both without corruption."""
+    from gan_deeplearning4j_amd.ops import gpu_ops
+
+    x = mk((8, 16, 8, 8), 44, 0.4)
+    wt = mk((16, 3, 4, 4), 45, 0.1)
+    wc = mk((16, 3, 5, 5), 46, 0.1)
+    img = gpu_ops.conv_transpose2d(x, wt, None, 2, 1, "tanh", 0.0)
+    y1 = gpu_ops.conv2d(img, wc, None, 2, 2, "identity", 0.0)
+    y2 = gpu_ops.conv2d(img.detach(), wc, None, 2, 2, "identity", 0.0)
+    assert torch.equal(y1, y2)
+    imgr = torch.tanh(F.conv_transpose2d(x.float().cpu(), wt.float().cpu(),
+                                         None, stride=2, padding=1))
+    yr = F.conv2d(imgr, wc.float().cpu(), None, stride=2, padding=2)
+    assert relerr(y1, yr) < 0.04
