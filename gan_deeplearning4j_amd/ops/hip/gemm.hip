@@ -392,7 +392,7 @@ template <int KT>  // 64 or 128
 __global__ __launch_bounds__(256, 2) void gemm_tn_kshort(
     const unsigned short* __restrict__ A, const unsigned short* __restrict__ B,
     unsigned short* __restrict__ C, const float* __restrict__ bias, int M,
-    int N, long lda, long ldb, int act, float slope) {
+    int N, long lda, long ldb, int act, float slope, int direct_epi) {
   constexpr int NT = KT / TN_BK;  // K-tiles
   __shared__ __attribute__((aligned(128))) char lds[2 * NT * TN_TILE_B];
   auto abuf = [&](int t) -> char* { return lds + t * TN_TILE_B; };
@@ -448,7 +448,7 @@ __global__ __launch_bounds__(256, 2) void gemm_tn_kshort(
     }
   }
 
-  if ((N & 7) == 0) {
+  if ((N & 7) == 0 && !direct_epi) {
     unsigned short* ctile = (unsigned short*)lds;  // [128][128] bf16
     __syncthreads();
     #pragma unroll
@@ -948,16 +948,21 @@ int launch_gemm_tn(const void* A, const void* B, void* C_bf16, float* C_f32,
   if (C_bf16 != nullptr && bn_part == nullptr && (K == 64 || K == 128) &&
       kshort_enabled()) {
     dim3 grid(ceil_div(M, TN_BM), ceil_div(N, TN_BN));
+    static int depi = -1;
+    if (depi < 0) {
+      const char* e = getenv("GDLJ_KSHORT_EPI");
+      depi = (e != nullptr && e[0] == '1') ? 1 : 0;  // probe: direct stores
+    }
     if (K == 64)
       hipLaunchKernelGGL((gemm_tn_kshort<64>), grid, dim3(256), 0, s,
                          (const unsigned short*)A, (const unsigned short*)B,
                          (unsigned short*)C_bf16, bias, M, N, lda, ldb, act,
-                         slope);
+                         slope, depi);
     else
       hipLaunchKernelGGL((gemm_tn_kshort<128>), grid, dim3(256), 0, s,
                          (const unsigned short*)A, (const unsigned short*)B,
                          (unsigned short*)C_bf16, bias, M, N, lda, ldb, act,
-                         slope);
+                         slope, depi);
     return (int)grid.x;
   }
   if (C_bf16 != nullptr && bn_part == nullptr &&
