@@ -448,7 +448,7 @@ __global__ __launch_bounds__(256, 2) void gemm_tn_kshort(
     }
   }
 
-  if ((N & 7) == 0 && !direct_epi) {
+  if ((N & 7) == 0 && direct_epi != 1) {
     unsigned short* ctile = (unsigned short*)lds;  // [128][128] bf16
     __syncthreads();
     #pragma unroll
@@ -474,9 +474,13 @@ __global__ __launch_bounds__(256, 2) void gemm_tn_kshort(
       int seg = piece & 15;
       int grow = m0 + row;
       int gcol = n0 + seg * 8;
-      if (grow < M && gcol + 8 <= N)
-        *(s16x8*)(&C[(long)grow * N + gcol]) =
-            *(const s16x8*)(ctile + row * 128 + seg * 8);
+      if (grow < M && gcol + 8 <= N) {
+        s16x8 v = *(const s16x8*)(ctile + row * 128 + seg * 8);
+        if (direct_epi == 2)  // probe: nontemporal C stream
+          __builtin_nontemporal_store(v, (s16x8*)(&C[(long)grow * N + gcol]));
+        else
+          *(s16x8*)(&C[(long)grow * N + gcol]) = v;
+      }
     }
     return;
   }
@@ -495,6 +499,130 @@ __global__ __launch_bounds__(256, 2) void gemm_tn_kshort(
                                               slope));
       }
     }
+}
+
+// 256-row stager for the wide K-short tile (2048 chunks per 64-K tile)
+DEV_INLINE void tn_stage256(const unsigned short* __restrict__ g, int row0,
+                            int nrows, long ldk, int k0, char* lds) {
+  const int t = threadIdx.x;
+  const int wid = t >> 6;
+  #pragma unroll
+  for (int i = 0; i < 8; ++i) {
+    int chunk = i * 256 + t;
+    int row = chunk >> 3;
+    int slot = chunk & 7;
+    int gslot = slot ^ (row & 7);
+    int grow = min(row0 + row, nrows - 1);
+    const unsigned short* src = g + (long)grow * ldk + k0 + gslot * 8;
+    char* dst = lds + (i * 256 + wid * 64) * 16;
+    GLDS16(src, dst);
+  }
+}
+
+// Wide K=128 single-shot tile: 256(M) x 64(N), 4 waves each owning a
+// 64x64 panel, 2 blocks/CU (A 64 KiB + B 16 KiB LDS) — half the block
+// count of the 128x128 kshort at the same per-thread register budget,
+// so the per-block stage-latency chain amortizes over 2x the MACs.
+__global__ __launch_bounds__(256, 2) void gemm_tn_kshort_wide(
+    const unsigned short* __restrict__ A, const unsigned short* __restrict__ B,
+    unsigned short* __restrict__ C, const float* __restrict__ bias, int M,
+    int N, long lda, long ldb, int act, float slope) {
+  // 80 KiB dynamic LDS (static __shared__ caps at 64 KiB; the launcher
+  // raises MaxDynamicSharedMemorySize): 2 blocks/CU at exactly 160 KiB
+  extern __shared__ __attribute__((aligned(16))) char lds[];
+  auto abuf = [&](int t) -> char* { return lds + t * 32768; };       // 2x32K
+  auto bbuf = [&](int t) -> char* { return lds + 65536 + t * 8192; };
+
+  int nwgx = gridDim.x;
+  int bidx = blockIdx.x;
+  if (nwgx >= 8) {
+    int q = nwgx / 8, r = nwgx % 8;
+    int xcd = bidx % 8, idx = bidx / 8;
+    bidx = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + idx;
+  }
+  const int m0 = bidx * 256;
+  const int n0 = blockIdx.y * 64;
+  const int wid = threadIdx.x >> 6;
+  const int lane = threadIdx.x & 63;
+  const int fr = lane & 15, fq = lane >> 4;
+
+  f32x4 acc[4][4];
+  #pragma unroll
+  for (int i = 0; i < 4; ++i)
+    #pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  #pragma unroll
+  for (int t = 0; t < 2; ++t) {
+    tn_stage256(A, m0, M, lda, t * 64, abuf(t));
+    // B: 64 rows x 64 k = 512 chunks (2 per thread)
+    const int th = threadIdx.x;
+    #pragma unroll
+    for (int i = 0; i < 2; ++i) {
+      int chunk = i * 256 + th;
+      int row = chunk >> 3;
+      int slot = chunk & 7;
+      int gslot = slot ^ (row & 7);
+      int grow = min(n0 + row, N - 1);
+      const unsigned short* src = B + (long)grow * ldb + t * 64 + gslot * 8;
+      char* dst = bbuf(t) + (i * 256 + wid * 64) * 16;
+      GLDS16(src, dst);
+    }
+  }
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __syncthreads();
+
+  #pragma unroll
+  for (int t = 0; t < 2; ++t) {
+    const char* Al = abuf(t);
+    const char* Bl = bbuf(t);
+    #pragma unroll
+    for (int kc = 0; kc < 2; ++kc) {
+      bf16x8 a[4], b[4];
+      #pragma unroll
+      for (int mi = 0; mi < 4; ++mi)
+        a[mi] = tn_frag(Al, wid * 64 + mi * 16 + fr, kc * 4 + fq);
+      #pragma unroll
+      for (int ni = 0; ni < 4; ++ni)
+        b[ni] = tn_frag(Bl, ni * 16 + fr, kc * 4 + fq);
+      #pragma unroll
+      for (int mi = 0; mi < 4; ++mi)
+        #pragma unroll
+        for (int ni = 0; ni < 4; ++ni)
+          acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a[mi], b[ni], acc[mi][ni], 0, 0, 0);
+    }
+  }
+
+  // epilogue: ctile [256][64] bf16 = 32 KiB
+  unsigned short* ctile = (unsigned short*)lds;
+  __syncthreads();
+  #pragma unroll
+  for (int mi = 0; mi < 4; ++mi) {
+    #pragma unroll
+    for (int ni = 0; ni < 4; ++ni) {
+      int lc = ni * 16 + fr;
+      float bv = bias != nullptr ? bias[min(n0 + lc, N - 1)] : 0.f;
+      #pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int lr = wid * 64 + mi * 16 + fq * 4 + r;
+        ctile[lr * 64 + lc] = f2bf(act_fwd(acc[mi][ni][r] + bv, act, slope));
+      }
+    }
+  }
+  __syncthreads();
+  const int t = threadIdx.x;
+  #pragma unroll
+  for (int i = 0; i < 8; ++i) {
+    int piece = i * 256 + t;  // 2048 pieces = 256 rows x 8 segs
+    int row = piece >> 3;
+    int seg = piece & 7;
+    int grow = m0 + row;
+    int gcol = n0 + seg * 8;
+    if (grow < M && gcol + 8 <= N)
+      *(s16x8*)(&C[(long)grow * N + gcol]) =
+          *(const s16x8*)(ctile + row * 64 + seg * 8);
+  }
 }
 
 // ---------------------------------------------------------------------------
@@ -947,6 +1075,27 @@ int launch_gemm_tn(const void* A, const void* B, void* C_bf16, float* C_f32,
                    int act, float slope, float* bn_part, hipStream_t s) {
   if (C_bf16 != nullptr && bn_part == nullptr && (K == 64 || K == 128) &&
       kshort_enabled()) {
+    static int kwide = -1;
+    if (kwide < 0) {
+      const char* e = getenv("GDLJ_KSHORT_WIDE");
+      kwide = (e != nullptr && e[0] == '0') ? 0 : 1;
+    }
+    if (kwide && K == 128 && (N & 63) == 0 && M >= 16 * 256) {
+      static int attr_done = 0;
+      if (!attr_done) {
+        (void)hipFuncSetAttribute(
+            reinterpret_cast<const void*>(&gemm_tn_kshort_wide),
+            hipFuncAttributeMaxDynamicSharedMemorySize, 5 * 16384);
+        attr_done = 1;
+      }
+      dim3 gridw(ceil_div(M, 256), N / 64);
+      hipLaunchKernelGGL(gemm_tn_kshort_wide, gridw, dim3(256), 5 * 16384,
+                         s, (const unsigned short*)A,
+                         (const unsigned short*)B,
+                         (unsigned short*)C_bf16, bias, M, N, lda, ldb, act,
+                         slope);
+      return (int)gridw.x;
+    }
     dim3 grid(ceil_div(M, TN_BM), ceil_div(N, TN_BN));
     static int depi = -1;
     if (depi < 0) {
