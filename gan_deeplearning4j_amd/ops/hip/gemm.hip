@@ -1075,10 +1075,13 @@ int launch_gemm_tn(const void* A, const void* B, void* C_bf16, float* C_f32,
                    int act, float slope, float* bn_part, hipStream_t s) {
   if (C_bf16 != nullptr && bn_part == nullptr && (K == 64 || K == 128) &&
       kshort_enabled()) {
+    // 256x64 wide tile measured -10% vs the 128x128 kshort (fewer
+    // blocks did NOT amortize: the wide tile halves B-reuse per byte of
+    // LDS and the 80 KiB footprint stiffens scheduling) — opt-in probe.
     static int kwide = -1;
     if (kwide < 0) {
       const char* e = getenv("GDLJ_KSHORT_WIDE");
-      kwide = (e != nullptr && e[0] == '0') ? 0 : 1;
+      kwide = (e != nullptr && e[0] == '1') ? 1 : 0;
     }
     if (kwide && K == 128 && (N & 63) == 0 && M >= 16 * 256) {
       static int attr_done = 0;
