@@ -388,11 +388,12 @@ __global__ __launch_bounds__(256, 2) void gemm_tn_core(
 // the C-write stream + per-block overhead, so the win is structural
 // overhead removal, not MFMA scheduling.
 // ---------------------------------------------------------------------------
-template <int KT>  // 64 or 128
+template <int KT, bool GATHER_A = false>  // KT = 64 or 128
 __global__ __launch_bounds__(256, 2) void gemm_tn_kshort(
     const unsigned short* __restrict__ A, const unsigned short* __restrict__ B,
     unsigned short* __restrict__ C, const float* __restrict__ bias, int M,
-    int N, long lda, long ldb, int act, float slope, int direct_epi) {
+    int N, long lda, long ldb, int act, float slope, int direct_epi,
+    ConvGather ga, const unsigned short* __restrict__ zp) {
   constexpr int NT = KT / TN_BK;  // K-tiles
   __shared__ __attribute__((aligned(128))) char lds[2 * NT * TN_TILE_B];
   auto abuf = [&](int t) -> char* { return lds + t * TN_TILE_B; };
@@ -418,9 +419,14 @@ __global__ __launch_bounds__(256, 2) void gemm_tn_kshort(
     #pragma unroll
     for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
 
+  TnGatherStager gs;
+  if (GATHER_A) gs.init(ga, m0, M);
   #pragma unroll
   for (int t = 0; t < NT; ++t) {
-    tn_stage(A, m0, M, lda, t * TN_BK, abuf(t));
+    if (GATHER_A)
+      gs.stage(A, ga, zp, t * TN_BK, abuf(t));
+    else
+      tn_stage(A, m0, M, lda, t * TN_BK, abuf(t));
     tn_stage(B, n0, N, ldb, t * TN_BK, bbuf(t));
   }
   asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
@@ -475,11 +481,18 @@ __global__ __launch_bounds__(256, 2) void gemm_tn_kshort(
       int grow = m0 + row;
       int gcol = n0 + seg * 8;
       if (grow < M && gcol + 8 <= N) {
+        long crow = grow;
+        if (GATHER_A && ga.mode == 2) {
+          int n2, h2, w2;
+          np_decode(ga, (unsigned)grow, n2, h2, w2);
+          crow = ((long)n2 * ga.oH + h2 * ga.stride + ga.oqh) * ga.oW +
+                 w2 * ga.stride + ga.oqw;
+        }
         s16x8 v = *(const s16x8*)(ctile + row * 128 + seg * 8);
         if (direct_epi == 2)  // probe: nontemporal C stream
-          __builtin_nontemporal_store(v, (s16x8*)(&C[(long)grow * N + gcol]));
+          __builtin_nontemporal_store(v, (s16x8*)(&C[crow * N + gcol]));
         else
-          *(s16x8*)(&C[(long)grow * N + gcol]) = v;
+          *(s16x8*)(&C[crow * N + gcol]) = v;
       }
     }
     return;
@@ -495,7 +508,14 @@ __global__ __launch_bounds__(256, 2) void gemm_tn_kshort(
       for (int r = 0; r < 4; ++r) {
         int row = m0 + wr * 64 + mi * 16 + fq * 4 + r;
         if (row >= M) continue;
-        C[(long)row * N + col] = f2bf(act_fwd(acc[mi][ni][r] + bv, act,
+        long crow = row;
+        if (GATHER_A && ga.mode == 2) {
+          int n2, h2, w2;
+          np_decode(ga, (unsigned)row, n2, h2, w2);
+          crow = ((long)n2 * ga.oH + h2 * ga.stride + ga.oqh) * ga.oW +
+                 w2 * ga.stride + ga.oqw;
+        }
+        C[crow * N + col] = f2bf(act_fwd(acc[mi][ni][r] + bv, act,
                                               slope));
       }
     }
@@ -1105,16 +1125,17 @@ int launch_gemm_tn(const void* A, const void* B, void* C_bf16, float* C_f32,
       const char* e = getenv("GDLJ_KSHORT_EPI");
       depi = (e != nullptr && e[0] == '1') ? 1 : 0;  // probe: direct stores
     }
+    ConvGather kdummy{};
     if (K == 64)
       hipLaunchKernelGGL((gemm_tn_kshort<64>), grid, dim3(256), 0, s,
                          (const unsigned short*)A, (const unsigned short*)B,
                          (unsigned short*)C_bf16, bias, M, N, lda, ldb, act,
-                         slope, depi);
+                         slope, depi, kdummy, nullptr);
     else
       hipLaunchKernelGGL((gemm_tn_kshort<128>), grid, dim3(256), 0, s,
                          (const unsigned short*)A, (const unsigned short*)B,
                          (unsigned short*)C_bf16, bias, M, N, lda, ldb, act,
-                         slope, depi);
+                         slope, depi, kdummy, nullptr);
     return (int)grid.x;
   }
   if (C_bf16 != nullptr && bn_part == nullptr &&
@@ -1149,6 +1170,22 @@ int launch_gemm_tn_gather(const void* img, const void* B, void* C_bf16,
                           int act, float slope, ConvGather ga,
                           const void* zero_page, float* bn_part,
                           hipStream_t s) {
+  if (bn_part == nullptr && (K == 64 || K == 128) && kshort_enabled()) {
+    dim3 grid(ceil_div(M, TN_BM), ceil_div(N, TN_BN));
+    if (K == 64)
+      hipLaunchKernelGGL((gemm_tn_kshort<64, true>), grid, dim3(256), 0, s,
+                         (const unsigned short*)img,
+                         (const unsigned short*)B, (unsigned short*)C_bf16,
+                         bias, M, N, 0, ldb, act, slope, 0, ga,
+                         (const unsigned short*)zero_page);
+    else
+      hipLaunchKernelGGL((gemm_tn_kshort<128, true>), grid, dim3(256), 0, s,
+                         (const unsigned short*)img,
+                         (const unsigned short*)B, (unsigned short*)C_bf16,
+                         bias, M, N, 0, ldb, act, slope, 0, ga,
+                         (const unsigned short*)zero_page);
+    return (int)grid.x;
+  }
   if (bn_part == nullptr && gemm_tn_8p_eligible(M, N, K)) {
     return launch_gemm_tn_8p(img, B, C_bf16, bias, M, N, K, 0, ldb, act,
                              slope, 1, ga, zero_page, s);
