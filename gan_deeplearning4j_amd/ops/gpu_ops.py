@@ -130,12 +130,19 @@ _chan_pad: dict = {}
 def deposit_chan_pad(trimmed: torch.Tensor, padded: torch.Tensor) -> None:
     if len(_chan_pad) >= 4:
         _chan_pad.pop(next(iter(_chan_pad)))
-    _chan_pad[trimmed.data_ptr()] = padded
+    _chan_pad[trimmed.data_ptr()] = (padded, trimmed.numel())
 
 
 def take_chan_pad(t: torch.Tensor, c8: int):
-    p = _chan_pad.get(t.data_ptr())
-    if p is not None and p.shape[-1] == c8 and p.shape[0] == t.shape[0]:
+    ent = _chan_pad.get(t.data_ptr())
+    if ent is None:
+        return None
+    p, numel = ent
+    # the key (a base data_ptr) can also belong to a DIFFERENT view of
+    # the same storage: require the consumer's logical element count and
+    # batch to match the deposited trim exactly
+    if (p.shape[-1] == c8 and p.shape[0] == t.shape[0]
+            and t.numel() == numel):
         return p
     return None
 
