@@ -247,6 +247,7 @@ DEV_INLINE bf16x8 frag(const char* opb, int rowblk, int kc, int fr, int fq) {
 //   1 = partial s_waitcnt lgkmcnt(8) before phase-1's first barrier
 //   2 = static young-half setprio instead of per-cluster flips (T5 static)
 //   4 = n-major XCD grid decomposition (B-panel L2 affinity)
+//   16 = prefetch the B ni-half-1 fragments in phase 1 (16/0/8/0 reads)
 // TWEAK=2 (static young-half priority, no per-cluster flips) measured
 // +4.9% within-probe over the flip form at square-4k (tools/ab_gemm8p.py:
 // 1188 -> 1247 TF median of 12 interleaved rounds; n-major decomposition
@@ -371,6 +372,13 @@ __global__ __launch_bounds__(512, 2) void gemm_tn_8p(
       blo[ni][0] = frag<SWZ>(Bb, wc * 4 + ni, 0, fr, fq);
       blo[ni][1] = frag<SWZ>(Bb, wc * 4 + ni, 1, fr, fq);
     }
+    if (TWEAK & 16) {
+      #pragma unroll
+      for (int ni = 0; ni < 2; ++ni) {
+        bhi[ni][0] = frag<SWZ>(Bb, wc * 4 + 2 + ni, 0, fr, fq);
+        bhi[ni][1] = frag<SWZ>(Bb, wc * 4 + 2 + ni, 1, fr, fq);
+      }
+    }
     stage_a(t + 1, 1);
     if (TWEAK & 1)  // drain most of the 12-read burst before the barrier
       asm volatile("s_waitcnt lgkmcnt(8)" ::: "memory");
@@ -379,11 +387,13 @@ __global__ __launch_bounds__(512, 2) void gemm_tn_8p(
                 if (DEEP && BNT == 256)
                     asm volatile("s_waitcnt vmcnt(6)" ::: "memory"));
 
-    // phase 2: B(ni half 1) reads; stage R3(t+1)
-    #pragma unroll
-    for (int ni = 0; ni < 2; ++ni) {
-      bhi[ni][0] = frag<SWZ>(Bb, wc * 4 + 2 + ni, 0, fr, fq);
-      bhi[ni][1] = frag<SWZ>(Bb, wc * 4 + 2 + ni, 1, fr, fq);
+    // phase 2: B(ni half 1) reads (unless phase-1 prefetched); stage R3
+    if (!(TWEAK & 16)) {
+      #pragma unroll
+      for (int ni = 0; ni < 2; ++ni) {
+        bhi[ni][0] = frag<SWZ>(Bb, wc * 4 + 2 + ni, 0, fr, fq);
+        bhi[ni][1] = frag<SWZ>(Bb, wc * 4 + 2 + ni, 1, fr, fq);
+      }
     }
     stage_b(t + 1, 1);
     P8_BAR_MFMA(0, 1, bhi, );
@@ -631,6 +641,7 @@ int launch_gemm_tn_8p_tweak(int tweak, const void* A, const void* B, void* C,
     case 4: P8_TW(4); break;
     case 5: P8_TW(5); break;
     case 7: P8_TW(7); break;
+    case 18: P8_TW(18); break;  // 16|2: bhi prefetch + static priority
     case 8:  // DEEP pipeline re-test under the static-priority regime
       (void)hipFuncSetAttribute(
           reinterpret_cast<const void*>(&gemm_tn_8p<false, 2, true, 256, 2>),

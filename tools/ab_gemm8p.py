@@ -25,7 +25,7 @@ def main():
     torch.manual_seed(0)
     A = (torch.rand(n, n, device="cuda") * 2 - 1).to(torch.bfloat16)
     B = (torch.rand(n, n, device="cuda") * 2 - 1).to(torch.bfloat16)
-    variants = [0, 2, 8]  # base-flips vs static vs static+DEEP
+    variants = [2, 18]  # static vs static+bhi-prefetch
     # numerics sanity vs variant 0
     base = e.gemm_tn_8p_tweak(A, B, 0)
     for v in variants[1:]:
