@@ -1163,6 +1163,22 @@ int launch_gemm_tn(const void* A, const void* B, void* C_bf16, float* C_f32,
   return (int)grid.x;
 }
 
+// conv_direct.hip: LDS-image direct conv for the small-C forward family
+extern "C" int launch_conv_direct_smallc(const void*, const void*, void*,
+                                         const float*, const void*, int, int,
+                                         int, int, int, int, int, int, int,
+                                         int, int, int, int, float,
+                                         hipStream_t);
+
+static int direct_conv_enabled() {
+  static int v = -1;
+  if (v < 0) {
+    const char* e = getenv("GDLJ_DIRECT_CONV");
+    v = (e != nullptr && e[0] == '0') ? 0 : 1;
+  }
+  return v;
+}
+
 // implicit-GEMM conv forward / gathered-A TN: A is an NHWC image; M = the
 // number of patch positions; K = kpad (zero page covers k >= rsc).
 int launch_gemm_tn_gather(const void* img, const void* B, void* C_bf16,
@@ -1170,6 +1186,12 @@ int launch_gemm_tn_gather(const void* img, const void* B, void* C_bf16,
                           int act, float slope, ConvGather ga,
                           const void* zero_page, float* bn_part,
                           hipStream_t s) {
+  if (bn_part == nullptr && ga.mode == 0 && ldb == K &&
+      direct_conv_enabled() &&
+      launch_conv_direct_smallc(img, B, C_bf16, bias, zero_page, ga.N, ga.H,
+                                ga.W, ga.C, ga.Ho, ga.Wo, ga.R, ga.S,
+                                ga.stride, ga.pad, N, K, act, slope, s))
+    return ceil_div(M, 128);
   if (bn_part == nullptr && (K == 64 || K == 128) && kshort_enabled()) {
     dim3 grid(ceil_div(M, TN_BM), ceil_div(N, TN_BN));
     if (K == 64)

@@ -135,6 +135,27 @@ def test_conv2d_fwd_bwd_vs_torch():
     assert relerr(b.grad, br.grad) < 0.04
 
 
+@pytest.mark.parametrize("cin,h,cout,r,stride,pad", [
+    (3, 64, 64, 5, 2, 2),    # dcgan64 conv1: the direct small-C kernel
+    (3, 32, 64, 5, 2, 2),    # 16x16 grid variant (hoperblk=8)
+    (8, 32, 48, 3, 1, 1),    # stride-1, Kout<64, C exactly 8
+    (3, 28, 64, 5, 2, 2),    # Ho*Wo=196 not /128 -> gather fallback
+])
+def test_conv2d_direct_smallc_vs_torch(cin, h, cout, r, stride, pad):
+    # conv_direct.hip eligibility covers the first three rows; the last
+    # proves the fallback keeps working on ineligible geometry.
+    from gan_deeplearning4j_amd.ops import gpu_ops
+
+    x = mk((4, cin, h, h), 40, 0.5)
+    w = mk((cout, cin, r, r), 41, 0.2)
+    b = torch.randn(cout, device=DEV, dtype=torch.bfloat16)
+    y = gpu_ops.conv2d(x, w, b, stride, pad, "lrelu", 0.2)
+    yr = F.leaky_relu(
+        F.conv2d(x.float().cpu(), w.float().cpu(), b.float().cpu(),
+                 stride=stride, padding=pad), 0.2)
+    assert relerr(y, yr) < 0.03
+
+
 def test_conv_transpose2d_fwd_bwd_vs_torch():
     from gan_deeplearning4j_amd.ops import gpu_ops
 
