@@ -17,13 +17,18 @@ SHAPES = [
 
 
 def run_one():
+    sys.path.insert(0, os.path.dirname(os.path.dirname(
+        os.path.abspath(__file__))))
     import torch
     from gan_deeplearning4j_amd.ops import gpu_ops
 
     for name, nb, cin, h, cout, r, stride, pad in SHAPES:
-        x = (torch.randn(nb, cin, h, h) * 0.5).to("cuda", torch.bfloat16)
-        w = (torch.randn(cout, cin, r, r) * 0.2).to("cuda", torch.bfloat16)
-        b = torch.randn(cout, device="cuda", dtype=torch.bfloat16)
+        g = torch.Generator().manual_seed(7)
+        x = (torch.randn(nb, cin, h, h, generator=g) * 0.5).to(
+            "cuda", torch.bfloat16)
+        w = (torch.randn(cout, cin, r, r, generator=g) * 0.2).to(
+            "cuda", torch.bfloat16)
+        b = torch.randn(cout, generator=g).to("cuda", torch.bfloat16)
         for _ in range(3):
             y = gpu_ops.conv2d(x, w, b, stride, pad, "lrelu", 0.2)
         torch.cuda.synchronize()
