@@ -155,6 +155,11 @@ import os
 
 PARITY_STRIDED = os.environ.get("GDLJ_PARITY", "0") == "1"
 
+
+def _dgrad_direct_on() -> bool:
+    """Fused direct strided dgrad (conv_dgrad_direct.hip), default on."""
+    return os.environ.get("GDLJ_DGRAD_DIRECT", "1") != "0"
+
 # FP8 conv-forward mode (BASELINE config 4: DCGAN-128 fp8 MFMA path).
 # When enabled, conv forwards with C % 16 == 0 quantize activations and
 # weights to e4m3 (per-tensor dynamic scale) and run the fp8 MFMA implicit
@@ -514,10 +519,47 @@ class _Conv2d(torch.autograd.Function):
                     Ko8, H, W, C, R, S, stride, pad, 0, 0.0)
                 dx = _as_nchw_view(dx2d.view(N, H, W, C)).to(ctx.dtypes[0])
             else:
-                # strided default: dcol GEMM + col2im gather
+                # strided default: dcol GEMM + col2im gather; when the
+                # geometry qualifies, the fused direct kernel
+                # (conv_dgrad_direct.hip) replaces BOTH in one pass
                 rsc8 = R * S * C8
                 wt = _packed(wp, "wt", lambda: _pad_k(
                     wp[:, :rsc8].t().contiguous()))   # [rsc8, kout_pad]
+                pact = ctx.prev_act
+                if pact is not None and (
+                        C8 != C
+                        or os.environ.get("GDLJ_NO_ACT_FUSE") == "1"):
+                    pact = None
+                res = None
+                if not _fp8_bwd_on() and _dgrad_direct_on():
+                    p_code, p_slope, p_bias = (
+                        pact if pact is not None else (0, 0.0, None))
+                    res = ext.conv_dgrad_direct(
+                        dpre8, wt, xh if pact is not None else None,
+                        _zp(dpre.device), N, H, W, C8, Ho, Wo, R, S,
+                        stride, pad, p_code, p_slope,
+                        pact is not None and p_bias is not False
+                        and p_bias is not None)
+                if res:
+                    dxh = res[0]
+                    if pact is not None:
+                        dx = _as_nchw_view(dxh).to(ctx.dtypes[0])
+                        deposit_act_fused(
+                            dx, (pact[0], C,
+                                 res[1] if len(res) > 1 else None))
+                    elif C8 != C and dxh.dtype == ctx.dtypes[0]:
+                        trimmed = dxh[..., :C]
+                        dx = _as_nchw_view(trimmed)
+                        deposit_chan_pad(trimmed, dxh)
+                    else:
+                        if C8 != C:
+                            dxh = dxh[..., :C].contiguous()
+                        dx = _as_nchw_view(dxh).to(ctx.dtypes[0])
+                    if (ctx.has_bias and ctx.needs_input_grad[2]
+                            and db is None):
+                        db = ext.col_sum(dpre).to(ctx.dtypes[2])
+                    return (dx, dw, db, None, None, None, None, None,
+                            None)
                 if _fp8_bwd_on():
                     wtq, _, iwt = _packed(wp, "wt_fp8", lambda: tuple(
                         ext.fp8_quantize(_pad_k128(wt))))

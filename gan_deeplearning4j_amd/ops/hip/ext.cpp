@@ -68,6 +68,12 @@ int launch_gemm_tn_gather(const void*, const void*, void*, const float*,
                           const void*, float*, hipStream_t);
 int launch_col2im_dact(const void*, void*, ConvGeom, const void*, int,
                        float, float*, hipStream_t);
+int conv_dgrad_direct_eligible(int, int, int, int, long, int, int, int,
+                               int);
+void launch_conv_dgrad_direct(const void*, const void*, void*, const void*,
+                              float*, const void*, int, int, int, int, int,
+                              int, int, long, int, int, int, int, float,
+                              int, hipStream_t);
 int launch_col2im_stats(const void*, void*, ConvGeom, const float*, int,
                         float, float*, hipStream_t);
 void launch_gemm_nt(const void*, const void*, float*, int, int, int, long,
@@ -419,6 +425,55 @@ std::vector<torch::Tensor> col2im_dact(
     launch_col_sum_sum2(part.data_ptr<float>(), gx, (int)C,
                         db.data_ptr<float>(), s);
   return {out, db};
+}
+
+// Direct parity-decomposed strided dgrad (conv_dgrad_direct.hip):
+// replaces dcol GEMM + col2im(_dact) when eligible.  dpre is the 2D
+// post-act-bwd gradient [N*Ho*Wo][Ko8]; wt the [R*S*C8][ldw] transposed
+// weight pack; y0 (optional) the producer's activation output for the
+// fused act backward.  Returns {} when ineligible (caller falls back),
+// else {dx_nhwc} or {dx_nhwc, producer_db}.
+std::vector<torch::Tensor> conv_dgrad_direct(
+    torch::Tensor dpre, torch::Tensor wt, c10::optional<torch::Tensor> y0,
+    torch::Tensor zero_page, int64_t N, int64_t H, int64_t W, int64_t C8,
+    int64_t Ho, int64_t Wo, int64_t R, int64_t S, int64_t stride,
+    int64_t pad, int64_t act, double slope, bool want_bias) {
+  check_bf16(dpre, "dpre");
+  check_bf16(wt, "wt");
+  int64_t Ko8 = dpre.size(1);
+  long ldw = (long)wt.size(1);
+  int el = conv_dgrad_direct_eligible((int)H, (int)W, (int)C8, (int)Ko8,
+                                      ldw, (int)R, (int)S, (int)stride,
+                                      (int)pad);
+  if (el == 0) return {};
+  TORCH_CHECK(dpre.size(0) == N * Ho * Wo, "dpre rows");
+  TORCH_CHECK(wt.size(0) >= R * S * C8, "wt rows");
+  auto st = cur_stream();
+  torch::Tensor out = torch::empty({N, H, W, C8}, dpre.options());
+  const void* y0p = nullptr;
+  float* part_p = nullptr;
+  torch::Tensor part, db;
+  if (y0.has_value() && y0->defined()) {
+    check_bf16(*y0, "y0");
+    TORCH_CHECK(y0->numel() == N * H * W * C8, "y0 shape");
+    y0p = y0->data_ptr();
+    if (want_bias) {
+      auto f32 = dpre.options().dtype(torch::kFloat32);
+      part = torch::zeros({2048, C8}, f32);
+      part_p = part.data_ptr<float>();
+      db = torch::empty({C8}, f32);
+    }
+  }
+  launch_conv_dgrad_direct(dpre.data_ptr(), wt.data_ptr(), out.data_ptr(),
+                           y0p, part_p, zero_page.data_ptr(), (int)N,
+                           (int)H, (int)W, (int)C8, (int)Ho, (int)Wo,
+                           (int)Ko8, ldw, (int)R, (int)S, (int)pad,
+                           (int)act, (float)slope, el, st);
+  if (part_p != nullptr) {
+    launch_col_sum_sum2(part_p, 2048, (int)C8, db.data_ptr<float>(), st);
+    return {out, db};
+  }
+  return {out};
 }
 
 // col2im with fused BN stats: returns {y_nhwc, sum, sumsq}
@@ -975,6 +1030,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("bn_bwd", &bn_bwd);
   mod.def("bn_bwd_act", &bn_bwd_act);
   mod.def("col2im_dact", &col2im_dact);
+  mod.def("conv_dgrad_direct", &conv_dgrad_direct,
+          "direct parity-decomposed strided dgrad ({} when ineligible)");
   mod.def("fused_adam", &fused_adam);
   mod.def("fused_rmsprop", &fused_rmsprop);
   mod.def("csv_load", &csv_load, "multithreaded CSV -> fp32 tensor");

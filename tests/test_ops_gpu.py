@@ -156,6 +156,69 @@ def test_conv2d_direct_smallc_vs_torch(cin, h, cout, r, stride, pad):
     assert relerr(y, yr) < 0.03
 
 
+@pytest.mark.parametrize("cin,h,cout,r", [
+    (64, 32, 128, 5),    # conv2 class: BM=128, one c-tile, one co-chunk
+    (128, 16, 256, 5),   # conv3 class: BM=64, 2 c-tiles, 2 co-chunks
+    (64, 32, 128, 4),    # even taps: Rc/Sc differ per parity class
+])
+def test_conv_dgrad_direct_vs_torch(cin, h, cout, r):
+    # conv_dgrad_direct.hip (parity-decomposed fused dgrad) vs torch,
+    # plus bit-consistency vs the dcol+col2im fallback path.
+    import os
+    from gan_deeplearning4j_amd.ops import gpu_ops
+
+    stride, pad = 2, 2
+    x = mk((4, cin, h, h), 50, 0.5).requires_grad_(True)
+    w = mk((cout, cin, r, r), 51, 0.2).requires_grad_(True)
+    y = gpu_ops.conv2d(x, w, None, stride, pad, "identity")
+    gout = mk(y.shape, 52)
+    y.backward(gout)
+    xr = x.detach().float().cpu().requires_grad_(True)
+    wr = w.detach().float().cpu().requires_grad_(True)
+    yr = F.conv2d(xr, wr, None, stride=stride, padding=pad)
+    yr.backward(gout.float().cpu())
+    assert relerr(x.grad, xr.grad) < 0.04
+    assert relerr(w.grad, wr.grad) < 0.04
+
+    # fallback consistency (the gate is re-read per call)
+    x2 = x.detach().clone().requires_grad_(True)
+    os.environ["GDLJ_DGRAD_DIRECT"] = "0"
+    try:
+        y2 = gpu_ops.conv2d(x2, w.detach(), None, stride, pad, "identity")
+        y2.backward(gout)
+    finally:
+        os.environ.pop("GDLJ_DGRAD_DIRECT")
+    assert relerr(x.grad, x2.grad) < 0.02
+
+
+def test_conv_dgrad_direct_pact_fusion():
+    # producer act-backward fusion: dx must come back multiplied by
+    # lrelu'(xh) with the producer bias partials deposited
+    from gan_deeplearning4j_amd.ops import gpu_ops
+    from gan_deeplearning4j_amd.ops.gpu_ops import take_act_fused
+
+    cin, h, cout, r, stride, pad = 64, 32, 128, 5, 2, 2
+    xh = F.leaky_relu(mk((4, cin, h, h), 60, 0.5), 0.2)
+    x = xh.detach().requires_grad_(True)
+    w = mk((cout, cin, r, r), 61, 0.2)
+    y = gpu_ops.conv2d(x, w, None, stride, pad, "identity",
+                       prev_act=(3, 0.2, True))
+    gout = mk(y.shape, 62)
+    y.backward(gout)
+    fused = take_act_fused(x.grad)
+    assert fused is not None and fused[0] == 3
+    x2 = x.detach().clone().requires_grad_(True)
+    y2 = gpu_ops.conv2d(x2, w, None, stride, pad, "identity")
+    y2.backward(gout)
+    ref = x2.grad.float() * torch.where(
+        x.detach().float() > 0, 1.0, 0.2)
+    assert relerr(x.grad, ref) < 0.03
+    # bias partials = column sums of the fused dx
+    db = fused[2].float().cpu()
+    ref_db = ref.sum(dim=(0, 2, 3)).cpu()
+    assert (db[:cin] - ref_db).abs().max() / ref_db.abs().max() < 0.02
+
+
 def test_conv_transpose2d_fwd_bwd_vs_torch():
     from gan_deeplearning4j_amd.ops import gpu_ops
 
