@@ -113,13 +113,18 @@ __global__ __launch_bounds__(256, 2) void conv_dgrad_direct(
   const int lo_w = wb0 - (Sc - 1);        // dy col of region col 0
 
   // ---- stage the dy region: [rgn_rows][rgn_cols][Ko8], zero halo ----
+  // Source-side chunk swizzle (slot = j ^ (cc & mask)): without it the
+  // A-fragment reads of 16 consecutive-cc pixels land 256B apart =
+  // same LDS bank (16-way conflict); the xor spreads them 8-wide.
   const int co_chunks = Ko8 >> 3;
+  const int cmask = co_chunks - 1;
   const int ncell = rgn_rows * rgn_cols * co_chunks;
   for (int cell = threadIdx.x; cell < ncell; cell += 256) {
     int rc_ = cell / co_chunks;
-    int j = cell - rc_ * co_chunks;
+    int slot = cell - rc_ * co_chunks;
     int rr = rc_ / rgn_cols;
     int cc = rc_ - rr * rgn_cols;
+    int j = slot ^ (cc & cmask);
     int hig = lo_h + rr;
     int wig = lo_w + cc;
     const unsigned short* src = zp;
@@ -182,20 +187,23 @@ __global__ __launch_bounds__(256, 2) void conv_dgrad_direct(
     // region coordinates for this tap: rr = pxh + (Rc-1) - rc,
     // cc = pxw + (Sc-1) - sc  (always in-grid; halo cells are zero)
     long abase[MI];
+    int axor[MI];
     #pragma unroll
     for (int mi = 0; mi < MI; ++mi) {
       int rr = pxh[mi] + (Rc - 1) - rc_;
       int cc = pxw[mi] + (Sc - 1) - sc_;
       abase[mi] = ((long)rr * rgn_cols + cc) * Ko8 * 2;
+      axor[mi] = cc & cmask;
     }
     const char* Wl = wbuf(cur);
     #pragma unroll
     for (int kc = 0; kc < 4; ++kc) {
-      int k = j * 128 + kc * 32 + fq * 8;  // co within Ko8
+      int jc = j * 16 + kc * 4 + fq;       // 16B co-chunk within Ko8
       bf16x8 a[MI], b[4];
       #pragma unroll
       for (int mi = 0; mi < MI; ++mi)
-        a[mi] = *(const bf16x8*)(rgn + abase[mi] + k * 2);
+        a[mi] = *(const bf16x8*)(rgn + abase[mi] +
+                                 ((jc ^ axor[mi]) << 4));
       int sub = kc >> 1, kslot = (kc & 1) * 4 + fq;
       #pragma unroll
       for (int ni = 0; ni < 4; ++ni)
@@ -274,8 +282,10 @@ int conv_dgrad_direct_eligible(int H, int W, int C8, int Ko8, long ldw,
     int rows_c = bm / Wc;
     int rgn_rows = rows_c + ((R + 1) >> 1) - 1;
     int rgn_cols = Wc + ((S + 1) >> 1) - 1;
+    // cap at 2 blocks/CU (160 KiB LDS): 1-block/CU shapes measured
+    // slower than their dcol path (stage latency unhidden)
     int lds_b = rgn_rows * rgn_cols * Ko8 * 2 + WBUF_B;
-    if (lds_b > 120 * 1024) continue;
+    if (lds_b > 80 * 1024) continue;
     return bm == 128 ? lds_b : -lds_b;
   }
   return 0;

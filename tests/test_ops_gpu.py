@@ -192,31 +192,44 @@ def test_conv_dgrad_direct_vs_torch(cin, h, cout, r):
 
 
 def test_conv_dgrad_direct_pact_fusion():
-    # producer act-backward fusion: dx must come back multiplied by
-    # lrelu'(xh) with the producer bias partials deposited
+    # real producer->consumer chain: conv_a(lrelu+bias) -> conv_b; the
+    # consumer's dgrad folds the producer's act backward + bias-grad
+    # reduction (deposit consumed by the producer's backward). Checked
+    # against the col2im_dact fallback AND a plain torch fp32 chain.
+    import os
     from gan_deeplearning4j_amd.ops import gpu_ops
-    from gan_deeplearning4j_amd.ops.gpu_ops import take_act_fused
 
-    cin, h, cout, r, stride, pad = 64, 32, 128, 5, 2, 2
-    xh = F.leaky_relu(mk((4, cin, h, h), 60, 0.5), 0.2)
-    x = xh.detach().requires_grad_(True)
-    w = mk((cout, cin, r, r), 61, 0.2)
-    y = gpu_ops.conv2d(x, w, None, stride, pad, "identity",
-                       prev_act=(3, 0.2, True))
-    gout = mk(y.shape, 62)
-    y.backward(gout)
-    fused = take_act_fused(x.grad)
-    assert fused is not None and fused[0] == 3
-    x2 = x.detach().clone().requires_grad_(True)
-    y2 = gpu_ops.conv2d(x2, w, None, stride, pad, "identity")
-    y2.backward(gout)
-    ref = x2.grad.float() * torch.where(
-        x.detach().float() > 0, 1.0, 0.2)
-    assert relerr(x.grad, ref) < 0.03
-    # bias partials = column sums of the fused dx
-    db = fused[2].float().cpu()
-    ref_db = ref.sum(dim=(0, 2, 3)).cpu()
-    assert (db[:cin] - ref_db).abs().max() / ref_db.abs().max() < 0.02
+    gout_cpu = torch.randn(4, 128, 16, 16,
+                           generator=torch.Generator().manual_seed(72))
+
+    def run(gate):
+        os.environ["GDLJ_DGRAD_DIRECT"] = gate
+        try:
+            x0 = mk((4, 64, 32, 32), 70, 0.5).requires_grad_(True)
+            w1 = mk((64, 64, 3, 3), 71, 0.2).requires_grad_(True)
+            b1 = mk((64,), 73).requires_grad_(True)
+            w2 = mk((128, 64, 5, 5), 74, 0.2).requires_grad_(True)
+            xh = gpu_ops.conv2d(x0, w1, b1, 1, 1, "lrelu", 0.2)
+            y = gpu_ops.conv2d(xh, w2, None, 2, 2, "identity",
+                               prev_act=(3, 0.2, True))
+            y.backward(gout_cpu.to(DEV, torch.bfloat16))
+            return x0.grad, w1.grad, b1.grad, w2.grad
+        finally:
+            os.environ.pop("GDLJ_DGRAD_DIRECT")
+
+    r1 = run("1")
+    r0 = run("0")
+    for a, b in zip(r1, r0):
+        assert relerr(a, b) < 0.02
+    x0 = mk((4, 64, 32, 32), 70, 0.5).float().cpu().requires_grad_(True)
+    w1 = mk((64, 64, 3, 3), 71, 0.2).float().cpu().requires_grad_(True)
+    b1 = mk((64,), 73).float().cpu().requires_grad_(True)
+    w2 = mk((128, 64, 5, 5), 74, 0.2).float().cpu().requires_grad_(True)
+    y = F.conv2d(F.leaky_relu(F.conv2d(x0, w1, b1, padding=1), 0.2),
+                 w2, None, stride=2, padding=2)
+    y.backward(gout_cpu)
+    for a, b in zip(r1, (x0.grad, w1.grad, b1.grad, w2.grad)):
+        assert relerr(a, b) < 0.06
 
 
 def test_conv_transpose2d_fwd_bwd_vs_torch():
