@@ -259,9 +259,32 @@ __global__ __launch_bounds__(256, 2) void conv_dgrad_direct(
     *(s16x8*)(dx + addr) = v;
   }
   if (y0 != nullptr && part != nullptr) {
-    float* prow = part + (long)(blockIdx.x & 2047) * C8 + cy * 64 + seg * 8;
+    // reduce the 32 per-thread partial vectors of each seg in LDS
+    // first: naive per-lane atomics (2048 lane-ops per block, 32-way
+    // duplicate addresses) serialized across ALL concurrent blocks and
+    // cost ~12 ms/call at the DCGAN-64 conv2 shape.
+    float* red = (float*)rgn;  // region is dead after the MFMA loop
+    const int rrow = t2 >> 3;  // 32 contributor rows per seg
     #pragma unroll
-    for (int jj = 0; jj < 8; ++jj) atomicAdd(&prow[jj], bsum[jj]);
+    for (int jj = 0; jj < 8; ++jj)
+      red[(rrow * 8 + seg) * 8 + jj] = bsum[jj];
+    __syncthreads();
+    for (int off = 16; off > 0; off >>= 1) {
+      if (rrow < off) {
+        #pragma unroll
+        for (int jj = 0; jj < 8; ++jj)
+          red[(rrow * 8 + seg) * 8 + jj] +=
+              red[((rrow + off) * 8 + seg) * 8 + jj];
+      }
+      __syncthreads();
+    }
+    if (rrow == 0) {
+      float* prow =
+          part + (long)(blockIdx.x & 2047) * C8 + cy * 64 + seg * 8;
+      #pragma unroll
+      for (int jj = 0; jj < 8; ++jj)
+        atomicAdd(&prow[jj], red[seg * 8 + jj]);
+    }
   }
 }
 
