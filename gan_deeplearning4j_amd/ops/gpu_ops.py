@@ -690,24 +690,46 @@ class _ConvTranspose2d(torch.autograd.Function):
                 w2a = _packed(w, "w2a", lambda: _pad_k(
                     _bf(w.detach().permute(2, 3, 1, 0))
                     .reshape(R * S * Cout, Cin)))
-            # strided convT dcol: K = Cin <= 1024 (thin, memory-bound):
-            # fp8 measured -2% end-to-end at dcgan128 b2048
-            # (profiles/fp8_bwd_ab.md mechanism) — opt-in only
-            if FP8_CONV and os.environ.get("GDLJ_FP8_CONVT") == "1":
-                w2aq, _, iw2 = _packed(w, "w2a_fp8", lambda: tuple(
-                    ext.fp8_quantize(_pad_k128(w2a))))
-                xq2, ix2 = _quant_delayed(_pad_k128(x2d), w, "xT")
-                col = ext.gemm_tn_fp8(xq2, w2aq, ix2, iw2, None, 0, 0.0)
+            # parity-direct convT forward (conv_dgrad_direct.hip, same
+            # kernel as the strided dgrad with a bias+act-fwd epilogue
+            # and fused BN stats) — no col buffer / col2im pass
+            res = None
+            if _dgrad_direct_on() and not (
+                    FP8_CONV
+                    and os.environ.get("GDLJ_FP8_CONVT") == "1"):
+                res = ext.conv_transpose_fwd_direct(
+                    x2d, w2a, bias, _zp(x.device), N, Hi, Wi, Ho, Wo,
+                    Co8, R, S, stride, pad, act, slope,
+                    bool(emit_stats and Cout % 8 == 0))
+            if res:
+                yh = res[0]
+                if len(res) > 1:
+                    stats = (res[1], res[2])
             else:
-                col = ext.gemm_tn(x2d, w2a, None, 0, 0.0, False)  # [NPin,RSCo8]
-            if emit_stats and Cout % 8 == 0:
+                # strided convT dcol: K = Cin <= 1024 (thin,
+                # memory-bound): fp8 measured -2% end-to-end at
+                # dcgan128 b2048 (profiles/fp8_bwd_ab.md) — opt-in only
+                if FP8_CONV and os.environ.get("GDLJ_FP8_CONVT") == "1":
+                    w2aq, _, iw2 = _packed(w, "w2a_fp8", lambda: tuple(
+                        ext.fp8_quantize(_pad_k128(w2a))))
+                    xq2, ix2 = _quant_delayed(_pad_k128(x2d), w, "xT")
+                    col = ext.gemm_tn_fp8(xq2, w2aq, ix2, iw2, None, 0,
+                                          0.0)
+                else:
+                    col = ext.gemm_tn(x2d, w2a, None, 0, 0.0,
+                                      False)  # [NPin, RSCo8]
+            if res and len(res) > 1:
+                pass                           # fused stats already set
+            elif not res and emit_stats and Cout % 8 == 0:
                 yh, ssum, ssq = ext.col2im_stats(
                     col, N, Ho, Wo, Cout, Hi, Wi, R, S, stride, pad,
                     R * S * Cout, bias, act, slope)
                 stats = (ssum, ssq)
             else:
-                yh = ext.col2im(col, N, Ho, Wo, Co8, Hi, Wi, R, S, stride,
-                                pad, R * S * Co8, bias, act, slope)
+                if not res:
+                    yh = ext.col2im(col, N, Ho, Wo, Co8, Hi, Wi, R, S,
+                                    stride, pad, R * S * Co8, bias, act,
+                                    slope)
                 if Co8 != Cout:
                     if act in (0, 1):  # identity/tanh: act(0 + 0-bias) == 0
                         yh_pad = yh

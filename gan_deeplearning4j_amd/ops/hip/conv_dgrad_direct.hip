@@ -65,7 +65,13 @@ DEV_INLINE bf16x8 cdd_wfrag(const char* lds, int row, int kslot) {
   return *(const bf16x8*)(lds + byte);
 }
 
-// MI = pixel fragments per wave (BM = MI * 64)
+// MI = pixel fragments per wave (BM = MI * 64).
+//
+// The same kernel serves the strided TRANSPOSED-conv forward (G side):
+// y[ho,wo,co] = sum over parity-valid taps of x[(ho+p-r)/2][ci] *
+// W[ci,co,r,s] is the identical gather; pass x as `dy`, the
+// [R*S*Co8][Cin] "w2a" pack as `Wt`, bias_fwd + act for the fused
+// epilogue, and stats_part ([2048][2*C8]) for fused BN statistics.
 template <int MI>
 __global__ __launch_bounds__(256, 2) void conv_dgrad_direct(
     const unsigned short* __restrict__ dy,   // [N][Ho][Wo][Ko8]
@@ -73,6 +79,8 @@ __global__ __launch_bounds__(256, 2) void conv_dgrad_direct(
     unsigned short* __restrict__ dx,         // [N][H][W][C8]
     const unsigned short* __restrict__ y0,   // producer act out (or null)
     float* __restrict__ part,                // [2048][C8] f32 (or null)
+    const float* __restrict__ bias_fwd,      // convT fwd bias (or null)
+    float* __restrict__ stats_part,          // [2048][2*C8] f32 (or null)
     const unsigned short* __restrict__ zp,   // 16B zero page
     int Nb, int H, int W, int C8, int Ho, int Wo, int Ko8, long ldw,
     int R, int S, int pad, int act, float slope,
@@ -218,17 +226,21 @@ __global__ __launch_bounds__(256, 2) void conv_dgrad_direct(
     __syncthreads();  // all waves done with wbuf(cur) before restage
   }
 
-  // ---- epilogue: ctile in the W-buffer area, optional act' + bias ----
+  // ---- epilogue: ctile in the W-buffer area; dgrad mode applies the
+  // producer act' downstream, convT-fwd mode applies bias+act here ----
   unsigned short* ctile = (unsigned short*)wbuf(0);  // [BM][64]
   #pragma unroll
   for (int mi = 0; mi < MI; ++mi) {
     #pragma unroll
     for (int ni = 0; ni < 4; ++ni) {
       int lc = ni * 16 + fr;
+      float bv = bias_fwd != nullptr ? bias_fwd[cy * 64 + lc] : 0.f;
       #pragma unroll
       for (int r = 0; r < 4; ++r) {
         int lr = wid * (MI * 16) + mi * 16 + fq * 4 + r;
-        ctile[lr * 64 + lc] = f2bf(acc[mi][ni][r]);
+        float v = acc[mi][ni][r];
+        if (bias_fwd != nullptr) v = act_fwd(v + bv, act, slope);
+        ctile[lr * 64 + lc] = f2bf(v);
       }
     }
   }
@@ -236,6 +248,8 @@ __global__ __launch_bounds__(256, 2) void conv_dgrad_direct(
   const int t2 = threadIdx.x;
   const int seg = t2 & 7;            // fixed c-block per thread
   float bsum[8] = {0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f};
+  float ssum[8] = {0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f};
+  float ssq[8] = {0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f};
   #pragma unroll
   for (int i = 0; i < BM / 32; ++i) {
     int piece = i * 256 + t2;        // BM*8 pieces = BM rows x 8 segs
@@ -256,18 +270,27 @@ __global__ __launch_bounds__(256, 2) void conv_dgrad_direct(
         bsum[jj] += d;
       }
     }
+    if (stats_part != nullptr) {
+      #pragma unroll
+      for (int jj = 0; jj < 8; ++jj) {
+        float yv = bf2f((unsigned short)v[jj]);
+        ssum[jj] += yv;
+        ssq[jj] += yv * yv;
+      }
+    }
     *(s16x8*)(dx + addr) = v;
   }
-  if (y0 != nullptr && part != nullptr) {
-    // reduce the 32 per-thread partial vectors of each seg in LDS
-    // first: naive per-lane atomics (2048 lane-ops per block, 32-way
-    // duplicate addresses) serialized across ALL concurrent blocks and
-    // cost ~12 ms/call at the DCGAN-64 conv2 shape.
-    float* red = (float*)rgn;  // region is dead after the MFMA loop
-    const int rrow = t2 >> 3;  // 32 contributor rows per seg
+  // Block-level seg reductions in LDS before the global atomics: naive
+  // per-lane atomics (2048 lane-ops per block, 32-way duplicate
+  // addresses) serialized across ALL concurrent blocks and cost
+  // ~12 ms/call at the DCGAN-64 conv2 shape.  All conditions are
+  // block-uniform, so the inner barriers are safe.
+  float* red = (float*)rgn;  // region is dead after the MFMA loop
+  const int rrow = t2 >> 3;  // 32 contributor rows per seg
+  auto reduce_and_add = [&](const float* vec, float* dst) {
     #pragma unroll
     for (int jj = 0; jj < 8; ++jj)
-      red[(rrow * 8 + seg) * 8 + jj] = bsum[jj];
+      red[(rrow * 8 + seg) * 8 + jj] = vec[jj];
     __syncthreads();
     for (int off = 16; off > 0; off >>= 1) {
       if (rrow < off) {
@@ -279,12 +302,19 @@ __global__ __launch_bounds__(256, 2) void conv_dgrad_direct(
       __syncthreads();
     }
     if (rrow == 0) {
-      float* prow =
-          part + (long)(blockIdx.x & 2047) * C8 + cy * 64 + seg * 8;
       #pragma unroll
       for (int jj = 0; jj < 8; ++jj)
-        atomicAdd(&prow[jj], red[seg * 8 + jj]);
+        atomicAdd(&dst[seg * 8 + jj], red[seg * 8 + jj]);
     }
+    __syncthreads();  // red is reused by the next reduction
+  };
+  if (y0 != nullptr && part != nullptr)
+    reduce_and_add(bsum,
+                   part + (long)(blockIdx.x & 2047) * C8 + cy * 64);
+  if (stats_part != nullptr) {
+    float* srow = stats_part + (long)(blockIdx.x & 2047) * 2 * C8;
+    reduce_and_add(ssum, srow + cy * 64);
+    reduce_and_add(ssq, srow + C8 + cy * 64);
   }
 }
 
@@ -315,11 +345,12 @@ int conv_dgrad_direct_eligible(int H, int W, int C8, int Ko8, long ldw,
 }
 
 void launch_conv_dgrad_direct(const void* dy, const void* Wt, void* dx,
-                              const void* y0, float* part, const void* zp,
-                              int Nb, int H, int W, int C8, int Ho, int Wo,
-                              int Ko8, long ldw, int R, int S, int pad,
-                              int act, float slope, int lds_flag,
-                              hipStream_t s) {
+                              const void* y0, float* part,
+                              const float* bias_fwd, float* stats_part,
+                              const void* zp, int Nb, int H, int W, int C8,
+                              int Ho, int Wo, int Ko8, long ldw, int R,
+                              int S, int pad, int act, float slope,
+                              int lds_flag, hipStream_t s) {
   using namespace cdd;
   int bm = lds_flag > 0 ? 128 : 64;
   int lds_b = lds_flag > 0 ? lds_flag : -lds_flag;
@@ -342,6 +373,7 @@ void launch_conv_dgrad_direct(const void* dy, const void* Wt, void* dx,
     hipLaunchKernelGGL((conv_dgrad_direct<2>), grid, dim3(256), lds_b, s,
                        (const unsigned short*)dy, (const unsigned short*)Wt,
                        (unsigned short*)dx, (const unsigned short*)y0, part,
+                       bias_fwd, stats_part,
                        (const unsigned short*)zp, Nb, H, W, C8, Ho, Wo, Ko8,
                        ldw, R, S, pad, act, slope, Hc, Wc, rows_c, rgn_rows,
                        rgn_cols, make_fastdiv(Wc));
@@ -349,6 +381,7 @@ void launch_conv_dgrad_direct(const void* dy, const void* Wt, void* dx,
     hipLaunchKernelGGL((conv_dgrad_direct<1>), grid, dim3(256), lds_b, s,
                        (const unsigned short*)dy, (const unsigned short*)Wt,
                        (unsigned short*)dx, (const unsigned short*)y0, part,
+                       bias_fwd, stats_part,
                        (const unsigned short*)zp, Nb, H, W, C8, Ho, Wo, Ko8,
                        ldw, R, S, pad, act, slope, Hc, Wc, rows_c, rgn_rows,
                        rgn_cols, make_fastdiv(Wc));

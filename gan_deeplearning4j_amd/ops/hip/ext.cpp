@@ -71,9 +71,9 @@ int launch_col2im_dact(const void*, void*, ConvGeom, const void*, int,
 int conv_dgrad_direct_eligible(int, int, int, int, long, int, int, int,
                                int);
 void launch_conv_dgrad_direct(const void*, const void*, void*, const void*,
-                              float*, const void*, int, int, int, int, int,
-                              int, int, long, int, int, int, int, float,
-                              int, hipStream_t);
+                              float*, const float*, float*, const void*,
+                              int, int, int, int, int, int, int, long, int,
+                              int, int, int, float, int, hipStream_t);
 int launch_col2im_stats(const void*, void*, ConvGeom, const float*, int,
                         float, float*, hipStream_t);
 void launch_gemm_nt(const void*, const void*, float*, int, int, int, long,
@@ -465,15 +465,75 @@ std::vector<torch::Tensor> conv_dgrad_direct(
     }
   }
   launch_conv_dgrad_direct(dpre.data_ptr(), wt.data_ptr(), out.data_ptr(),
-                           y0p, part_p, zero_page.data_ptr(), (int)N,
-                           (int)H, (int)W, (int)C8, (int)Ho, (int)Wo,
-                           (int)Ko8, ldw, (int)R, (int)S, (int)pad,
-                           (int)act, (float)slope, el, st);
+                           y0p, part_p, nullptr, nullptr,
+                           zero_page.data_ptr(), (int)N, (int)H, (int)W,
+                           (int)C8, (int)Ho, (int)Wo, (int)Ko8, ldw,
+                           (int)R, (int)S, (int)pad, (int)act,
+                           (float)slope, el, st);
   if (part_p != nullptr) {
     launch_col_sum_sum2(part_p, 2048, (int)C8, db.data_ptr<float>(), st);
     return {out, db};
   }
   return {out};
+}
+
+// Strided transposed-conv FORWARD through the same parity-direct
+// kernel: x2d is the flattened input [N*Hi*Wi][Cin-pad], w2a the
+// [R*S*Co8][Cin-pad] pack (gpu_ops "w2a" cache), bias f32 [Co8].
+// Returns {} when ineligible, {y_nhwc} or {y_nhwc, sum, sumsq}
+// (fused BN statistics) otherwise.
+std::vector<torch::Tensor> conv_transpose_fwd_direct(
+    torch::Tensor x2d, torch::Tensor w2a, c10::optional<torch::Tensor> bias,
+    torch::Tensor zero_page, int64_t N, int64_t Hi, int64_t Wi,
+    int64_t Ho, int64_t Wo, int64_t Co8, int64_t R, int64_t S,
+    int64_t stride, int64_t pad, int64_t act, double slope,
+    bool want_stats) {
+  check_bf16(x2d, "x2d");
+  check_bf16(w2a, "w2a");
+  int64_t Kin = x2d.size(1);       // padded Cin = contraction K
+  long ldw = (long)w2a.size(1);
+  int el = conv_dgrad_direct_eligible((int)Ho, (int)Wo, (int)Co8, (int)Kin,
+                                      ldw, (int)R, (int)S, (int)stride,
+                                      (int)pad);
+  if (el == 0) return {};
+  TORCH_CHECK(x2d.size(0) == N * Hi * Wi, "x2d rows");
+  TORCH_CHECK(w2a.size(0) >= R * S * Co8, "w2a rows");
+  auto st = cur_stream();
+  torch::Tensor y = torch::empty({N, Ho, Wo, Co8}, x2d.options());
+  const float* bias_p = nullptr;
+  if (bias.has_value() && bias->defined() && bias->numel() > 0) {
+    check_f32(*bias, "bias");
+    TORCH_CHECK(bias->numel() >= Co8, "bias length");
+    bias_p = bias->data_ptr<float>();
+  }
+  torch::Tensor zero_bias;
+  if (bias_p == nullptr) {
+    // the kernel keys "apply act_fwd" off the bias pointer
+    zero_bias = torch::zeros({Co8},
+                             x2d.options().dtype(torch::kFloat32));
+    bias_p = zero_bias.data_ptr<float>();
+  }
+  float* stats_p = nullptr;
+  torch::Tensor part, ssum, ssq;
+  if (want_stats) {
+    auto f32 = x2d.options().dtype(torch::kFloat32);
+    part = torch::zeros({2048, 2 * Co8}, f32);
+    stats_p = part.data_ptr<float>();
+    ssum = torch::empty({Co8}, f32);
+    ssq = torch::empty({Co8}, f32);
+  }
+  launch_conv_dgrad_direct(x2d.data_ptr(), w2a.data_ptr(), y.data_ptr(),
+                           nullptr, nullptr, bias_p, stats_p,
+                           zero_page.data_ptr(), (int)N, (int)Ho, (int)Wo,
+                           (int)Co8, (int)Hi, (int)Wi, (int)Kin, ldw,
+                           (int)R, (int)S, (int)pad, (int)act,
+                           (float)slope, el, st);
+  if (stats_p != nullptr) {
+    launch_bn_stats_sum2(stats_p, 2048, (int)Co8, ssum.data_ptr<float>(),
+                         ssq.data_ptr<float>(), st);
+    return {y, ssum, ssq};
+  }
+  return {y};
 }
 
 // col2im with fused BN stats: returns {y_nhwc, sum, sumsq}
@@ -1032,6 +1092,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("col2im_dact", &col2im_dact);
   mod.def("conv_dgrad_direct", &conv_dgrad_direct,
           "direct parity-decomposed strided dgrad ({} when ineligible)");
+  mod.def("conv_transpose_fwd_direct", &conv_transpose_fwd_direct,
+          "strided convT forward via the parity-direct kernel");
   mod.def("fused_adam", &fused_adam);
   mod.def("fused_rmsprop", &fused_rmsprop);
   mod.def("csv_load", &csv_load, "multithreaded CSV -> fp32 tensor");

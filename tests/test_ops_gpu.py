@@ -232,6 +232,53 @@ def test_conv_dgrad_direct_pact_fusion():
         assert relerr(a, b) < 0.06
 
 
+@pytest.mark.parametrize("cin,hi,cout,r", [
+    (256, 8, 128, 4),    # BM=64 class tile, 2 co-chunks
+    (128, 16, 64, 4),    # BM=128 class tile
+    (256, 8, 128, 5),    # odd taps: per-class Rc/Sc differ
+])
+def test_conv_transpose_fwd_direct_vs_torch(cin, hi, cout, r):
+    # convT forward through the parity-direct kernel (bias+act fused),
+    # vs torch and vs the col2im fallback
+    import os
+    from gan_deeplearning4j_amd.ops import gpu_ops
+
+    stride, pad = 2, 1 if r == 4 else 2
+    x = mk((4, cin, hi, hi), 80, 0.5)
+    w = mk((cin, cout, r, r), 81, 0.1)
+    b = torch.randn(cout, device=DEV, dtype=torch.bfloat16)
+    y = gpu_ops.conv_transpose2d(x, w, b, stride, pad, "tanh")
+    yr = torch.tanh(F.conv_transpose2d(
+        x.float().cpu(), w.float().cpu(), b.float().cpu(),
+        stride=stride, padding=pad))
+    assert relerr(y, yr) < 0.04
+    os.environ["GDLJ_DGRAD_DIRECT"] = "0"
+    try:
+        y2 = gpu_ops.conv_transpose2d(x, w, b, stride, pad, "tanh")
+    finally:
+        os.environ.pop("GDLJ_DGRAD_DIRECT")
+    assert relerr(y, y2) < 0.02
+
+
+def test_conv_transpose_fwd_direct_stats():
+    # the fused BN-statistics epilogue vs direct sums over the output
+    from gan_deeplearning4j_amd.ops import gpu_ops
+    from gan_deeplearning4j_amd.ops.gpu_ops import take_bn_stats
+
+    x = mk((4, 256, 8, 8), 82, 0.5)
+    w = mk((256, 128, 4, 4), 83, 0.1)
+    y = gpu_ops.conv_transpose2d(x, w, None, 2, 1, "identity",
+                                 emit_stats=True)
+    stats = take_bn_stats(y)
+    assert stats is not None
+    yf = y.float()
+    ssum, ssq = stats[0].float().cpu(), stats[1].float().cpu()
+    rsum = yf.sum(dim=(0, 2, 3)).cpu()
+    rsq = (yf * yf).sum(dim=(0, 2, 3)).cpu()
+    assert (ssum - rsum).abs().max() / rsum.abs().max() < 0.02
+    assert (ssq - rsq).abs().max() / rsq.abs().max() < 0.02
+
+
 def test_conv_transpose2d_fwd_bwd_vs_torch():
     from gan_deeplearning4j_amd.ops import gpu_ops
 
