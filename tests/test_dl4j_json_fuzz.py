@@ -100,3 +100,50 @@ def test_random_graph_roundtrip(chain, n_in, merge_in, out_spec, lr):
         y1 = g.output(*xs)
         y2 = g2.output(*xs)
     assert torch.allclose(y1, y2, atol=1e-6), (y1 - y2).abs().max()
+
+
+@settings(max_examples=8, deadline=None)
+@given(
+    chain=chain,
+    n_in=st.integers(2, 12),
+    lr=st.sampled_from([0.002, 0.0]),
+)
+def test_random_graph_zip_roundtrip(chain, n_in, lr):
+    # full ModelSerializer zip round trip (DL4J layout + coefficients +
+    # updater state) over the same random graph family
+    import tempfile
+    from pathlib import Path
+
+    from gan_deeplearning4j_amd.graph.serialization import ModelSerializer
+
+    cfg = _cfg()
+    gb = GraphBuilder(seed=666, optim_cfg=cfg.optim)
+    gb.add_inputs("in_a")
+    gb.set_input_types(InputType.feed_forward(n_in))
+    prev, width = "in_a", n_in
+    for i, (kind, n, act) in enumerate(chain):
+        name = f"l_{i}_{kind}"
+        if kind == "dense":
+            gb.add_layer(name, DenseLayer(width, n, act, lr=lr), prev)
+            width = n
+        elif kind == "bn":
+            gb.add_layer(name, BatchNormLayer(width, lr=lr), prev)
+        else:
+            gb.add_layer(name, ActivationLayer(act), prev)
+        prev = name
+    gb.add_layer("out_0", OutputLayer(width, 3, "softmax", "mcxent", lr=lr),
+                 prev)
+    gb.set_outputs("out_0")
+    g = gb.build().init()
+
+    with tempfile.TemporaryDirectory() as td:
+        p = Path(td) / "m.zip"
+        ModelSerializer.write_model(g, p, save_updater=True)
+        g2 = ModelSerializer.restore_computation_graph(p)
+    assert g2.layer_names() == g.layer_names()
+    torch.manual_seed(1)
+    x = torch.randn(3, n_in)
+    g.eval()
+    g2.eval()
+    with torch.no_grad():
+        assert torch.allclose(g.output(x), g2.output(x), atol=1e-6)
