@@ -160,7 +160,10 @@ def bce_with_logits_loss(
         from . import gpu_ops
 
         return gpu_ops.bce_with_logits(logits, labels)
-    return F.binary_cross_entropy_with_logits(logits.float(), labels.float())
+    if logits.dtype != torch.float64:
+        logits = logits.float()
+    return F.binary_cross_entropy_with_logits(
+        logits, labels.to(logits.dtype))
 
 
 def softmax_cross_entropy(
@@ -171,8 +174,10 @@ def softmax_cross_entropy(
         from . import gpu_ops
 
         return gpu_ops.softmax_cross_entropy(logits, onehot)
-    logp = F.log_softmax(logits.float(), dim=1)
-    return -(onehot.float() * logp).sum(dim=1).mean()
+    if logits.dtype != torch.float64:
+        logits = logits.float()
+    logp = F.log_softmax(logits, dim=1)
+    return -(onehot.to(logits.dtype) * logp).sum(dim=1).mean()
 
 
 def mse_loss(pred: torch.Tensor, target: torch.Tensor) -> torch.Tensor:
