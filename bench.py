@@ -104,7 +104,7 @@ def main():
         torch.distributed.all_reduce(t, op=torch.distributed.ReduceOp.MAX)
         elapsed = float(t.item())
 
-    n_gpus = world if use_gpu else 1
+    n_gpus = world  # == --gpus under the driver contract (CPU dry-runs incl.)
     global_batch = per_gpu_batch * world
     images_per_sec = global_batch * args.steps / elapsed
     size = m.image_height
