@@ -23,6 +23,7 @@ def _check_json_line(out: str, n_gpus_expected: int):
                 "dtype", "data", "config"):
         assert key in d, f"missing {key}"
     assert d["value"] > 0
+    assert d["n_gpus"] == n_gpus_expected
     assert d["data"] == "synthetic"
     assert d["scaling"] == "weak"
     assert "global_batch" in d["config"]
