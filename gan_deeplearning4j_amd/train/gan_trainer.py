@@ -286,6 +286,11 @@ class ReferenceProtocolTrainer:
         for g in (self.dis, self.gen, self.gan, self.cv):
             broadcast_parameters(g)
 
+        # init-time summary + shape smoke forward, like the reference
+        # prints after each graph init (Java:167-170, 223-225, 312-314,
+        # 365-368): summary() then output(randn(10, n_in)).shape
+        self.smoke_shapes = self._init_smoke()
+
         self.metrics = MetricsLogger(str(self.out_dir),
                                      cfg.train.print_every)
         torch.manual_seed(cfg.train.seed)
@@ -296,6 +301,37 @@ class ReferenceProtocolTrainer:
         self.soft_fake = (std * torch.randn(b, 1, generator=self._cpu_gen))
         self.soft_real = (1 + std * torch.randn(b, 1, generator=self._cpu_gen))
         self.batch_counter = 0
+
+    def _init_smoke(self) -> dict:
+        """Reference parity: after init, print each graph's summary()
+        and a 10-sample forward's output shape (Java:167-170, 223-225,
+        312-314, 365-368). Returns {name: shape} for tests."""
+        import logging
+
+        log = logging.getLogger("gan_deeplearning4j_amd")
+        nf = self.cfg.data.num_features
+        zs = self.cfg.model.z_size
+        shapes = {}
+        specs = [("dis", self.dis, nf), ("gen", self.gen, zs),
+                 ("gan", self.gan, zs), ("cv", self.cv, nf)]
+        dt = (torch.bfloat16 if self.device.type == "cuda"
+              else torch.float32)
+        # dedicated generator: must not consume from the label-noise /
+        # training RNG streams (golden trajectories depend on them)
+        smoke_gen = torch.Generator().manual_seed(
+            self.cfg.train.seed + 10)
+        for name, g, n_in in specs:
+            log.debug("%s summary:\n%s", name, g.summary())
+            x = torch.randn(10, n_in, generator=smoke_gen).to(
+                self.device, dt)
+            g.eval()
+            with torch.no_grad():
+                y = g.output(x)
+            g.train()
+            shapes[name] = tuple(y.shape)
+            log.info("%s smoke forward: (10, %d) -> %s", name, n_in,
+                     shapes[name])
+        return shapes
 
     # ----------------------------------------------------------- resume
     def resume(self) -> bool:

@@ -73,6 +73,14 @@ def test_reference_protocol_two_iterations(tmp_path):
 
     tr = ReferenceProtocolTrainer(cfg, device=torch.device("cpu"),
                                   out_dir=str(tmp_path / "out"))
+    # init-time shape smoke (reference Java:167-170, 223-225, 312-314,
+    # 365-368): D/CV map features->1/10 classes, G/GAN map z->image/1
+    nf = cfg.data.num_features
+    assert tr.smoke_shapes["dis"] == (10, 1)
+    assert tr.smoke_shapes["cv"] == (10, cfg.data.num_classes)
+    assert tr.smoke_shapes["gen"][0] == 10
+    assert int(np.prod(tr.smoke_shapes["gen"][1:])) == nf
+    assert tr.smoke_shapes["gan"] == (10, 1)
     last = tr.run(train_it, test_it)
     assert tr.batch_counter == 2
     assert np.isfinite(last["loss_d"]) and np.isfinite(last["loss_g"])
