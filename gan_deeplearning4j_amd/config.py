@@ -86,6 +86,9 @@ class TrainConfig:
     use_gpu: bool = True               # Java:92 useGpu
     reference_semantics: bool = False  # exact frozen-copy + weight-sync protocol
     train_classifier: bool = True      # transfer-learned classifier path
+    ema_decay: float = 0.0             # >0: keep an fp32 EMA of G's params
+    #                                    (framework extension; 0 = off,
+    #                                    matching the reference's protocol)
 
 
 @dataclass

@@ -18,6 +18,7 @@ Two protocols over the same op library:
 from __future__ import annotations
 
 import logging
+from contextlib import contextmanager
 from pathlib import Path
 from typing import Optional
 
@@ -113,6 +114,13 @@ class GanTrainer:
         import os as _os
 
         self._d_concat = _os.environ.get("GDLJ_D_CONCAT") == "1"
+        # fp32 EMA of G's params (framework extension; sampling/serving
+        # quality knob standard in production GAN trainers). Updated
+        # after every G step — inside the captured graph when capturing.
+        self.ema_decay = float(getattr(cfg.train, "ema_decay", 0.0))
+        self._ema = ([p.detach().float().clone()
+                      for p in self.gen.parameters()]
+                     if self.ema_decay > 0 else None)
         # hipGraph capture of the whole training step (one replay per step;
         # removes launch/dispatch host overhead). GPU-only; falls back to
         # eager if capture fails.
@@ -196,6 +204,11 @@ class GanTrainer:
         self.gen.updater.step()
         for p in self.dis.parameters():
             p.requires_grad_(True)
+        if self._ema is not None:
+            with torch.no_grad():
+                d = self.ema_decay
+                for e, p in zip(self._ema, self.gen.parameters()):
+                    e.mul_(d).add_(p.detach().to(e.dtype), alpha=1.0 - d)
 
         # losses stay on-device (no .item() sync in the hot loop); callers
         # float() them when they actually need host values
@@ -249,9 +262,32 @@ class GanTrainer:
             self._graph_out = self._step_eager(self._static_real)
         log.info("training step captured as hipGraph")
 
+    @contextmanager
+    def ema_weights(self):
+        """Temporarily swap the fp32 EMA weights into G (for sampling /
+        serving / checkpointing the smoothed generator). No-op when
+        ema_decay == 0. The packed-kernel caches key on tensor versions,
+        so they rebuild for the swapped weights and again on restore."""
+        if self._ema is None:
+            yield self.gen
+            return
+        backup = [p.detach().clone() for p in self.gen.parameters()]
+        with torch.no_grad():
+            for p, e in zip(self.gen.parameters(), self._ema):
+                p.copy_(e.to(p.dtype))
+        try:
+            yield self.gen
+        finally:
+            with torch.no_grad():
+                for p, b in zip(self.gen.parameters(), backup):
+                    p.copy_(b)
+
     @torch.no_grad()
-    def sample_grid(self, n: int = 10) -> torch.Tensor:
+    def sample_grid(self, n: int = 10, ema: bool = False) -> torch.Tensor:
         z = latent_grid(n, self.z_size, self.device, self.dtype)
+        if ema and self._ema is not None:
+            with self.ema_weights():
+                return self.gen.output(z).float().cpu()
         return self.gen.output(z).float().cpu()
 
 

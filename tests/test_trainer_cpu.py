@@ -40,6 +40,38 @@ def test_fast_trainer_mlp():
     assert grid.shape[0] == 16
 
 
+def test_ema_generator():
+    # fp32 EMA of G's params: exact recursion check over two steps,
+    # swap-in/swap-out restores the live weights bit-exactly
+    cfg = preset("mlp_tabular_cpu")
+    cfg.train.use_gpu = False
+    cfg.train.ema_decay = 0.5
+    gen, dis = build_mlp_gan(cfg, hidden=16)
+    tr = GanTrainer(gen, dis, cfg, device=torch.device("cpu"))
+    ema_expect = [p.detach().float().clone() for p in tr.gen.parameters()]
+    x = torch.rand(8, cfg.data.num_features)
+    for _ in range(2):
+        tr.step(x)
+        for e, p in zip(ema_expect, tr.gen.parameters()):
+            e.mul_(0.5).add_(p.detach().float(), alpha=0.5)
+    for e, a in zip(ema_expect, tr._ema):
+        assert torch.allclose(e, a, atol=1e-7)
+    live = [p.detach().clone() for p in tr.gen.parameters()]
+    with tr.ema_weights():
+        for p, e in zip(tr.gen.parameters(), tr._ema):
+            assert torch.allclose(p.detach().float(), e, atol=1e-6)
+    for p, b in zip(tr.gen.parameters(), live):
+        assert torch.equal(p.detach(), b)
+    # default off
+    cfg2 = preset("mlp_tabular_cpu")
+    cfg2.train.use_gpu = False
+    gen2, dis2 = build_mlp_gan(cfg2, hidden=16)
+    tr2 = GanTrainer(gen2, dis2, cfg2, device=torch.device("cpu"))
+    assert tr2._ema is None
+    with tr2.ema_weights() as g:
+        assert g is tr2.gen
+
+
 def test_fast_trainer_dcgan28_step():
     cfg = preset("dcgan28")
     cfg.train.use_gpu = False
