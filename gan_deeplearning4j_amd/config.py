@@ -89,6 +89,9 @@ class TrainConfig:
     ema_decay: float = 0.0             # >0: keep an fp32 EMA of G's params
     #                                    (framework extension; 0 = off,
     #                                    matching the reference's protocol)
+    loss_type: str = "bce"             # adversarial objective of the fast
+    #                                    trainer: bce (reference XENT) |
+    #                                    lsgan (least-squares) | hinge
 
 
 @dataclass

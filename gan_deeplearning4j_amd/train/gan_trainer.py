@@ -114,6 +114,14 @@ class GanTrainer:
         import os as _os
 
         self._d_concat = _os.environ.get("GDLJ_D_CONCAT") == "1"
+        # adversarial objective (framework extension; the reference is
+        # sigmoid+XENT = bce): lsgan = least-squares on logits, hinge =
+        # margin loss (label softening applies to bce/lsgan targets only)
+        self.loss_type = getattr(cfg.train, "loss_type", "bce")
+        if self.loss_type not in ("bce", "lsgan", "hinge"):
+            raise ValueError(f"unknown loss_type {self.loss_type!r}")
+        if self.loss_type != "bce":
+            self._d_concat = False  # concat path is bce-only
         # fp32 EMA of G's params (framework extension; sampling/serving
         # quality knob standard in production GAN trainers). Updated
         # after every G step — inside the captured graph when capturing.
@@ -184,8 +192,15 @@ class GanTrainer:
         else:
             d_real = self.dis(real)
             d_fake = self.dis(fake.detach())
-            loss_d = OF.bce_with_logits_loss(d_real, soft_real) + \
-                OF.bce_with_logits_loss(d_fake, soft_fake)
+            if self.loss_type == "lsgan":
+                loss_d = OF.mse_loss(d_real, soft_real) + \
+                    OF.mse_loss(d_fake, soft_fake)
+            elif self.loss_type == "hinge":
+                loss_d = torch.relu(1.0 - d_real.float()).mean() + \
+                    torch.relu(1.0 + d_fake.float()).mean()
+            else:
+                loss_d = OF.bce_with_logits_loss(d_real, soft_real) + \
+                    OF.bce_with_logits_loss(d_fake, soft_fake)
         loss_d.backward()
         self.d_reducer.finish()
         self.dis.updater.step()
@@ -196,9 +211,15 @@ class GanTrainer:
         self.gen.updater.zero_grad()
         self.g_reducer.prepare()
         g_logits = self.dis(fake)
-        loss_g = OF.bce_with_logits_loss(
-            g_logits, torch.ones(n, 1, device=self.device)
-        )
+        if self.loss_type == "lsgan":
+            loss_g = OF.mse_loss(g_logits,
+                                 torch.ones(n, 1, device=self.device))
+        elif self.loss_type == "hinge":
+            loss_g = -g_logits.float().mean()
+        else:
+            loss_g = OF.bce_with_logits_loss(
+                g_logits, torch.ones(n, 1, device=self.device)
+            )
         loss_g.backward()
         self.g_reducer.finish()
         self.gen.updater.step()
