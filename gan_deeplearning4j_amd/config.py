@@ -92,6 +92,8 @@ class TrainConfig:
     loss_type: str = "bce"             # adversarial objective of the fast
     #                                    trainer: bce (reference XENT) |
     #                                    lsgan (least-squares) | hinge
+    augment: str = ""                  # DiffAugment policy for D inputs,
+    #                                    e.g. "translate,cutout" ("" = off)
 
 
 @dataclass

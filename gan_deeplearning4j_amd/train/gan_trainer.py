@@ -120,8 +120,14 @@ class GanTrainer:
         self.loss_type = getattr(cfg.train, "loss_type", "bce")
         if self.loss_type not in ("bce", "lsgan", "hinge"):
             raise ValueError(f"unknown loss_type {self.loss_type!r}")
-        if self.loss_type != "bce":
-            self._d_concat = False  # concat path is bce-only
+        # DiffAugment policy for D inputs (train/augment.py; "" = off)
+        self.augment = getattr(cfg.train, "augment", "")
+        if self.augment:
+            from .augment import diff_augment as _da
+
+            _da(torch.zeros(1, 1, 8, 8), self.augment)  # validate policy
+        if self.loss_type != "bce" or self.augment:
+            self._d_concat = False  # concat path is plain-bce only
         # fp32 EMA of G's params (framework extension; sampling/serving
         # quality knob standard in production GAN trainers). Updated
         # after every G step — inside the captured graph when capturing.
@@ -190,8 +196,15 @@ class GanTrainer:
             # x2 keeps the gradient scale of mean(real)+mean(fake)
             loss_d = OF.bce_with_logits_loss(d_all, lab) * 2
         else:
-            d_real = self.dis(real)
-            d_fake = self.dis(fake.detach())
+            real_in, fake_in = real, fake.detach()
+            use_aug = bool(self.augment) and real.dim() == 4  # images only
+            if use_aug:
+                from .augment import diff_augment
+
+                real_in = diff_augment(real_in, self.augment)
+                fake_in = diff_augment(fake_in, self.augment)
+            d_real = self.dis(real_in)
+            d_fake = self.dis(fake_in)
             if self.loss_type == "lsgan":
                 loss_d = OF.mse_loss(d_real, soft_real) + \
                     OF.mse_loss(d_fake, soft_fake)
@@ -210,7 +223,12 @@ class GanTrainer:
             p.requires_grad_(False)
         self.gen.updater.zero_grad()
         self.g_reducer.prepare()
-        g_logits = self.dis(fake)
+        g_in = fake
+        if self.augment and fake.dim() == 4:
+            from .augment import diff_augment
+
+            g_in = diff_augment(g_in, self.augment)
+        g_logits = self.dis(g_in)
         if self.loss_type == "lsgan":
             loss_g = OF.mse_loss(g_logits,
                                  torch.ones(n, 1, device=self.device))
