@@ -116,3 +116,46 @@ def feature_extractor_eval(
         "n_train": n_train,
         "n_test": n_test,
     }
+
+
+# --------------------------------------------------------------- MMD metric
+def mmd2(x: torch.Tensor, y: torch.Tensor,
+         scales=(0.5, 1.0, 2.0, 4.0)) -> float:
+    """Unbiased squared Maximum Mean Discrepancy between two sample sets
+    [n,d] / [m,d], RBF mixture kernel with median-heuristic bandwidth.
+
+    A model-free GAN quality metric: near 0 when x and y come from the
+    same distribution, positive when they differ. Typically called on
+    frozen-discriminator features of real vs generated batches
+    (`feature_mmd`), the no-external-model analog of feature-space FID.
+    """
+    x = x.float().reshape(x.shape[0], -1)
+    y = y.float().reshape(y.shape[0], -1)
+    n, m = x.shape[0], y.shape[0]
+    if n < 2 or m < 2:
+        raise ValueError("mmd2 needs at least 2 samples per set")
+    dxx = torch.cdist(x, x) ** 2
+    dyy = torch.cdist(y, y) ** 2
+    dxy = torch.cdist(x, y) ** 2
+    pooled = torch.cat([dxy.reshape(-1), dxx.reshape(-1),
+                        dyy.reshape(-1)])  # symmetric in (x, y)
+    bw = pooled.median().clamp_min(1e-12)
+
+    def kmean(d2, unbias):
+        k = sum(torch.exp(-d2 / (2.0 * s * bw)) for s in scales)
+        if unbias:  # drop the k(z,z) diagonal
+            nn = d2.shape[0]
+            return (k.sum() - k.diagonal().sum()) / (nn * (nn - 1))
+        return k.mean()
+
+    return float(kmean(dxx, True) + kmean(dyy, True) - 2 * kmean(dxy, False))
+
+
+@torch.no_grad()
+def feature_mmd(dis, real: torch.Tensor, fake: torch.Tensor,
+                feat_layer: str = "d_dense_feat") -> float:
+    """MMD^2 between real and generated batches in the frozen
+    discriminator's feature space (activations at `feat_layer`)."""
+    fr = dis.feed_forward(real, upto=feat_layer)
+    ff = dis.feed_forward(fake, upto=feat_layer)
+    return mmd2(fr, ff)

@@ -41,3 +41,42 @@ def test_graph_feed_forward_extracts_features():
     assert feats.shape == (4, 32)
     acts = dis.feed_forward(x)
     assert "d_out" in acts and "d_dense_feat" in acts
+
+
+def test_mmd2_separates_distributions():
+    import torch
+    from gan_deeplearning4j_amd.train.feature_eval import mmd2
+
+    g = torch.Generator().manual_seed(0)
+    a = torch.randn(128, 16, generator=g)
+    b = torch.randn(128, 16, generator=g)
+    c = torch.randn(128, 16, generator=g) + 2.0
+    same = mmd2(a, b)
+    diff = mmd2(a, c)
+    assert abs(same) < 0.02
+    assert diff > 10 * max(abs(same), 1e-4)
+    # symmetry and input validation
+    assert abs(mmd2(a, c) - mmd2(c, a)) < 1e-5
+    import pytest
+    with pytest.raises(ValueError):
+        mmd2(a[:1], b)
+
+
+def test_feature_mmd_on_discriminator():
+    import torch
+    from gan_deeplearning4j_amd.config import preset
+    from gan_deeplearning4j_amd.models import build_dcgan
+    from gan_deeplearning4j_amd.train.feature_eval import feature_mmd
+
+    cfg = preset("dcgan28")
+    cfg.train.use_gpu = False
+    cfg.model.base_width = 8
+    gen, dis = build_dcgan(cfg)
+    g = torch.Generator().manual_seed(1)
+    real = torch.rand(32, 1, 28, 28, generator=g) * 2 - 1
+    near = real + 0.01 * torch.randn(32, 1, 28, 28, generator=g)
+    far = torch.zeros(32, 1, 28, 28)
+    m_near = feature_mmd(dis, real, near)
+    m_far = feature_mmd(dis, real, far)
+    assert m_far > m_near >= 0 or m_near < 0.05
+    assert m_far > 0.05
