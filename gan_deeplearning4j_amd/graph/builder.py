@@ -505,6 +505,33 @@ class ComputationGraph(nn.Module):
         lines.append(f"total params: {total}")
         return "\n".join(lines)
 
+    def to_dot(self) -> str:
+        """Graphviz DOT of the vertex topology (inputs, layers with
+        param counts, preprocessor edges, outputs) — render with any
+        `dot` tool; complements the tabular summary()."""
+        lines = ["digraph ComputationGraph {", "  rankdir=TB;"]
+        for name in self.input_names:
+            lines.append(
+                f'  "{name}" [shape=oval, style=filled, '
+                f'fillcolor=lightblue];')
+        for name in self._topo:
+            layer = self.layers[name]
+            n = layer.n_params()
+            shape = ("doubleoctagon" if name in self.output_names
+                     else "box")
+            label = f"{name}\\n{type(layer).__name__}"
+            if n:
+                label += f"\\n{n:,} params"
+            lines.append(f'  "{name}" [shape={shape}, label="{label}"];')
+            for s in self._vertex_inputs[name]:
+                edge = f'  "{s}" -> "{name}"'
+                if name in self.preprocessors:
+                    pname = type(self.preprocessors[name]).__name__
+                    edge += f' [label="{pname}"]'
+                lines.append(edge + ";")
+        lines.append("}")
+        return "\n".join(lines)
+
     # ------------------------------------------------------------ misc
     def clone(self) -> "ComputationGraph":
         import copy

@@ -68,3 +68,26 @@ def test_flat_and_image_inputs_agree():
     y_flat = d.output(x)
     y_img = d.output(x.reshape(4, 1, 28, 28))
     assert torch.allclose(y_flat, y_img, atol=1e-6)
+
+
+def test_to_dot_export():
+    import torch
+    from gan_deeplearning4j_amd.config import GanConfig
+    from gan_deeplearning4j_amd.models import build_cgan
+    from gan_deeplearning4j_amd.models.reference_protocol import (
+        build_discriminator,
+    )
+
+    cfg = GanConfig()
+    cfg.train.use_gpu = False
+    d = build_discriminator(cfg)
+    dot = d.to_dot()
+    assert dot.startswith("digraph") and dot.rstrip().endswith("}")
+    # every vertex and input appears; outputs get the double shape
+    for name in d.layer_names():
+        assert f'"{name}"' in dot
+    assert "doubleoctagon" in dot
+    gen, _ = build_cgan(cfg)
+    dot2 = gen.to_dot()
+    assert '"g_z" -> "g_merge"' in dot2 and '"g_label" -> "g_merge"' in dot2
+    assert "FeedForwardToCnnPreProcessor" in dot2  # preprocessor edge label
