@@ -1,4 +1,5 @@
-from .csv_reader import CSVRecordReader, RecordReaderDataSetIterator, DataSet
+from .csv_reader import (CSVRecordReader, DataSet,
+                         RecordReaderDataSetIterator, TensorDataSetIterator)
 from .synthetic import (
     pixel_lattice_images,
     transactions_tabular,
@@ -8,6 +9,7 @@ from .synthetic import (
 __all__ = [
     "CSVRecordReader",
     "RecordReaderDataSetIterator",
+    "TensorDataSetIterator",
     "DataSet",
     "pixel_lattice_images",
     "transactions_tabular",

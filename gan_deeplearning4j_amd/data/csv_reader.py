@@ -131,3 +131,50 @@ class RecordReaderDataSetIterator:
         if self.drop_last:
             return n // self.batch_size
         return (n + self.batch_size - 1) // self.batch_size
+
+
+class TensorDataSetIterator:
+    """DataSet minibatches straight from in-memory tensors — skips the
+    CSV round trip for programmatic pipelines (ND4J's
+    ListDataSetIterator analog). Labels may be class indices [N] (one-
+    hot encoded with num_classes) or already-one-hot [N, C].
+    """
+
+    def __init__(self, features: torch.Tensor, labels: torch.Tensor,
+                 batch_size: int, num_classes: int = 0,
+                 shuffle: bool = False, seed: int = 0,
+                 drop_last: bool = False):
+        if features.shape[0] != labels.shape[0]:
+            raise ValueError("features/labels row mismatch")
+        self.features = features
+        if labels.dim() == 1:
+            if num_classes <= 0:
+                raise ValueError("index labels need num_classes")
+            oh = torch.zeros(labels.shape[0], num_classes)
+            oh[torch.arange(labels.shape[0]), labels.long()] = 1.0
+            labels = oh
+        self.labels = labels
+        self.batch_size = batch_size
+        self.shuffle = shuffle
+        self.seed = seed
+        self.drop_last = drop_last
+        self._epoch = 0
+
+    def __iter__(self) -> Iterator[DataSet]:
+        n = self.features.shape[0]
+        order = torch.arange(n)
+        if self.shuffle:
+            g = torch.Generator().manual_seed(self.seed + self._epoch)
+            order = torch.randperm(n, generator=g)
+        self._epoch += 1
+        for start in range(0, n, self.batch_size):
+            idx = order[start:start + self.batch_size]
+            if self.drop_last and len(idx) < self.batch_size:
+                break
+            yield DataSet(self.features[idx], self.labels[idx])
+
+    def num_batches(self) -> int:
+        n = self.features.shape[0]
+        if self.drop_last:
+            return n // self.batch_size
+        return (n + self.batch_size - 1) // self.batch_size

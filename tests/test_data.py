@@ -64,3 +64,33 @@ def test_native_csv_loader_matches_numpy(tmp_path):
     b = np.loadtxt(p, delimiter=",", dtype=np.float32)
     assert a.shape == b.shape
     assert np.allclose(a, b, atol=1e-5)
+
+
+def test_tensor_dataset_iterator():
+    import pytest
+    import torch
+    from gan_deeplearning4j_amd.data import TensorDataSetIterator
+
+    x = torch.arange(20, dtype=torch.float32).reshape(10, 2)
+    y = torch.tensor([0, 1, 2, 0, 1, 2, 0, 1, 2, 0])
+    it = TensorDataSetIterator(x, y, batch_size=4, num_classes=3)
+    batches = list(it)
+    assert [b.num_examples() for b in batches] == [4, 4, 2]
+    assert batches[0].labels.shape == (4, 3)
+    assert torch.equal(batches[0].features, x[:4])
+    assert batches[0].labels[0].argmax() == 0
+    # drop_last + shuffle determinism per epoch
+    it2 = TensorDataSetIterator(x, y, 4, 3, shuffle=True, seed=7,
+                                drop_last=True)
+    e1 = [b.features.clone() for b in it2]
+    assert len(e1) == 2 and it2.num_batches() == 2
+    e2 = [b.features for b in it2]
+    assert not all(torch.equal(a, b) for a, b in zip(e1, e2))  # reshuffled
+    # one-hot passthrough + validation
+    oh = torch.eye(3)[y]
+    it3 = TensorDataSetIterator(x, oh, 5)
+    assert next(iter(it3)).labels.shape == (5, 3)
+    with pytest.raises(ValueError):
+        TensorDataSetIterator(x, y, 4)      # index labels, no num_classes
+    with pytest.raises(ValueError):
+        TensorDataSetIterator(x, y[:5], 4, 3)
