@@ -21,6 +21,7 @@ import argparse
 import base64
 import tempfile
 import threading
+import time
 from pathlib import Path
 from typing import Optional
 
@@ -207,6 +208,39 @@ def create_app(generator: Optional[ComputationGraph] = None,
 
     app = FastAPI(title="gan_deeplearning4j_amd serving")
 
+    # Prometheus metrics (per-app registry so tests can build many apps)
+    try:
+        from prometheus_client import (CONTENT_TYPE_LATEST,
+                                       CollectorRegistry, Counter,
+                                       Histogram, generate_latest)
+
+        _reg = CollectorRegistry()
+        _req_count = Counter("serve_requests_total", "requests served",
+                             ["endpoint", "model"], registry=_reg)
+        _req_err = Counter("serve_errors_total", "request errors",
+                           ["endpoint"], registry=_reg)
+        _req_lat = Histogram("serve_request_seconds", "request latency",
+                             ["endpoint"], registry=_reg)
+
+        @app.get("/metrics")
+        def metrics():
+            from fastapi import Response
+
+            return Response(generate_latest(_reg),
+                            media_type=CONTENT_TYPE_LATEST)
+    except ImportError:  # pragma: no cover - metrics become no-ops
+        class _Null:
+            def labels(self, *a, **k):
+                return self
+
+            def inc(self, *a):
+                pass
+
+            def observe(self, *a):
+                pass
+
+        _req_count = _req_err = _req_lat = _Null()
+
     @app.get("/healthz")
     def healthz():
         return {
@@ -238,10 +272,14 @@ def create_app(generator: Optional[ComputationGraph] = None,
         name = req.model or "default"
         gen_ep = gen_eps.get(name)
         if gen_ep is None:
+            _req_err.labels("generate").inc()
             raise HTTPException(404, f"no generator {name!r} loaded")
         gen_mb = gen_mbs[name]
         if req.n > max_batch:
+            _req_err.labels("generate").inc()
             raise HTTPException(400, f"n > max_batch ({max_batch})")
+        _req_count.labels("generate", name).inc()
+        _t0 = time.monotonic()
         it = gen_ep.graph.input_types[gen_ep.graph.input_names[0]]
         z_size = it.shape(1)[1]
         g = torch.Generator().manual_seed(req.seed) if req.seed is not None \
@@ -249,6 +287,7 @@ def create_app(generator: Optional[ComputationGraph] = None,
         z = torch.randn(req.n, z_size, generator=g)
         # concurrent requests coalesce into one padded graph replay
         samples = await gen_mb.submit(z)
+        _req_lat.labels("generate").observe(time.monotonic() - _t0)
         if req.format == "png_base64":
             from .utils.imaging import save_image_grid
 
@@ -265,7 +304,9 @@ def create_app(generator: Optional[ComputationGraph] = None,
     @app.post("/discriminate")
     def discriminate(req: DiscriminateReq):
         if dis_ep is None:
+            _req_err.labels("discriminate").inc()
             raise HTTPException(404, "no discriminator loaded")
+        _req_count.labels("discriminate", "default").inc()
         x = torch.tensor(req.inputs, dtype=torch.float32)
         if x.ndim == 1:
             x = x.unsqueeze(0)

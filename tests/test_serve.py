@@ -189,3 +189,18 @@ def test_discriminate_concurrent_consistency(client):
     with concurrent.futures.ThreadPoolExecutor(max_workers=8) as ex:
         for i, scores in ex.map(hit, list(range(8)) * 4):
             assert scores == pytest.approx(want[i], abs=1e-5), i
+
+
+def test_metrics_endpoint(client):
+    # prometheus metrics: counters move with requests
+    r = client.post("/generate", json={"n": 2})
+    assert r.status_code == 200
+    m = client.get("/metrics")
+    assert m.status_code == 200
+    body = m.text
+    assert 'serve_requests_total{endpoint="generate",model="default"}' in body
+    assert "serve_request_seconds" in body
+    # errors counted
+    client.post("/generate", json={"n": 10_000})
+    body2 = client.get("/metrics").text
+    assert 'serve_errors_total{endpoint="generate"}' in body2
