@@ -301,6 +301,51 @@ class GanTrainer:
             self._graph_out = self._step_eager(self._static_real)
         log.info("training step captured as hipGraph")
 
+    # ------------------------------------------------- save / resume
+    def save(self, out_dir) -> None:
+        """Checkpoint both graphs (DL4J zip layout + updater state),
+        the EMA state and the step counter. The reference protocol
+        checkpoints per-iteration (Java:605-618); this is the fast
+        trainer's equivalent."""
+        out = Path(out_dir)
+        out.mkdir(parents=True, exist_ok=True)
+        ModelSerializer.write_model(self.gen, out / "gen_model.zip",
+                                    save_updater=True)
+        ModelSerializer.write_model(self.dis, out / "dis_model.zip",
+                                    save_updater=True)
+        state = {"it": self.it}
+        if self._ema is not None:
+            state["ema"] = [e.cpu() for e in self._ema]
+        torch.save(state, out / "trainer_state.pt")
+
+    def resume(self, out_dir) -> bool:
+        """Restore params, buffers, updater state, EMA and the step
+        counter IN PLACE (tensor identities survive, so captured
+        graphs, reducers and packed-weight caches stay valid — the
+        caches re-key on the bumped tensor versions)."""
+        out = Path(out_dir)
+        paths = [out / "gen_model.zip", out / "dis_model.zip",
+                 out / "trainer_state.pt"]
+        if not all(p.exists() for p in paths):
+            return False
+        for live, path in ((self.gen, paths[0]), (self.dis, paths[1])):
+            g2 = ModelSerializer.restore_computation_graph(
+                path, load_updater=True)
+            g2.to_device(self.device, self.dtype)
+            with torch.no_grad():
+                for p, p2 in zip(live.parameters(), g2.parameters()):
+                    p.copy_(p2)
+                for b, b2 in zip(live.buffers(), g2.buffers()):
+                    b.copy_(b2)
+            live.updater.load_state_dict(g2.updater.state_dict())
+        state = torch.load(out / "trainer_state.pt", weights_only=True)
+        self.it = int(state["it"])
+        if self._ema is not None and "ema" in state:
+            with torch.no_grad():
+                for e, s in zip(self._ema, state["ema"]):
+                    e.copy_(s.to(e.device))
+        return True
+
     @contextmanager
     def ema_weights(self):
         """Temporarily swap the fp32 EMA weights into G (for sampling /
