@@ -82,8 +82,11 @@ def run_fast(cfg: GanConfig):
     train_it, _ = _iterators(cfg, rank)
     metrics = MetricsLogger(cfg.train.out_dir, cfg.train.print_every,
                             is_main())
+    # crash recovery: pick up from the last full checkpoint if present
+    if tr.resume(cfg.train.out_dir):
+        log.info("resumed fast trainer at iteration %d", tr.it)
     m = cfg.model
-    it = 0
+    it = tr.it
     while it < cfg.train.num_iterations:
         for ds in train_it:
             if it >= cfg.train.num_iterations:
@@ -97,11 +100,7 @@ def run_fast(cfg: GanConfig):
             metrics.step(it, loss_d=float(out["loss_d"]),
                          loss_g=float(out["loss_g"]), images=out["images"])
             if is_main() and it % cfg.train.save_every == 0:
-                from .graph.serialization import ModelSerializer
-
-                out_dir = Path(cfg.train.out_dir)
-                ModelSerializer.write_model(tr.gen, out_dir / "gen_model.zip")
-                ModelSerializer.write_model(tr.dis, out_dir / "dis_model.zip")
+                tr.save(cfg.train.out_dir)
     if is_main():
         import numpy as np
 
