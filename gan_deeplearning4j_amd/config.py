@@ -68,6 +68,10 @@ class OptimConfig:
     grad_clip: float = 1.0                  # ClipElementWiseAbsoluteValue(1.0), Java:123-124
     l2: float = 1e-4                        # Java:125 l2(0.0001)
     weight_init: str = "xavier"             # Java:127 WeightInit.XAVIER
+    lr_schedule: str = ""                   # framework extension: "" (fixed,
+    #                                         reference semantics) | "linear"
+    #                                         | "cosine" over num_iterations
+    lr_warmup_steps: int = 0                # linear warmup before the decay
 
 
 @dataclass

@@ -60,6 +60,11 @@ class Updater:
         self.grad_clip = grad_clip
         self.l2 = l2
         self.t = 0
+        # global LR multiplier (schedules; trainers set it per step).
+        # NOTE: reaches the fused kernel as a scalar argument, so it is
+        # frozen inside a captured graph — trainers disable capture
+        # when a schedule is active.
+        self.lr_scale = 1.0
         # device-side step counter (hipGraph-replayable bias correction)
         self._t_dev: Optional[torch.Tensor] = None
 
@@ -132,10 +137,10 @@ class Updater:
             s.v.mul_(self.beta2).addcmul_(g, g, value=1 - self.beta2)
             mhat = s.m / (1 - self.beta1 ** self.t)
             vhat = s.v / (1 - self.beta2 ** self.t)
-            w.add_(-s.lr * mhat / (vhat.sqrt() + self.eps))
+            w.add_(-s.lr * self.lr_scale * mhat / (vhat.sqrt() + self.eps))
         elif self.kind == "rmsprop":
             s.v.mul_(self.rms_decay).addcmul_(g, g, value=1 - self.rms_decay)
-            w.add_(-s.lr * g / (s.v.sqrt() + self.eps))
+            w.add_(-s.lr * self.lr_scale * g / (s.v.sqrt() + self.eps))
         else:
             raise KeyError(self.kind)
         if s.master is not None:
@@ -151,7 +156,7 @@ class Updater:
             master=s.master,
             m=s.m,
             v=s.v,
-            lr=s.lr,
+            lr=s.lr * self.lr_scale,
             beta1=self.beta1,
             beta2=self.beta2,
             rms_decay=self.rms_decay,

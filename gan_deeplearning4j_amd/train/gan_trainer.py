@@ -135,10 +135,19 @@ class GanTrainer:
         self._ema = ([p.detach().float().clone()
                       for p in self.gen.parameters()]
                      if self.ema_decay > 0 else None)
+        # LR schedule (framework extension): warmup + linear/cosine decay
+        # over num_iterations. Incompatible with graph capture (lr is a
+        # scalar kernel argument, frozen at capture) -> forces eager.
+        self.lr_schedule = getattr(cfg.optim, "lr_schedule", "")
+        self.lr_warmup = int(getattr(cfg.optim, "lr_warmup_steps", 0))
+        if self.lr_schedule not in ("", "linear", "cosine"):
+            raise ValueError(f"unknown lr_schedule {self.lr_schedule!r}")
+        self._horizon = max(int(cfg.train.num_iterations), 1)
         # hipGraph capture of the whole training step (one replay per step;
         # removes launch/dispatch host overhead). GPU-only; falls back to
         # eager if capture fails.
-        self.capture = capture and self.device.type == "cuda"
+        self.capture = capture and self.device.type == "cuda" \
+            and not self.lr_schedule and self.lr_warmup == 0
         self._graph = None
         self._graph_failed = False
         self._static_real = None
@@ -169,9 +178,29 @@ class GanTrainer:
         z = torch.randn(n, self.z_size, generator=self._g)
         return z.to(self.device, self.dtype)
 
+    def lr_scale_at(self, it: int) -> float:
+        """Schedule multiplier for 1-based step `it` (1.0 when off)."""
+        if not self.lr_schedule and self.lr_warmup == 0:
+            return 1.0
+        if self.lr_warmup > 0 and it <= self.lr_warmup:
+            return it / self.lr_warmup
+        if not self.lr_schedule:
+            return 1.0
+        span = max(self._horizon - self.lr_warmup, 1)
+        prog = min(max(it - self.lr_warmup, 0) / span, 1.0)
+        if self.lr_schedule == "linear":
+            return 1.0 - prog
+        import math
+
+        return 0.5 * (1.0 + math.cos(math.pi * prog))  # cosine
+
     def step(self, real: torch.Tensor) -> dict:
         """One alternating D+G update on a batch of real images."""
         self.it += 1
+        if self.lr_schedule or self.lr_warmup:
+            scale = self.lr_scale_at(self.it)
+            self.gen.updater.lr_scale = scale
+            self.dis.updater.lr_scale = scale
         if self.capture and not self._graph_failed:
             return self._step_graphed(real)
         return self._step_eager(real)

@@ -1,3 +1,4 @@
+import pytest
 import torch
 
 from gan_deeplearning4j_amd.config import OptimConfig
@@ -92,3 +93,44 @@ def test_bf16_master_weights():
     assert slot.master.dtype == torch.float32
     # param tracks master rounded to bf16
     assert torch.equal(p.detach(), slot.master.to(torch.bfloat16))
+
+
+def test_lr_schedule_scales():
+    import torch
+    from gan_deeplearning4j_amd.config import preset
+    from gan_deeplearning4j_amd.models import build_mlp_gan
+    from gan_deeplearning4j_amd.train import GanTrainer
+
+    cfg = preset("mlp_tabular_cpu")
+    cfg.train.use_gpu = False
+    cfg.train.num_iterations = 6
+    cfg.optim.lr_schedule = "linear"
+    cfg.optim.lr_warmup_steps = 2
+    gen, dis = build_mlp_gan(cfg, hidden=16)
+    tr = GanTrainer(gen, dis, cfg, device=torch.device("cpu"))
+    scales = [tr.lr_scale_at(i) for i in range(1, 7)]
+    assert scales[0] == 0.5 and scales[1] == 1.0       # warmup
+    assert scales[-1] == 0.0                           # decayed to zero
+    assert all(a >= b for a, b in zip(scales[1:], scales[2:]))
+    # cosine endpoints
+    cfg.optim.lr_schedule = "cosine"
+    cfg.optim.lr_warmup_steps = 0
+    tr2 = GanTrainer(*build_mlp_gan(cfg, hidden=16), cfg,
+                     device=torch.device("cpu"))
+    assert abs(tr2.lr_scale_at(6)) < 1e-9
+    assert tr2.lr_scale_at(3) == pytest.approx(0.5)
+    # a zero-scale step leaves parameters untouched
+    x = torch.rand(8, cfg.data.num_features)
+    for _ in range(6):
+        tr2.step(x)
+    w = torch.cat([p.detach().reshape(-1).clone()
+                   for p in tr2.gen.parameters()])
+    tr2.step(x)   # it=7 -> prog clamps to 1 -> scale 0
+    w2 = torch.cat([p.detach().reshape(-1)
+                    for p in tr2.gen.parameters()])
+    assert torch.equal(w2, w)  # only BN running stats may move
+    # invalid schedule rejected
+    cfg.optim.lr_schedule = "step"
+    with pytest.raises(ValueError):
+        GanTrainer(*build_mlp_gan(cfg, hidden=16), cfg,
+                   device=torch.device("cpu"))
