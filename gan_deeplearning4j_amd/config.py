@@ -98,6 +98,7 @@ class TrainConfig:
     #                                    lsgan (least-squares) | hinge
     augment: str = ""                  # DiffAugment policy for D inputs,
     #                                    e.g. "translate,cutout" ("" = off)
+    d_steps_per_g: int = 1             # n-critic: D updates per G update
 
 
 @dataclass

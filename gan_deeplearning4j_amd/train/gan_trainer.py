@@ -135,6 +135,8 @@ class GanTrainer:
         self._ema = ([p.detach().float().clone()
                       for p in self.gen.parameters()]
                      if self.ema_decay > 0 else None)
+        # n-critic: D updates per G update (1 = the reference recipe)
+        self.d_steps = max(1, int(getattr(cfg.train, "d_steps_per_g", 1)))
         # LR schedule (framework extension): warmup + linear/cosine decay
         # over num_iterations. Incompatible with graph capture (lr is a
         # scalar kernel argument, frozen at capture) -> forces eager.
@@ -178,6 +180,30 @@ class GanTrainer:
         z = torch.randn(n, self.z_size, generator=self._g)
         return z.to(self.device, self.dtype)
 
+    def _d_loss(self, real, fake, soft_real, soft_fake):
+        if self._d_concat:
+            both = torch.cat([real, fake.detach()], dim=0)
+            d_all = self.dis(both)
+            lab = torch.cat([soft_real, soft_fake], dim=0)
+            # x2 keeps the gradient scale of mean(real)+mean(fake)
+            return OF.bce_with_logits_loss(d_all, lab) * 2
+        real_in, fake_in = real, fake.detach()
+        if self.augment and real.dim() == 4:  # images only
+            from .augment import diff_augment
+
+            real_in = diff_augment(real_in, self.augment)
+            fake_in = diff_augment(fake_in, self.augment)
+        d_real = self.dis(real_in)
+        d_fake = self.dis(fake_in)
+        if self.loss_type == "lsgan":
+            return OF.mse_loss(d_real, soft_real) + \
+                OF.mse_loss(d_fake, soft_fake)
+        if self.loss_type == "hinge":
+            return torch.relu(1.0 - d_real.float()).mean() + \
+                torch.relu(1.0 + d_fake.float()).mean()
+        return OF.bce_with_logits_loss(d_real, soft_real) + \
+            OF.bce_with_logits_loss(d_fake, soft_fake)
+
     def lr_scale_at(self, it: int) -> float:
         """Schedule multiplier for 1-based step `it` (1.0 when off)."""
         if not self.lr_schedule and self.lr_warmup == 0:
@@ -210,43 +236,22 @@ class GanTrainer:
         real = real.to(self.device, self.dtype)
         soft_real, soft_fake = self._labels(n)
 
-        z = self.sample_z(n)
         self.gen.train()
         self.dis.train()
-        fake = self.gen(z)
 
-        # ---- D step -------------------------------------------------
-        self.dis.updater.zero_grad()
-        self.d_reducer.prepare()
-        if self._d_concat:
-            both = torch.cat([real, fake.detach()], dim=0)
-            d_all = self.dis(both)
-            lab = torch.cat([soft_real, soft_fake], dim=0)
-            # x2 keeps the gradient scale of mean(real)+mean(fake)
-            loss_d = OF.bce_with_logits_loss(d_all, lab) * 2
-        else:
-            real_in, fake_in = real, fake.detach()
-            use_aug = bool(self.augment) and real.dim() == 4  # images only
-            if use_aug:
-                from .augment import diff_augment
-
-                real_in = diff_augment(real_in, self.augment)
-                fake_in = diff_augment(fake_in, self.augment)
-            d_real = self.dis(real_in)
-            d_fake = self.dis(fake_in)
-            if self.loss_type == "lsgan":
-                loss_d = OF.mse_loss(d_real, soft_real) + \
-                    OF.mse_loss(d_fake, soft_fake)
-            elif self.loss_type == "hinge":
-                loss_d = torch.relu(1.0 - d_real.float()).mean() + \
-                    torch.relu(1.0 + d_fake.float()).mean()
-            else:
-                loss_d = OF.bce_with_logits_loss(d_real, soft_real) + \
-                    OF.bce_with_logits_loss(d_fake, soft_fake)
-        loss_d.backward()
-        self.d_reducer.finish()
-        self.dis.updater.step()
-
+        # ---- D step(s): train.d_steps_per_g > 1 is the n-critic
+        # recipe (fresh z per D update; the G step reuses the last
+        # fake). d_steps == 1 reproduces the reference flow exactly. --
+        fake = loss_d = None
+        for _ in range(self.d_steps):
+            z = self.sample_z(n)
+            fake = self.gen(z)
+            self.dis.updater.zero_grad()
+            self.d_reducer.prepare()
+            loss_d = self._d_loss(real, fake, soft_real, soft_fake)
+            loss_d.backward()
+            self.d_reducer.finish()
+            self.dis.updater.step()
         # ---- G step (stop-gradient freeze of D: no D wgrad compute) --
         for p in self.dis.parameters():
             p.requires_grad_(False)

@@ -172,3 +172,18 @@ def test_reference_weight_sync_semantics(tmp_path):
         tr.gen.get_layer("gen_dense_layer_2").get_param("W"),
         tr.gan.get_layer("gan_dense_layer_2").get_param("W"),
     )
+
+
+def test_n_critic_d_steps():
+    cfg = preset("mlp_tabular_cpu")
+    cfg.train.use_gpu = False
+    cfg.train.d_steps_per_g = 3
+    gen, dis = build_mlp_gan(cfg, hidden=16)
+    tr = GanTrainer(gen, dis, cfg, device=torch.device("cpu"))
+    x = torch.rand(8, cfg.data.num_features)
+    out = tr.step(x)
+    assert tr.dis.updater.t == 3     # three D updates
+    assert tr.gen.updater.t == 1     # one G update
+    assert np.isfinite(float(out["loss_d"]))
+    tr.step(x)
+    assert tr.dis.updater.t == 6 and tr.gen.updater.t == 2
