@@ -180,7 +180,18 @@ class GanTrainer:
         z = torch.randn(n, self.z_size, generator=self._g)
         return z.to(self.device, self.dtype)
 
-    def _d_loss(self, real, fake, soft_real, soft_fake):
+    def _d_loss(self, real, fake, soft_real, soft_fake, cond=()):
+        if cond:
+            d_real = self.dis(real, *cond)
+            d_fake = self.dis(fake.detach(), *cond)
+            if self.loss_type == "lsgan":
+                return OF.mse_loss(d_real, soft_real) + \
+                    OF.mse_loss(d_fake, soft_fake)
+            if self.loss_type == "hinge":
+                return torch.relu(1.0 - d_real.float()).mean() + \
+                    torch.relu(1.0 + d_fake.float()).mean()
+            return OF.bce_with_logits_loss(d_real, soft_real) + \
+                OF.bce_with_logits_loss(d_fake, soft_fake)
         if self._d_concat:
             both = torch.cat([real, fake.detach()], dim=0)
             d_all = self.dis(both)
@@ -220,20 +231,29 @@ class GanTrainer:
 
         return 0.5 * (1.0 + math.cos(math.pi * prog))  # cosine
 
-    def step(self, real: torch.Tensor) -> dict:
-        """One alternating D+G update on a batch of real images."""
+    def step(self, real: torch.Tensor,
+             labels: Optional[torch.Tensor] = None) -> dict:
+        """One alternating D+G update on a batch of real images.
+
+        labels: conditioning input for multi-input (cGAN) graphs —
+        passed as the second graph input to both G and D. Conditional
+        steps always run eager (capture is keyed to one input shape).
+        """
         self.it += 1
         if self.lr_schedule or self.lr_warmup:
             scale = self.lr_scale_at(self.it)
             self.gen.updater.lr_scale = scale
             self.dis.updater.lr_scale = scale
-        if self.capture and not self._graph_failed:
+        if labels is None and self.capture and not self._graph_failed:
             return self._step_graphed(real)
-        return self._step_eager(real)
+        return self._step_eager(real, labels)
 
-    def _step_eager(self, real: torch.Tensor) -> dict:
+    def _step_eager(self, real: torch.Tensor,
+                    labels: Optional[torch.Tensor] = None) -> dict:
         n = real.shape[0]
         real = real.to(self.device, self.dtype)
+        cond = () if labels is None else (
+            labels.to(self.device, self.dtype),)
         soft_real, soft_fake = self._labels(n)
 
         self.gen.train()
@@ -245,10 +265,10 @@ class GanTrainer:
         fake = loss_d = None
         for _ in range(self.d_steps):
             z = self.sample_z(n)
-            fake = self.gen(z)
+            fake = self.gen(z, *cond)
             self.dis.updater.zero_grad()
             self.d_reducer.prepare()
-            loss_d = self._d_loss(real, fake, soft_real, soft_fake)
+            loss_d = self._d_loss(real, fake, soft_real, soft_fake, cond)
             loss_d.backward()
             self.d_reducer.finish()
             self.dis.updater.step()
@@ -258,11 +278,11 @@ class GanTrainer:
         self.gen.updater.zero_grad()
         self.g_reducer.prepare()
         g_in = fake
-        if self.augment and fake.dim() == 4:
+        if self.augment and fake.dim() == 4 and not cond:
             from .augment import diff_augment
 
             g_in = diff_augment(g_in, self.augment)
-        g_logits = self.dis(g_in)
+        g_logits = self.dis(g_in, *cond)
         if self.loss_type == "lsgan":
             loss_g = OF.mse_loss(g_logits,
                                  torch.ones(n, 1, device=self.device))

@@ -98,3 +98,23 @@ def test_cgan_serialization_roundtrip(tmp_path):
     z = torch.randn(2, cfg.model.z_size)
     y = torch.eye(cfg.data.num_classes)[torch.tensor([3, 7])]
     assert torch.allclose(gen.output(z, y), g2.output(z, y), atol=1e-6)
+
+
+def test_cgan_trains_through_gan_trainer():
+    # conditional graphs train through the SAME fast trainer via
+    # step(real, labels=onehot) — reducers/losses/EMA all apply
+    import numpy as np
+    from gan_deeplearning4j_amd.train import GanTrainer
+
+    cfg, gen, dis = _models()
+    cfg.train.use_gpu = False
+    tr = GanTrainer(gen, dis, cfg, device=torch.device("cpu"))
+    n, ncls = 8, cfg.data.num_classes
+    real = torch.rand(n, 1, 28, 28)
+    y = torch.eye(ncls)[torch.randint(0, ncls, (n,))]
+    w0 = gen.params_flat().clone()
+    out = tr.step(real, labels=y)
+    assert np.isfinite(float(out["loss_d"]))
+    assert np.isfinite(float(out["loss_g"]))
+    assert tr.gen.updater.t == 1 and tr.dis.updater.t == 1
+    assert not torch.allclose(gen.params_flat(), w0)
